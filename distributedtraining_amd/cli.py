@@ -1,0 +1,112 @@
+"""CLI entry points: miner | validator | averager | bench.
+
+Replaces the reference's neurons/{miner,validator,averager}.py process
+entry points with explicit subcommands over the file
+transport — the multi-process plumbing mode. The rccl rank mode is driven
+by bench.py / torchrun instead.
+
+Usage:
+  python -m distributedtraining_amd.cli miner     --comm.root /tmp/ex --hotkey m0 --steps 200
+  python -m distributedtraining_amd.cli validator --comm.root /tmp/ex --rounds 1
+  python -m distributedtraining_amd.cli averager  --comm.root /tmp/ex --rounds 1
+"""
+
+from __future__ import annotations
+
+import argparse
+import logging
+import sys
+
+import torch
+
+from . import config as cfg_mod
+from .config import Config, ModelConfig
+from .models import build_model
+from .parallel.flat import FlatParams
+from .registry import FileRegistry
+from .roles.averager import ParameterizedAverager
+from .roles.miner import DeltaLoop
+from .roles.validator import DeltaValidator
+from .store import FileStore
+from .utils.data import synthetic_batches, synthetic_eval_set
+
+
+def _setup(cfg: Config, hotkey: str):
+    torch.manual_seed(cfg.seed)
+    model_cfg = cfg.model
+    model = build_model(model_cfg)
+    device = torch.device(cfg.comm.device)
+    model.to(device)
+    fp = FlatParams(model, device=device)
+    store = FileStore(cfg.comm.root, hotkey=hotkey)
+    registry = FileRegistry(cfg.comm.root,
+                            epoch_length=cfg.validate.epoch_length,
+                            ema_alpha=cfg.validate.score_ema_alpha)
+    return model, fp, store, registry
+
+
+def main(argv=None) -> int:
+    argv = argv if argv is not None else sys.argv[1:]
+    logging.basicConfig(level=logging.INFO)
+    top = argparse.ArgumentParser("distributedtraining_amd")
+    top.add_argument("role", choices=["miner", "validator", "averager"])
+    top.add_argument("--hotkey", default=None)
+    top.add_argument("--steps", type=int, default=100)
+    top.add_argument("--rounds", type=int, default=1)
+    top.add_argument("--tiny", action="store_true",
+                     help="use the tiny test-scale model")
+    ns, rest = top.parse_known_args(argv)
+    cfg = cfg_mod.from_args(rest)
+    if ns.tiny:
+        cfg.model = ModelConfig.gpt2_tiny()
+    hotkey = ns.hotkey or ns.role
+
+    model, fp, store, registry = _setup(cfg, hotkey)
+
+    if ns.role == "miner":
+        # publish the initial base if none exists yet (first process up)
+        if store.pull_model() is None:
+            store.push_model({"format": "dta-base-v1",
+                              "flat_master": fp.master.cpu(),
+                              "spec": fp.spec})
+        data = synthetic_batches(cfg.model.vocab_size, cfg.train.batch_size,
+                                 cfg.train.seq_len, seed=cfg.seed + hash(hotkey) % 1000)
+        miner = DeltaLoop(model, fp, data, cfg.train, store=store,
+                          registry=registry, hotkey=hotkey)
+        miner.maybe_pull_base()
+        miner.train(ns.steps)
+        miner.last_push_step = -10**9  # force a final push
+        miner.maybe_push_delta()
+        print(f"miner {hotkey}: {ns.steps} steps, "
+              f"avg loss {miner.average_loss():.4f}")
+        return 0
+
+    ev = synthetic_eval_set(cfg.model.vocab_size, cfg.validate.n_eval_batches,
+                            cfg.validate.batch_size,
+                            min(cfg.validate.seq_len, cfg.model.n_positions))
+    if ns.role == "validator":
+        sd = store.pull_model()
+        if sd is not None:
+            fp.load_flat_master(sd["flat_master"])
+        validator = DeltaValidator(model, fp, ev, cfg.validate, store=store,
+                                   registry=registry)
+        for _ in range(ns.rounds):
+            scores = validator.validate_and_score()
+            print("scores:", {k: round(v, 4) for k, v in scores.items()})
+        return 0
+
+    if ns.role == "averager":
+        sd = store.pull_model()
+        if sd is not None:
+            fp.load_flat_master(sd["flat_master"])
+        averager = ParameterizedAverager(model, fp, cfg.average, store=store,
+                                         registry=registry)
+        for _ in range(ns.rounds):
+            averager.run_round(ev)
+            print("averager: merged + published new base")
+        return 0
+    return 1
+
+
+if __name__ == "__main__":
+    sys.exit(main())

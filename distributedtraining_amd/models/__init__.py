@@ -1,0 +1,17 @@
+from ..config import ModelConfig
+from .gpt2 import GPT2LM, CausalLMOutput
+from .llama import LlamaLM
+from .mlp import FeedforwardNN
+
+__all__ = ["build_model", "GPT2LM", "LlamaLM", "FeedforwardNN",
+           "CausalLMOutput"]
+
+
+def build_model(cfg: ModelConfig):
+    if cfg.family == "gpt2":
+        return GPT2LM(cfg)
+    if cfg.family == "llama":
+        return LlamaLM(cfg)
+    if cfg.family == "mlp":
+        return FeedforwardNN()
+    raise ValueError(f"unknown model family {cfg.family!r}")

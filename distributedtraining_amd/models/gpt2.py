@@ -1,0 +1,121 @@
+"""Native GPT-2 built on the framework's own op surface.
+
+The reference outsources its entire model to HF transformers
+(`AutoModelForCausalLM.from_pretrained("openai-community/gpt2")`,
+/root/reference/neurons/miner.py:60-62). Here the architecture is
+reimplemented directly so the hot path runs on our CDNA4 kernels:
+
+* fused causal flash attention (ops.causal_attention)
+* fused LayerNorm (ops.layer_norm)
+* tanh-GELU (ops.gelu — GPT-2's "gelu_new")
+* fused token+position embedding gather (ops.embedding_fwd)
+* fused log-softmax cross-entropy over the 50k vocab (ops.cross_entropy_loss)
+
+Projection GEMMs go through F.linear → hipBLASLt (library GEMMs).
+Numerics parity with transformers.GPT2LMHeadModel is covered by
+tests/test_gpt2_parity.py.
+"""
+
+from __future__ import annotations
+
+import math
+from dataclasses import dataclass
+from typing import Optional
+
+import torch
+import torch.nn as nn
+import torch.nn.functional as F
+
+from .. import ops
+from ..config import ModelConfig
+
+
+class GPT2Block(nn.Module):
+    def __init__(self, cfg: ModelConfig):
+        super().__init__()
+        E = cfg.n_embd
+        self.n_head = cfg.n_head
+        self.ln_1_w = nn.Parameter(torch.ones(E))
+        self.ln_1_b = nn.Parameter(torch.zeros(E))
+        self.attn_qkv_w = nn.Parameter(torch.empty(3 * E, E))
+        self.attn_qkv_b = nn.Parameter(torch.zeros(3 * E))
+        self.attn_proj_w = nn.Parameter(torch.empty(E, E))
+        self.attn_proj_b = nn.Parameter(torch.zeros(E))
+        self.ln_2_w = nn.Parameter(torch.ones(E))
+        self.ln_2_b = nn.Parameter(torch.zeros(E))
+        self.mlp_fc_w = nn.Parameter(torch.empty(4 * E, E))
+        self.mlp_fc_b = nn.Parameter(torch.zeros(4 * E))
+        self.mlp_proj_w = nn.Parameter(torch.empty(E, 4 * E))
+        self.mlp_proj_b = nn.Parameter(torch.zeros(E))
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        B, S, E = x.shape
+        H = self.n_head
+        h = ops.layer_norm(x, self.ln_1_w, self.ln_1_b)
+        qkv = F.linear(h, self.attn_qkv_w, self.attn_qkv_b)
+        q, k, v = qkv.split(E, dim=-1)
+        q = q.view(B, S, H, E // H).transpose(1, 2)
+        k = k.view(B, S, H, E // H).transpose(1, 2)
+        v = v.view(B, S, H, E // H).transpose(1, 2)
+        o = ops.causal_attention(q, k, v)
+        o = o.transpose(1, 2).reshape(B, S, E)
+        x = x + F.linear(o, self.attn_proj_w, self.attn_proj_b)
+        h = ops.layer_norm(x, self.ln_2_w, self.ln_2_b)
+        h = ops.gelu(F.linear(h, self.mlp_fc_w, self.mlp_fc_b))
+        x = x + F.linear(h, self.mlp_proj_w, self.mlp_proj_b)
+        return x
+
+
+@dataclass
+class CausalLMOutput:
+    loss: Optional[torch.Tensor]
+    logits: Optional[torch.Tensor]
+
+
+class GPT2LM(nn.Module):
+    """GPT-2 causal LM with tied embeddings and internal shifted-CE loss
+    (matching the reference's `labels=` call contract,
+    training_manager.py:380-385)."""
+
+    def __init__(self, cfg: ModelConfig):
+        super().__init__()
+        assert cfg.family == "gpt2"
+        self.cfg = cfg
+        self.wte = nn.Parameter(torch.empty(cfg.vocab_size, cfg.n_embd))
+        self.wpe = nn.Parameter(torch.empty(cfg.n_positions, cfg.n_embd))
+        self.blocks = nn.ModuleList(GPT2Block(cfg) for _ in range(cfg.n_layer))
+        self.ln_f_w = nn.Parameter(torch.ones(cfg.n_embd))
+        self.ln_f_b = nn.Parameter(torch.zeros(cfg.n_embd))
+        self.reset_parameters()
+
+    def reset_parameters(self) -> None:
+        std = 0.02
+        resid_std = std / math.sqrt(2 * self.cfg.n_layer)
+        with torch.no_grad():
+            self.wte.normal_(0, std)
+            self.wpe.normal_(0, std / 2)
+            for b in self.blocks:
+                b.attn_qkv_w.normal_(0, std)
+                b.attn_proj_w.normal_(0, resid_std)
+                b.mlp_fc_w.normal_(0, std)
+                b.mlp_proj_w.normal_(0, resid_std)
+
+    def forward(self, input_ids: torch.Tensor,
+                attention_mask: Optional[torch.Tensor] = None,
+                labels: Optional[torch.Tensor] = None,
+                return_logits: Optional[bool] = None) -> CausalLMOutput:
+        x = ops.embedding_fwd(input_ids, self.wte, self.wpe)
+        for blk in self.blocks:
+            x = blk(x)
+        x = ops.layer_norm(x, self.ln_f_w, self.ln_f_b)
+        if labels is None:
+            logits = F.linear(x, self.wte)
+            return CausalLMOutput(loss=None, logits=logits)
+        # shifted CE: predict token t+1 from position t (HF semantics)
+        logits = F.linear(x[:, :-1, :], self.wte)
+        tgt = labels[:, 1:].contiguous().view(-1)
+        loss = ops.cross_entropy_loss(
+            logits.reshape(-1, self.cfg.vocab_size), tgt)
+        return CausalLMOutput(
+            loss=loss,
+            logits=logits if (return_logits or not self.training) else None)

@@ -1,0 +1,124 @@
+"""Native Llama-3-family causal LM on the framework's op surface.
+
+The reference has no Llama path (it trains GPT-2 only); this is the scale
+model called for by BASELINE.json config #4 (Llama-3-8B bf16, 8 miners,
+sized for 288 GB HBM3E). Same kernel stack as GPT-2 with the Llama op
+variants: RMSNorm, RoPE, SwiGLU, grouped-query attention.
+"""
+
+from __future__ import annotations
+
+import math
+from typing import Optional
+
+import torch
+import torch.nn as nn
+import torch.nn.functional as F
+
+from .. import ops
+from ..config import ModelConfig
+from .gpt2 import CausalLMOutput
+
+
+def rope_tables(n_pos: int, head_dim: int, theta: float,
+                device=None) -> tuple:
+    inv = 1.0 / (theta ** (torch.arange(0, head_dim, 2,
+                                        dtype=torch.float32, device=device)
+                           / head_dim))
+    t = torch.arange(n_pos, dtype=torch.float32, device=device)
+    freqs = torch.outer(t, inv)  # [S, D/2]
+    return freqs.cos(), freqs.sin()
+
+
+class LlamaBlock(nn.Module):
+    def __init__(self, cfg: ModelConfig):
+        super().__init__()
+        E = cfg.n_embd
+        self.n_head = cfg.n_head
+        self.n_kv = cfg.n_kv_head or cfg.n_head
+        self.head_dim = E // cfg.n_head
+        I = cfg.intermediate_size or 4 * E
+        kv_dim = self.n_kv * self.head_dim
+        self.norm_eps = cfg.norm_eps
+        self.attn_norm_w = nn.Parameter(torch.ones(E))
+        self.q_w = nn.Parameter(torch.empty(E, E))
+        self.k_w = nn.Parameter(torch.empty(kv_dim, E))
+        self.v_w = nn.Parameter(torch.empty(kv_dim, E))
+        self.o_w = nn.Parameter(torch.empty(E, E))
+        self.mlp_norm_w = nn.Parameter(torch.ones(E))
+        self.gate_w = nn.Parameter(torch.empty(I, E))
+        self.up_w = nn.Parameter(torch.empty(I, E))
+        self.down_w = nn.Parameter(torch.empty(E, I))
+
+    def forward(self, x: torch.Tensor, cos: torch.Tensor,
+                sin: torch.Tensor) -> torch.Tensor:
+        B, S, E = x.shape
+        D = self.head_dim
+        h = ops.rms_norm(x, self.attn_norm_w, self.norm_eps)
+        q = F.linear(h, self.q_w).view(B, S, self.n_head, D).transpose(1, 2)
+        k = F.linear(h, self.k_w).view(B, S, self.n_kv, D).transpose(1, 2)
+        v = F.linear(h, self.v_w).view(B, S, self.n_kv, D).transpose(1, 2)
+        q = ops.rope(q, cos, sin)
+        k = ops.rope(k, cos, sin)
+        o = ops.causal_attention(q, k, v)
+        o = o.transpose(1, 2).reshape(B, S, E)
+        x = x + F.linear(o, self.o_w)
+        h = ops.rms_norm(x, self.mlp_norm_w, self.norm_eps)
+        h = ops.swiglu(F.linear(h, self.gate_w), F.linear(h, self.up_w))
+        x = x + F.linear(h, self.down_w)
+        return x
+
+
+class LlamaLM(nn.Module):
+    def __init__(self, cfg: ModelConfig):
+        super().__init__()
+        assert cfg.family == "llama"
+        self.cfg = cfg
+        self.tok_emb = nn.Parameter(torch.empty(cfg.vocab_size, cfg.n_embd))
+        self.blocks = nn.ModuleList(LlamaBlock(cfg) for _ in range(cfg.n_layer))
+        self.final_norm_w = nn.Parameter(torch.ones(cfg.n_embd))
+        if cfg.tie_word_embeddings:
+            self.lm_head_w = None
+        else:
+            self.lm_head_w = nn.Parameter(torch.empty(cfg.vocab_size, cfg.n_embd))
+        cos, sin = rope_tables(cfg.n_positions, cfg.n_embd // cfg.n_head,
+                               cfg.rope_theta)
+        self.register_buffer("rope_cos", cos, persistent=False)
+        self.register_buffer("rope_sin", sin, persistent=False)
+        self.reset_parameters()
+
+    def reset_parameters(self) -> None:
+        std = 0.02
+        with torch.no_grad():
+            self.tok_emb.normal_(0, std)
+            if self.lm_head_w is not None:
+                self.lm_head_w.normal_(0, std)
+            for b in self.blocks:
+                for w in (b.q_w, b.k_w, b.v_w, b.gate_w, b.up_w):
+                    w.normal_(0, std)
+                for w in (b.o_w, b.down_w):
+                    w.normal_(0, std / math.sqrt(2 * self.cfg.n_layer))
+
+    def _head(self):
+        return self.lm_head_w if self.lm_head_w is not None else self.tok_emb
+
+    def forward(self, input_ids: torch.Tensor,
+                attention_mask: Optional[torch.Tensor] = None,
+                labels: Optional[torch.Tensor] = None,
+                return_logits: Optional[bool] = None) -> CausalLMOutput:
+        S = input_ids.shape[1]
+        cos = self.rope_cos[:S]
+        sin = self.rope_sin[:S]
+        x = ops.embedding_fwd(input_ids, self.tok_emb, None)
+        for blk in self.blocks:
+            x = blk(x, cos, sin)
+        x = ops.rms_norm(x, self.final_norm_w, self.cfg.norm_eps)
+        if labels is None:
+            return CausalLMOutput(loss=None, logits=F.linear(x, self._head()))
+        logits = F.linear(x[:, :-1, :], self._head())
+        tgt = labels[:, 1:].contiguous().view(-1)
+        loss = ops.cross_entropy_loss(
+            logits.reshape(-1, self.cfg.vocab_size), tgt)
+        return CausalLMOutput(
+            loss=loss,
+            logits=logits if (return_logits or not self.training) else None)

@@ -1,0 +1,395 @@
+"""Functional op surface for the MI355X-native model stack.
+
+Every op has exactly two execution paths:
+
+* **CPU** — a plain fp32 PyTorch composition. This is the numerics gold the
+  GPU kernels are tested against, and what CPU-only plumbing/tests run on.
+* **ROCm GPU** — a hand-written gfx950 HIP kernel behind a
+  ``torch.autograd.Function`` (see hip/*.hip). There is no eager fallback on
+  GPU: missing extension ⇒ loud error (backend.require_ext).
+
+Plain projection GEMMs (QKV/out/MLP/LM-head matmuls) intentionally go
+through ``torch.nn.functional.linear`` → hipBLASLt, per the library-GEMM /
+hand-written-fused-op split; everything fused or memory-bound is ours.
+
+Reference behavior being reimplemented: the implicit per-step op set of
+GPT-2 training in /root/reference.
+"""
+
+from __future__ import annotations
+
+import math
+from typing import Optional, Tuple
+
+import torch
+import torch.nn.functional as F
+
+from . import backend
+from .backend import has_ext, require_ext, use_hip
+
+__all__ = [
+    "layer_norm", "rms_norm", "gelu", "swiglu", "causal_attention",
+    "cross_entropy_loss", "embedding_fwd", "rope", "adamw_step",
+    "delta_sub", "axpy_", "weighted_merge", "grad_merge_weights",
+    "has_nan", "l2norm",
+]
+
+
+# --------------------------------------------------------------------------
+# LayerNorm
+# --------------------------------------------------------------------------
+class _LayerNormFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, w, b, eps):
+        m = require_ext()
+        x2 = x.contiguous()
+        y, mean, rstd = m.layernorm_fwd(x2.view(-1, x2.shape[-1]), w, b, eps)
+        ctx.save_for_backward(x2, w, mean, rstd)
+        return y.view_as(x2)
+
+    @staticmethod
+    def backward(ctx, dy):
+        m = require_ext()
+        x, w, mean, rstd = ctx.saved_tensors
+        N = x.shape[-1]
+        dx, dw, db = m.layernorm_bwd(dy.contiguous().view(-1, N),
+                                     x.view(-1, N), w, mean, rstd)
+        return dx.view_as(x), dw, db, None
+
+
+def layer_norm(x: torch.Tensor, w: torch.Tensor, b: torch.Tensor,
+               eps: float = 1e-5) -> torch.Tensor:
+    if use_hip(x):
+        return _LayerNormFn.apply(x, w, b, eps)
+    return F.layer_norm(x, (x.shape[-1],), w, b, eps)
+
+
+# --------------------------------------------------------------------------
+# RMSNorm (Llama family)
+# --------------------------------------------------------------------------
+class _RMSNormFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, w, eps):
+        m = require_ext()
+        x2 = x.contiguous()
+        y, rstd = m.rmsnorm_fwd(x2.view(-1, x2.shape[-1]), w, eps)
+        ctx.save_for_backward(x2, w, rstd)
+        return y.view_as(x2)
+
+    @staticmethod
+    def backward(ctx, dy):
+        m = require_ext()
+        x, w, rstd = ctx.saved_tensors
+        N = x.shape[-1]
+        dx, dw = m.rmsnorm_bwd(dy.contiguous().view(-1, N), x.view(-1, N),
+                               w, rstd)
+        return dx.view_as(x), dw, None
+
+
+def rms_norm(x: torch.Tensor, w: torch.Tensor, eps: float = 1e-5) -> torch.Tensor:
+    if use_hip(x):
+        return _RMSNormFn.apply(x, w, eps)
+    xf = x.float()
+    y = xf * torch.rsqrt(xf.pow(2).mean(-1, keepdim=True) + eps)
+    return (y * w.float()).to(x.dtype)
+
+
+# --------------------------------------------------------------------------
+# GELU (tanh approximation — GPT-2 uses gelu_new)
+# --------------------------------------------------------------------------
+_GELU_C = math.sqrt(2.0 / math.pi)
+
+
+class _GeluFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x):
+        m = require_ext()
+        x2 = x.contiguous()
+        ctx.save_for_backward(x2)
+        return m.gelu_fwd(x2)
+
+    @staticmethod
+    def backward(ctx, dy):
+        m = require_ext()
+        (x,) = ctx.saved_tensors
+        return m.gelu_bwd(dy.contiguous(), x)
+
+
+def gelu(x: torch.Tensor) -> torch.Tensor:
+    if use_hip(x):
+        return _GeluFn.apply(x)
+    return F.gelu(x, approximate="tanh")
+
+
+# --------------------------------------------------------------------------
+# SwiGLU (Llama MLP): silu(gate) * up
+# --------------------------------------------------------------------------
+class _SwiGLUFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, gate, up):
+        m = require_ext()
+        gate, up = gate.contiguous(), up.contiguous()
+        ctx.save_for_backward(gate, up)
+        return m.swiglu_fwd(gate, up)
+
+    @staticmethod
+    def backward(ctx, dy):
+        m = require_ext()
+        gate, up = ctx.saved_tensors
+        dgate, dup = m.swiglu_bwd(dy.contiguous(), gate, up)
+        return dgate, dup
+
+
+def swiglu(gate: torch.Tensor, up: torch.Tensor) -> torch.Tensor:
+    if use_hip(gate):
+        return _SwiGLUFn.apply(gate, up)
+    return F.silu(gate) * up
+
+
+# --------------------------------------------------------------------------
+# Fused causal attention (flash-style): q,k,v [B,H,S,D] -> o [B,H,S,D]
+# --------------------------------------------------------------------------
+class _AttnFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, q, k, v, scale):
+        m = require_ext()
+        q, k, v = q.contiguous(), k.contiguous(), v.contiguous()
+        o, lse = m.attn_fwd(q, k, v, scale)
+        ctx.save_for_backward(q, k, v, o, lse)
+        ctx.scale = scale
+        return o
+
+    @staticmethod
+    def backward(ctx, do):
+        m = require_ext()
+        q, k, v, o, lse = ctx.saved_tensors
+        dq, dk, dv = m.attn_bwd(do.contiguous(), q, k, v, o, lse, ctx.scale)
+        return dq, dk, dv, None
+
+
+def causal_attention(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
+                     scale: Optional[float] = None) -> torch.Tensor:
+    """softmax(q kᵀ · scale + causal_mask) v over [B, H, S, D] tensors.
+
+    GQA: k/v may have fewer heads (H_kv dividing H); heads are grouped.
+    """
+    if scale is None:
+        scale = 1.0 / math.sqrt(q.shape[-1])
+    if use_hip(q):
+        if k.shape[1] != q.shape[1]:  # expand GQA heads for the kernel
+            rep = q.shape[1] // k.shape[1]
+            k = k.repeat_interleave(rep, dim=1)
+            v = v.repeat_interleave(rep, dim=1)
+        return _AttnFn.apply(q, k, v, scale)
+    return F.scaled_dot_product_attention(q, k, v, is_causal=True,
+                                          scale=scale, enable_gqa=True)
+
+
+# --------------------------------------------------------------------------
+# Fused log-softmax + cross entropy over the vocab dim
+# --------------------------------------------------------------------------
+class _CrossEntropyFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, logits, targets, ignore_index):
+        m = require_ext()
+        logits = logits.contiguous()
+        loss_sum, lse, count = m.ce_fwd(logits, targets, ignore_index)
+        ctx.save_for_backward(logits, targets, lse)
+        ctx.ignore_index = ignore_index
+        ctx.count = max(int(count), 1)
+        return loss_sum / ctx.count
+
+    @staticmethod
+    def backward(ctx, dloss):
+        m = require_ext()
+        logits, targets, lse = ctx.saved_tensors
+        scale = float(dloss) / ctx.count
+        dlogits = m.ce_bwd(logits, targets, lse, scale, ctx.ignore_index)
+        return dlogits, None, None
+
+
+def cross_entropy_loss(logits: torch.Tensor, targets: torch.Tensor,
+                       ignore_index: int = -100) -> torch.Tensor:
+    """Mean CE over non-ignored targets. logits [T,V], targets [T]."""
+    if use_hip(logits):
+        return _CrossEntropyFn.apply(logits, targets, ignore_index)
+    return F.cross_entropy(logits.float(), targets, ignore_index=ignore_index)
+
+
+# --------------------------------------------------------------------------
+# Embedding: fused token+position gather (and scatter-add backward)
+# --------------------------------------------------------------------------
+class _EmbeddingFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, ids, wte, wpe):
+        m = require_ext()
+        ids = ids.contiguous()
+        ctx.save_for_backward(ids)
+        ctx.vocab, ctx.npos = wte.shape[0], (wpe.shape[0] if wpe is not None else 0)
+        ctx.dim = wte.shape[1]
+        ctx.dtype = wte.dtype
+        return m.embedding_fwd(ids, wte, wpe if wpe is not None else torch.empty(0, dtype=wte.dtype, device=wte.device))
+
+    @staticmethod
+    def backward(ctx, dy):
+        m = require_ext()
+        (ids,) = ctx.saved_tensors
+        dwte, dwpe = m.embedding_bwd(dy.contiguous(), ids, ctx.vocab, ctx.npos)
+        return None, dwte, (dwpe if ctx.npos else None)
+
+
+def embedding_fwd(ids: torch.Tensor, wte: torch.Tensor,
+                  wpe: Optional[torch.Tensor] = None) -> torch.Tensor:
+    """x[b,s,:] = wte[ids[b,s]] (+ wpe[s] if given)."""
+    if use_hip(wte):
+        return _EmbeddingFn.apply(ids, wte, wpe)
+    x = F.embedding(ids, wte)
+    if wpe is not None:
+        x = x + wpe[: ids.shape[-1]].unsqueeze(0)
+    return x
+
+
+# --------------------------------------------------------------------------
+# RoPE (Llama): rotate q,k in-place-free
+# --------------------------------------------------------------------------
+class _RopeFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, cos, sin):
+        m = require_ext()
+        ctx.save_for_backward(cos, sin)
+        return m.rope_fwd(x.contiguous(), cos, sin)
+
+    @staticmethod
+    def backward(ctx, dy):
+        m = require_ext()
+        cos, sin = ctx.saved_tensors
+        return m.rope_bwd(dy.contiguous(), cos, sin), None, None
+
+
+def rope(x: torch.Tensor, cos: torch.Tensor, sin: torch.Tensor) -> torch.Tensor:
+    """Apply rotary embedding to x [B,H,S,D] with cos/sin [S,D/2] (fp32)."""
+    if use_hip(x):
+        return _RopeFn.apply(x, cos, sin)
+    xf = x.float()
+    d2 = x.shape[-1] // 2
+    x1, x2 = xf[..., :d2], xf[..., d2:]
+    c = cos[: x.shape[-2]].view(1, 1, x.shape[-2], d2)
+    s = sin[: x.shape[-2]].view(1, 1, x.shape[-2], d2)
+    out = torch.cat([x1 * c - x2 * s, x1 * s + x2 * c], dim=-1)
+    return out.to(x.dtype)
+
+
+# --------------------------------------------------------------------------
+# Fused AdamW over flat buffers (reference: optimizer.step(),
+# training_manager.py:391 — torch AdamW; here one kernel over one buffer)
+# --------------------------------------------------------------------------
+def adamw_step(master: torch.Tensor, grad: torch.Tensor, m: torch.Tensor,
+               v: torch.Tensor, out_bf16: Optional[torch.Tensor], step: int,
+               lr: float, beta1: float = 0.9, beta2: float = 0.999,
+               eps: float = 1e-8, weight_decay: float = 0.01) -> None:
+    """Decoupled AdamW (torch semantics): in-place update of fp32 master,
+    m, v; optionally writes the bf16 working copy."""
+    if use_hip(master):
+        require_ext().adamw_step(master, grad, m, v,
+                                 out_bf16 if out_bf16 is not None else master.new_empty(0).to(torch.bfloat16),
+                                 step, lr, beta1, beta2, eps, weight_decay)
+        return
+    g = grad.float()
+    master.mul_(1.0 - lr * weight_decay)
+    m.mul_(beta1).add_(g, alpha=1 - beta1)
+    v.mul_(beta2).addcmul_(g, g, value=1 - beta2)
+    bc1 = 1 - beta1 ** step
+    bc2 = 1 - beta2 ** step
+    denom = (v / bc2).sqrt_().add_(eps)
+    master.addcdiv_(m, denom, value=-lr / bc1)
+    if out_bf16 is not None:
+        out_bf16.copy_(master.to(out_bf16.dtype))
+
+
+# --------------------------------------------------------------------------
+# Delta / merge primitives over flat buffers
+# --------------------------------------------------------------------------
+def delta_sub(w: torch.Tensor, base: torch.Tensor,
+              out: Optional[torch.Tensor] = None) -> torch.Tensor:
+    """delta = w − base (reference: training_manager.py:417-421)."""
+    if out is None:
+        out = torch.empty_like(base)
+    if use_hip(w):
+        require_ext().delta_sub(w, base, out)
+        return out
+    torch.sub(w.float(), base.float(), out=out)
+    return out
+
+
+def axpy_(w: torch.Tensor, x: torch.Tensor, alpha: float = 1.0) -> torch.Tensor:
+    """w += alpha·x (reference delta apply: validation_logic.py:252-259)."""
+    if use_hip(w):
+        require_ext().axpy(w, x, alpha)
+        return w
+    return w.add_(x.to(w.dtype), alpha=alpha)
+
+
+def weighted_merge(base: torch.Tensor, deltas: torch.Tensor,
+                   W: torch.Tensor, offsets: torch.Tensor,
+                   out: Optional[torch.Tensor] = None) -> torch.Tensor:
+    """merged[e] = Σ_i W[i, seg(e)] · (base[e] + deltas[i,e]).
+
+    ``W`` is [N, S] per-(miner, parameter-tensor) merge weights; ``offsets``
+    [S+1] the flat-buffer segment boundaries. Reference semantics:
+    averaging_logic.py:422-470 (get_averaged_params over per-param weights).
+    """
+    N, P = deltas.shape
+    if out is None:
+        out = torch.empty_like(base)
+    if use_hip(base):
+        require_ext().weighted_merge(base, deltas, W, offsets.to(base.device), out)
+        return out
+    seg = _seg_ids(offsets, P)
+    acc = torch.zeros(P, dtype=torch.float32)
+    for i in range(N):
+        acc += W[i].float()[seg] * (base.float() + deltas[i].float())
+    out.copy_(acc.to(out.dtype))
+    return out
+
+
+def grad_merge_weights(g: torch.Tensor, base: torch.Tensor,
+                       deltas: torch.Tensor, merged: torch.Tensor,
+                       offsets: torch.Tensor) -> torch.Tensor:
+    """grad_W[i,j] = Σ_{e∈seg j} g[e] · (base[e]+deltas[i,e] − merged[e]).
+
+    The averager's manual meta-gradient (averaging_logic.py:512-522)."""
+    N, P = deltas.shape
+    S = offsets.numel() - 1
+    if use_hip(g):
+        return require_ext().grad_merge_weights(g, base, deltas, merged,
+                                                offsets.to(g.device))
+    seg = _seg_ids(offsets, P)
+    gw = torch.zeros(N, S, dtype=torch.float32)
+    diff_base = base.float() - merged.float()
+    gf = g.float()
+    for i in range(N):
+        contrib = gf * (diff_base + deltas[i].float())
+        gw[i] = torch.zeros(S).index_add_(0, seg, contrib)
+    return gw
+
+
+def _seg_ids(offsets: torch.Tensor, P: int) -> torch.Tensor:
+    seg = torch.zeros(P, dtype=torch.long)
+    off = offsets.tolist()
+    for j in range(len(off) - 1):
+        seg[off[j]:off[j + 1]] = j
+    return seg
+
+
+def has_nan(flat: torch.Tensor) -> bool:
+    """Reference: have_nans, averaging_logic.py:121-127."""
+    if use_hip(flat):
+        return bool(require_ext().has_nan(flat))
+    return bool(torch.isnan(flat).any().item())
+
+
+def l2norm(flat: torch.Tensor) -> float:
+    """Grad-norm for clip/normalize (reference: training_manager.py:181-196)."""
+    if use_hip(flat):
+        return float(require_ext().l2norm_sq(flat)) ** 0.5
+    return float(flat.float().pow(2).sum().item()) ** 0.5

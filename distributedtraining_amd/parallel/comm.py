@@ -1,0 +1,98 @@
+"""RCCL/xGMI communication plane (with gloo CPU twin).
+
+Replaces the reference's HF-Hub git transport with collectives (SURVEY.md
+§2.4 C1-C8 mapping):
+
+  C1/C3/C4 miner delta push + validator/averager fetch
+      -> ONE all-gather of the flat fp32 delta across ranks. On the 8-GPU
+         MI355X xGMI clique every GPU owns 7 direct 153 GB/s links, so RCCL's
+         all-gather moves a 0.5 GB GPT-2 delta in ~ms; deltas then sit
+         HBM-resident for merge/scoring (no disk, no per-miner downloads).
+  C2/C5 merged-base publication -> broadcast from the averager rank.
+  C6/C7/C8 registry/scores/membership -> tiny all-gather-object or the
+         in-process registry (registry.py).
+
+Backend "nccl" IS RCCL on ROCm. The gloo backend runs the identical code
+path on CPU for tests (world_size>1 via torchrun or spawned procs).
+"""
+
+from __future__ import annotations
+
+import datetime
+import os
+from typing import List, Optional
+
+import torch
+import torch.distributed as dist
+
+
+class CommPlane:
+    """Thin, explicit wrapper over one torch.distributed process group."""
+
+    def __init__(self, backend: Optional[str] = None,
+                 device: Optional[torch.device] = None):
+        self.rank = int(os.environ.get("RANK", "0"))
+        self.world_size = int(os.environ.get("WORLD_SIZE", "1"))
+        self.local_rank = int(os.environ.get("LOCAL_RANK", str(self.rank)))
+        if device is None:
+            if torch.cuda.is_available():
+                device = torch.device("cuda", self.local_rank)
+            else:
+                device = torch.device("cpu")
+        self.device = device
+        if self.world_size > 1 and not dist.is_initialized():
+            if backend is None:
+                backend = "nccl" if device.type == "cuda" else "gloo"
+            if device.type == "cuda":
+                torch.cuda.set_device(device)
+            dist.init_process_group(
+                backend=backend,
+                timeout=datetime.timedelta(minutes=10))
+        self.backend = backend
+
+    @property
+    def is_distributed(self) -> bool:
+        return self.world_size > 1
+
+    def barrier(self) -> None:
+        if self.is_distributed:
+            if self.device.type == "cuda":
+                dist.barrier(device_ids=[self.device.index])
+            else:
+                dist.barrier()
+
+    # -- C1/C3/C4: delta exchange -------------------------------------------
+    def all_gather_flat(self, flat: torch.Tensor) -> torch.Tensor:
+        """Gather each rank's flat tensor -> [world, P] (every rank gets all
+        deltas, HBM-resident)."""
+        if not self.is_distributed:
+            return flat.unsqueeze(0)
+        out = torch.empty(self.world_size * flat.numel(), dtype=flat.dtype,
+                          device=flat.device)
+        dist.all_gather_into_tensor(out, flat.contiguous())
+        return out.view(self.world_size, flat.numel())
+
+    # -- C2/C5: base publication --------------------------------------------
+    def broadcast_flat(self, flat: torch.Tensor, src: int = 0) -> torch.Tensor:
+        if self.is_distributed:
+            dist.broadcast(flat, src=src)
+        return flat
+
+    # -- C6/C7: small metadata (scores, hashes, membership) ------------------
+    def all_gather_object(self, obj) -> List:
+        if not self.is_distributed:
+            return [obj]
+        out = [None] * self.world_size
+        dist.all_gather_object(out, obj)
+        return out
+
+    def broadcast_object(self, obj, src: int = 0):
+        if not self.is_distributed:
+            return obj
+        box = [obj if self.rank == src else None]
+        dist.broadcast_object_list(box, src=src)
+        return box[0]
+
+    def close(self) -> None:
+        if dist.is_initialized():
+            dist.destroy_process_group()

@@ -1,0 +1,113 @@
+"""Local-SGD node orchestrator: N miner ranks on one MI355X box.
+
+This is the framework's first-class parallelism: each of
+the node's GPUs runs one miner doing communication-sparse local SGD on a
+shared base; every merge interval the per-rank weight deltas are exchanged
+with ONE RCCL all-gather over the xGMI clique, merged (uniform,
+score-weighted or meta-learned weights), and the merged base is installed
+on every rank — replacing the reference's miner→HF-hub→averager→HF-hub→miner
+round trip with two collectives.
+
+Merge placement: the all-gather leaves ALL deltas resident on EVERY rank,
+so the uniform/score-weighted merges are computed redundantly by each rank
+(deterministic kernels ⇒ identical bases, no broadcast needed); the
+meta-learned merge runs on rank 0 (it needs val-loss backward passes) and
+is broadcast (C2/C5).
+"""
+
+from __future__ import annotations
+
+import logging
+from typing import Callable, Dict, List, Optional
+
+import torch
+
+from ..config import Config
+from ..parallel.comm import CommPlane
+from ..parallel.flat import FlatParams
+from ..roles.averager import ParameterizedAverager
+from ..roles.miner import DeltaLoop
+from ..roles.validator import DeltaValidator
+from ..store import DeltaCheckpoint
+
+log = logging.getLogger(__name__)
+
+
+class LocalSGDNode:
+    def __init__(self, model, fp: FlatParams, data_iter, cfg: Config,
+                 comm: CommPlane, val_batches: Optional[List[dict]] = None,
+                 merge_strategy: Optional[str] = None):
+        self.comm = comm
+        self.cfg = cfg
+        self.fp = fp
+        self.model = model
+        self.val_batches = val_batches or []
+        self.merge_strategy = merge_strategy or cfg.average.strategy
+        self.miner = DeltaLoop(model, fp, data_iter, cfg.train,
+                               hotkey=f"rank{comm.rank}")
+        self.averager = ParameterizedAverager(model, fp, cfg.average)
+        self.merge_rounds = 0
+
+    def sync_initial_base(self) -> None:
+        """Rank 0's random init becomes the shared base on all ranks."""
+        self.comm.broadcast_flat(self.fp.master, src=0)
+        self.fp.sync_work_from_master()
+        self.miner.install_base(self.fp.master)
+
+    def train_steps(self, n: int) -> float:
+        last = float("nan")
+        for _ in range(n):
+            last = self.miner.train_step()
+        return last
+
+    # -- the delta exchange + merge (C1..C5) ---------------------------------
+    def merge_round(self, scores: Optional[List[float]] = None) -> None:
+        base = self.miner.base
+        delta = self.fp.make_delta(base)     # fused θ−θ_base
+        deltas = self.comm.all_gather_flat(delta.flat)   # [world, P] resident
+        strat = self.merge_strategy
+        if strat in ("mean", "uniform"):
+            W = self.averager._uniform_weights(deltas.shape[0])
+            merged = self.averager.merged_from(base, deltas, W)
+        elif strat == "score_weighted":
+            assert scores is not None
+            merged = self.averager.score_weighted_merge(base, deltas, scores)
+        elif strat == "parameterized":
+            if self.comm.rank == 0:
+                merged = self.averager.meta_learning(base, deltas,
+                                                     self.val_batches)
+            else:
+                merged = torch.empty_like(base)
+            self.comm.broadcast_flat(merged, src=0)
+        else:
+            raise ValueError(f"unknown merge strategy {strat!r}")
+        self.miner.install_base(merged)
+        self.merge_rounds += 1
+
+    # -- optional distributed validation (BASELINE config #3) ----------------
+    def validation_round(self) -> Dict[str, float]:
+        """Every rank scores every delta on its own eval set shard; rank 0's
+        normalized scores win (kept simple: identical eval data ⇒ identical
+        scores)."""
+        base = self.miner.base
+        delta = self.fp.make_delta(base)
+        deltas = self.comm.all_gather_flat(delta.flat)
+        saved = self.fp.master.clone()
+        validator = DeltaValidator(self.model, self.fp, self.val_batches,
+                                   self.cfg.validate)
+        ckpts = {f"rank{i}": DeltaCheckpoint(deltas[i], self.fp.spec, "")
+                 for i in range(deltas.shape[0])}
+        scores = validator.validate_and_score(ckpts)
+        self.fp.master.copy_(saved)
+        self.fp.sync_work_from_master()
+        return scores
+
+    def run(self, total_steps: int, merge_every: int) -> None:
+        self.sync_initial_base()
+        done = 0
+        while done < total_steps:
+            n = min(merge_every, total_steps - done)
+            self.train_steps(n)
+            done += n
+            if done < total_steps or total_steps % merge_every == 0:
+                self.merge_round()

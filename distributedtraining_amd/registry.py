@@ -1,0 +1,190 @@
+"""Coordination plane: membership, address registry, scores.
+
+Replaces the reference's Bittensor chain stack
+(/root/reference/hivetrain/btt_connector.py + chain_manager.py) with the
+shape the reference itself uses for local runs (LocalBittensorNetwork,
+btt_connector.py:530-671; LocalAddressStore, chain_manager.py:124-168):
+
+* ``Registry`` — in-process: membership table (hotkey -> address), scores
+  with EMA folding (btt_connector.py:310-356), weight-set gating by epoch
+  length (btt_connector.py:382-385), MAD outlier scoring
+  (btt_connector.py:387-426).
+* ``FileRegistry`` — the same interface persisted as JSON so independent
+  miner/validator/averager *processes* share one registry (multi-node /
+  plumbing mode).
+
+In rccl mode, membership is simply the launcher-known world size; ranks map
+to hotkeys "rank{r}".
+"""
+
+from __future__ import annotations
+
+import json
+import os
+import tempfile
+import threading
+import time
+from typing import Dict, List, Optional
+
+import numpy as np
+
+
+def mad_outlier_scores(metrics: Dict[str, List[float]],
+                       threshold: float = 2.0) -> Dict[str, int]:
+    """Median-absolute-deviation outlier detection over per-member metrics.
+
+    Semantics follow the reference (btt_connector.py:387-426): a member whose
+    median metric deviates from the population median by more than
+    ``threshold`` normalized MADs scores 0, else 1.  Members whose deviation
+    is undefined (zero MAD) are treated as outliers only if they differ from
+    the median at all.
+    """
+    if not metrics:
+        return {}
+    med_per = {k: float(np.median(v)) for k, v in metrics.items()}
+    values = np.array(list(med_per.values()), dtype=np.float64)
+    median = float(np.median(values))
+    # scale='normal' equivalent: MAD / 0.67448975
+    mad = float(np.median(np.abs(values - median))) / 0.6744897501960817
+    out: Dict[str, int] = {}
+    for k, v in med_per.items():
+        if mad == 0.0:
+            is_out = v != median
+        else:
+            is_out = abs(v - median) / mad > threshold
+        out[k] = 0 if is_out else 1
+    return out
+
+
+class Registry:
+    """In-process membership + scores registry (thread-safe)."""
+
+    def __init__(self, epoch_length: int = 100, ema_alpha: float = 0.333333):
+        self._lock = threading.Lock()
+        self._addresses: Dict[str, str] = {}     # hotkey -> exchange address
+        self._scores: Dict[str, float] = {}      # EMA-folded scores
+        self._last_weight_set: int = -10**12      # "block" of last weight set
+        self._block0 = time.monotonic()
+        self.epoch_length = epoch_length
+        self.ema_alpha = ema_alpha
+        self._metrics: Dict[str, List[float]] = {}
+
+    # -- membership / address store (reference chain_manager.py:71-115) -----
+    def store_address(self, hotkey: str, address: str) -> None:
+        with self._lock:
+            self._addresses[hotkey] = address
+
+    def retrieve_address(self, hotkey: str) -> Optional[str]:
+        with self._lock:
+            return self._addresses.get(hotkey)
+
+    def deregister(self, hotkey: str) -> None:
+        with self._lock:
+            self._addresses.pop(hotkey, None)
+            self._scores.pop(hotkey, None)
+
+    @property
+    def hotkeys(self) -> List[str]:
+        with self._lock:
+            return list(self._addresses.keys())
+
+    # -- scores (reference btt_connector.py:310-356) -------------------------
+    def current_block(self) -> int:
+        # the reference counts 12 s chain blocks; we count seconds since start
+        return int(time.monotonic() - self._block0)
+
+    def should_set_weights(self) -> bool:
+        return (self.current_block() - self._last_weight_set) > self.epoch_length
+
+    def set_weights(self, scores: Dict[str, float]) -> Dict[str, float]:
+        """EMA-fold new scores into the registry, alpha=0.333 (T=5)."""
+        with self._lock:
+            a = self.ema_alpha
+            for hk in set(list(self._scores) + list(scores)):
+                prev = self._scores.get(hk, 0.0)
+                self._scores[hk] = a * scores.get(hk, 0.0) + (1 - a) * prev
+            self._last_weight_set = self.current_block()
+            return dict(self._scores)
+
+    def get_weights(self) -> Dict[str, float]:
+        with self._lock:
+            return dict(self._scores)
+
+    # -- anomaly metrics (reference btt_connector.py:387-426) ----------------
+    def report_metric(self, hotkey: str, value: float) -> None:
+        with self._lock:
+            self._metrics.setdefault(hotkey, []).append(float(value))
+
+    def detect_metric_anomaly(self, threshold: float = 2.0) -> Dict[str, int]:
+        with self._lock:
+            return mad_outlier_scores(self._metrics, threshold)
+
+
+class FileRegistry(Registry):
+    """Registry persisted to a JSON file; safe across processes.
+
+    Mirrors the reference's LocalAddressStore (chain_manager.py:124-168) and
+    LocalBittensorNetwork JSON metagraph (btt_connector.py:558-571), unified
+    behind one interface.
+    """
+
+    def __init__(self, root: str, epoch_length: int = 100,
+                 ema_alpha: float = 0.333333):
+        super().__init__(epoch_length, ema_alpha)
+        os.makedirs(root, exist_ok=True)
+        self.path = os.path.join(root, "registry.json")
+        self._load()
+
+    def _load(self) -> None:
+        if os.path.exists(self.path):
+            try:
+                with open(self.path) as f:
+                    d = json.load(f)
+                self._addresses = d.get("addresses", {})
+                self._scores = d.get("scores", {})
+                self._last_weight_set = d.get("last_weight_set", -10**12)
+            except (json.JSONDecodeError, OSError):
+                pass  # concurrent writer; keep current state
+
+    def _save(self) -> None:
+        d = {"addresses": self._addresses, "scores": self._scores,
+             "last_weight_set": self._last_weight_set}
+        fd, tmp = tempfile.mkstemp(dir=os.path.dirname(self.path))
+        with os.fdopen(fd, "w") as f:
+            json.dump(d, f)
+        os.replace(tmp, self.path)  # atomic on POSIX
+
+    def store_address(self, hotkey: str, address: str) -> None:
+        with self._lock:
+            self._load_unlocked()
+            self._addresses[hotkey] = address
+            self._save()
+
+    def _load_unlocked(self):
+        # helper used while already holding the lock
+        if os.path.exists(self.path):
+            try:
+                with open(self.path) as f:
+                    d = json.load(f)
+                self._addresses.update(d.get("addresses", {}))
+                for k, v in d.get("scores", {}).items():
+                    self._scores.setdefault(k, v)
+            except (json.JSONDecodeError, OSError):
+                pass
+
+    def retrieve_address(self, hotkey: str) -> Optional[str]:
+        with self._lock:
+            self._load_unlocked()
+            return self._addresses.get(hotkey)
+
+    @property
+    def hotkeys(self) -> List[str]:
+        with self._lock:
+            self._load_unlocked()
+            return list(self._addresses.keys())
+
+    def set_weights(self, scores: Dict[str, float]) -> Dict[str, float]:
+        out = super().set_weights(scores)
+        with self._lock:
+            self._save()
+        return out

@@ -1,0 +1,137 @@
+"""Miner role: local-SGD training with periodic weight-delta publication.
+
+Reimplements the reference's production miner path, DeltaLoop
+(/root/reference/hivetrain/training_manager.py:345-433):
+
+  * snapshot base weights on every base-model pull (:349-351, :374-377)
+  * per-step forward/backward/AdamW (:380-392)
+  * every send interval: weight_delta = θ − θ_base, push (:405-427)
+  * on new base: pull, reinit optimizer, re-snapshot (:361-378)
+  * gradient staleness metric (:156-168)
+
+MI355X-native differences: θ and θ_base are flat fp32 buffers (one fused
+delta kernel instead of a per-param Python dict loop), the push is a
+DeltaCheckpoint to a FileStore or an RCCL all-gather (parallel/local_sgd.py),
+and intervals are step-based for determinism.
+"""
+
+from __future__ import annotations
+
+import logging
+import math
+import time
+from typing import Callable, Iterable, Optional
+
+import torch
+
+from ..config import TrainConfig
+from ..parallel.flat import FlatParams, FusedAdamW
+from ..registry import Registry
+from ..store import DeltaCheckpoint, FileStore
+
+log = logging.getLogger(__name__)
+
+
+class DeltaLoop:
+    def __init__(self, model, fp: FlatParams, data_iter: Iterable,
+                 cfg: TrainConfig, store: Optional[FileStore] = None,
+                 registry: Optional[Registry] = None, hotkey: str = "miner0",
+                 on_push: Optional[Callable] = None):
+        self.model = model
+        self.fp = fp
+        self.data = iter(data_iter)
+        self.cfg = cfg
+        self.store = store
+        self.registry = registry
+        self.hotkey = hotkey
+        self.on_push = on_push
+        self.opt = FusedAdamW(fp, lr=cfg.lr, betas=tuple(cfg.betas),
+                              eps=cfg.eps, weight_decay=cfg.weight_decay)
+        self.step_count = 0
+        self.total_loss = 0.0
+        self.total_examples = 0
+        self.base = fp.snapshot()
+        self.base_hash = fp.master_hash()
+        self.last_push_step = 0
+        self.last_base_time = time.time()
+        if registry is not None and store is not None:
+            registry.store_address(hotkey, store.my_address())
+
+    # -- base refresh (reference :361-378) ----------------------------------
+    def maybe_pull_base(self) -> bool:
+        if self.store is None:
+            return False
+        if not self.store.check_for_new_model():
+            return False
+        sd = self.store.pull_model(map_location="cpu")
+        if sd is None or "flat_master" not in sd:
+            return False
+        self.install_base(sd["flat_master"])
+        log.info("%s: pulled new base model", self.hotkey)
+        return True
+
+    def install_base(self, flat_fp32: torch.Tensor) -> None:
+        """New shared base: load, reinit optimizer, re-snapshot
+        (reference deliberately re-creates optimizer state, :371-373)."""
+        self.fp.load_flat_master(flat_fp32)
+        self.opt.reset_state()
+        self.opt.zero_grad()
+        self.base = self.fp.snapshot()
+        self.base_hash = self.fp.master_hash()
+        self.last_base_time = time.time()
+
+    # -- training ------------------------------------------------------------
+    def train_step(self, batch=None) -> float:
+        if batch is None:
+            batch = next(self.data)
+        input_ids = batch["input_ids"].to(self.fp.device, non_blocking=True)
+        labels = batch.get("labels", batch["input_ids"]).to(
+            self.fp.device, non_blocking=True)
+        out = self.model(input_ids=input_ids, labels=labels)
+        out.loss.backward()
+        self.opt.step()
+        self.opt.zero_grad()
+        self.step_count += 1
+        loss = float(out.loss.detach())
+        self.total_loss += loss * input_ids.shape[0]
+        self.total_examples += input_ids.shape[0]
+        return loss
+
+    # -- delta publication (reference :405-427) ------------------------------
+    def make_delta(self) -> DeltaCheckpoint:
+        return self.fp.make_delta(self.base, step=self.step_count,
+                                  base_hash=self.base_hash)
+
+    def maybe_push_delta(self) -> Optional[DeltaCheckpoint]:
+        if self.step_count - self.last_push_step < self.cfg.send_interval_steps:
+            return None
+        ckpt = self.make_delta()
+        if self.store is not None:
+            self.store.push_delta(ckpt)
+        if self.on_push is not None:
+            self.on_push(ckpt)
+        self.last_push_step = self.step_count
+        return ckpt
+
+    def gradient_staleness(self) -> float:
+        """Seconds since last base refresh (reference metric, :156-168)."""
+        return time.time() - self.last_base_time
+
+    def average_loss(self) -> float:
+        if self.total_examples == 0:
+            return float("nan")
+        return self.total_loss / self.total_examples
+
+    def perplexity(self) -> float:
+        return math.exp(min(self.average_loss(), 20.0))
+
+    def train(self, steps: int) -> float:
+        """Run the outer loop: step, poll base, push delta."""
+        last = float("nan")
+        for _ in range(steps):
+            if (self.cfg.pull_interval_steps and
+                    self.step_count % self.cfg.pull_interval_steps == 0):
+                self.maybe_pull_base()
+            last = self.train_step()
+            self.maybe_push_delta()
+        return last

@@ -1,0 +1,51 @@
+import math
+
+import pytest
+
+from distributedtraining_amd.registry import (FileRegistry, Registry,
+                                              mad_outlier_scores)
+
+
+def test_address_roundtrip():
+    r = Registry()
+    r.store_address("hk1", "/tmp/a")
+    r.store_address("hk2", "/tmp/b")
+    assert r.retrieve_address("hk1") == "/tmp/a"
+    assert r.retrieve_address("missing") is None
+    assert set(r.hotkeys) == {"hk1", "hk2"}
+    r.deregister("hk1")
+    assert r.retrieve_address("hk1") is None
+
+
+def test_score_ema_alpha():
+    # reference semantics: s' = a*new + (1-a)*old, a=0.333 (btt_connector.py:317)
+    r = Registry(ema_alpha=0.333333)
+    r.store_address("hk", "x")
+    out1 = r.set_weights({"hk": 1.0})
+    assert out1["hk"] == pytest.approx(0.333333)
+    out2 = r.set_weights({"hk": 1.0})
+    assert out2["hk"] == pytest.approx(0.333333 + 0.666667 * 0.333333, rel=1e-4)
+
+
+def test_should_set_weights_gating():
+    r = Registry(epoch_length=10**9)
+    assert r.should_set_weights()  # never set before
+    r.set_weights({})
+    assert not r.should_set_weights()
+
+
+def test_mad_outliers():
+    metrics = {f"m{i}": [1.0 + 0.01 * i] for i in range(8)}
+    metrics["bad"] = [100.0]
+    scores = mad_outlier_scores(metrics, threshold=2.0)
+    assert scores["bad"] == 0
+    assert all(scores[f"m{i}"] == 1 for i in range(8))
+
+
+def test_file_registry_cross_instance(tmp_path):
+    a = FileRegistry(str(tmp_path))
+    a.store_address("hk1", "addr1")
+    b = FileRegistry(str(tmp_path))
+    assert b.retrieve_address("hk1") == "addr1"
+    b.store_address("hk2", "addr2")
+    assert set(a.hotkeys) == {"hk1", "hk2"}
