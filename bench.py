@@ -1,0 +1,145 @@
+#!/usr/bin/env python3
+"""Flagship benchmark: miner local-SGD training throughput (tokens/s,
+whole node) on GPT-2-small, synthetic data, random init — the BASELINE.json
+headline metric for this framework (the reference publishes no numbers;
+BASELINE.md pins only the workload: GPT-2-small, seq 64, AdamW,
+periodic weight-delta exchange + merge).
+
+Per-rank work is fixed as GPUs grow (weak scaling): each GPU runs one miner
+(BASELINE config #2); every --merge-every steps the timed region includes
+the full delta exchange (RCCL all-gather over xGMI) + uniform merge +
+base reinstall — far MORE often than the reference's 800 s send interval.
+
+Launch (driver contract):
+  python bench.py --gpus 1 --steps 30 --warmup 10
+  python -m torch.distributed.run --nnodes=1 --nproc-per-node 8 \
+      --master-addr 127.0.0.1 bench.py --gpus 8 --steps 30 --warmup 10
+"""
+
+import argparse
+import json
+import os
+import sys
+import time
+
+import torch
+
+
+def main() -> int:
+    p = argparse.ArgumentParser()
+    p.add_argument("--gpus", type=int, default=1)
+    p.add_argument("--steps", type=int, default=30)
+    p.add_argument("--warmup", type=int, default=10)
+    p.add_argument("--batch-size", type=int, default=256)
+    p.add_argument("--seq-len", type=int, default=64)
+    p.add_argument("--merge-every", type=int, default=25)
+    p.add_argument("--model", default="gpt2-small",
+                   choices=["gpt2-small", "gpt2-tiny", "llama3-8b"])
+    p.add_argument("--merge-strategy", default="mean")
+    p.add_argument("--cpu", action="store_true",
+                   help="CPU smoke mode (tiny model, tests only)")
+    args = p.parse_args()
+
+    from distributedtraining_amd.config import Config, ModelConfig
+    from distributedtraining_amd.models import build_model
+    from distributedtraining_amd.parallel.comm import CommPlane
+    from distributedtraining_amd.parallel.flat import FlatParams
+    from distributedtraining_amd.parallel.local_sgd import LocalSGDNode
+    from distributedtraining_amd.utils.data import synthetic_batches
+
+    cfg = Config()
+    if args.cpu or args.model == "gpt2-tiny":
+        cfg.model = ModelConfig.gpt2_tiny()
+        args.batch_size = min(args.batch_size, 4)
+        args.seq_len = min(args.seq_len, 32)
+    elif args.model == "llama3-8b":
+        cfg.model = ModelConfig.llama3_8b()
+    else:
+        cfg.model = ModelConfig.gpt2_small()
+    cfg.train.batch_size = args.batch_size
+    cfg.train.seq_len = args.seq_len
+    cfg.train.send_interval_steps = 10**9   # exchange via merge_round below
+    cfg.train.pull_interval_steps = 0
+
+    use_gpu = torch.cuda.is_available() and not args.cpu
+    if not use_gpu and not args.cpu:
+        print("ERROR: no GPU and --cpu not given", file=sys.stderr)
+        return 1
+
+    comm = CommPlane(device=None if use_gpu else torch.device("cpu"))
+    device = comm.device
+    rank, world = comm.rank, comm.world_size
+
+    torch.manual_seed(1234)           # same init everywhere; rank0 broadcast
+    model = build_model(cfg.model).to(device)
+    fp = FlatParams(model)
+    data = synthetic_batches(cfg.model.vocab_size, args.batch_size,
+                             args.seq_len, seed=1000 + rank)
+    node = LocalSGDNode(model, fp, data, cfg, comm,
+                        merge_strategy=args.merge_strategy)
+    node.sync_initial_base()
+
+    def sync():
+        if use_gpu:
+            torch.cuda.synchronize()
+        comm.barrier()
+
+    def run_steps(n: int, base_step: int) -> None:
+        for i in range(n):
+            node.miner.train_step()
+            step = base_step + i + 1
+            if args.merge_every and step % args.merge_every == 0:
+                node.merge_round()
+
+    # warmup (untimed) — includes one merge to warm the collective path
+    run_steps(args.warmup, 0)
+    if world > 1 or True:
+        node.merge_round()
+
+    sync()
+    t0 = time.perf_counter()
+    run_steps(args.steps, 0)
+    sync()
+    elapsed = time.perf_counter() - t0
+
+    # MAX over ranks
+    t = torch.tensor([elapsed], dtype=torch.float64, device=device
+                     if comm.backend != "gloo" else "cpu")
+    if comm.is_distributed:
+        torch.distributed.all_reduce(t, op=torch.distributed.ReduceOp.MAX)
+    elapsed = float(t.item())
+
+    tokens = args.steps * args.batch_size * args.seq_len * world
+    value = tokens / elapsed
+    if rank == 0:
+        out = {
+            "metric": "tokens/sec (whole node, all miners), GPT-2-small local-SGD",
+            "value": value,
+            "unit": "tokens/s",
+            "n_gpus": world,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": elapsed / args.steps * 1000.0,
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "bf16" if use_gpu else "fp32",
+            "data": "synthetic",
+            "config": {
+                "model": args.model,
+                "global_batch": args.batch_size * world,
+                "seq_len": args.seq_len,
+                "parallelism": f"local_sgd_dp{world}",
+                "merge_every": args.merge_every,
+                "merge_strategy": args.merge_strategy,
+                "optimizer": "fused AdamW (lr 5e-4, reference config)",
+                "final_train_loss": node.miner.average_loss(),
+            },
+        }
+        print(json.dumps(out))
+    comm.close()
+    return 0
+
+
+if __name__ == "__main__":
+    sys.exit(main())

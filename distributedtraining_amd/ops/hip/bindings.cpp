@@ -1,0 +1,368 @@
+// Torch glue for the gfx950 kernel library (_dta_hip extension).
+// All tensor-shape/dtype validation lives here; kernels get raw pointers.
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+
+#include <vector>
+
+#include "dta_kernels.h"
+
+namespace {
+
+using torch::Tensor;
+
+inline hipStream_t stream() {
+  return at::hip::getCurrentHIPStream().stream();
+}
+inline const bf16_t* bfp(const Tensor& t) {
+  return reinterpret_cast<const bf16_t*>(t.data_ptr());
+}
+inline bf16_t* bfp_mut(Tensor& t) {
+  return reinterpret_cast<bf16_t*>(t.data_ptr());
+}
+
+void check_bf16(const Tensor& t, const char* name) {
+  TORCH_CHECK(t.is_cuda(), name, " must be on GPU");
+  TORCH_CHECK(t.scalar_type() == torch::kBFloat16, name, " must be bf16");
+  TORCH_CHECK(t.is_contiguous(), name, " must be contiguous");
+}
+void check_f32(const Tensor& t, const char* name) {
+  TORCH_CHECK(t.is_cuda(), name, " must be on GPU");
+  TORCH_CHECK(t.scalar_type() == torch::kFloat32, name, " must be fp32");
+  TORCH_CHECK(t.is_contiguous(), name, " must be contiguous");
+}
+
+// ---- norms ----------------------------------------------------------------
+std::vector<Tensor> layernorm_fwd(Tensor x, Tensor w, Tensor b, double eps) {
+  check_bf16(x, "x"); check_bf16(w, "w"); check_bf16(b, "b");
+  const int64_t rows = x.size(0);
+  const int cols = int(x.size(1));
+  TORCH_CHECK(cols % 8 == 0, "cols must be a multiple of 8");
+  auto y = torch::empty_like(x);
+  auto mean = torch::empty({rows}, x.options().dtype(torch::kFloat32));
+  auto rstd = torch::empty({rows}, x.options().dtype(torch::kFloat32));
+  launch_layernorm_fwd(bfp(x), bfp(w), bfp(b), bfp_mut(y),
+                       mean.data_ptr<float>(), rstd.data_ptr<float>(), rows,
+                       cols, float(eps), stream());
+  return {y, mean, rstd};
+}
+
+std::vector<Tensor> layernorm_bwd(Tensor dy, Tensor x, Tensor w, Tensor mean,
+                                  Tensor rstd) {
+  check_bf16(dy, "dy"); check_bf16(x, "x"); check_bf16(w, "w");
+  const int64_t rows = x.size(0);
+  const int cols = int(x.size(1));
+  auto dx = torch::empty_like(x);
+  auto dw32 = torch::zeros({cols}, x.options().dtype(torch::kFloat32));
+  auto db32 = torch::zeros({cols}, x.options().dtype(torch::kFloat32));
+  launch_layernorm_bwd(bfp(dy), bfp(x), bfp(w), mean.data_ptr<float>(),
+                       rstd.data_ptr<float>(), bfp_mut(dx),
+                       dw32.data_ptr<float>(), db32.data_ptr<float>(), rows,
+                       cols, stream());
+  return {dx, dw32.to(torch::kBFloat16), db32.to(torch::kBFloat16)};
+}
+
+std::vector<Tensor> rmsnorm_fwd(Tensor x, Tensor w, double eps) {
+  check_bf16(x, "x"); check_bf16(w, "w");
+  const int64_t rows = x.size(0);
+  const int cols = int(x.size(1));
+  TORCH_CHECK(cols % 8 == 0, "cols must be a multiple of 8");
+  auto y = torch::empty_like(x);
+  auto rstd = torch::empty({rows}, x.options().dtype(torch::kFloat32));
+  launch_rmsnorm_fwd(bfp(x), bfp(w), bfp_mut(y), rstd.data_ptr<float>(),
+                     rows, cols, float(eps), stream());
+  return {y, rstd};
+}
+
+std::vector<Tensor> rmsnorm_bwd(Tensor dy, Tensor x, Tensor w, Tensor rstd) {
+  check_bf16(dy, "dy"); check_bf16(x, "x"); check_bf16(w, "w");
+  const int64_t rows = x.size(0);
+  const int cols = int(x.size(1));
+  auto dx = torch::empty_like(x);
+  auto dw32 = torch::zeros({cols}, x.options().dtype(torch::kFloat32));
+  launch_rmsnorm_bwd(bfp(dy), bfp(x), bfp(w), rstd.data_ptr<float>(),
+                     bfp_mut(dx), dw32.data_ptr<float>(), rows, cols,
+                     stream());
+  return {dx, dw32.to(torch::kBFloat16)};
+}
+
+// ---- elementwise ----------------------------------------------------------
+Tensor gelu_fwd(Tensor x) {
+  auto y = torch::empty_like(x);
+  if (x.scalar_type() == torch::kBFloat16) {
+    check_bf16(x, "x");
+    launch_gelu_fwd(bfp(x), bfp_mut(y), x.numel(), stream());
+  } else {
+    check_f32(x, "x");
+    launch_gelu_fwd_f32(x.data_ptr<float>(), y.data_ptr<float>(), x.numel(),
+                        stream());
+  }
+  return y;
+}
+Tensor gelu_bwd(Tensor dy, Tensor x) {
+  auto dx = torch::empty_like(x);
+  if (x.scalar_type() == torch::kBFloat16) {
+    check_bf16(x, "x"); check_bf16(dy, "dy");
+    launch_gelu_bwd(bfp(dy), bfp(x), bfp_mut(dx), x.numel(), stream());
+  } else {
+    launch_gelu_bwd_f32(dy.data_ptr<float>(), x.data_ptr<float>(),
+                        dx.data_ptr<float>(), x.numel(), stream());
+  }
+  return dx;
+}
+Tensor swiglu_fwd(Tensor g, Tensor u) {
+  check_bf16(g, "gate"); check_bf16(u, "up");
+  auto y = torch::empty_like(g);
+  launch_swiglu_fwd(bfp(g), bfp(u), bfp_mut(y), g.numel(), stream());
+  return y;
+}
+std::vector<Tensor> swiglu_bwd(Tensor dy, Tensor g, Tensor u) {
+  check_bf16(dy, "dy");
+  auto dg = torch::empty_like(g);
+  auto du = torch::empty_like(u);
+  launch_swiglu_bwd(bfp(dy), bfp(g), bfp(u), bfp_mut(dg), bfp_mut(du),
+                    g.numel(), stream());
+  return {dg, du};
+}
+
+// ---- flat plane ------------------------------------------------------------
+void delta_sub(Tensor w, Tensor base, Tensor out) {
+  check_f32(w, "w"); check_f32(base, "base"); check_f32(out, "out");
+  launch_delta_sub(w.data_ptr<float>(), base.data_ptr<float>(),
+                   out.data_ptr<float>(), w.numel(), stream());
+}
+void axpy(Tensor w, Tensor x, double alpha) {
+  check_f32(w, "w"); check_f32(x, "x");
+  launch_axpy(w.data_ptr<float>(), x.data_ptr<float>(), float(alpha),
+              w.numel(), stream());
+}
+bool has_nan(Tensor x) {
+  check_f32(x, "x");
+  auto flag = torch::zeros({1}, x.options().dtype(torch::kInt32));
+  launch_nan_any(x.data_ptr<float>(), x.numel(), flag.data_ptr<int>(),
+                 stream());
+  return flag.item<int>() != 0;
+}
+double l2norm_sq(Tensor x) {
+  check_f32(x, "x");
+  auto out = torch::zeros({1}, x.options());
+  launch_l2norm_sq(x.data_ptr<float>(), x.numel(), out.data_ptr<float>(),
+                   stream());
+  return out.item<float>();
+}
+
+void adamw_step(Tensor master, Tensor grad, Tensor m, Tensor v,
+                Tensor out_bf16, int64_t step, double lr, double beta1,
+                double beta2, double eps, double wd) {
+  check_f32(master, "master"); check_f32(m, "m"); check_f32(v, "v");
+  const bool gb = grad.scalar_type() == torch::kBFloat16;
+  bf16_t* outp = out_bf16.numel() ? bfp_mut(out_bf16) : nullptr;
+  launch_adamw(master.data_ptr<float>(), grad.data_ptr(), gb,
+               m.data_ptr<float>(), v.data_ptr<float>(), outp, int(step),
+               float(lr), float(beta1), float(beta2), float(eps), float(wd),
+               master.numel(), stream());
+}
+
+// ---- merge plane -----------------------------------------------------------
+void weighted_merge(Tensor base, Tensor deltas, Tensor W, Tensor offsets,
+                    Tensor out) {
+  check_f32(base, "base"); check_f32(deltas, "deltas"); check_f32(out, "out");
+  auto Wc = W.to(base.device(), torch::kFloat32).contiguous();
+  auto offs = offsets.to(base.device()).contiguous();
+  const int n_models = int(deltas.size(0));
+  const int n_segs = int(offsets.numel() - 1);
+  launch_weighted_merge(base.data_ptr<float>(), deltas.data_ptr<float>(),
+                        Wc.data_ptr<float>(), offs.data_ptr<int64_t>(),
+                        n_models, n_segs, base.numel(),
+                        out.data_ptr<float>(), stream());
+}
+
+Tensor grad_merge_weights(Tensor g, Tensor base, Tensor deltas, Tensor merged,
+                          Tensor offsets) {
+  check_f32(base, "base"); check_f32(deltas, "deltas");
+  const int n_models = int(deltas.size(0));
+  const int n_segs = int(offsets.numel() - 1);
+  const int64_t P = base.numel();
+  // build boundary-aligned chunks host-side (<=1M elements each)
+  auto offs_cpu = offsets.to(torch::kCPU).contiguous();
+  const int64_t* op = offs_cpu.data_ptr<int64_t>();
+  std::vector<int64_t> chunks;
+  constexpr int64_t CH = 1 << 20;
+  for (int j = 0; j < n_segs; ++j)
+    for (int64_t s0 = op[j]; s0 < op[j + 1]; s0 += CH) {
+      chunks.push_back(s0);
+      chunks.push_back(std::min(s0 + CH, op[j + 1]));
+      chunks.push_back(j);
+    }
+  const int n_chunks = int(chunks.size() / 3);
+  auto ch = torch::from_blob(chunks.data(), {int64_t(chunks.size())},
+                             torch::kInt64)
+                .to(base.device());
+  auto gw = torch::zeros({n_models, n_segs},
+                         base.options().dtype(torch::kFloat32));
+  const bool gb = g.scalar_type() == torch::kBFloat16;
+  launch_grad_merge_weights(g.data_ptr(), gb, base.data_ptr<float>(),
+                            deltas.data_ptr<float>(),
+                            merged.data_ptr<float>(),
+                            ch.data_ptr<int64_t>(), n_models, n_chunks, P,
+                            gw.data_ptr<float>(), n_segs, stream());
+  return gw;
+}
+
+// ---- CE --------------------------------------------------------------------
+std::vector<Tensor> ce_fwd(Tensor logits, Tensor targets,
+                           int64_t ignore_index) {
+  check_bf16(logits, "logits");
+  TORCH_CHECK(targets.scalar_type() == torch::kInt64, "targets int64");
+  const int64_t rows = logits.size(0), vocab = logits.size(1);
+  auto lse = torch::empty({rows}, logits.options().dtype(torch::kFloat32));
+  auto loss = torch::zeros({1}, logits.options().dtype(torch::kFloat32));
+  auto count = torch::zeros({1}, logits.options().dtype(torch::kInt32));
+  launch_ce_fwd(bfp(logits), targets.data_ptr<int64_t>(), rows, vocab,
+                ignore_index, lse.data_ptr<float>(), loss.data_ptr<float>(),
+                count.data_ptr<int>(), stream());
+  return {loss.squeeze(0), lse, count.squeeze(0)};
+}
+
+Tensor ce_bwd(Tensor logits, Tensor targets, Tensor lse, double scale,
+              int64_t ignore_index) {
+  check_bf16(logits, "logits");
+  const int64_t rows = logits.size(0), vocab = logits.size(1);
+  auto dl = torch::empty_like(logits);
+  launch_ce_bwd(bfp(logits), targets.data_ptr<int64_t>(),
+                lse.data_ptr<float>(), float(scale), ignore_index,
+                bfp_mut(dl), rows, vocab, stream());
+  return dl;
+}
+
+// ---- embedding -------------------------------------------------------------
+Tensor embedding_fwd(Tensor ids, Tensor wte, Tensor wpe) {
+  check_bf16(wte, "wte");
+  TORCH_CHECK(ids.scalar_type() == torch::kInt64, "ids int64");
+  const bool has_wpe = wpe.numel() > 0;
+  const int dim = int(wte.size(1));
+  const int seq = int(ids.size(-1));
+  const int64_t n_tok = ids.numel();
+  auto out_sizes = ids.sizes().vec();
+  out_sizes.push_back(dim);
+  auto out = torch::empty(out_sizes, wte.options());
+  launch_embedding_fwd(ids.data_ptr<int64_t>(), bfp(wte),
+                       has_wpe ? bfp(wpe) : nullptr, bfp_mut(out), n_tok,
+                       seq, dim, has_wpe, stream());
+  return out;
+}
+
+std::vector<Tensor> embedding_bwd(Tensor dy, Tensor ids, int64_t vocab,
+                                  int64_t npos) {
+  check_bf16(dy, "dy");
+  const int dim = int(dy.size(-1));
+  const int seq = int(ids.size(-1));
+  const int64_t n_tok = ids.numel();
+  auto f32 = dy.options().dtype(torch::kFloat32);
+  auto dwte = torch::zeros({vocab, dim}, f32);
+  auto dwpe = npos ? torch::zeros({npos, dim}, f32) : torch::zeros({0}, f32);
+  launch_embedding_bwd(bfp(dy), ids.data_ptr<int64_t>(),
+                       dwte.data_ptr<float>(),
+                       npos ? dwpe.data_ptr<float>() : nullptr, n_tok, seq,
+                       dim, npos > 0, stream());
+  return {dwte.to(torch::kBFloat16),
+          npos ? dwpe.to(torch::kBFloat16) : dwpe};
+}
+
+// ---- rope ------------------------------------------------------------------
+Tensor rope_apply(Tensor x, Tensor cos_t, Tensor sin_t, bool backward) {
+  check_bf16(x, "x");
+  // x: [B,H,S,D] or [BH,S,D]
+  const int hd = int(x.size(-1));
+  const int seq = int(x.size(-2));
+  const int64_t bh = x.numel() / (int64_t(seq) * hd);
+  auto y = torch::empty_like(x);
+  auto c = cos_t.contiguous();
+  auto sn = sin_t.contiguous();
+  launch_rope(bfp(x), c.data_ptr<float>(), sn.data_ptr<float>(), bfp_mut(y),
+              bh, seq, hd, backward, stream());
+  return y;
+}
+Tensor rope_fwd(Tensor x, Tensor c, Tensor s) { return rope_apply(x, c, s, false); }
+Tensor rope_bwd(Tensor x, Tensor c, Tensor s) { return rope_apply(x, c, s, true); }
+
+// ---- attention -------------------------------------------------------------
+std::vector<Tensor> attn_fwd(Tensor q, Tensor k, Tensor v, double scale) {
+  check_bf16(q, "q"); check_bf16(k, "k"); check_bf16(v, "v");
+  const int hd = int(q.size(-1));
+  const int seq = int(q.size(-2));
+  const int64_t bh = q.numel() / (int64_t(seq) * hd);
+  TORCH_CHECK(hd == 32 || hd == 64 || hd == 128, "head dim must be 32/64/128");
+  TORCH_CHECK(seq % 16 == 0, "seq must be a multiple of 16");
+  auto o = torch::empty_like(q);
+  auto lse = torch::empty({bh, seq}, q.options().dtype(torch::kFloat32));
+  launch_attn_fwd(bfp(q), bfp(k), bfp(v), bfp_mut(o),
+                  lse.data_ptr<float>(), bh, seq, hd, float(scale), stream());
+  return {o, lse};
+}
+
+std::vector<Tensor> attn_bwd(Tensor dout, Tensor q, Tensor k, Tensor v,
+                             Tensor o, Tensor lse, double scale) {
+  check_bf16(dout, "dout");
+  const int hd = int(q.size(-1));
+  const int seq = int(q.size(-2));
+  const int64_t bh = q.numel() / (int64_t(seq) * hd);
+  auto delta = torch::empty({bh, seq}, q.options().dtype(torch::kFloat32));
+  launch_attn_delta(bfp(dout), bfp(o), delta.data_ptr<float>(), bh * seq, hd,
+                    stream());
+  auto dq = torch::empty_like(q);
+  auto dk = torch::empty_like(k);
+  auto dv = torch::empty_like(v);
+  launch_attn_bwd_dq(bfp(dout), bfp(q), bfp(k), bfp(v),
+                     lse.data_ptr<float>(), delta.data_ptr<float>(),
+                     bfp_mut(dq), bh, seq, hd, float(scale), stream());
+  launch_attn_bwd_dkv(bfp(dout), bfp(q), bfp(k), bfp(v),
+                      lse.data_ptr<float>(), delta.data_ptr<float>(),
+                      bfp_mut(dk), bfp_mut(dv), bh, seq, hd, float(scale),
+                      stream());
+  return {dq, dk, dv};
+}
+
+// ---- mfma self-test --------------------------------------------------------
+Tensor mfma_selftest_16(Tensor A, Tensor B) {
+  check_bf16(A, "A"); check_bf16(B, "B");
+  auto D = torch::zeros({16, 16}, A.options().dtype(torch::kFloat32));
+  launch_mfma_probe_16(bfp(A), bfp(B), D.data_ptr<float>(), stream());
+  return D;
+}
+Tensor mfma_selftest_32(Tensor A, Tensor B) {
+  check_bf16(A, "A"); check_bf16(B, "B");
+  auto D = torch::zeros({32, 32}, A.options().dtype(torch::kFloat32));
+  launch_mfma_probe_32(bfp(A), bfp(B), D.data_ptr<float>(), stream());
+  return D;
+}
+
+}  // namespace
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("layernorm_fwd", &layernorm_fwd);
+  m.def("layernorm_bwd", &layernorm_bwd);
+  m.def("rmsnorm_fwd", &rmsnorm_fwd);
+  m.def("rmsnorm_bwd", &rmsnorm_bwd);
+  m.def("gelu_fwd", &gelu_fwd);
+  m.def("gelu_bwd", &gelu_bwd);
+  m.def("swiglu_fwd", &swiglu_fwd);
+  m.def("swiglu_bwd", &swiglu_bwd);
+  m.def("delta_sub", &delta_sub);
+  m.def("axpy", &axpy);
+  m.def("has_nan", &has_nan);
+  m.def("l2norm_sq", &l2norm_sq);
+  m.def("adamw_step", &adamw_step);
+  m.def("weighted_merge", &weighted_merge);
+  m.def("grad_merge_weights", &grad_merge_weights);
+  m.def("ce_fwd", &ce_fwd);
+  m.def("ce_bwd", &ce_bwd);
+  m.def("embedding_fwd", &embedding_fwd);
+  m.def("embedding_bwd", &embedding_bwd);
+  m.def("rope_fwd", &rope_fwd);
+  m.def("rope_bwd", &rope_bwd);
+  m.def("attn_fwd", &attn_fwd);
+  m.def("attn_bwd", &attn_bwd);
+  m.def("mfma_selftest_16", &mfma_selftest_16);
+  m.def("mfma_selftest_32", &mfma_selftest_32);
+}

@@ -1,0 +1,105 @@
+// Launcher declarations for the gfx950 kernel library.
+// Raw-pointer interfaces; the torch glue lives in bindings.cpp only.
+#pragma once
+
+#include <hip/hip_runtime.h>
+#include <cstdint>
+
+using bf16_t = unsigned short;  // raw bf16 bits at the ABI boundary
+
+// ---- elementwise ----------------------------------------------------------
+void launch_gelu_fwd(const bf16_t* x, bf16_t* y, int64_t n, hipStream_t s);
+void launch_gelu_bwd(const bf16_t* dy, const bf16_t* x, bf16_t* dx, int64_t n,
+                     hipStream_t s);
+void launch_gelu_fwd_f32(const float* x, float* y, int64_t n, hipStream_t s);
+void launch_gelu_bwd_f32(const float* dy, const float* x, float* dx,
+                         int64_t n, hipStream_t s);
+void launch_swiglu_fwd(const bf16_t* g, const bf16_t* u, bf16_t* y, int64_t n,
+                       hipStream_t s);
+void launch_swiglu_bwd(const bf16_t* dy, const bf16_t* g, const bf16_t* u,
+                       bf16_t* dg, bf16_t* du, int64_t n, hipStream_t s);
+
+// ---- flat-plane ops -------------------------------------------------------
+void launch_delta_sub(const float* w, const float* base, float* out,
+                      int64_t n, hipStream_t s);
+void launch_axpy(float* w, const float* x, float alpha, int64_t n,
+                 hipStream_t s);
+void launch_nan_any(const float* x, int64_t n, int* flag, hipStream_t s);
+void launch_l2norm_sq(const float* x, int64_t n, float* out, hipStream_t s);
+void launch_adamw(float* master, const void* grad, bool grad_is_bf16,
+                  float* m, float* v, bf16_t* out_bf16, int step, float lr,
+                  float beta1, float beta2, float eps, float wd, int64_t n,
+                  hipStream_t s);
+
+// ---- merge plane ----------------------------------------------------------
+void launch_weighted_merge(const float* base, const float* deltas,  // [N,P]
+                           const float* W,                          // [N,S]
+                           const int64_t* offsets, int n_models, int n_segs,
+                           int64_t P, float* out, hipStream_t s);
+// chunks: [n_chunks,3] int64 (start,end,seg) rows, none crossing a segment
+void launch_grad_merge_weights(const void* g, bool g_is_bf16,
+                               const float* base, const float* deltas,
+                               const float* merged, const int64_t* chunks,
+                               int n_models, int n_chunks, int64_t P,
+                               float* grad_w,  // [N,S] pre-zeroed
+                               int n_segs, hipStream_t s);
+
+// ---- norms ----------------------------------------------------------------
+void launch_layernorm_fwd(const bf16_t* x, const bf16_t* w, const bf16_t* b,
+                          bf16_t* y, float* mean, float* rstd, int64_t rows,
+                          int cols, float eps, hipStream_t s);
+void launch_layernorm_bwd(const bf16_t* dy, const bf16_t* x, const bf16_t* w,
+                          const float* mean, const float* rstd, bf16_t* dx,
+                          float* dw, float* db,  // fp32, pre-zeroed
+                          int64_t rows, int cols, hipStream_t s);
+void launch_rmsnorm_fwd(const bf16_t* x, const bf16_t* w, bf16_t* y,
+                        float* rstd, int64_t rows, int cols, float eps,
+                        hipStream_t s);
+void launch_rmsnorm_bwd(const bf16_t* dy, const bf16_t* x, const bf16_t* w,
+                        const float* rstd, bf16_t* dx, float* dw,
+                        int64_t rows, int cols, hipStream_t s);
+
+// ---- fused cross entropy --------------------------------------------------
+void launch_ce_fwd(const bf16_t* logits, const int64_t* targets,
+                   int64_t rows, int64_t vocab, int64_t ignore_index,
+                   float* lse, float* loss_sum, int* count, hipStream_t s);
+void launch_ce_bwd(const bf16_t* logits, const int64_t* targets,
+                   const float* lse, float scale, int64_t ignore_index,
+                   bf16_t* dlogits, int64_t rows, int64_t vocab,
+                   hipStream_t s);
+
+// ---- embedding ------------------------------------------------------------
+void launch_embedding_fwd(const int64_t* ids, const bf16_t* wte,
+                          const bf16_t* wpe, bf16_t* out, int64_t n_tok,
+                          int seq_len, int dim, bool has_wpe, hipStream_t s);
+void launch_embedding_bwd(const bf16_t* dy, const int64_t* ids,
+                          float* dwte_f32, float* dwpe_f32, int64_t n_tok,
+                          int seq_len, int dim, bool has_wpe, hipStream_t s);
+void launch_f32_to_bf16(const float* x, bf16_t* y, int64_t n, hipStream_t s);
+
+// ---- rope -----------------------------------------------------------------
+void launch_rope(const bf16_t* x, const float* cos_t, const float* sin_t,
+                 bf16_t* y, int64_t bh, int seq, int hd, bool backward,
+                 hipStream_t s);
+
+// ---- attention ------------------------------------------------------------
+void launch_attn_fwd(const bf16_t* q, const bf16_t* k, const bf16_t* v,
+                     bf16_t* o, float* lse, int64_t bh, int seq, int hd,
+                     float scale, hipStream_t s);
+void launch_attn_delta(const bf16_t* dout, const bf16_t* o, float* delta,
+                       int64_t rows, int hd, hipStream_t s);
+void launch_attn_bwd_dq(const bf16_t* dout, const bf16_t* q, const bf16_t* k,
+                        const bf16_t* v, const float* lse, const float* delta,
+                        bf16_t* dq, int64_t bh, int seq, int hd, float scale,
+                        hipStream_t s);
+void launch_attn_bwd_dkv(const bf16_t* dout, const bf16_t* q, const bf16_t* k,
+                         const bf16_t* v, const float* lse, const float* delta,
+                         bf16_t* dk, bf16_t* dv, int64_t bh, int seq, int hd,
+                         float scale, hipStream_t s);
+
+// ---- mfma layout self-test ------------------------------------------------
+// D[32,32] = A[32,16] x B[16,32] and D[16,16] = A[16,32] x B[32,16]
+void launch_mfma_probe_32(const bf16_t* A, const bf16_t* B, float* D,
+                          hipStream_t s);
+void launch_mfma_probe_16(const bf16_t* A, const bf16_t* B, float* D,
+                          hipStream_t s);

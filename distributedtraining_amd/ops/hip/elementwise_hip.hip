@@ -1,0 +1,264 @@
+#include "hip/hip_runtime.h"
+// Elementwise kernels: GELU (tanh approx), SwiGLU, flat-plane delta ops.
+// All memory-bound: vectorized bf16x8 / float4 loads (Guideline 13),
+// grid-stride with capped grid (Guideline 11).
+#include "dta_common.h"
+#include "dta_kernels.h"
+
+namespace {
+
+constexpr float GELU_C = 0.7978845608028654f;   // sqrt(2/pi)
+constexpr float GELU_A = 0.044715f;
+
+DEV float gelu_f(float x) {
+  float u = GELU_C * (x + GELU_A * x * x * x);
+  return 0.5f * x * (1.0f + tanhf(u));
+}
+DEV float gelu_df(float x) {
+  float x2 = x * x;
+  float u = GELU_C * x * (1.0f + GELU_A * x2);
+  float t = tanhf(u);
+  float sech2 = 1.0f - t * t;
+  return 0.5f * (1.0f + t) + 0.5f * x * sech2 * GELU_C * (1.0f + 3.0f * GELU_A * x2);
+}
+DEV float silu_f(float x) { return x / (1.0f + expf(-x)); }
+DEV float silu_df(float x) {
+  float s = 1.0f / (1.0f + expf(-x));
+  return s * (1.0f + x * (1.0f - s));
+}
+
+// -- generic vectorized 1-in-1-out bf16 map ---------------------------------
+template <float (*F)(float)>
+__global__ void map_bf16_k(const ushort* __restrict__ x, ushort* __restrict__ y,
+                           int64_t n) {
+  int64_t i = (int64_t(blockIdx.x) * blockDim.x + threadIdx.x) * 8;
+  const int64_t stride = int64_t(gridDim.x) * blockDim.x * 8;
+  for (; i + 8 <= n; i += stride) {
+    s16x8 vx = *reinterpret_cast<const s16x8*>(x + i);
+    s16x8 vy;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) vy[j] = f2bf(F(bf2f(ushort(vx[j]))));
+    *reinterpret_cast<s16x8*>(y + i) = vy;
+  }
+  if (i < n && i + 8 > n)
+    for (; i < n; ++i) y[i] = f2bf(F(bf2f(x[i])));
+}
+
+template <float (*F)(float)>
+__global__ void map_f32_k(const float* __restrict__ x, float* __restrict__ y,
+                          int64_t n) {
+  int64_t i = (int64_t(blockIdx.x) * blockDim.x + threadIdx.x) * 4;
+  const int64_t stride = int64_t(gridDim.x) * blockDim.x * 4;
+  for (; i + 4 <= n; i += stride) {
+    f32x4 vx = *reinterpret_cast<const f32x4*>(x + i);
+    f32x4 vy;
+#pragma unroll
+    for (int j = 0; j < 4; ++j) vy[j] = F(vx[j]);
+    *reinterpret_cast<f32x4*>(y + i) = vy;
+  }
+  if (i < n && i + 4 > n)
+    for (; i < n; ++i) y[i] = F(x[i]);
+}
+
+// dx = dy * f'(x)
+template <float (*DF)(float)>
+__global__ void map_grad_bf16_k(const ushort* __restrict__ dy,
+                                const ushort* __restrict__ x,
+                                ushort* __restrict__ dx, int64_t n) {
+  int64_t i = (int64_t(blockIdx.x) * blockDim.x + threadIdx.x) * 8;
+  const int64_t stride = int64_t(gridDim.x) * blockDim.x * 8;
+  for (; i + 8 <= n; i += stride) {
+    s16x8 vdy = *reinterpret_cast<const s16x8*>(dy + i);
+    s16x8 vx = *reinterpret_cast<const s16x8*>(x + i);
+    s16x8 o;
+#pragma unroll
+    for (int j = 0; j < 8; ++j)
+      o[j] = f2bf(bf2f(ushort(vdy[j])) * DF(bf2f(ushort(vx[j]))));
+    *reinterpret_cast<s16x8*>(dx + i) = o;
+  }
+  if (i < n && i + 8 > n)
+    for (; i < n; ++i) dx[i] = f2bf(bf2f(dy[i]) * DF(bf2f(x[i])));
+}
+
+template <float (*DF)(float)>
+__global__ void map_grad_f32_k(const float* __restrict__ dy,
+                               const float* __restrict__ x,
+                               float* __restrict__ dx, int64_t n) {
+  int64_t i = int64_t(blockIdx.x) * blockDim.x + threadIdx.x;
+  const int64_t stride = int64_t(gridDim.x) * blockDim.x;
+  for (; i < n; i += stride) dx[i] = dy[i] * DF(x[i]);
+}
+
+__global__ void swiglu_fwd_k(const ushort* __restrict__ g,
+                             const ushort* __restrict__ u,
+                             ushort* __restrict__ y, int64_t n) {
+  int64_t i = (int64_t(blockIdx.x) * blockDim.x + threadIdx.x) * 8;
+  const int64_t stride = int64_t(gridDim.x) * blockDim.x * 8;
+  for (; i + 8 <= n; i += stride) {
+    s16x8 vg = *reinterpret_cast<const s16x8*>(g + i);
+    s16x8 vu = *reinterpret_cast<const s16x8*>(u + i);
+    s16x8 o;
+#pragma unroll
+    for (int j = 0; j < 8; ++j)
+      o[j] = f2bf(silu_f(bf2f(ushort(vg[j]))) * bf2f(ushort(vu[j])));
+    *reinterpret_cast<s16x8*>(y + i) = o;
+  }
+  if (i < n && i + 8 > n)
+    for (; i < n; ++i) y[i] = f2bf(silu_f(bf2f(g[i])) * bf2f(u[i]));
+}
+
+__global__ void swiglu_bwd_k(const ushort* __restrict__ dy,
+                             const ushort* __restrict__ g,
+                             const ushort* __restrict__ u,
+                             ushort* __restrict__ dg, ushort* __restrict__ du,
+                             int64_t n) {
+  int64_t i = (int64_t(blockIdx.x) * blockDim.x + threadIdx.x) * 8;
+  const int64_t stride = int64_t(gridDim.x) * blockDim.x * 8;
+  for (; i + 8 <= n; i += stride) {
+    s16x8 vdy = *reinterpret_cast<const s16x8*>(dy + i);
+    s16x8 vg = *reinterpret_cast<const s16x8*>(g + i);
+    s16x8 vu = *reinterpret_cast<const s16x8*>(u + i);
+    s16x8 og, ou;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      float fdy = bf2f(ushort(vdy[j])), fg = bf2f(ushort(vg[j])),
+            fu = bf2f(ushort(vu[j]));
+      og[j] = f2bf(fdy * fu * silu_df(fg));
+      ou[j] = f2bf(fdy * silu_f(fg));
+    }
+    *reinterpret_cast<s16x8*>(dg + i) = og;
+    *reinterpret_cast<s16x8*>(du + i) = ou;
+  }
+  if (i < n && i + 8 > n)
+    for (; i < n; ++i) {
+      float fdy = bf2f(dy[i]), fg = bf2f(g[i]), fu = bf2f(u[i]);
+      dg[i] = f2bf(fdy * fu * silu_df(fg));
+      du[i] = f2bf(fdy * silu_f(fg));
+    }
+}
+
+// -- flat-plane: delta = w - base; w += a*x; nan scan; sq-norm --------------
+__global__ void delta_sub_k(const float* __restrict__ w,
+                            const float* __restrict__ base,
+                            float* __restrict__ out, int64_t n) {
+  int64_t i = (int64_t(blockIdx.x) * blockDim.x + threadIdx.x) * 4;
+  const int64_t stride = int64_t(gridDim.x) * blockDim.x * 4;
+  for (; i + 4 <= n; i += stride) {
+    f32x4 a = *reinterpret_cast<const f32x4*>(w + i);
+    f32x4 b = *reinterpret_cast<const f32x4*>(base + i);
+    f32x4 o = {a[0] - b[0], a[1] - b[1], a[2] - b[2], a[3] - b[3]};
+    *reinterpret_cast<f32x4*>(out + i) = o;
+  }
+  if (i < n && i + 4 > n)
+    for (; i < n; ++i) out[i] = w[i] - base[i];
+}
+
+__global__ void axpy_k(float* __restrict__ w, const float* __restrict__ x,
+                       float alpha, int64_t n) {
+  int64_t i = (int64_t(blockIdx.x) * blockDim.x + threadIdx.x) * 4;
+  const int64_t stride = int64_t(gridDim.x) * blockDim.x * 4;
+  for (; i + 4 <= n; i += stride) {
+    f32x4 a = *reinterpret_cast<const f32x4*>(w + i);
+    f32x4 b = *reinterpret_cast<const f32x4*>(x + i);
+    f32x4 o = {fmaf(alpha, b[0], a[0]), fmaf(alpha, b[1], a[1]),
+               fmaf(alpha, b[2], a[2]), fmaf(alpha, b[3], a[3])};
+    *reinterpret_cast<f32x4*>(w + i) = o;
+  }
+  if (i < n && i + 4 > n)
+    for (; i < n; ++i) w[i] = fmaf(alpha, x[i], w[i]);
+}
+
+__global__ void nan_any_k(const float* __restrict__ x, int64_t n,
+                          int* __restrict__ flag) {
+  int64_t i = (int64_t(blockIdx.x) * blockDim.x + threadIdx.x) * 4;
+  const int64_t stride = int64_t(gridDim.x) * blockDim.x * 4;
+  bool bad = false;
+  for (; i + 4 <= n; i += stride) {
+    f32x4 a = *reinterpret_cast<const f32x4*>(x + i);
+    bad |= (isnan(a[0]) || isnan(a[1]) || isnan(a[2]) || isnan(a[3]));
+  }
+  if (i < n && i + 4 > n)
+    for (; i < n; ++i) bad |= isnan(x[i]);
+  if (__any(bad) && (threadIdx.x & 63) == 0) atomicOr(flag, 1);
+}
+
+__global__ void l2norm_sq_k(const float* __restrict__ x, int64_t n,
+                            float* __restrict__ out) {
+  __shared__ float lds[16];
+  int64_t i = (int64_t(blockIdx.x) * blockDim.x + threadIdx.x) * 4;
+  const int64_t stride = int64_t(gridDim.x) * blockDim.x * 4;
+  float acc = 0.f;
+  for (; i + 4 <= n; i += stride) {
+    f32x4 a = *reinterpret_cast<const f32x4*>(x + i);
+    acc = fmaf(a[0], a[0], acc); acc = fmaf(a[1], a[1], acc);
+    acc = fmaf(a[2], a[2], acc); acc = fmaf(a[3], a[3], acc);
+  }
+  if (i < n && i + 4 > n)
+    for (; i < n; ++i) acc = fmaf(x[i], x[i], acc);
+  acc = block_sum<16>(acc, lds);
+  if (threadIdx.x == 0) atomicAdd(out, acc);
+}
+
+__global__ void f32_to_bf16_k(const float* __restrict__ x,
+                              ushort* __restrict__ y, int64_t n) {
+  int64_t i = (int64_t(blockIdx.x) * blockDim.x + threadIdx.x) * 8;
+  const int64_t stride = int64_t(gridDim.x) * blockDim.x * 8;
+  for (; i + 8 <= n; i += stride) {
+    f32x4 a = *reinterpret_cast<const f32x4*>(x + i);
+    f32x4 b = *reinterpret_cast<const f32x4*>(x + i + 4);
+    s16x8 o;
+#pragma unroll
+    for (int j = 0; j < 4; ++j) { o[j] = f2bf(a[j]); o[4 + j] = f2bf(b[j]); }
+    *reinterpret_cast<s16x8*>(y + i) = o;
+  }
+  if (i < n && i + 8 > n)
+    for (; i < n; ++i) y[i] = f2bf(x[i]);
+}
+
+}  // namespace
+
+#define LAUNCH_EW(kernel, n, ...)                                          \
+  do {                                                                     \
+    const int block_ = 256;                                                \
+   hipLaunchKernelGGL(( kernel), dim3(elementwise_grid(n, block_, 8)), dim3(block_), 0, s, __VA_ARGS__); \
+  } while (0)
+
+void launch_gelu_fwd(const bf16_t* x, bf16_t* y, int64_t n, hipStream_t s) {
+  LAUNCH_EW(map_bf16_k<gelu_f>, n, x, y, n);
+}
+void launch_gelu_bwd(const bf16_t* dy, const bf16_t* x, bf16_t* dx, int64_t n,
+                     hipStream_t s) {
+  LAUNCH_EW(map_grad_bf16_k<gelu_df>, n, dy, x, dx, n);
+}
+void launch_gelu_fwd_f32(const float* x, float* y, int64_t n, hipStream_t s) {
+  LAUNCH_EW(map_f32_k<gelu_f>, n, x, y, n);
+}
+void launch_gelu_bwd_f32(const float* dy, const float* x, float* dx,
+                         int64_t n, hipStream_t s) {
+  LAUNCH_EW(map_grad_f32_k<gelu_df>, n, dy, x, dx, n);
+}
+void launch_swiglu_fwd(const bf16_t* g, const bf16_t* u, bf16_t* y, int64_t n,
+                       hipStream_t s) {
+  LAUNCH_EW(swiglu_fwd_k, n, g, u, y, n);
+}
+void launch_swiglu_bwd(const bf16_t* dy, const bf16_t* g, const bf16_t* u,
+                       bf16_t* dg, bf16_t* du, int64_t n, hipStream_t s) {
+  LAUNCH_EW(swiglu_bwd_k, n, dy, g, u, dg, du, n);
+}
+void launch_delta_sub(const float* w, const float* base, float* out,
+                      int64_t n, hipStream_t s) {
+  LAUNCH_EW(delta_sub_k, n, w, base, out, n);
+}
+void launch_axpy(float* w, const float* x, float alpha, int64_t n,
+                 hipStream_t s) {
+  LAUNCH_EW(axpy_k, n, w, x, alpha, n);
+}
+void launch_nan_any(const float* x, int64_t n, int* flag, hipStream_t s) {
+  LAUNCH_EW(nan_any_k, n, x, n, flag);
+}
+void launch_l2norm_sq(const float* x, int64_t n, float* out, hipStream_t s) {
+  LAUNCH_EW(l2norm_sq_k, n, x, n, out);
+}
+void launch_f32_to_bf16(const float* x, bf16_t* y, int64_t n, hipStream_t s) {
+  LAUNCH_EW(f32_to_bf16_k, n, x, y, n);
+}

@@ -1,0 +1,91 @@
+// Fused token+position embedding gather (fwd) and scatter-add (bwd).
+// Replaces the reference's transformers wte/wpe lookup.
+// fwd: out[t,:] = wte[ids[t]] + wpe[t % seq]; one wave per token row.
+// bwd: fp32 atomicAdd scatter into dwte/dwpe workspaces (bf16 atomics
+// don't exist; the fp32 accumulate also preserves precision), cast after.
+#include "dta_common.h"
+#include "dta_kernels.h"
+
+namespace {
+
+constexpr int ROW_WAVES = 4;
+
+template <bool HAS_WPE>
+__global__ void emb_fwd_k(const int64_t* __restrict__ ids,
+                          const ushort* __restrict__ wte,
+                          const ushort* __restrict__ wpe,
+                          ushort* __restrict__ out, int64_t n_tok,
+                          int seq_len, int dim) {
+  const int lane = threadIdx.x & 63, wid = threadIdx.x >> 6;
+  const int nchunk = dim >> 3;
+  for (int64_t t = int64_t(blockIdx.x) * ROW_WAVES + wid; t < n_tok;
+       t += int64_t(gridDim.x) * ROW_WAVES) {
+    const int64_t id = ids[t];
+    const ushort* te = wte + id * dim;
+    const ushort* pe = HAS_WPE ? wpe + int64_t(t % seq_len) * dim : nullptr;
+    ushort* o = out + t * dim;
+    for (int c = lane; c < nchunk; c += 64) {
+      s16x8 v = *reinterpret_cast<const s16x8*>(te + c * 8);
+      if (HAS_WPE) {
+        s16x8 p = *reinterpret_cast<const s16x8*>(pe + c * 8);
+#pragma unroll
+        for (int j = 0; j < 8; ++j)
+          v[j] = f2bf(bf2f(ushort(v[j])) + bf2f(ushort(p[j])));
+      }
+      *reinterpret_cast<s16x8*>(o + c * 8) = v;
+    }
+  }
+}
+
+template <bool HAS_WPE>
+__global__ void emb_bwd_k(const ushort* __restrict__ dy,
+                          const int64_t* __restrict__ ids,
+                          float* __restrict__ dwte, float* __restrict__ dwpe,
+                          int64_t n_tok, int seq_len, int dim) {
+  const int lane = threadIdx.x & 63, wid = threadIdx.x >> 6;
+  for (int64_t t = int64_t(blockIdx.x) * ROW_WAVES + wid; t < n_tok;
+       t += int64_t(gridDim.x) * ROW_WAVES) {
+    const int64_t id = ids[t];
+    const ushort* g = dy + t * dim;
+    float* te = dwte + id * dim;
+    float* pe = HAS_WPE ? dwpe + int64_t(t % seq_len) * dim : nullptr;
+    for (int c = lane; c * 2 < dim; c += 64) {
+      float f0 = bf2f(g[c * 2]);
+      float f1 = bf2f(g[c * 2 + 1]);
+      atomicAdd(te + c * 2, f0);
+      atomicAdd(te + c * 2 + 1, f1);
+      if (HAS_WPE) {
+        atomicAdd(pe + c * 2, f0);
+        atomicAdd(pe + c * 2 + 1, f1);
+      }
+    }
+  }
+}
+
+}  // namespace
+
+void launch_embedding_fwd(const int64_t* ids, const bf16_t* wte,
+                          const bf16_t* wpe, bf16_t* out, int64_t n_tok,
+                          int seq_len, int dim, bool has_wpe, hipStream_t s) {
+  int64_t want = (n_tok + ROW_WAVES - 1) / ROW_WAVES;
+  const int grid = int(want < 4096 ? (want > 0 ? want : 1) : 4096);
+  if (has_wpe)
+    emb_fwd_k<true><<<grid, 256, 0, s>>>(ids, wte, wpe, out, n_tok, seq_len,
+                                         dim);
+  else
+    emb_fwd_k<false><<<grid, 256, 0, s>>>(ids, wte, nullptr, out, n_tok,
+                                          seq_len, dim);
+}
+
+void launch_embedding_bwd(const bf16_t* dy, const int64_t* ids,
+                          float* dwte_f32, float* dwpe_f32, int64_t n_tok,
+                          int seq_len, int dim, bool has_wpe, hipStream_t s) {
+  int64_t want = (n_tok + ROW_WAVES - 1) / ROW_WAVES;
+  const int grid = int(want < 2048 ? (want > 0 ? want : 1) : 2048);
+  if (has_wpe)
+    emb_bwd_k<true><<<grid, 256, 0, s>>>(dy, ids, dwte_f32, dwpe_f32, n_tok,
+                                         seq_len, dim);
+  else
+    emb_bwd_k<false><<<grid, 256, 0, s>>>(dy, ids, dwte_f32, nullptr, n_tok,
+                                          seq_len, dim);
+}

@@ -1,0 +1,333 @@
+"""GPU numerics: every CDNA4 HIP kernel vs a plain PyTorch fp32 reference.
+
+Run on the MI355X box: python -m pytest tests -m gpu -x -q
+"""
+
+import math
+
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+DEV = "cuda:0"
+
+
+def _ext():
+    from distributedtraining_amd.ops.backend import require_ext
+    return require_ext()
+
+
+def _rand_bf16(*shape, seed=0, scale=1.0):
+    g = torch.Generator(device="cpu").manual_seed(seed)
+    return (torch.randn(*shape, generator=g) * scale).to(DEV, torch.bfloat16)
+
+
+# ---------------------------------------------------------------------------
+# MFMA fragment-layout self-test: identity x asymmetric matrix (guide §3:
+# always A=I with ASYMMETRIC B to catch transposed C-writes).
+# ---------------------------------------------------------------------------
+def test_mfma_layout_16x16x32():
+    m = _ext()
+    A = torch.zeros(16, 32)
+    A[:, :16] = torch.eye(16)
+    B = torch.arange(32 * 16, dtype=torch.float32).reshape(32, 16) / 100.0
+    D = m.mfma_selftest_16(A.to(DEV, torch.bfloat16).contiguous(),
+                           B.to(DEV, torch.bfloat16).contiguous())
+    ref = A.to(torch.bfloat16).float() @ B.to(torch.bfloat16).float()
+    torch.testing.assert_close(D.cpu(), ref, rtol=1e-2, atol=1e-2)
+    # also a full random check
+    A2 = torch.randn(16, 32)
+    B2 = torch.randn(32, 16)
+    D2 = m.mfma_selftest_16(A2.to(DEV, torch.bfloat16).contiguous(),
+                            B2.to(DEV, torch.bfloat16).contiguous())
+    ref2 = A2.to(torch.bfloat16).float() @ B2.to(torch.bfloat16).float()
+    torch.testing.assert_close(D2.cpu(), ref2, rtol=2e-2, atol=2e-2)
+
+
+def test_mfma_layout_32x32x16():
+    m = _ext()
+    A = torch.zeros(32, 16)
+    A[:16, :] = torch.eye(16)
+    B = torch.arange(16 * 32, dtype=torch.float32).reshape(16, 32) / 100.0
+    D = m.mfma_selftest_32(A.to(DEV, torch.bfloat16).contiguous(),
+                           B.to(DEV, torch.bfloat16).contiguous())
+    ref = A.to(torch.bfloat16).float() @ B.to(torch.bfloat16).float()
+    torch.testing.assert_close(D.cpu(), ref, rtol=1e-2, atol=1e-2)
+
+
+# ---------------------------------------------------------------------------
+# LayerNorm / RMSNorm
+# ---------------------------------------------------------------------------
+@pytest.mark.parametrize("rows,cols", [(128, 768), (64, 64), (33, 4096)])
+def test_layernorm_fwd_bwd(rows, cols):
+    m = _ext()
+    x = _rand_bf16(rows, cols, seed=1)
+    w = _rand_bf16(cols, seed=2, scale=0.5)
+    b = _rand_bf16(cols, seed=3, scale=0.5)
+    y, mean, rstd = m.layernorm_fwd(x, w, b, 1e-5)
+    ref = torch.nn.functional.layer_norm(
+        x.float(), (cols,), w.float(), b.float(), 1e-5)
+    torch.testing.assert_close(y.float(), ref, rtol=2e-2, atol=2e-2)
+
+    dy = _rand_bf16(rows, cols, seed=4)
+    dx, dw, db = m.layernorm_bwd(dy, x, w, mean, rstd)
+    xr = x.float().detach().requires_grad_(True)
+    wr = w.float().detach().requires_grad_(True)
+    br = b.float().detach().requires_grad_(True)
+    torch.nn.functional.layer_norm(xr, (cols,), wr, br, 1e-5).backward(
+        dy.float())
+    torch.testing.assert_close(dx.float(), xr.grad, rtol=5e-2, atol=5e-2)
+    torch.testing.assert_close(dw.float(), wr.grad, rtol=5e-2,
+                               atol=0.3 * math.sqrt(rows / 128))
+    torch.testing.assert_close(db.float(), br.grad, rtol=5e-2,
+                               atol=0.3 * math.sqrt(rows / 128))
+
+
+@pytest.mark.parametrize("rows,cols", [(128, 768), (64, 4096)])
+def test_rmsnorm_fwd_bwd(rows, cols):
+    m = _ext()
+    x = _rand_bf16(rows, cols, seed=1)
+    w = _rand_bf16(cols, seed=2, scale=0.5)
+    y, rstd = m.rmsnorm_fwd(x, w, 1e-5)
+    xf = x.float()
+    ref = xf * torch.rsqrt(xf.pow(2).mean(-1, keepdim=True) + 1e-5) * w.float()
+    torch.testing.assert_close(y.float(), ref, rtol=2e-2, atol=2e-2)
+
+    dy = _rand_bf16(rows, cols, seed=4)
+    xr = xf.detach().requires_grad_(True)
+    wr = w.float().detach().requires_grad_(True)
+    (xr * torch.rsqrt(xr.pow(2).mean(-1, keepdim=True) + 1e-5) * wr).backward(
+        dy.float())
+    dx, dw = m.rmsnorm_bwd(dy, x, w, rstd)
+    torch.testing.assert_close(dx.float(), xr.grad, rtol=5e-2, atol=5e-2)
+    torch.testing.assert_close(dw.float(), wr.grad, rtol=5e-2, atol=0.3)
+
+
+# ---------------------------------------------------------------------------
+# GELU / SwiGLU
+# ---------------------------------------------------------------------------
+def test_gelu():
+    m = _ext()
+    x = _rand_bf16(1000, seed=5, scale=2.0)
+    y = m.gelu_fwd(x)
+    ref = torch.nn.functional.gelu(x.float(), approximate="tanh")
+    torch.testing.assert_close(y.float(), ref, rtol=2e-2, atol=2e-2)
+    dy = _rand_bf16(1000, seed=6)
+    xr = x.float().detach().requires_grad_(True)
+    torch.nn.functional.gelu(xr, approximate="tanh").backward(dy.float())
+    dx = m.gelu_bwd(dy, x)
+    torch.testing.assert_close(dx.float(), xr.grad, rtol=3e-2, atol=3e-2)
+
+
+def test_swiglu():
+    m = _ext()
+    g = _rand_bf16(1024, seed=7)
+    u = _rand_bf16(1024, seed=8)
+    y = m.swiglu_fwd(g, u)
+    ref = torch.nn.functional.silu(g.float()) * u.float()
+    torch.testing.assert_close(y.float(), ref, rtol=2e-2, atol=2e-2)
+    dy = _rand_bf16(1024, seed=9)
+    gr = g.float().detach().requires_grad_(True)
+    ur = u.float().detach().requires_grad_(True)
+    (torch.nn.functional.silu(gr) * ur).backward(dy.float())
+    dg, du = m.swiglu_bwd(dy, g, u)
+    torch.testing.assert_close(dg.float(), gr.grad, rtol=3e-2, atol=3e-2)
+    torch.testing.assert_close(du.float(), ur.grad, rtol=3e-2, atol=3e-2)
+
+
+# ---------------------------------------------------------------------------
+# Attention
+# ---------------------------------------------------------------------------
+@pytest.mark.parametrize("B,H,S,D", [(2, 3, 64, 64), (1, 2, 512, 64),
+                                     (2, 2, 128, 128), (1, 1, 48, 32)])
+def test_attention_fwd_bwd(B, H, S, D):
+    m = _ext()
+    scale = 1.0 / math.sqrt(D)
+    q = _rand_bf16(B, H, S, D, seed=10)
+    k = _rand_bf16(B, H, S, D, seed=11)
+    v = _rand_bf16(B, H, S, D, seed=12)
+    o, lse = m.attn_fwd(q, k, v, scale)
+    ref = torch.nn.functional.scaled_dot_product_attention(
+        q.float(), k.float(), v.float(), is_causal=True, scale=scale)
+    torch.testing.assert_close(o.float(), ref, rtol=3e-2, atol=3e-2)
+
+    do = _rand_bf16(B, H, S, D, seed=13)
+    qr = q.float().detach().requires_grad_(True)
+    kr = k.float().detach().requires_grad_(True)
+    vr = v.float().detach().requires_grad_(True)
+    torch.nn.functional.scaled_dot_product_attention(
+        qr, kr, vr, is_causal=True, scale=scale).backward(do.float())
+    dq, dk, dv = m.attn_bwd(do, q, k, v, o, lse, scale)
+    torch.testing.assert_close(dq.float(), qr.grad, rtol=5e-2, atol=5e-2)
+    torch.testing.assert_close(dk.float(), kr.grad, rtol=5e-2, atol=5e-2)
+    torch.testing.assert_close(dv.float(), vr.grad, rtol=5e-2, atol=5e-2)
+
+
+# ---------------------------------------------------------------------------
+# Cross entropy
+# ---------------------------------------------------------------------------
+@pytest.mark.parametrize("rows,vocab", [(64, 50257), (128, 512), (32, 1000)])
+def test_cross_entropy(rows, vocab):
+    m = _ext()
+    logits = _rand_bf16(rows, vocab, seed=14, scale=3.0)
+    g = torch.Generator().manual_seed(15)
+    targets = torch.randint(0, vocab, (rows,), generator=g).to(DEV)
+    targets[::7] = -100  # ignore_index stripes
+    loss_sum, lse, count = m.ce_fwd(logits, targets, -100)
+    ref = torch.nn.functional.cross_entropy(logits.float(), targets,
+                                            ignore_index=-100,
+                                            reduction="sum")
+    n_valid = int((targets != -100).sum())
+    assert int(count) == n_valid
+    torch.testing.assert_close(loss_sum.float().cpu(), ref.cpu(),
+                               rtol=1e-2, atol=1e-2 * rows)
+
+    lr = logits.float().detach().requires_grad_(True)
+    torch.nn.functional.cross_entropy(lr, targets, ignore_index=-100).backward()
+    dl = m.ce_bwd(logits, targets, lse, 1.0 / n_valid, -100)
+    torch.testing.assert_close(dl.float(), lr.grad, rtol=5e-2,
+                               atol=1e-4)
+
+
+# ---------------------------------------------------------------------------
+# Embedding
+# ---------------------------------------------------------------------------
+def test_embedding_fwd_bwd():
+    m = _ext()
+    V, P, E, B, S = 512, 128, 64, 4, 32
+    wte = _rand_bf16(V, E, seed=16)
+    wpe = _rand_bf16(P, E, seed=17)
+    g = torch.Generator().manual_seed(18)
+    ids = torch.randint(0, V, (B, S), generator=g).to(DEV)
+    out = m.embedding_fwd(ids, wte, wpe)
+    ref = wte.float()[ids] + wpe.float()[:S].unsqueeze(0)
+    torch.testing.assert_close(out.float(), ref, rtol=2e-2, atol=2e-2)
+
+    dy = _rand_bf16(B, S, E, seed=19)
+    dwte, dwpe = m.embedding_bwd(dy, ids, V, P)
+    wr = wte.float().detach().requires_grad_(True)
+    pr = wpe.float().detach().requires_grad_(True)
+    (wr[ids] + pr[:S].unsqueeze(0)).backward(dy.float())
+    torch.testing.assert_close(dwte.float(), wr.grad, rtol=3e-2, atol=3e-2)
+    torch.testing.assert_close(dwpe.float(), pr.grad, rtol=3e-2, atol=3e-2)
+
+
+# ---------------------------------------------------------------------------
+# RoPE
+# ---------------------------------------------------------------------------
+def test_rope_fwd_bwd():
+    from distributedtraining_amd.models.llama import rope_tables
+    m = _ext()
+    B, H, S, D = 2, 4, 32, 64
+    cos, sin = rope_tables(S, D, 10000.0)
+    cos, sin = cos.to(DEV), sin.to(DEV)
+    x = _rand_bf16(B, H, S, D, seed=20)
+    y = m.rope_fwd(x, cos, sin)
+    xf = x.float()
+    d2 = D // 2
+    x1, x2 = xf[..., :d2], xf[..., d2:]
+    c = cos.view(1, 1, S, d2)
+    s = sin.view(1, 1, S, d2)
+    ref = torch.cat([x1 * c - x2 * s, x1 * s + x2 * c], dim=-1)
+    torch.testing.assert_close(y.float(), ref, rtol=2e-2, atol=2e-2)
+    # bwd is the inverse rotation: rope_bwd(rope_fwd(x)) == x
+    x_rt = m.rope_bwd(y, cos, sin)
+    torch.testing.assert_close(x_rt.float(), xf, rtol=3e-2, atol=3e-2)
+
+
+# ---------------------------------------------------------------------------
+# Flat plane: adamw / delta / axpy / nan / norm / merge / grad_W
+# ---------------------------------------------------------------------------
+def test_adamw_matches_cpu_reference():
+    m = _ext()
+    from distributedtraining_amd import ops
+    P = 4099  # odd tail on purpose
+    g0 = torch.Generator().manual_seed(21)
+    master = torch.randn(P, generator=g0)
+    mm = torch.zeros(P)
+    vv = torch.zeros(P)
+    master_g = master.to(DEV)
+    m_g = mm.to(DEV)
+    v_g = vv.to(DEV)
+    wout = torch.empty(P, dtype=torch.bfloat16, device=DEV)
+    for t in range(1, 4):
+        grad = torch.randn(P, generator=g0)
+        ops.adamw_step(master, grad, mm, vv, None, t, 1e-2, 0.9, 0.999,
+                       1e-8, 0.05)  # CPU reference path
+        m.adamw_step(master_g, grad.to(DEV, torch.bfloat16), m_g, v_g, wout,
+                     t, 1e-2, 0.9, 0.999, 1e-8, 0.05)
+    torch.testing.assert_close(master_g.cpu(), master, rtol=3e-2, atol=1e-3)
+    torch.testing.assert_close(wout.float().cpu(), master, rtol=3e-2,
+                               atol=2e-2)
+
+
+def test_flat_primitives():
+    m = _ext()
+    P = 1 << 20 | 3
+    g0 = torch.Generator().manual_seed(22)
+    w = torch.randn(P, generator=g0).to(DEV)
+    base = torch.randn(P, generator=g0).to(DEV)
+    out = torch.empty_like(w)
+    m.delta_sub(w, base, out)
+    torch.testing.assert_close(out, w - base)
+    w2 = base.clone()
+    m.axpy(w2, out, 1.0)
+    torch.testing.assert_close(w2, w, rtol=1e-6, atol=1e-6)
+    assert not m.has_nan(w)
+    w[12345] = float("nan")
+    assert m.has_nan(w)
+    v = torch.full((1000,), 2.0, device=DEV)
+    assert abs(m.l2norm_sq(v) - 4000.0) < 1.0
+
+
+def test_merge_and_grad_w_match_cpu():
+    m = _ext()
+    from distributedtraining_amd import ops
+    g0 = torch.Generator().manual_seed(23)
+    segs = [1000, 37, 4096, 123]
+    P = sum(segs)
+    offsets = torch.tensor([0] + list(torch.tensor(segs).cumsum(0)))
+    N = 3
+    base = torch.randn(P, generator=g0)
+    deltas = torch.randn(N, P, generator=g0)
+    W = torch.rand(N, len(segs), generator=g0)
+    cpu = ops.weighted_merge(base, deltas, W, offsets)
+    out = torch.empty(P, device=DEV)
+    m.weighted_merge(base.to(DEV), deltas.to(DEV), W.to(DEV),
+                     offsets.to(DEV), out)
+    torch.testing.assert_close(out.cpu(), cpu, rtol=1e-4, atol=1e-4)
+
+    g = torch.randn(P, generator=g0)
+    cpu_gw = ops.grad_merge_weights(g, base, deltas, cpu, offsets)
+    gw = m.grad_merge_weights(g.to(DEV), base.to(DEV), deltas.to(DEV),
+                              out, offsets.to(DEV))
+    torch.testing.assert_close(gw.cpu(), cpu_gw, rtol=1e-3, atol=1e-2)
+
+
+# ---------------------------------------------------------------------------
+# End-to-end: one GPT-2 miner step on GPU + merge kernels through the roles
+# ---------------------------------------------------------------------------
+def test_gpu_miner_step_and_merge():
+    import distributedtraining_amd.ops as ops
+    from distributedtraining_amd.config import Config, ModelConfig
+    from distributedtraining_amd.models import build_model
+    from distributedtraining_amd.parallel.flat import FlatParams
+    from distributedtraining_amd.roles.miner import DeltaLoop
+    from distributedtraining_amd.utils.data import synthetic_batches
+
+    cfg = Config()
+    cfg.model = ModelConfig.gpt2_tiny()
+    torch.manual_seed(0)
+    model = build_model(cfg.model).to(DEV)
+    fp = FlatParams(model)
+    assert fp.work.dtype == torch.bfloat16
+    data = synthetic_batches(cfg.model.vocab_size, 4, 32, seed=1)
+    loop = DeltaLoop(model, fp, data, cfg.train)
+    l0 = loop.train_step()
+    for _ in range(10):
+        l1 = loop.train_step()
+    assert math.isfinite(l0) and math.isfinite(l1)
+    d = loop.make_delta()
+    assert float(d.flat.abs().sum()) > 0
+    assert not ops.has_nan(d.flat)
