@@ -149,37 +149,42 @@ def swiglu(gate: torch.Tensor, up: torch.Tensor) -> torch.Tensor:
 # --------------------------------------------------------------------------
 # Fused causal attention (flash-style): q,k,v [B,H,S,D] -> o [B,H,S,D]
 # --------------------------------------------------------------------------
+def _dense_last(t: torch.Tensor) -> torch.Tensor:
+    # kernel requirement: innermost head_dim contiguous; any batch/head/seq
+    # strides are fine (no transpose copies)
+    return t if t.stride(-1) == 1 else t.contiguous()
+
+
 class _AttnFn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, q, k, v, scale):
         m = require_ext()
-        q, k, v = q.contiguous(), k.contiguous(), v.contiguous()
-        o, lse = m.attn_fwd(q, k, v, scale)
-        ctx.save_for_backward(q, k, v, o, lse)
+        q, k, v = _dense_last(q), _dense_last(k), _dense_last(v)
+        o_bshd, lse = m.attn_fwd(q, k, v, scale)
+        ctx.save_for_backward(q, k, v, o_bshd, lse)
         ctx.scale = scale
-        return o
+        return o_bshd.permute(0, 2, 1, 3)  # [B,H,S,D] view, zero-copy
 
     @staticmethod
     def backward(ctx, do):
         m = require_ext()
-        q, k, v, o, lse = ctx.saved_tensors
-        dq, dk, dv = m.attn_bwd(do.contiguous(), q, k, v, o, lse, ctx.scale)
-        return dq, dk, dv, None
+        q, k, v, o_bshd, lse = ctx.saved_tensors
+        dq_bshd, dk, dv = m.attn_bwd(_dense_last(do), q, k, v, o_bshd, lse,
+                                     ctx.scale)
+        return dq_bshd.permute(0, 2, 1, 3), dk, dv, None
 
 
 def causal_attention(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
                      scale: Optional[float] = None) -> torch.Tensor:
     """softmax(q kᵀ · scale + causal_mask) v over [B, H, S, D] tensors.
 
-    GQA: k/v may have fewer heads (H_kv dividing H); heads are grouped.
+    GQA: k/v may have fewer heads (H_kv dividing H) — handled natively by
+    the kernel (no repeat_interleave). Inputs may be permuted views of
+    [B,S,H*D] projections; only the head_dim must be contiguous.
     """
     if scale is None:
         scale = 1.0 / math.sqrt(q.shape[-1])
     if use_hip(q):
-        if k.shape[1] != q.shape[1]:  # expand GQA heads for the kernel
-            rep = q.shape[1] // k.shape[1]
-            k = k.repeat_interleave(rep, dim=1)
-            v = v.repeat_interleave(rep, dim=1)
         return _AttnFn.apply(q, k, v, scale)
     return F.scaled_dot_product_attention(q, k, v, is_causal=True,
                                           scale=scale, enable_gqa=True)

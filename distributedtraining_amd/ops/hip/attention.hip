@@ -3,50 +3,43 @@
 // the reference's transformers attention (SURVEY.md §2.2 op table; miner
 // seq 64, validator seq 512).
 //
-// Structure (v1, correctness-first):
+// Structure (v2):
 //   * one 64-lane wave owns a 16-row Q tile; blocks of 4 independent waves
-//     (no __syncthreads in the loop), grid = BH x ceil(S/64).
+//     (no __syncthreads in the loop), grid = (ceil(S/64), B*H).
+//   * STRIDE-AWARE addressing: q/k/v/o are consumed as [B,H,S,D] *views*
+//     of the projection output [B,S,H*D] (and dq/dk/dv written the same
+//     way) so the model layer does zero transpose/contiguous copies.
+//   * GQA native: kv head = h / group; no repeat_interleave materialization.
 //   * K/V tiles are read directly from global memory — at these sequence
-//     lengths a (b,h)'s K/V fit in L2 (guide §5 common-mistake 7: LDS
-//     staging of L2-resident data is pure overhead).
+//     lengths a (b,h)'s K/V fit in L2 (guide §5 common-mistake 7).
 //   * swapped QK^T: mfma(A=K_tile, B=Q^T) puts a query's scores in lanes
-//     sharing (lane&15) so the softmax row-reduce is two shfl_xor ops.
-//   * P is redistributed score->A-fragment via 16 shfls per 32-key tile
-//     and fed to mfma(P, V) accumulating O in f32.
+//     sharing (lane&15) so the softmax row-reduce is two shfl_xor ops;
+//     exp via the single-instruction exp2 path.
 //
-// Fragment maps (verified on-device by mfma_selftest, see bindings.cpp):
+// Fragment maps (verified on-device by mfma_selftest, tests/test_ops_gpu.py):
 //   mfma_f32_16x16x32_bf16: A[i][k]: i=lane&15, k=8*(lane>>4)+j (j=0..7)
 //                           B[k][n]: n=lane&15, k=8*(lane>>4)+j
 //                           C/D[i][j]: col=lane&15, row=4*(lane>>4)+reg
-//
-// Backward is the standard flash recomputation split into a dQ kernel
-// (grid over Q tiles) and a dK/dV kernel (grid over KV tiles), with
-// delta_row = sum_d dO*O precomputed.
 #include "dta_common.h"
 #include "dta_kernels.h"
 
 namespace {
 
-using mfma16 = f32x4;
-
 DEV f32x4 mfma_bf16(bf16x8 a, bf16x8 b, f32x4 c) {
   return __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, c, 0, 0, 0);
 }
 
-// load an A/B fragment whose per-lane 8 elements are CONTIGUOUS in memory:
-// row r(lane), starting col c0(lane) — one 16B read.
 DEV bf16x8 load_frag_row(const ushort* base, int64_t row_stride, int row,
                          int col0) {
-  const s16x8 v = *reinterpret_cast<const s16x8*>(base + int64_t(row) * row_stride + col0);
+  const s16x8 v =
+      *reinterpret_cast<const s16x8*>(base + int64_t(row) * row_stride + col0);
   union { s16x8 s; bf16x8 b; } u;
   u.s = v;
   return u.b;
 }
 
-// load a B fragment with per-lane elements STRIDED by rows (V-style):
-// element j comes from row r0+j, fixed col. 8 scalar reads (L2-resident).
 DEV bf16x8 load_frag_col(const ushort* base, int64_t row_stride, int row0,
-                         int col, int row_max /*exclusive*/) {
+                         int col, int row_max) {
   bf16x8 out;
 #pragma unroll
   for (int j = 0; j < 8; ++j) {
@@ -71,86 +64,86 @@ DEV bf16x8 pack_bf16x8(const float* f) {
 }
 
 // Redistribute a 32-key score tile from C layout (q=lane&15,
-// key=16*ks+4*(lane>>4)+r held in p0/p1) into the A-fragment layout
+// key=16*ks+4*(lane>>4)+r in p0/p1) into the A-fragment layout
 // (q=lane&15, k=8*(lane>>4)+j) — two shfl rounds per j.
 DEV bf16x8 scores_to_afrag(const f32x4& p0, const f32x4& p1, int lane) {
   const int g = lane >> 4;
   float pa[8];
 #pragma unroll
   for (int j = 0; j < 8; ++j) {
-    const int kt = 8 * g + j;                    // the key this lane wants
+    const int kt = 8 * g + j;
     const int src = (lane & 15) + 16 * ((kt >> 2) & 3);
-    const float va = __shfl(p0[j & 3], src);     // keys 0..15 live in p0
-    const float vb = __shfl(p1[j & 3], src);     // keys 16..31 live in p1
+    const float va = __shfl(p0[j & 3], src);
+    const float vb = __shfl(p1[j & 3], src);
     pa[j] = (kt < 16) ? va : vb;
   }
   return pack_bf16x8(pa);
 }
 
 // ---------------- forward ----------------
-template <int DTILES>  // D = 16*DTILES (4 -> 64, 8 -> 128)
+template <int DTILES>  // D = 16*DTILES
 __global__ void attn_fwd_k(const ushort* __restrict__ q,
                            const ushort* __restrict__ k,
                            const ushort* __restrict__ v,
                            ushort* __restrict__ o, float* __restrict__ lse,
-                           int seq, float scale) {
-  constexpr int D = 16 * DTILES;
-  constexpr int DSL = DTILES / 2;  // 32-wide d slices
+                           AttnGeom geo) {
   const int lane = threadIdx.x & 63;
   const int wid = threadIdx.x >> 6;
-  const int64_t bh = blockIdx.y;
-  const int q0 = blockIdx.x * 64 + wid * 16;  // this wave's q-tile base
+  const int64_t bhid = blockIdx.y;
+  const int b = int(bhid) / geo.H, h = int(bhid) % geo.H;
+  const int hk = h / geo.grp;
+  const int seq = geo.seq;
+  const int q0 = blockIdx.x * 64 + wid * 16;
   if (q0 >= seq) return;
-  const ushort* qp = q + bh * int64_t(seq) * D;
-  const ushort* kp = k + bh * int64_t(seq) * D;
-  const ushort* vp = v + bh * int64_t(seq) * D;
+  const ushort* qp = q + b * geo.qb + h * geo.qh;
+  const ushort* kp = k + b * geo.kb + hk * geo.kh;
+  const ushort* vp = v + b * geo.vb + hk * geo.vh;
+  const float sc2 = geo.scale * LOG2E;  // fold scale into the exp2 argument
 
-  // Q fragments (B operand of the swapped QK^T): 16B per lane per slice
-  bf16x8 qb[DSL];
+  bf16x8 qb_[DTILES / 2];
   const int qrow = q0 + (lane & 15);
   const int qr_ld = qrow < seq ? qrow : seq - 1;
 #pragma unroll
-  for (int sl = 0; sl < DSL; ++sl)
-    qb[sl] = load_frag_row(qp, D, qr_ld, 32 * sl + 8 * (lane >> 4));
+  for (int sl = 0; sl < DTILES / 2; ++sl)
+    qb_[sl] = load_frag_row(qp, geo.qs, qr_ld, 32 * sl + 8 * (lane >> 4));
 
   f32x4 acc[DTILES];
 #pragma unroll
   for (int t = 0; t < DTILES; ++t) acc[t] = f32x4{0, 0, 0, 0};
-  float m_run = -INFINITY, l_run = 0.f;
+  float m_run = -INFINITY, l_run = 0.f;  // in exp2 units
 
-  const int kv_end = min(seq, q0 + 16);  // causal: keys <= max q row
+  const int g = lane >> 4;
+  const int kv_end = min(seq, q0 + 16);
   for (int kv0 = 0; kv0 < kv_end; kv0 += 32) {
-    // scores S^T subtiles: p0 = keys kv0..kv0+15, p1 = +16..31
     f32x4 p0 = {0, 0, 0, 0}, p1 = {0, 0, 0, 0};
 #pragma unroll
-    for (int sl = 0; sl < DSL; ++sl) {
+    for (int sl = 0; sl < DTILES / 2; ++sl) {
       const int c0 = 32 * sl + 8 * (lane >> 4);
       const int kr0 = kv0 + (lane & 15);
       const int kr1 = kr0 + 16;
-      bf16x8 ka = load_frag_row(kp, D, kr0 < seq ? kr0 : seq - 1, c0);
-      p0 = mfma_bf16(ka, qb[sl], p0);
-      bf16x8 kb = load_frag_row(kp, D, kr1 < seq ? kr1 : seq - 1, c0);
-      p1 = mfma_bf16(kb, qb[sl], p1);
+      bf16x8 ka = load_frag_row(kp, geo.ks, kr0 < seq ? kr0 : seq - 1, c0);
+      p0 = mfma_bf16(ka, qb_[sl], p0);
+      bf16x8 kb2 = load_frag_row(kp, geo.ks, kr1 < seq ? kr1 : seq - 1, c0);
+      p1 = mfma_bf16(kb2, qb_[sl], p1);
     }
-    // causal + bounds mask, scale; layout: q=lane&15, key=16ks+4g+r
-    const int g = lane >> 4;
     float mx = -INFINITY;
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
       const int k0a = kv0 + 4 * g + r, k1a = k0a + 16;
-      p0[r] = (k0a <= qrow && k0a < seq) ? p0[r] * scale : -INFINITY;
-      p1[r] = (k1a <= qrow && k1a < seq) ? p1[r] * scale : -INFINITY;
+      p0[r] = (k0a <= qrow && k0a < seq) ? p0[r] * sc2 : -INFINITY;
+      p1[r] = (k1a <= qrow && k1a < seq) ? p1[r] * sc2 : -INFINITY;
       mx = fmaxf(mx, fmaxf(p0[r], p1[r]));
     }
     mx = fmaxf(mx, __shfl_xor(mx, 16));
-    mx = fmaxf(mx, __shfl_xor(mx, 32));   // all 4 groups now share row max
+    mx = fmaxf(mx, __shfl_xor(mx, 32));
     const float m_new = fmaxf(m_run, mx);
-    const float alpha = (m_run == -INFINITY) ? 0.f : expf(m_run - m_new);
+    const float alpha =
+        (m_run == -INFINITY) ? 0.f : __builtin_exp2f(m_run - m_new);
     float psum = 0.f;
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
-      p0[r] = (p0[r] == -INFINITY) ? 0.f : expf(p0[r] - m_new);
-      p1[r] = (p1[r] == -INFINITY) ? 0.f : expf(p1[r] - m_new);
+      p0[r] = (p0[r] == -INFINITY) ? 0.f : __builtin_exp2f(p0[r] - m_new);
+      p1[r] = (p1[r] == -INFINITY) ? 0.f : __builtin_exp2f(p1[r] - m_new);
       psum += p0[r] + p1[r];
     }
     psum += __shfl_xor(psum, 16);
@@ -158,67 +151,66 @@ __global__ void attn_fwd_k(const ushort* __restrict__ q,
     l_run = l_run * alpha + psum;
     m_run = m_new;
 
-    // O rescale: lane holds rows q=4g+r -> pull alpha from lane (4g+r)
     bf16x8 pa = scores_to_afrag(p0, p1, lane);
+    float a_r[4];
+#pragma unroll
+    for (int r = 0; r < 4; ++r) a_r[r] = __shfl(alpha, 4 * g + r);
 #pragma unroll
     for (int t = 0; t < DTILES; ++t) {
-      const float a_r0 = __shfl(alpha, 4 * g + 0);
-      const float a_r1 = __shfl(alpha, 4 * g + 1);
-      const float a_r2 = __shfl(alpha, 4 * g + 2);
-      const float a_r3 = __shfl(alpha, 4 * g + 3);
-      acc[t][0] *= a_r0; acc[t][1] *= a_r1;
-      acc[t][2] *= a_r2; acc[t][3] *= a_r3;
-      bf16x8 vb = load_frag_col(vp, D, kv0 + 8 * g, 16 * t + (lane & 15),
-                                seq);
+      acc[t][0] *= a_r[0]; acc[t][1] *= a_r[1];
+      acc[t][2] *= a_r[2]; acc[t][3] *= a_r[3];
+      bf16x8 vb = load_frag_col(vp, geo.vs, kv0 + 8 * g,
+                                16 * t + (lane & 15), seq);
       acc[t] = mfma_bf16(pa, vb, acc[t]);
     }
   }
 
-  // epilogue: O /= l, store; lse = m + log(l)
-  const int g = lane >> 4;
-  const float l_q = (lane < 16) ? l_run : 0.f;  // canonical copy at g=0
-  const float m_q = m_run;
-  if (lane < 16 && qrow < seq) lse[bh * seq + qrow] = m_q + logf(l_run);
+  // epilogue: O /= l; lse stored in NATURAL-log units = (m + log2 l)*ln2
+  if (lane < 16 && qrow < seq)
+    lse[bhid * seq + qrow] = (m_run + __builtin_log2f(l_run)) * LN2;
+  float inv[4];
 #pragma unroll
-  for (int t = 0; t < DTILES; ++t) {
-    float inv[4];
+  for (int r = 0; r < 4; ++r) {
+    float lr = __shfl(l_run, 4 * g + r);
+    inv[r] = lr > 0.f ? 1.0f / lr : 0.f;
+  }
+  ushort* op = o + b * geo.ob + h * geo.oh;
 #pragma unroll
-    for (int r = 0; r < 4; ++r) {
-      float lr = __shfl(l_run, 4 * g + r);
-      inv[r] = lr > 0.f ? 1.0f / lr : 0.f;
-    }
-    ushort* op = o + bh * int64_t(seq) * D;
+  for (int t = 0; t < DTILES; ++t)
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
       const int orow = q0 + 4 * g + r;
       if (orow < seq)
-        op[int64_t(orow) * D + 16 * t + (lane & 15)] = f2bf(acc[t][r] * inv[r]);
+        op[int64_t(orow) * geo.os_ + 16 * t + (lane & 15)] =
+            f2bf(acc[t][r] * inv[r]);
     }
-  }
 }
 
 // ---------------- delta = rowsum(dO * O) ----------------
 __global__ void attn_delta_k(const ushort* __restrict__ dout,
                              const ushort* __restrict__ o,
-                             float* __restrict__ delta, int64_t rows,
-                             int hd) {
+                             float* __restrict__ delta, AttnGeom geo) {
   const int lane = threadIdx.x & 63, wid = threadIdx.x >> 6;
-  for (int64_t r = int64_t(blockIdx.x) * 4 + wid; r < rows;
-       r += int64_t(gridDim.x) * 4) {
-    const ushort* a = dout + r * hd;
-    const ushort* b = o + r * hd;
+  const int hd = geo.hd;
+  const int64_t rows = int64_t(geo.B) * geo.H * geo.seq;
+  for (int64_t rr = int64_t(blockIdx.x) * 4 + wid; rr < rows;
+       rr += int64_t(gridDim.x) * 4) {
+    const int s_ = int(rr % geo.seq);
+    const int h = int((rr / geo.seq) % geo.H);
+    const int b = int(rr / (int64_t(geo.seq) * geo.H));
+    const ushort* a = dout + b * geo.db_ + h * geo.dh + int64_t(s_) * geo.ds;
+    const ushort* c = o + b * geo.ob + h * geo.oh + int64_t(s_) * geo.os_;
     float acc = 0.f;
     for (int i = lane * 2; i + 2 <= hd; i += 128) {
-      acc = fmaf(bf2f(a[i]), bf2f(b[i]), acc);
-      acc = fmaf(bf2f(a[i + 1]), bf2f(b[i + 1]), acc);
+      acc = fmaf(bf2f(a[i]), bf2f(c[i]), acc);
+      acc = fmaf(bf2f(a[i + 1]), bf2f(c[i + 1]), acc);
     }
     acc = wave_sum(acc);
-    if (lane == 0) delta[r] = acc;
+    if (lane == 0) delta[rr] = acc;
   }
 }
 
 // ---------------- backward dQ ----------------
-// dQ[q,d] = scale * sum_k P(q,k) * (dP(q,k) - delta_q) * K[k,d]
 template <int DTILES>
 __global__ void attn_bwd_dq_k(const ushort* __restrict__ dout,
                               const ushort* __restrict__ q,
@@ -226,118 +218,122 @@ __global__ void attn_bwd_dq_k(const ushort* __restrict__ dout,
                               const ushort* __restrict__ v,
                               const float* __restrict__ lse,
                               const float* __restrict__ delta,
-                              ushort* __restrict__ dq, int seq, float scale) {
-  constexpr int D = 16 * DTILES;
-  constexpr int DSL = DTILES / 2;
+                              ushort* __restrict__ dq, AttnGeom geo) {
   const int lane = threadIdx.x & 63;
   const int wid = threadIdx.x >> 6;
-  const int64_t bh = blockIdx.y;
+  const int64_t bhid = blockIdx.y;
+  const int b = int(bhid) / geo.H, h = int(bhid) % geo.H;
+  const int hk = h / geo.grp;
+  const int seq = geo.seq;
   const int q0 = blockIdx.x * 64 + wid * 16;
   if (q0 >= seq) return;
-  const ushort* qp = q + bh * int64_t(seq) * D;
-  const ushort* kp = k + bh * int64_t(seq) * D;
-  const ushort* vp = v + bh * int64_t(seq) * D;
-  const ushort* dop = dout + bh * int64_t(seq) * D;
+  const ushort* qp = q + b * geo.qb + h * geo.qh;
+  const ushort* kp = k + b * geo.kb + hk * geo.kh;
+  const ushort* vp = v + b * geo.vb + hk * geo.vh;
+  const ushort* dop = dout + b * geo.db_ + h * geo.dh;
 
   const int qrow = q0 + (lane & 15);
   const int qr_ld = qrow < seq ? qrow : seq - 1;
-  bf16x8 qb[DSL], dob[DSL];
+  bf16x8 qb_[DTILES / 2], dob[DTILES / 2];
 #pragma unroll
-  for (int sl = 0; sl < DSL; ++sl) {
+  for (int sl = 0; sl < DTILES / 2; ++sl) {
     const int c0 = 32 * sl + 8 * (lane >> 4);
-    qb[sl] = load_frag_row(qp, D, qr_ld, c0);
-    dob[sl] = load_frag_row(dop, D, qr_ld, c0);
+    qb_[sl] = load_frag_row(qp, geo.qs, qr_ld, c0);
+    dob[sl] = load_frag_row(dop, geo.ds, qr_ld, c0);
   }
-  const float lse_q = lse[bh * seq + qr_ld];
-  const float dlt_q = delta[bh * seq + qr_ld];
+  const float lse_q = lse[bhid * seq + qr_ld] * LOG2E;  // exp2 units
+  const float dlt_q = delta[bhid * seq + qr_ld];
+  const float sc2 = geo.scale * LOG2E;
 
   f32x4 acc[DTILES];
 #pragma unroll
   for (int t = 0; t < DTILES; ++t) acc[t] = f32x4{0, 0, 0, 0};
 
+  const int g = lane >> 4;
   const int kv_end = min(seq, q0 + 16);
   for (int kv0 = 0; kv0 < kv_end; kv0 += 32) {
     f32x4 s0 = {0, 0, 0, 0}, s1 = {0, 0, 0, 0};
     f32x4 dp0 = {0, 0, 0, 0}, dp1 = {0, 0, 0, 0};
 #pragma unroll
-    for (int sl = 0; sl < DSL; ++sl) {
+    for (int sl = 0; sl < DTILES / 2; ++sl) {
       const int c0 = 32 * sl + 8 * (lane >> 4);
       const int kr0 = kv0 + (lane & 15), kr1 = kr0 + 16;
-      bf16x8 ka = load_frag_row(kp, D, kr0 < seq ? kr0 : seq - 1, c0);
-      bf16x8 kb = load_frag_row(kp, D, kr1 < seq ? kr1 : seq - 1, c0);
-      bf16x8 va = load_frag_row(vp, D, kr0 < seq ? kr0 : seq - 1, c0);
-      bf16x8 vb2 = load_frag_row(vp, D, kr1 < seq ? kr1 : seq - 1, c0);
-      s0 = mfma_bf16(ka, qb[sl], s0);
-      s1 = mfma_bf16(kb, qb[sl], s1);
+      bf16x8 ka = load_frag_row(kp, geo.ks, kr0 < seq ? kr0 : seq - 1, c0);
+      bf16x8 kb2 = load_frag_row(kp, geo.ks, kr1 < seq ? kr1 : seq - 1, c0);
+      bf16x8 va = load_frag_row(vp, geo.vs, kr0 < seq ? kr0 : seq - 1, c0);
+      bf16x8 vb2 = load_frag_row(vp, geo.vs, kr1 < seq ? kr1 : seq - 1, c0);
+      s0 = mfma_bf16(ka, qb_[sl], s0);
+      s1 = mfma_bf16(kb2, qb_[sl], s1);
       dp0 = mfma_bf16(va, dob[sl], dp0);
       dp1 = mfma_bf16(vb2, dob[sl], dp1);
     }
-    const int g = lane >> 4;
-    // lse/delta for this lane's q (= lane&15) are already per-lane
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
       const int k0a = kv0 + 4 * g + r, k1a = k0a + 16;
       const bool v0 = (k0a <= qrow && k0a < seq);
       const bool v1 = (k1a <= qrow && k1a < seq);
-      const float P0 = v0 ? expf(s0[r] * scale - lse_q) : 0.f;
-      const float P1 = v1 ? expf(s1[r] * scale - lse_q) : 0.f;
+      const float P0 = v0 ? __builtin_exp2f(s0[r] * sc2 - lse_q) : 0.f;
+      const float P1 = v1 ? __builtin_exp2f(s1[r] * sc2 - lse_q) : 0.f;
       s0[r] = P0 * (dp0[r] - dlt_q);
       s1[r] = P1 * (dp1[r] - dlt_q);
     }
     bf16x8 dsa = scores_to_afrag(s0, s1, lane);
 #pragma unroll
     for (int t = 0; t < DTILES; ++t) {
-      bf16x8 kb2 = load_frag_col(kp, D, kv0 + 8 * g, 16 * t + (lane & 15),
-                                 seq);
-      acc[t] = mfma_bf16(dsa, kb2, acc[t]);
+      bf16x8 kcb = load_frag_col(kp, geo.ks, kv0 + 8 * g,
+                                 16 * t + (lane & 15), seq);
+      acc[t] = mfma_bf16(dsa, kcb, acc[t]);
     }
   }
-  const int g = lane >> 4;
-  ushort* dqp = dq + bh * int64_t(seq) * D;
+  ushort* dqp = dq + b * geo.ob + h * geo.oh;  // dq uses o-geometry strides
 #pragma unroll
   for (int t = 0; t < DTILES; ++t)
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
       const int orow = q0 + 4 * g + r;
       if (orow < seq)
-        dqp[int64_t(orow) * D + 16 * t + (lane & 15)] =
-            f2bf(acc[t][r] * scale);
+        dqp[int64_t(orow) * geo.os_ + 16 * t + (lane & 15)] =
+            f2bf(acc[t][r] * geo.scale);
     }
 }
 
 // ---------------- backward dK/dV ----------------
-// wave owns a 16-row KV tile; iterates q tiles of 32 (q >= kv0).
-// dV[k,d] = sum_q P^T(k,q) dO[q,d];  dK[k,d] = scale*sum_q dS^T(k,q) Q[q,d]
-// P^T from S = mfma(Q_sub, K^T) -> C: (k=lane&15, q=4g+r).
-template <int DTILES>
+// wave owns 16 KV rows; iterates q tiles of 32. With GQA (grp>1) each
+// kv head is walked once per ATTACHED q head (blockIdx.y covers B*H) and
+// results are accumulated with fp32 atomics into dk/dv.
+template <int DTILES, bool ATOMIC>
 __global__ void attn_bwd_dkv_k(const ushort* __restrict__ dout,
                                const ushort* __restrict__ q,
                                const ushort* __restrict__ k,
                                const ushort* __restrict__ v,
                                const float* __restrict__ lse,
                                const float* __restrict__ delta,
+                               float* __restrict__ dk32,
+                               float* __restrict__ dv32,
                                ushort* __restrict__ dk,
-                               ushort* __restrict__ dv, int seq,
-                               float scale) {
-  constexpr int D = 16 * DTILES;
-  constexpr int DSL = DTILES / 2;
+                               ushort* __restrict__ dv, AttnGeom geo) {
   const int lane = threadIdx.x & 63;
   const int wid = threadIdx.x >> 6;
-  const int64_t bh = blockIdx.y;
+  const int64_t bhid = blockIdx.y;
+  const int b = int(bhid) / geo.H, h = int(bhid) % geo.H;
+  const int hk = h / geo.grp;
+  const int seq = geo.seq;
   const int kv0 = blockIdx.x * 64 + wid * 16;
   if (kv0 >= seq) return;
-  const ushort* qp = q + bh * int64_t(seq) * D;
-  const ushort* kp = k + bh * int64_t(seq) * D;
-  const ushort* vp = v + bh * int64_t(seq) * D;
-  const ushort* dop = dout + bh * int64_t(seq) * D;
+  const ushort* qp = q + b * geo.qb + h * geo.qh;
+  const ushort* kp = k + b * geo.kb + hk * geo.kh;
+  const ushort* vp = v + b * geo.vb + hk * geo.vh;
+  const ushort* dop = dout + b * geo.db_ + h * geo.dh;
+  const float sc2 = geo.scale * LOG2E;
 
-  // K fragments as B operand of mfma(Q, K^T): (k=lane&15, d=8g+j)
   const int krow = kv0 + (lane & 15);
   const int kr_ld = krow < seq ? krow : seq - 1;
-  bf16x8 kb[DSL];
+  bf16x8 kb_[DTILES / 2], vbf[DTILES / 2];
 #pragma unroll
-  for (int sl = 0; sl < DSL; ++sl)
-    kb[sl] = load_frag_row(kp, D, kr_ld, 32 * sl + 8 * (lane >> 4));
+  for (int sl = 0; sl < DTILES / 2; ++sl) {
+    kb_[sl] = load_frag_row(kp, geo.ks, kr_ld, 32 * sl + 8 * (lane >> 4));
+    vbf[sl] = load_frag_row(vp, geo.vs, kr_ld, 32 * sl + 8 * (lane >> 4));
+  }
 
   f32x4 acck[DTILES], accv[DTILES];
 #pragma unroll
@@ -347,78 +343,68 @@ __global__ void attn_bwd_dkv_k(const ushort* __restrict__ dout,
   }
 
   const int g = lane >> 4;
-  const int q_start = (kv0 / 32) * 32;  // first q tile with q >= kv0
+  const int q_start = (kv0 / 32) * 32;
   for (int q0 = q_start; q0 < seq; q0 += 32) {
     f32x4 s0 = {0, 0, 0, 0}, s1 = {0, 0, 0, 0};
+    f32x4 dp0 = {0, 0, 0, 0}, dp1 = {0, 0, 0, 0};
 #pragma unroll
-    for (int sl = 0; sl < DSL; ++sl) {
+    for (int sl = 0; sl < DTILES / 2; ++sl) {
       const int c0 = 32 * sl + 8 * (lane >> 4);
       const int qr0 = q0 + (lane & 15), qr1 = qr0 + 16;
-      bf16x8 qa = load_frag_row(qp, D, qr0 < seq ? qr0 : seq - 1, c0);
-      bf16x8 qa1 = load_frag_row(qp, D, qr1 < seq ? qr1 : seq - 1, c0);
-      // S^T? no: mfma(A=Q[16q][32d], B=K^T) gives (k=lane&15, q=4g+r)
-      s0 = mfma_bf16(qa, kb[sl], s0);
-      s1 = mfma_bf16(qa1, kb[sl], s1);
+      bf16x8 qa = load_frag_row(qp, geo.qs, qr0 < seq ? qr0 : seq - 1, c0);
+      bf16x8 qa1 = load_frag_row(qp, geo.qs, qr1 < seq ? qr1 : seq - 1, c0);
+      s0 = mfma_bf16(qa, kb_[sl], s0);
+      s1 = mfma_bf16(qa1, kb_[sl], s1);
+      bf16x8 doa = load_frag_row(dop, geo.ds, qr0 < seq ? qr0 : seq - 1, c0);
+      bf16x8 doa1 = load_frag_row(dop, geo.ds, qr1 < seq ? qr1 : seq - 1, c0);
+      dp0 = mfma_bf16(doa, vbf[sl], dp0);
+      dp1 = mfma_bf16(doa1, vbf[sl], dp1);
     }
-    // lane holds (k=krow, q = q0 + 4g + r (+16 for s1))
     f32x4 p0, p1, ds0, ds1;
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
       const int qa0 = q0 + 4 * g + r, qa1 = qa0 + 16;
       const bool v0 = (qa0 >= krow && qa0 < seq);
       const bool v1 = (qa1 >= krow && qa1 < seq);
-      const float lse0 = lse[bh * seq + (v0 ? qa0 : 0)];
-      const float lse1 = lse[bh * seq + (v1 ? qa1 : 0)];
-      p0[r] = v0 ? expf(s0[r] * scale - lse0) : 0.f;
-      p1[r] = v1 ? expf(s1[r] * scale - lse1) : 0.f;
-    }
-    // dP^T(k,q) = sum_d V[k,d] dO[q,d]: mfma(A=dO[16q][32d], B=V^T)
-    f32x4 dp0 = {0, 0, 0, 0}, dp1 = {0, 0, 0, 0};
-    bf16x8 vbf[DSL];
-#pragma unroll
-    for (int sl = 0; sl < DSL; ++sl)
-      vbf[sl] = load_frag_row(vp, D, kr_ld, 32 * sl + 8 * (lane >> 4));
-#pragma unroll
-    for (int sl = 0; sl < DSL; ++sl) {
-      const int c0 = 32 * sl + 8 * (lane >> 4);
-      const int qr0 = q0 + (lane & 15), qr1 = qr0 + 16;
-      bf16x8 doa = load_frag_row(dop, D, qr0 < seq ? qr0 : seq - 1, c0);
-      bf16x8 doa1 = load_frag_row(dop, D, qr1 < seq ? qr1 : seq - 1, c0);
-      dp0 = mfma_bf16(doa, vbf[sl], dp0);
-      dp1 = mfma_bf16(doa1, vbf[sl], dp1);
-    }
-#pragma unroll
-    for (int r = 0; r < 4; ++r) {
-      const int qa0 = q0 + 4 * g + r, qa1 = qa0 + 16;
-      const float d0 = delta[bh * seq + ((qa0 < seq) ? qa0 : 0)];
-      const float d1 = delta[bh * seq + ((qa1 < seq) ? qa1 : 0)];
+      const float l0 = lse[bhid * seq + (v0 ? qa0 : 0)] * LOG2E;
+      const float l1 = lse[bhid * seq + (v1 ? qa1 : 0)] * LOG2E;
+      p0[r] = v0 ? __builtin_exp2f(s0[r] * sc2 - l0) : 0.f;
+      p1[r] = v1 ? __builtin_exp2f(s1[r] * sc2 - l1) : 0.f;
+      const float d0 = delta[bhid * seq + ((qa0 < seq) ? qa0 : 0)];
+      const float d1 = delta[bhid * seq + ((qa1 < seq) ? qa1 : 0)];
       ds0[r] = p0[r] * (dp0[r] - d0);
       ds1[r] = p1[r] * (dp1[r] - d1);
     }
-    // A fragments (k=lane&15, q=8g+j) from (k=lane&15, q=16ks+4g+r)
     bf16x8 pa = scores_to_afrag(p0, p1, lane);
     bf16x8 dsa = scores_to_afrag(ds0, ds1, lane);
 #pragma unroll
     for (int t = 0; t < DTILES; ++t) {
-      bf16x8 dob = load_frag_col(dop, D, q0 + 8 * g, 16 * t + (lane & 15),
-                                 seq);
-      accv[t] = mfma_bf16(pa, dob, accv[t]);
-      bf16x8 qcb = load_frag_col(qp, D, q0 + 8 * g, 16 * t + (lane & 15),
-                                 seq);
+      bf16x8 dob2 = load_frag_col(dop, geo.ds, q0 + 8 * g,
+                                  16 * t + (lane & 15), seq);
+      accv[t] = mfma_bf16(pa, dob2, accv[t]);
+      bf16x8 qcb = load_frag_col(qp, geo.qs, q0 + 8 * g,
+                                 16 * t + (lane & 15), seq);
       acck[t] = mfma_bf16(dsa, qcb, acck[t]);
     }
   }
-  ushort* dkp = dk + bh * int64_t(seq) * D;
-  ushort* dvp = dv + bh * int64_t(seq) * D;
+  constexpr int D = 16 * DTILES;
 #pragma unroll
   for (int t = 0; t < DTILES; ++t)
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
       const int orow = kv0 + 4 * g + r;
-      if (orow < seq) {
-        dkp[int64_t(orow) * D + 16 * t + (lane & 15)] =
-            f2bf(acck[t][r] * scale);
-        dvp[int64_t(orow) * D + 16 * t + (lane & 15)] = f2bf(accv[t][r]);
+      if (orow >= seq) continue;
+      const int col = 16 * t + (lane & 15);
+      if (ATOMIC) {
+        // fp32 accumulation across the grp q-heads sharing this kv head
+        int64_t off = ((int64_t(b) * (geo.H / geo.grp) + hk) * seq + orow) * D + col;
+        atomicAdd(dk32 + off, acck[t][r] * geo.scale);
+        atomicAdd(dv32 + off, accv[t][r]);
+      } else {
+        ushort* dkp = dk + b * geo.kb + hk * geo.kh;
+        ushort* dvp = dv + b * geo.vb + hk * geo.vh;
+        dkp[int64_t(orow) * geo.ks + col] = f2bf(acck[t][r] * geo.scale);
+        dvp[int64_t(orow) * geo.vs + col] = f2bf(accv[t][r]);
       }
     }
 }
@@ -430,8 +416,8 @@ __global__ void mfma_probe16_k(const ushort* A, const ushort* B, float* D) {
 #pragma unroll
   for (int j = 0; j < 8; ++j) {
     union { ushort s; __bf16 v; } ca, cb;
-    ca.s = A[(lane & 15) * 32 + 8 * (lane >> 4) + j];   // A[16][32]
-    cb.s = B[(8 * (lane >> 4) + j) * 16 + (lane & 15)]; // B[32][16]
+    ca.s = A[(lane & 15) * 32 + 8 * (lane >> 4) + j];
+    cb.s = B[(8 * (lane >> 4) + j) * 16 + (lane & 15)];
     a[j] = ca.v;
     b[j] = cb.v;
   }
@@ -449,8 +435,8 @@ __global__ void mfma_probe32_k(const ushort* A, const ushort* B, float* D) {
 #pragma unroll
   for (int j = 0; j < 8; ++j) {
     union { ushort s; __bf16 v; } ca, cb;
-    ca.s = A[(lane & 31) * 16 + 8 * (lane >> 5) + j];   // A[32][16]
-    cb.s = B[(8 * (lane >> 5) + j) * 32 + (lane & 31)]; // B[16][32]
+    ca.s = A[(lane & 31) * 16 + 8 * (lane >> 5) + j];
+    cb.s = B[(8 * (lane >> 5) + j) * 32 + (lane & 31)];
     a[j] = ca.v;
     b[j] = cb.v;
   }
@@ -466,55 +452,56 @@ __global__ void mfma_probe32_k(const ushort* A, const ushort* B, float* D) {
 }  // namespace
 
 void launch_attn_fwd(const bf16_t* q, const bf16_t* k, const bf16_t* v,
-                     bf16_t* o, float* lse, int64_t bh, int seq, int hd,
-                     float scale, hipStream_t s) {
-  dim3 grid((seq + 63) / 64, bh);
-  if (hd == 64)
-    attn_fwd_k<4><<<grid, 256, 0, s>>>(q, k, v, o, lse, seq, scale);
-  else if (hd == 128)
-    attn_fwd_k<8><<<grid, 256, 0, s>>>(q, k, v, o, lse, seq, scale);
-  else if (hd == 32)
-    attn_fwd_k<2><<<grid, 256, 0, s>>>(q, k, v, o, lse, seq, scale);
+                     bf16_t* o, float* lse, const AttnGeom& geo,
+                     hipStream_t s) {
+  dim3 grid((geo.seq + 63) / 64, int64_t(geo.B) * geo.H);
+  const int hd = geo.hd;
+  if (hd == 64) attn_fwd_k<4><<<grid, 256, 0, s>>>(q, k, v, o, lse, geo);
+  else if (hd == 128) attn_fwd_k<8><<<grid, 256, 0, s>>>(q, k, v, o, lse, geo);
+  else if (hd == 32) attn_fwd_k<2><<<grid, 256, 0, s>>>(q, k, v, o, lse, geo);
 }
 
 void launch_attn_delta(const bf16_t* dout, const bf16_t* o, float* delta,
-                       int64_t rows, int hd, hipStream_t s) {
+                       const AttnGeom& geo, hipStream_t s) {
+  int64_t rows = int64_t(geo.B) * geo.H * geo.seq;
   int64_t want = (rows + 3) / 4;
   const int grid = int(want < 4096 ? (want > 0 ? want : 1) : 4096);
-  attn_delta_k<<<grid, 256, 0, s>>>(dout, o, delta, rows, hd);
+  attn_delta_k<<<grid, 256, 0, s>>>(dout, o, delta, geo);
 }
 
 void launch_attn_bwd_dq(const bf16_t* dout, const bf16_t* q, const bf16_t* k,
                         const bf16_t* v, const float* lse, const float* delta,
-                        bf16_t* dq, int64_t bh, int seq, int hd, float scale,
-                        hipStream_t s) {
-  dim3 grid((seq + 63) / 64, bh);
-  if (hd == 64)
-    attn_bwd_dq_k<4><<<grid, 256, 0, s>>>(dout, q, k, v, lse, delta, dq, seq,
-                                          scale);
-  else if (hd == 128)
-    attn_bwd_dq_k<8><<<grid, 256, 0, s>>>(dout, q, k, v, lse, delta, dq, seq,
-                                          scale);
-  else if (hd == 32)
-    attn_bwd_dq_k<2><<<grid, 256, 0, s>>>(dout, q, k, v, lse, delta, dq, seq,
-                                          scale);
+                        bf16_t* dq, const AttnGeom& geo, hipStream_t s) {
+  dim3 grid((geo.seq + 63) / 64, int64_t(geo.B) * geo.H);
+  const int hd = geo.hd;
+  if (hd == 64) attn_bwd_dq_k<4><<<grid, 256, 0, s>>>(dout, q, k, v, lse, delta, dq, geo);
+  else if (hd == 128) attn_bwd_dq_k<8><<<grid, 256, 0, s>>>(dout, q, k, v, lse, delta, dq, geo);
+  else if (hd == 32) attn_bwd_dq_k<2><<<grid, 256, 0, s>>>(dout, q, k, v, lse, delta, dq, geo);
 }
 
 void launch_attn_bwd_dkv(const bf16_t* dout, const bf16_t* q,
                          const bf16_t* k, const bf16_t* v, const float* lse,
-                         const float* delta, bf16_t* dk, bf16_t* dv,
-                         int64_t bh, int seq, int hd, float scale,
+                         const float* delta, float* dk32, float* dv32,
+                         bf16_t* dk, bf16_t* dv, const AttnGeom& geo,
                          hipStream_t s) {
-  dim3 grid((seq + 63) / 64, bh);
-  if (hd == 64)
-    attn_bwd_dkv_k<4><<<grid, 256, 0, s>>>(dout, q, k, v, lse, delta, dk, dv,
-                                           seq, scale);
-  else if (hd == 128)
-    attn_bwd_dkv_k<8><<<grid, 256, 0, s>>>(dout, q, k, v, lse, delta, dk, dv,
-                                           seq, scale);
-  else if (hd == 32)
-    attn_bwd_dkv_k<2><<<grid, 256, 0, s>>>(dout, q, k, v, lse, delta, dk, dv,
-                                           seq, scale);
+  dim3 grid((geo.seq + 63) / 64, int64_t(geo.B) * geo.H);
+  const int hd = geo.hd;
+  const bool at = true;  // fp32 accumulation path (GQA heads fold in)
+#define DKV(D_)                                                              \
+  do {                                                                       \
+    if (at)                                                                  \
+      attn_bwd_dkv_k<D_, true><<<grid, 256, 0, s>>>(dout, q, k, v, lse,      \
+                                                    delta, dk32, dv32, dk,   \
+                                                    dv, geo);                \
+    else                                                                     \
+      attn_bwd_dkv_k<D_, false><<<grid, 256, 0, s>>>(dout, q, k, v, lse,     \
+                                                     delta, dk32, dv32, dk,  \
+                                                     dv, geo);               \
+  } while (0)
+  if (hd == 64) DKV(4);
+  else if (hd == 128) DKV(8);
+  else if (hd == 32) DKV(2);
+#undef DKV
 }
 
 void launch_mfma_probe_16(const bf16_t* A, const bf16_t* B, float* D,

@@ -287,40 +287,73 @@ Tensor rope_fwd(Tensor x, Tensor c, Tensor s) { return rope_apply(x, c, s, false
 Tensor rope_bwd(Tensor x, Tensor c, Tensor s) { return rope_apply(x, c, s, true); }
 
 // ---- attention -------------------------------------------------------------
+// q/k/v: [B,H(,Hk),S,D] views, innermost stride 1 (checked). Output o (and
+// dq) are allocated [B,S,H,D] contiguous — the caller permutes back, so a
+// transformer layer does zero transpose copies around attention.
+static void check_attn_view(const Tensor& t, const char* n) {
+  TORCH_CHECK(t.is_cuda() && t.scalar_type() == torch::kBFloat16, n,
+              " must be bf16 on GPU");
+  TORCH_CHECK(t.dim() == 4 && t.stride(3) == 1, n,
+              " must be a [B,H,S,D] view with contiguous D");
+}
+
+static AttnGeom make_geom(const Tensor& q, const Tensor& k, const Tensor& v,
+                          const Tensor& o_bshd, const Tensor& dout,
+                          double scale) {
+  AttnGeom g{};
+  g.B = int(q.size(0)); g.H = int(q.size(1));
+  g.seq = int(q.size(2)); g.hd = int(q.size(3));
+  const int hk = int(k.size(1));
+  TORCH_CHECK(g.H % hk == 0, "H must be a multiple of H_kv");
+  g.grp = g.H / hk;
+  g.scale = float(scale);
+  g.qb = q.stride(0); g.qh = q.stride(1); g.qs = q.stride(2);
+  g.kb = k.stride(0); g.kh = k.stride(1); g.ks = k.stride(2);
+  g.vb = v.stride(0); g.vh = v.stride(1); g.vs = v.stride(2);
+  // o_bshd is [B,S,H,D]: strides (batch, head, seq) = (0th, 2nd, 1st)
+  g.ob = o_bshd.stride(0); g.oh = o_bshd.stride(2); g.os_ = o_bshd.stride(1);
+  if (dout.defined()) {
+    g.db_ = dout.stride(0); g.dh = dout.stride(1); g.ds = dout.stride(2);
+  }
+  return g;
+}
+
 std::vector<Tensor> attn_fwd(Tensor q, Tensor k, Tensor v, double scale) {
-  check_bf16(q, "q"); check_bf16(k, "k"); check_bf16(v, "v");
-  const int hd = int(q.size(-1));
-  const int seq = int(q.size(-2));
-  const int64_t bh = q.numel() / (int64_t(seq) * hd);
+  check_attn_view(q, "q"); check_attn_view(k, "k"); check_attn_view(v, "v");
+  const int hd = int(q.size(3));
   TORCH_CHECK(hd == 32 || hd == 64 || hd == 128, "head dim must be 32/64/128");
-  TORCH_CHECK(seq % 16 == 0, "seq must be a multiple of 16");
-  auto o = torch::empty_like(q);
-  auto lse = torch::empty({bh, seq}, q.options().dtype(torch::kFloat32));
+  const int64_t B = q.size(0), H = q.size(1), S = q.size(2);
+  auto o = torch::empty({B, S, H, hd}, q.options());
+  auto lse = torch::empty({B * H, S}, q.options().dtype(torch::kFloat32));
+  AttnGeom geo = make_geom(q, k, v, o, Tensor(), scale);
   launch_attn_fwd(bfp(q), bfp(k), bfp(v), bfp_mut(o),
-                  lse.data_ptr<float>(), bh, seq, hd, float(scale), stream());
-  return {o, lse};
+                  lse.data_ptr<float>(), geo, stream());
+  return {o, lse};  // caller views o as [B,H,S,D] via permute(0,2,1,3)
 }
 
 std::vector<Tensor> attn_bwd(Tensor dout, Tensor q, Tensor k, Tensor v,
-                             Tensor o, Tensor lse, double scale) {
-  check_bf16(dout, "dout");
-  const int hd = int(q.size(-1));
-  const int seq = int(q.size(-2));
-  const int64_t bh = q.numel() / (int64_t(seq) * hd);
-  auto delta = torch::empty({bh, seq}, q.options().dtype(torch::kFloat32));
-  launch_attn_delta(bfp(dout), bfp(o), delta.data_ptr<float>(), bh * seq, hd,
+                             Tensor o_bshd, Tensor lse, double scale) {
+  check_attn_view(dout, "dout");
+  const int64_t B = q.size(0), H = q.size(1), S = q.size(2);
+  const int hd = int(q.size(3));
+  const int64_t Hk = k.size(1);
+  AttnGeom geo = make_geom(q, k, v, o_bshd, dout, scale);
+  auto delta = torch::empty({B * H, S}, lse.options());
+  launch_attn_delta(bfp(dout), bfp(o_bshd), delta.data_ptr<float>(), geo,
                     stream());
-  auto dq = torch::empty_like(q);
-  auto dk = torch::empty_like(k);
-  auto dv = torch::empty_like(v);
+  auto dq = torch::empty({B, S, H, hd}, q.options());
   launch_attn_bwd_dq(bfp(dout), bfp(q), bfp(k), bfp(v),
                      lse.data_ptr<float>(), delta.data_ptr<float>(),
-                     bfp_mut(dq), bh, seq, hd, float(scale), stream());
+                     bfp_mut(dq), geo, stream());
+  // dk/dv accumulate in fp32 [B,Hk,S,D] (GQA heads fold in), then cast
+  auto dk32 = torch::zeros({B, Hk, S, hd},
+                           q.options().dtype(torch::kFloat32));
+  auto dv32 = torch::zeros_like(dk32);
   launch_attn_bwd_dkv(bfp(dout), bfp(q), bfp(k), bfp(v),
                       lse.data_ptr<float>(), delta.data_ptr<float>(),
-                      bfp_mut(dk), bfp_mut(dv), bh, seq, hd, float(scale),
-                      stream());
-  return {dq, dk, dv};
+                      dk32.data_ptr<float>(), dv32.data_ptr<float>(),
+                      nullptr, nullptr, geo, stream());
+  return {dq, dk32.to(torch::kBFloat16), dv32.to(torch::kBFloat16)};
 }
 
 // ---- mfma self-test --------------------------------------------------------

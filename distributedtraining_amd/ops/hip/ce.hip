@@ -1,12 +1,12 @@
 // Fused log-softmax + cross-entropy over the vocab dimension, fwd + bwd.
 // This is the dominant memory cost of GPT-2-small training (logits are
-// [tokens, 50257] bf16); each pass streams the logits exactly once at
-// HBM rate. Replaces the reference's transformers-internal CE
+// [tokens, 50257] bf16). Replaces the reference's transformers-internal CE
 // (`labels=` path, training_manager.py:380-385).
 //
-// fwd: one block (4 waves) per row: online max+sumexp in one pass;
-//      loss_row = lse - x_target; atomicAdd of the block's loss into
-//      loss_sum; lse saved for backward.
+// fwd: one block per row, TWO phases (max pass, then sum-exp pass — the
+//      second pass hits the row in L2/L3); exp/log via the single-
+//      instruction exp2/log2 path (a fused online version was measured
+//      3x slower: the per-element rescale branch serializes on expf).
 // bwd: dlogits = scale_row * (softmax - onehot), one streaming pass.
 #include "dta_common.h"
 #include "dta_kernels.h"
@@ -23,35 +23,46 @@ __global__ void ce_fwd_k(const ushort* __restrict__ logits,
   __shared__ float lds[16];
   for (int64_t row = blockIdx.x; row < rows; row += gridDim.x) {
     const ushort* xr = logits + row * vocab;
-    const int64_t tgt = targets[row];
-    // online max + sumexp (per-thread, then block-combined)
-    float m = -INFINITY, ssum = 0.f;
+    // phase 1: row max (branch-free fmax chain, 4 partials for ILP)
+    float m0 = -INFINITY, m1 = -INFINITY, m2 = -INFINITY, m3 = -INFINITY;
     int64_t i = int64_t(threadIdx.x) * 8;
     const int64_t stride = int64_t(CE_BLOCK) * 8;
     for (; i + 8 <= vocab; i += stride) {
       s16x8 vx = *reinterpret_cast<const s16x8*>(xr + i);
-#pragma unroll
-      for (int j = 0; j < 8; ++j) {
-        float f = bf2f(ushort(vx[j]));
-        if (f > m) { ssum *= expf(m - f); m = f; }
-        ssum += expf(f - m);
-      }
+      m0 = fmaxf(m0, fmaxf(bf2f(ushort(vx[0])), bf2f(ushort(vx[1]))));
+      m1 = fmaxf(m1, fmaxf(bf2f(ushort(vx[2])), bf2f(ushort(vx[3]))));
+      m2 = fmaxf(m2, fmaxf(bf2f(ushort(vx[4])), bf2f(ushort(vx[5]))));
+      m3 = fmaxf(m3, fmaxf(bf2f(ushort(vx[6])), bf2f(ushort(vx[7]))));
     }
     if (i < vocab && i + 8 > vocab)
-      for (; i < vocab; ++i) {
-        float f = bf2f(xr[i]);
-        if (f > m) { ssum *= expf(m - f); m = f; }
-        ssum += expf(f - m);
-      }
-    // combine across block: M = max; S = sum of s_t * exp(m_t - M)
-    float M = block_max<16>(m, lds);
-    float S = block_sum<16>(ssum * expf(m - M), lds);
-    float l = M + logf(S);
+      for (; i < vocab; ++i) m0 = fmaxf(m0, bf2f(xr[i]));
+    float m = fmaxf(fmaxf(m0, m1), fmaxf(m2, m3));
+    const float M = block_max<16>(m, lds);
+    // phase 2: sum exp2((x-M)*log2e) — row now L2-resident
+    const float mb = M * LOG2E;
+    float s0 = 0.f, s1 = 0.f, s2 = 0.f, s3 = 0.f;
+    i = int64_t(threadIdx.x) * 8;
+    for (; i + 8 <= vocab; i += stride) {
+      s16x8 vx = *reinterpret_cast<const s16x8*>(xr + i);
+      s0 += __builtin_exp2f(bf2f(ushort(vx[0])) * LOG2E - mb) +
+            __builtin_exp2f(bf2f(ushort(vx[1])) * LOG2E - mb);
+      s1 += __builtin_exp2f(bf2f(ushort(vx[2])) * LOG2E - mb) +
+            __builtin_exp2f(bf2f(ushort(vx[3])) * LOG2E - mb);
+      s2 += __builtin_exp2f(bf2f(ushort(vx[4])) * LOG2E - mb) +
+            __builtin_exp2f(bf2f(ushort(vx[5])) * LOG2E - mb);
+      s3 += __builtin_exp2f(bf2f(ushort(vx[6])) * LOG2E - mb) +
+            __builtin_exp2f(bf2f(ushort(vx[7])) * LOG2E - mb);
+    }
+    if (i < vocab && i + 8 > vocab)
+      for (; i < vocab; ++i)
+        s0 += __builtin_exp2f(bf2f(xr[i]) * LOG2E - mb);
+    const float S = block_sum<16>(s0 + s1 + s2 + s3, lds);
+    const float l = M + fast_log(S);
     if (threadIdx.x == 0) {
       lse[row] = l;
+      const int64_t tgt = targets[row];
       if (tgt != ignore_index) {
-        float xt = bf2f(xr[tgt]);
-        atomicAdd(loss_sum, l - xt);
+        atomicAdd(loss_sum, l - bf2f(xr[tgt]));
         atomicAdd(count, 1);
       }
     }
@@ -68,7 +79,7 @@ __global__ void ce_bwd_k(const ushort* __restrict__ logits,
     const ushort* xr = logits + row * vocab;
     ushort* dxr = dlogits + row * vocab;
     const int64_t tgt = targets[row];
-    const float l = lse[row];
+    const float lb = lse[row] * LOG2E;
     const float sc = (tgt == ignore_index) ? 0.f : scale;
     int64_t i = int64_t(threadIdx.x) * 8;
     const int64_t stride = int64_t(CE_BLOCK) * 8;
@@ -77,15 +88,14 @@ __global__ void ce_bwd_k(const ushort* __restrict__ logits,
       s16x8 o;
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
-        float p = expf(bf2f(ushort(vx[j])) - l);
-        float g = sc * (p - ((i + j) == tgt ? 1.f : 0.f));
-        o[j] = f2bf(g);
+        float p = __builtin_exp2f(bf2f(ushort(vx[j])) * LOG2E - lb);
+        o[j] = f2bf(sc * (p - ((i + j) == tgt ? 1.f : 0.f)));
       }
       *reinterpret_cast<s16x8*>(dxr + i) = o;
     }
     if (i < vocab && i + 8 > vocab)
       for (; i < vocab; ++i) {
-        float p = expf(bf2f(xr[i]) - l);
+        float p = __builtin_exp2f(bf2f(xr[i]) * LOG2E - lb);
         dxr[i] = f2bf(sc * (p - (i == tgt ? 1.f : 0.f)));
       }
   }
@@ -96,7 +106,7 @@ __global__ void ce_bwd_k(const ushort* __restrict__ logits,
 void launch_ce_fwd(const bf16_t* logits, const int64_t* targets, int64_t rows,
                    int64_t vocab, int64_t ignore_index, float* lse,
                    float* loss_sum, int* count, hipStream_t s) {
-  const int grid = int(rows < 2048 ? (rows > 0 ? rows : 1) : 2048);
+  const int grid = int(rows < 4096 ? (rows > 0 ? rows : 1) : 4096);
   ce_fwd_k<<<grid, CE_BLOCK, 0, s>>>(logits, targets, rows, vocab,
                                      ignore_index, lse, loss_sum, count);
 }
@@ -105,7 +115,7 @@ void launch_ce_bwd(const bf16_t* logits, const int64_t* targets,
                    const float* lse, float scale, int64_t ignore_index,
                    bf16_t* dlogits, int64_t rows, int64_t vocab,
                    hipStream_t s) {
-  const int grid = int(rows < 2048 ? (rows > 0 ? rows : 1) : 2048);
+  const int grid = int(rows < 4096 ? (rows > 0 ? rows : 1) : 4096);
   ce_bwd_k<<<grid, CE_BLOCK, 0, s>>>(logits, targets, lse, scale,
                                      ignore_index, dlogits, rows, vocab);
 }

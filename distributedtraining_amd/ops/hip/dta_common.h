@@ -19,20 +19,28 @@ typedef __attribute__((ext_vector_type(8))) short    s16x8;   // 8 bf16 = 16 B
 typedef __attribute__((ext_vector_type(8))) __bf16   bf16x8;
 typedef __attribute__((ext_vector_type(4))) __bf16   bf16x4;
 
-// ---- bf16 <-> f32 ---------------------------------------------------------
+// ---- bf16 <-> f32 (hardware v_cvt; never bit-twiddle these in software) ---
 DEV float bf2f(ushort u) {
-  union { uint32_t u32; float f; } x;
-  x.u32 = uint32_t(u) << 16;
-  return x.f;
+  union { ushort s; __bf16 b; } x;
+  x.s = u;
+  return static_cast<float>(x.b);
 }
 DEV ushort f2bf(float f) {
-  union { float f; uint32_t u32; } x;
-  x.f = f;
-  uint32_t u = x.u32;
-  uint32_t lsb = (u >> 16) & 1;          // round to nearest even
-  u += 0x7fff + lsb;
-  if ((x.u32 & 0x7fffffff) > 0x7f800000) return ushort((x.u32 >> 16) | 0x40);
-  return ushort(u >> 16);
+  union { ushort s; __bf16 b; } x;
+  x.b = static_cast<__bf16>(f);
+  return x.s;
+}
+
+// ---- fast transcendentals (exp2/log2 are single hardware instructions) ----
+#define LOG2E 1.4426950408889634f
+#define LN2 0.6931471805599453f
+DEV float fast_exp(float x) { return __builtin_exp2f(x * LOG2E); }
+DEV float fast_log(float x) { return __builtin_log2f(x) * LN2; }
+DEV float fast_tanh(float x) {
+  // tanh(x) = (e-1)/(e+1), e = exp2(2x*log2e); clamp so e never overflows
+  float xc = fminf(fmaxf(x, -10.0f), 10.0f);
+  float e = __builtin_exp2f(2.0f * LOG2E * xc);
+  return (e - 1.0f) / (e + 1.0f);
 }
 
 // ---- wave reductions (64-lane) --------------------------------------------

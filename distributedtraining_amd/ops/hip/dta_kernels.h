@@ -83,19 +83,31 @@ void launch_rope(const bf16_t* x, const float* cos_t, const float* sin_t,
                  hipStream_t s);
 
 // ---- attention ------------------------------------------------------------
+// Strided geometry: q/k/v/o/dout are [B,H,S,D]-shaped views with arbitrary
+// batch/head/seq strides (innermost D contiguous); kv tensors have H/grp
+// heads (GQA). lse/delta are [B*H, S] fp32 contiguous.
+struct AttnGeom {
+  int B, H, grp, seq, hd;
+  float scale;
+  int64_t qb, qh, qs;      // q strides (batch, head, seq)
+  int64_t kb, kh, ks;      // k strides
+  int64_t vb, vh, vs;      // v strides
+  int64_t ob, oh, os_;     // o (and dq) strides
+  int64_t db_, dh, ds;     // dout strides
+};
 void launch_attn_fwd(const bf16_t* q, const bf16_t* k, const bf16_t* v,
-                     bf16_t* o, float* lse, int64_t bh, int seq, int hd,
-                     float scale, hipStream_t s);
+                     bf16_t* o, float* lse, const AttnGeom& geo,
+                     hipStream_t s);
 void launch_attn_delta(const bf16_t* dout, const bf16_t* o, float* delta,
-                       int64_t rows, int hd, hipStream_t s);
+                       const AttnGeom& geo, hipStream_t s);
 void launch_attn_bwd_dq(const bf16_t* dout, const bf16_t* q, const bf16_t* k,
                         const bf16_t* v, const float* lse, const float* delta,
-                        bf16_t* dq, int64_t bh, int seq, int hd, float scale,
-                        hipStream_t s);
-void launch_attn_bwd_dkv(const bf16_t* dout, const bf16_t* q, const bf16_t* k,
-                         const bf16_t* v, const float* lse, const float* delta,
-                         bf16_t* dk, bf16_t* dv, int64_t bh, int seq, int hd,
-                         float scale, hipStream_t s);
+                        bf16_t* dq, const AttnGeom& geo, hipStream_t s);
+void launch_attn_bwd_dkv(const bf16_t* dout, const bf16_t* q,
+                         const bf16_t* k, const bf16_t* v, const float* lse,
+                         const float* delta, float* dk32, float* dv32,
+                         bf16_t* dk, bf16_t* dv, const AttnGeom& geo,
+                         hipStream_t s);
 
 // ---- mfma layout self-test ------------------------------------------------
 // D[32,32] = A[32,16] x B[16,32] and D[16,16] = A[16,32] x B[32,16]

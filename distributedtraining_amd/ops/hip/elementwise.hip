@@ -11,18 +11,18 @@ constexpr float GELU_A = 0.044715f;
 
 DEV float gelu_f(float x) {
   float u = GELU_C * (x + GELU_A * x * x * x);
-  return 0.5f * x * (1.0f + tanhf(u));
+  return 0.5f * x * (1.0f + fast_tanh(u));
 }
 DEV float gelu_df(float x) {
   float x2 = x * x;
   float u = GELU_C * x * (1.0f + GELU_A * x2);
-  float t = tanhf(u);
+  float t = fast_tanh(u);
   float sech2 = 1.0f - t * t;
   return 0.5f * (1.0f + t) + 0.5f * x * sech2 * GELU_C * (1.0f + 3.0f * GELU_A * x2);
 }
-DEV float silu_f(float x) { return x / (1.0f + expf(-x)); }
+DEV float silu_f(float x) { return x / (1.0f + fast_exp(-x)); }
 DEV float silu_df(float x) {
-  float s = 1.0f / (1.0f + expf(-x));
+  float s = 1.0f / (1.0f + fast_exp(-x));
   return s * (1.0f + x * (1.0f - s));
 }
 

@@ -1,18 +1,22 @@
 #include "hip/hip_runtime.h"
 // Fused LayerNorm / RMSNorm forward + backward.
-// One 64-lane wave per row, values register-resident (single HBM pass);
-// bf16x8 vectorized loads (Guideline 13). Lane i owns column chunks
-// i, i+64, ... so dγ/dβ accumulate in registers across the block's rows
-// and flush with one atomicAdd per element at the end (Guideline 12).
 //
-// Replaces the implicit LayerNorm of the reference's GPT-2 forward/backward
+// fwd: one 64-lane wave per row, values register-resident, bf16x8 loads
+//      (Guideline 13); one HBM pass.
+// bwd: split in two streaming kernels (measured: a fused version spent
+//      ~95% of its time in a 2048-way atomicAdd flush of dγ/dβ):
+//        * dx kernel — pure streaming, no reductions across rows;
+//        * dγ/dβ kernel — column-parallel over ~256 row stripes,
+//          ≤256 atomic adds per output address (Guideline 12).
+//
+// Replaces the implicit LayerNorm of the reference's GPT-2 fwd/bwd
 // (transformers internals; SURVEY.md §2.2 op table).
 #include "dta_common.h"
 #include "dta_kernels.h"
 
 namespace {
 
-constexpr int ROW_WAVES = 4;  // rows processed concurrently per block
+constexpr int ROW_WAVES = 4;
 
 // ---- forward --------------------------------------------------------------
 template <int ITERS, bool RMS>
@@ -23,7 +27,7 @@ __global__ void norm_fwd_k(const ushort* __restrict__ x,
                            float* __restrict__ rstd, int64_t rows, int cols,
                            float eps) {
   const int lane = threadIdx.x & 63, wid = threadIdx.x >> 6;
-  const int nchunk = cols >> 3;  // bf16x8 chunks per row
+  const int nchunk = cols >> 3;
   for (int64_t row = int64_t(blockIdx.x) * ROW_WAVES + wid; row < rows;
        row += int64_t(gridDim.x) * ROW_WAVES) {
     const ushort* xr = x + row * cols;
@@ -78,27 +82,19 @@ __global__ void norm_fwd_k(const ushort* __restrict__ x,
   }
 }
 
-// ---- backward -------------------------------------------------------------
-// LN: xh=(x-mu)*rs; dx = rs*(dyw - mean(dyw) - xh*mean(dyw*xh)), dyw=dy*w
-// RMS: xh=x*rs;      dx = rs*(dyw - xh*mean(dyw*xh))
-// dw[c] += dy*xh ; db[c] += dy (LN only); accumulated over rows in regs.
+// ---- backward dx (streaming, no cross-row state) --------------------------
+// LN:  dx = rs*(dyw - mean(dyw) - xh*mean(dyw*xh)),  dyw = dy*w
+// RMS: dx = rs*(dyw - xh*mean(dyw*xh))
 template <int ITERS, bool RMS>
-__global__ void norm_bwd_k(const ushort* __restrict__ dy,
-                           const ushort* __restrict__ x,
-                           const ushort* __restrict__ w,
-                           const float* __restrict__ mean,
-                           const float* __restrict__ rstd,
-                           ushort* __restrict__ dx, float* __restrict__ dw,
-                           float* __restrict__ db, int64_t rows, int cols) {
+__global__ void norm_bwd_dx_k(const ushort* __restrict__ dy,
+                              const ushort* __restrict__ x,
+                              const ushort* __restrict__ w,
+                              const float* __restrict__ mean,
+                              const float* __restrict__ rstd,
+                              ushort* __restrict__ dx, int64_t rows,
+                              int cols) {
   const int lane = threadIdx.x & 63, wid = threadIdx.x >> 6;
   const int nchunk = cols >> 3;
-  float dw_acc[ITERS][8];
-  float db_acc[ITERS][8];
-#pragma unroll
-  for (int it = 0; it < ITERS; ++it)
-#pragma unroll
-    for (int j = 0; j < 8; ++j) { dw_acc[it][j] = 0.f; db_acc[it][j] = 0.f; }
-
   float wv[ITERS][8];
 #pragma unroll
   for (int it = 0; it < ITERS; ++it) {
@@ -109,15 +105,14 @@ __global__ void norm_bwd_k(const ushort* __restrict__ dy,
       for (int j = 0; j < 8; ++j) wv[it][j] = bf2f(ushort(vw[j]));
     }
   }
-
   for (int64_t row = int64_t(blockIdx.x) * ROW_WAVES + wid; row < rows;
        row += int64_t(gridDim.x) * ROW_WAVES) {
     const ushort* xr = x + row * cols;
     const ushort* dyr = dy + row * cols;
     const float mu = RMS ? 0.f : mean[row];
     const float rs = rstd[row];
-    float xh[ITERS][8], dyv[ITERS][8];
-    float s1 = 0.f, s2 = 0.f;  // sum(dyw), sum(dyw*xh)
+    float xh[ITERS][8], dyw[ITERS][8];
+    float s1 = 0.f, s2 = 0.f;
 #pragma unroll
     for (int it = 0; it < ITERS; ++it) {
       const int c = lane + it * 64;
@@ -127,16 +122,15 @@ __global__ void norm_bwd_k(const ushort* __restrict__ dy,
 #pragma unroll
         for (int j = 0; j < 8; ++j) {
           float xhj = (bf2f(ushort(vx[j])) - mu) * rs;
-          float dyj = bf2f(ushort(vdy[j]));
-          float dywj = dyj * wv[it][j];
+          float dywj = bf2f(ushort(vdy[j])) * wv[it][j];
           xh[it][j] = xhj;
-          dyv[it][j] = dyj;
+          dyw[it][j] = dywj;
           s1 += dywj;
           s2 = fmaf(dywj, xhj, s2);
         }
       } else {
 #pragma unroll
-        for (int j = 0; j < 8; ++j) { xh[it][j] = 0.f; dyv[it][j] = 0.f; }
+        for (int j = 0; j < 8; ++j) { xh[it][j] = 0.f; dyw[it][j] = 0.f; }
       }
     }
     s1 = wave_sum(s1) / cols;
@@ -149,29 +143,42 @@ __global__ void norm_bwd_k(const ushort* __restrict__ dy,
         s16x8 o;
 #pragma unroll
         for (int j = 0; j < 8; ++j) {
-          float dywj = dyv[it][j] * wv[it][j];
-          float v = RMS ? (dywj - xh[it][j] * s2)
-                        : (dywj - s1 - xh[it][j] * s2);
+          float v = RMS ? (dyw[it][j] - xh[it][j] * s2)
+                        : (dyw[it][j] - s1 - xh[it][j] * s2);
           o[j] = f2bf(rs * v);
-          dw_acc[it][j] = fmaf(dyv[it][j], xh[it][j], dw_acc[it][j]);
-          db_acc[it][j] += dyv[it][j];
         }
         *reinterpret_cast<s16x8*>(dxr + c * 8) = o;
       }
     }
   }
-  // flush per-lane column accumulators (fp32 atomics, low contention)
-#pragma unroll
-  for (int it = 0; it < ITERS; ++it) {
-    const int c = lane + it * 64;
-    if (c < nchunk) {
-#pragma unroll
-      for (int j = 0; j < 8; ++j) {
-        atomicAdd(dw + c * 8 + j, dw_acc[it][j]);
-        if (!RMS) atomicAdd(db + c * 8 + j, db_acc[it][j]);
-      }
-    }
+}
+
+// ---- backward dγ/dβ (column-parallel row reduction) ------------------------
+// grid = (ceil(cols/256), N_STRIPES); thread owns one column, walks its
+// stripe's rows; one atomicAdd per (stripe, column).
+template <bool RMS>
+__global__ void norm_bwd_dwdb_k(const ushort* __restrict__ dy,
+                                const ushort* __restrict__ x,
+                                const float* __restrict__ mean,
+                                const float* __restrict__ rstd,
+                                float* __restrict__ dw,
+                                float* __restrict__ db, int64_t rows,
+                                int cols) {
+  const int c = blockIdx.x * blockDim.x + threadIdx.x;
+  if (c >= cols) return;
+  const int64_t r0 = (rows * blockIdx.y) / gridDim.y;
+  const int64_t r1 = (rows * (blockIdx.y + 1)) / gridDim.y;
+  float dw_acc = 0.f, db_acc = 0.f;
+  for (int64_t r = r0; r < r1; ++r) {
+    const float mu = RMS ? 0.f : mean[r];
+    const float rs = rstd[r];
+    const float xv = bf2f(x[r * cols + c]);
+    const float dyv = bf2f(dy[r * cols + c]);
+    dw_acc = fmaf(dyv, (xv - mu) * rs, dw_acc);
+    db_acc += dyv;
   }
+  atomicAdd(dw + c, dw_acc);
+  if (!RMS) atomicAdd(db + c, db_acc);
 }
 
 template <bool RMS>
@@ -191,7 +198,6 @@ void dispatch_fwd(const ushort* x, const ushort* w, const ushort* b,
   switch (iters) {
     CASE_F(1) CASE_F(2) CASE_F(3) CASE_F(4) CASE_F(6) CASE_F(8) CASE_F(16)
     default: {
-      // generic fallback for odd sizes: round up to next supported
       if (iters <= 6) {hipLaunchKernelGGL(( norm_fwd_k<6, RMS>), dim3(grid), dim3(blk), 0, s, x, w, b, y, mean, rstd, rows, cols, eps); }
       else if (iters <= 8) {hipLaunchKernelGGL(( norm_fwd_k<8, RMS>), dim3(grid), dim3(blk), 0, s, x, w, b, y, mean, rstd, rows, cols, eps); }
       else {hipLaunchKernelGGL(( norm_fwd_k<16, RMS>), dim3(grid), dim3(blk), 0, s, x, w, b, y, mean, rstd, rows, cols, eps); }
@@ -207,20 +213,25 @@ void dispatch_bwd(const ushort* dy, const ushort* x, const ushort* w,
   const int nchunk = cols >> 3;
   const int iters = (nchunk + 63) / 64;
   int64_t want = (rows + ROW_WAVES - 1) / ROW_WAVES;
-  const int grid = int(want < 512 ? (want > 0 ? want : 1) : 512);
+  const int grid = int(want < 4096 ? (want > 0 ? want : 1) : 4096);
   const dim3 blk(64 * ROW_WAVES);
-#define CASE_B(I)                                                          \
-  case I:                                                                  \
-   hipLaunchKernelGGL(( norm_bwd_k<I, RMS>), dim3(grid), dim3(blk), 0, s, dy, x, w, mean, rstd, dx, dw,  \
-                                            db, rows, cols);               \
+#define CASE_B(I)                                                            \
+  case I:                                                                    \
+   hipLaunchKernelGGL(( norm_bwd_dx_k<I, RMS>), dim3(grid), dim3(blk), 0, s, dy, x, w, mean, rstd, dx,     \
+                                               rows, cols);                  \
     break;
   switch (iters) {
     CASE_B(1) CASE_B(2) CASE_B(3) CASE_B(4) CASE_B(6) CASE_B(8)
     default:
-     hipLaunchKernelGGL(( norm_bwd_k<8, RMS>), dim3(grid), dim3(blk), 0, s, dy, x, w, mean, rstd, dx, dw,
-                                              db, rows, cols);
+     hipLaunchKernelGGL(( norm_bwd_dx_k<8, RMS>), dim3(grid), dim3(blk), 0, s, dy, x, w, mean, rstd, dx,
+                                                 rows, cols);
   }
 #undef CASE_B
+  int stripes = int(rows >= 8192 ? 256 : (rows + 31) / 32);
+  if (stripes < 1) stripes = 1;
+  dim3 g2((cols + 255) / 256, stripes);
+ hipLaunchKernelGGL(( norm_bwd_dwdb_k<RMS>), dim3(g2), dim3(256), 0, s, dy, x, mean, rstd, dw, db, rows,
+                                          cols);
 }
 
 }  // namespace
