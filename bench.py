@@ -22,6 +22,18 @@ import os
 import sys
 import time
 
+# hipBLASLt algorithm selection via TunableOp: pre-tuned entries for the
+# flagship shapes live in tuning/tunableop<dev>.csv (committed); unknown
+# shapes are tuned during the untimed warmup. Must be set before torch
+# loads its backends.
+_here = os.path.dirname(os.path.abspath(__file__))
+if os.path.exists(os.path.join(_here, "tuning", "tunableop0.csv")):
+    os.environ.setdefault("PYTORCH_TUNABLEOP_ENABLED", "1")
+    os.environ.setdefault("PYTORCH_TUNABLEOP_TUNING", "1")
+    os.environ.setdefault("PYTORCH_TUNABLEOP_FILENAME",
+                          os.path.join(_here, "tuning", "tunableop.csv"))
+    os.environ.setdefault("PYTORCH_TUNABLEOP_VERBOSE", "0")
+
 import torch
 
 
@@ -30,7 +42,7 @@ def main() -> int:
     p.add_argument("--gpus", type=int, default=1)
     p.add_argument("--steps", type=int, default=30)
     p.add_argument("--warmup", type=int, default=10)
-    p.add_argument("--batch-size", type=int, default=256)
+    p.add_argument("--batch-size", type=int, default=512)
     p.add_argument("--seq-len", type=int, default=64)
     p.add_argument("--merge-every", type=int, default=25)
     p.add_argument("--model", default="gpt2-small",

@@ -402,10 +402,10 @@ __global__ void attn_bwd_dkv_k(const ushort* __restrict__ dout,
         atomicAdd(dk32 + off, acck[t][r] * geo.scale);
         atomicAdd(dv32 + off, accv[t][r]);
       } else {
-        ushort* dkp = dk + b * geo.kb + hk * geo.kh;
-        ushort* dvp = dv + b * geo.vb + hk * geo.vh;
-        dkp[int64_t(orow) * geo.ks + col] = f2bf(acck[t][r] * geo.scale);
-        dvp[int64_t(orow) * geo.vs + col] = f2bf(accv[t][r]);
+        ushort* dkp = dk + b * geo.gkb + hk * geo.gkh;
+        ushort* dvp = dv + b * geo.gkb + hk * geo.gkh;
+        dkp[int64_t(orow) * geo.gks + col] = f2bf(acck[t][r] * geo.scale);
+        dvp[int64_t(orow) * geo.gks + col] = f2bf(accv[t][r]);
       }
     }
 }
@@ -487,7 +487,8 @@ void launch_attn_bwd_dkv(const bf16_t* dout, const bf16_t* q,
                          hipStream_t s) {
   dim3 grid((geo.seq + 63) / 64, int64_t(geo.B) * geo.H);
   const int hd = geo.hd;
-  const bool at = true;  // fp32 accumulation path (GQA heads fold in)
+  const bool at = geo.grp > 1;  // GQA folds heads via fp32 atomics;
+                                // grp==1 stores bf16 directly
 #define DKV(D_)                                                              \
   do {                                                                       \
     if (at)                                                                  \

@@ -345,7 +345,18 @@ std::vector<Tensor> attn_bwd(Tensor dout, Tensor q, Tensor k, Tensor v,
   launch_attn_bwd_dq(bfp(dout), bfp(q), bfp(k), bfp(v),
                      lse.data_ptr<float>(), delta.data_ptr<float>(),
                      bfp_mut(dq), geo, stream());
-  // dk/dv accumulate in fp32 [B,Hk,S,D] (GQA heads fold in), then cast
+  if (geo.grp == 1) {
+    // direct bf16 store into [B,S,Hk,D]; caller permutes to [B,Hk,S,D]
+    auto dk = torch::empty({B, S, Hk, hd}, q.options());
+    auto dv = torch::empty({B, S, Hk, hd}, q.options());
+    geo.gkb = dk.stride(0); geo.gkh = dk.stride(2); geo.gks = dk.stride(1);
+    launch_attn_bwd_dkv(bfp(dout), bfp(q), bfp(k), bfp(v),
+                        lse.data_ptr<float>(), delta.data_ptr<float>(),
+                        nullptr, nullptr, bfp_mut(dk), bfp_mut(dv), geo,
+                        stream());
+    return {dq, dk.permute({0, 2, 1, 3}), dv.permute({0, 2, 1, 3})};
+  }
+  // GQA: accumulate in fp32 [B,Hk,S,D] (q heads fold in), then cast
   auto dk32 = torch::zeros({B, Hk, S, hd},
                            q.options().dtype(torch::kFloat32));
   auto dv32 = torch::zeros_like(dk32);

@@ -94,6 +94,7 @@ struct AttnGeom {
   int64_t vb, vh, vs;      // v strides
   int64_t ob, oh, os_;     // o (and dq) strides
   int64_t db_, dh, ds;     // dout strides
+  int64_t gkb, gkh, gks;   // dk/dv output strides (grp==1 direct-store path)
 };
 void launch_attn_fwd(const bf16_t* q, const bf16_t* k, const bf16_t* v,
                      bf16_t* o, float* lse, const AttnGeom& geo,

@@ -139,29 +139,39 @@ def test_swiglu():
 # ---------------------------------------------------------------------------
 # Attention
 # ---------------------------------------------------------------------------
-@pytest.mark.parametrize("B,H,S,D", [(2, 3, 64, 64), (1, 2, 512, 64),
-                                     (2, 2, 128, 128), (1, 1, 48, 32)])
-def test_attention_fwd_bwd(B, H, S, D):
-    m = _ext()
+@pytest.mark.parametrize("B,H,Hkv,S,D", [(2, 3, 3, 64, 64),
+                                         (1, 2, 2, 512, 64),
+                                         (2, 2, 2, 128, 128),
+                                         (1, 1, 1, 48, 32),
+                                         (2, 8, 2, 64, 128)])  # GQA
+def test_attention_fwd_bwd(B, H, Hkv, S, D):
+    """Through the autograd wrapper (as the models call it), strided
+    [B,H,S,D] views included."""
+    from distributedtraining_amd import ops
     scale = 1.0 / math.sqrt(D)
-    q = _rand_bf16(B, H, S, D, seed=10)
-    k = _rand_bf16(B, H, S, D, seed=11)
-    v = _rand_bf16(B, H, S, D, seed=12)
-    o, lse = m.attn_fwd(q, k, v, scale)
+    q = _rand_bf16(B, S, H * D, seed=10).view(B, S, H, D) \
+        .transpose(1, 2).requires_grad_(True)     # strided view like models
+    k = _rand_bf16(B, S, Hkv * D, seed=11).view(B, S, Hkv, D) \
+        .transpose(1, 2).requires_grad_(True)
+    v = _rand_bf16(B, S, Hkv * D, seed=12).view(B, S, Hkv, D) \
+        .transpose(1, 2).requires_grad_(True)
+    o = ops.causal_attention(q, k, v, scale)
     ref = torch.nn.functional.scaled_dot_product_attention(
-        q.float(), k.float(), v.float(), is_causal=True, scale=scale)
+        q.detach().float(), k.detach().float(), v.detach().float(),
+        is_causal=True, scale=scale, enable_gqa=True)
     torch.testing.assert_close(o.float(), ref, rtol=3e-2, atol=3e-2)
 
     do = _rand_bf16(B, H, S, D, seed=13)
-    qr = q.float().detach().requires_grad_(True)
-    kr = k.float().detach().requires_grad_(True)
-    vr = v.float().detach().requires_grad_(True)
+    o.backward(do)
+    qr = q.detach().float().requires_grad_(True)
+    kr = k.detach().float().requires_grad_(True)
+    vr = v.detach().float().requires_grad_(True)
     torch.nn.functional.scaled_dot_product_attention(
-        qr, kr, vr, is_causal=True, scale=scale).backward(do.float())
-    dq, dk, dv = m.attn_bwd(do, q, k, v, o, lse, scale)
-    torch.testing.assert_close(dq.float(), qr.grad, rtol=5e-2, atol=5e-2)
-    torch.testing.assert_close(dk.float(), kr.grad, rtol=5e-2, atol=5e-2)
-    torch.testing.assert_close(dv.float(), vr.grad, rtol=5e-2, atol=5e-2)
+        qr, kr, vr, is_causal=True, scale=scale,
+        enable_gqa=True).backward(do.float())
+    torch.testing.assert_close(q.grad.float(), qr.grad, rtol=5e-2, atol=5e-2)
+    torch.testing.assert_close(k.grad.float(), kr.grad, rtol=5e-2, atol=6e-2)
+    torch.testing.assert_close(v.grad.float(), vr.grad, rtol=5e-2, atol=5e-2)
 
 
 # ---------------------------------------------------------------------------
