@@ -48,6 +48,8 @@ def main() -> int:
     p.add_argument("--model", default="gpt2-small",
                    choices=["gpt2-small", "gpt2-tiny", "llama3-8b"])
     p.add_argument("--merge-strategy", default="mean")
+    p.add_argument("--no-graph", action="store_true",
+                   help="disable hipGraph step capture (eager launches)")
     p.add_argument("--cpu", action="store_true",
                    help="CPU smoke mode (tiny model, tests only)")
     args = p.parse_args()
@@ -96,9 +98,27 @@ def main() -> int:
             torch.cuda.synchronize()
         comm.barrier()
 
+    # Device-resident batch pool: synthetic data staged once, recycled in a
+    # ring (the timed region still does a fresh device copy per step).
+    pool = [{k: t.to(device) for k, t in next(node.miner.data).items()}
+            for _ in range(8)]
+    graphed = None
+    if use_gpu and not args.no_graph:
+        try:
+            from distributedtraining_amd.parallel.graphstep import (
+                GraphedMinerStep)
+            graphed = GraphedMinerStep(node.miner, pool, warmup=3)
+        except Exception as e:  # pragma: no cover - graph unsupported
+            print(f"WARN: hipGraph capture failed ({e!r}); eager path",
+                  file=sys.stderr)
+            graphed = None
+
     def run_steps(n: int, base_step: int) -> None:
         for i in range(n):
-            node.miner.train_step()
+            if graphed is not None:
+                graphed.step(pool[i % len(pool)])
+            else:
+                node.miner.train_step(pool[i % len(pool)])
             step = base_step + i + 1
             if args.merge_every and step % args.merge_every == 0:
                 node.merge_round()

@@ -49,20 +49,13 @@ class GPT2Block(nn.Module):
         self.mlp_proj_b = nn.Parameter(torch.zeros(E))
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
-        B, S, E = x.shape
-        H = self.n_head
         h = ops.layer_norm(x, self.ln_1_w, self.ln_1_b)
-        qkv = F.linear(h, self.attn_qkv_w, self.attn_qkv_b)
-        q, k, v = qkv.split(E, dim=-1)
-        q = q.view(B, S, H, E // H).transpose(1, 2)
-        k = k.view(B, S, H, E // H).transpose(1, 2)
-        v = v.view(B, S, H, E // H).transpose(1, 2)
-        o = ops.causal_attention(q, k, v)
-        o = o.transpose(1, 2).reshape(B, S, E)
-        x = x + F.linear(o, self.attn_proj_w, self.attn_proj_b)
+        qkv = ops.linear(h, self.attn_qkv_w, self.attn_qkv_b)
+        o = ops.qkv_attention(qkv, self.n_head)
+        x = x + ops.linear(o, self.attn_proj_w, self.attn_proj_b)
         h = ops.layer_norm(x, self.ln_2_w, self.ln_2_b)
-        h = ops.gelu(F.linear(h, self.mlp_fc_w, self.mlp_fc_b))
-        x = x + F.linear(h, self.mlp_proj_w, self.mlp_proj_b)
+        h = ops.gelu(ops.linear(h, self.mlp_fc_w, self.mlp_fc_b))
+        x = x + ops.linear(h, self.mlp_proj_w, self.mlp_proj_b)
         return x
 
 
@@ -109,10 +102,10 @@ class GPT2LM(nn.Module):
             x = blk(x)
         x = ops.layer_norm(x, self.ln_f_w, self.ln_f_b)
         if labels is None:
-            logits = F.linear(x, self.wte)
+            logits = ops.linear(x, self.wte)
             return CausalLMOutput(loss=None, logits=logits)
         # shifted CE: predict token t+1 from position t (HF semantics)
-        logits = F.linear(x[:, :-1, :], self.wte)
+        logits = ops.linear(x[:, :-1, :].contiguous(), self.wte)
         tgt = labels[:, 1:].contiguous().view(-1)
         loss = ops.cross_entropy_loss(
             logits.reshape(-1, self.cfg.vocab_size), tgt)

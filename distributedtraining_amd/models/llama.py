@@ -55,17 +55,17 @@ class LlamaBlock(nn.Module):
         B, S, E = x.shape
         D = self.head_dim
         h = ops.rms_norm(x, self.attn_norm_w, self.norm_eps)
-        q = F.linear(h, self.q_w).view(B, S, self.n_head, D).transpose(1, 2)
-        k = F.linear(h, self.k_w).view(B, S, self.n_kv, D).transpose(1, 2)
-        v = F.linear(h, self.v_w).view(B, S, self.n_kv, D).transpose(1, 2)
+        q = ops.linear(h, self.q_w).view(B, S, self.n_head, D).transpose(1, 2)
+        k = ops.linear(h, self.k_w).view(B, S, self.n_kv, D).transpose(1, 2)
+        v = ops.linear(h, self.v_w).view(B, S, self.n_kv, D).transpose(1, 2)
         q = ops.rope(q, cos, sin)
         k = ops.rope(k, cos, sin)
         o = ops.causal_attention(q, k, v)
         o = o.transpose(1, 2).reshape(B, S, E)
-        x = x + F.linear(o, self.o_w)
+        x = x + ops.linear(o, self.o_w)
         h = ops.rms_norm(x, self.mlp_norm_w, self.norm_eps)
-        h = ops.swiglu(F.linear(h, self.gate_w), F.linear(h, self.up_w))
-        x = x + F.linear(h, self.down_w)
+        h = ops.swiglu(ops.linear(h, self.gate_w), ops.linear(h, self.up_w))
+        x = x + ops.linear(h, self.down_w)
         return x
 
 
@@ -114,8 +114,8 @@ class LlamaLM(nn.Module):
             x = blk(x, cos, sin)
         x = ops.rms_norm(x, self.final_norm_w, self.cfg.norm_eps)
         if labels is None:
-            return CausalLMOutput(loss=None, logits=F.linear(x, self._head()))
-        logits = F.linear(x[:, :-1, :], self._head())
+            return CausalLMOutput(loss=None, logits=ops.linear(x, self._head()))
+        logits = ops.linear(x[:, :-1, :].contiguous(), self._head())
         tgt = labels[:, 1:].contiguous().view(-1)
         loss = ops.cross_entropy_loss(
             logits.reshape(-1, self.cfg.vocab_size), tgt)

@@ -29,10 +29,50 @@ from .backend import has_ext, require_ext, use_hip
 
 __all__ = [
     "layer_norm", "rms_norm", "gelu", "swiglu", "causal_attention",
-    "cross_entropy_loss", "embedding_fwd", "rope", "adamw_step",
-    "delta_sub", "axpy_", "weighted_merge", "grad_merge_weights",
-    "has_nan", "l2norm",
+    "qkv_attention", "linear", "cross_entropy_loss", "embedding_fwd",
+    "rope", "adamw_step", "delta_sub", "axpy_", "weighted_merge",
+    "grad_merge_weights", "has_nan", "l2norm",
 ]
+
+
+# --------------------------------------------------------------------------
+# Linear (library GEMM via hipBLASLt) with fast dbias backward
+# --------------------------------------------------------------------------
+class _LinearFn(torch.autograd.Function):
+    """F.linear semantics, but the bias gradient is computed by our
+    vectorized colsum kernel instead of torch's generic reduce_kernel
+    (~4x faster at the GPT-2 shapes; 48 dbias reductions per step)."""
+
+    @staticmethod
+    def forward(ctx, x, w, b):
+        x2 = x.reshape(-1, x.shape[-1])
+        ctx.save_for_backward(x2, w)
+        ctx.has_b = b is not None
+        ctx.xshape = x.shape
+        y = torch.addmm(b, x2, w.t()) if b is not None else x2.mm(w.t())
+        return y.view(*x.shape[:-1], w.shape[0])
+
+    @staticmethod
+    def backward(ctx, dy):
+        x2, w = ctx.saved_tensors
+        dy2 = dy.reshape(-1, dy.shape[-1])
+        if dy2.stride(-1) != 1:
+            dy2 = dy2.contiguous()
+        dx = dy2.mm(w).view(ctx.xshape)
+        dw = dy2.t().mm(x2)
+        db = None
+        if ctx.has_b:
+            db = require_ext().colsum(dy2).to(w.dtype)
+        return dx, dw, db
+
+
+def linear(x: torch.Tensor, w: torch.Tensor,
+           b: Optional[torch.Tensor] = None) -> torch.Tensor:
+    """y = x @ w.T (+ b). hipBLASLt GEMM with fused bias epilogue forward;
+    custom colsum dbias backward on GPU."""
+    if use_hip(x):
+        return _LinearFn.apply(x, w, b)
+    return F.linear(x, w, b)
 
 
 # --------------------------------------------------------------------------
@@ -191,25 +231,102 @@ def causal_attention(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
 
 
 # --------------------------------------------------------------------------
+# Packed-QKV causal attention: one [B,S,(H+2Hk)·D] projection in, [B,S,H·D]
+# out. The kernels read q/k/v as strided views of the packed buffer and the
+# backward stores dq/dk/dv directly into one dqkv buffer — no split-cat, no
+# transpose copies anywhere around attention (torch's CatArrayBatchedCopy
+# was 12 launches/step on GPT-2).
+# --------------------------------------------------------------------------
+def _qkv_views(t: torch.Tensor, H: int, Hk: int, D: int):
+    B, S, Fdim = t.shape
+    sb, ss = S * Fdim, Fdim
+    o0 = t.storage_offset()
+    q = t.as_strided((B, H, S, D), (sb, D, ss, 1), o0)
+    k = t.as_strided((B, Hk, S, D), (sb, D, ss, 1), o0 + H * D)
+    v = t.as_strided((B, Hk, S, D), (sb, D, ss, 1), o0 + (H + Hk) * D)
+    return q, k, v
+
+
+class _QKVAttnFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, qkv, H, Hk, scale):
+        m = require_ext()
+        B, S, Fdim = qkv.shape
+        D = Fdim // (H + 2 * Hk)
+        q, k, v = _qkv_views(qkv, H, Hk, D)
+        o_bshd, lse = m.attn_fwd(q, k, v, scale)   # [B,S,H,D] contiguous
+        ctx.save_for_backward(qkv, o_bshd, lse)
+        ctx.geom = (H, Hk, D, scale)
+        return o_bshd.view(B, S, H * D)
+
+    @staticmethod
+    def backward(ctx, do):
+        m = require_ext()
+        qkv, o_bshd, lse = ctx.saved_tensors
+        H, Hk, D, scale = ctx.geom
+        B, S, Fdim = qkv.shape
+        q, k, v = _qkv_views(qkv, H, Hk, D)
+        do4 = do.view(B, S, H, D).permute(0, 2, 1, 3)
+        if do4.stride(3) != 1:
+            do4 = do.contiguous().view(B, S, H, D).permute(0, 2, 1, 3)
+        dqkv = torch.empty_like(qkv)
+        sb, ss = S * Fdim, Fdim
+        o0 = dqkv.storage_offset()
+        dq_v = dqkv.as_strided((B, S, H, D), (sb, ss, D, 1), o0)
+        dk_v = dqkv.as_strided((B, S, Hk, D), (sb, ss, D, 1), o0 + H * D)
+        dv_v = dqkv.as_strided((B, S, Hk, D), (sb, ss, D, 1),
+                               o0 + (H + Hk) * D)
+        m.attn_bwd_packed(do4, q, k, v, o_bshd, lse, scale, dq_v, dk_v, dv_v)
+        return dqkv, None, None, None
+
+
+def qkv_attention(qkv: torch.Tensor, n_head: int,
+                  n_kv_head: Optional[int] = None,
+                  scale: Optional[float] = None) -> torch.Tensor:
+    """Causal attention over a packed qkv projection [B,S,(H+2Hk)·D] →
+    [B,S,H·D]. GQA when n_kv_head < n_head."""
+    Hk = n_kv_head or n_head
+    B, S, Fdim = qkv.shape
+    D = Fdim // (n_head + 2 * Hk)
+    if scale is None:
+        scale = 1.0 / math.sqrt(D)
+    if use_hip(qkv):
+        return _QKVAttnFn.apply(qkv, n_head, Hk, scale)
+    E = n_head * D
+    kvd = Hk * D
+    q = qkv[..., :E].view(B, S, n_head, D).transpose(1, 2)
+    k = qkv[..., E:E + kvd].view(B, S, Hk, D).transpose(1, 2)
+    v = qkv[..., E + kvd:].view(B, S, Hk, D).transpose(1, 2)
+    o = F.scaled_dot_product_attention(q, k, v, is_causal=True, scale=scale,
+                                       enable_gqa=True)
+    return o.transpose(1, 2).reshape(B, S, E)
+
+
+# --------------------------------------------------------------------------
 # Fused log-softmax + cross entropy over the vocab dim
 # --------------------------------------------------------------------------
 class _CrossEntropyFn(torch.autograd.Function):
+    """Sync-free on GPU: the non-ignored count stays a device scalar and the
+    backward scale (dloss/count) is computed on device, so the whole loss
+    path is hipGraph-capturable."""
+
     @staticmethod
     def forward(ctx, logits, targets, ignore_index):
         m = require_ext()
         logits = logits.contiguous()
         loss_sum, lse, count = m.ce_fwd(logits, targets, ignore_index)
-        ctx.save_for_backward(logits, targets, lse)
+        countf = count.clamp(min=1).to(torch.float32)
+        ctx.save_for_backward(logits, targets, lse, countf)
         ctx.ignore_index = ignore_index
-        ctx.count = max(int(count), 1)
-        return loss_sum / ctx.count
+        return loss_sum / countf
 
     @staticmethod
     def backward(ctx, dloss):
         m = require_ext()
-        logits, targets, lse = ctx.saved_tensors
-        scale = float(dloss) / ctx.count
-        dlogits = m.ce_bwd(logits, targets, lse, scale, ctx.ignore_index)
+        logits, targets, lse, countf = ctx.saved_tensors
+        scale_dev = (dloss.to(torch.float32) / countf).reshape(1)
+        dlogits = m.ce_bwd(logits, targets, lse, scale_dev, 0.0,
+                           ctx.ignore_index)
         return dlogits, None, None
 
 
@@ -291,13 +408,17 @@ def rope(x: torch.Tensor, cos: torch.Tensor, sin: torch.Tensor) -> torch.Tensor:
 def adamw_step(master: torch.Tensor, grad: torch.Tensor, m: torch.Tensor,
                v: torch.Tensor, out_bf16: Optional[torch.Tensor], step: int,
                lr: float, beta1: float = 0.9, beta2: float = 0.999,
-               eps: float = 1e-8, weight_decay: float = 0.01) -> None:
+               eps: float = 1e-8, weight_decay: float = 0.01,
+               bc: Optional[torch.Tensor] = None) -> None:
     """Decoupled AdamW (torch semantics): in-place update of fp32 master,
-    m, v; optionally writes the bf16 working copy."""
+    m, v; optionally writes the bf16 working copy. ``bc`` (fp32[2] device
+    buffer from adamw_tick) replaces host bias correction for the
+    hipGraph-capturable path."""
     if use_hip(master):
         require_ext().adamw_step(master, grad, m, v,
                                  out_bf16 if out_bf16 is not None else master.new_empty(0).to(torch.bfloat16),
-                                 step, lr, beta1, beta2, eps, weight_decay)
+                                 step, lr, beta1, beta2, eps, weight_decay,
+                                 bc if bc is not None else master.new_empty(0))
         return
     g = grad.float()
     master.mul_(1.0 - lr * weight_decay)
@@ -309,6 +430,13 @@ def adamw_step(master: torch.Tensor, grad: torch.Tensor, m: torch.Tensor,
     master.addcdiv_(m, denom, value=-lr / bc1)
     if out_bf16 is not None:
         out_bf16.copy_(master.to(out_bf16.dtype))
+
+
+def adamw_tick(t_dev: torch.Tensor, bc_dev: torch.Tensor,
+               beta1: float, beta2: float) -> None:
+    """Advance the device-side AdamW step counter and refresh the bias
+    correction buffer (graph-capturable; see adamw_step's ``bc``)."""
+    require_ext().adamw_tick(t_dev, bc_dev, beta1, beta2)
 
 
 # --------------------------------------------------------------------------

@@ -7,14 +7,19 @@
 
 namespace {
 
+// bc_p: optional device pointer {bc1, bc2} maintained by adamw_tick_k so
+// the step is hipGraph-capturable (step counter lives on device); if null
+// the host-computed bc1/bc2 scalars are used.
 template <bool GRAD_BF16, bool WRITE_BF16>
 __global__ void adamw_k(float* __restrict__ w, const void* __restrict__ gp,
                         float* __restrict__ m, float* __restrict__ v,
                         ushort* __restrict__ wout, float lr, float beta1,
                         float beta2, float eps, float wd, float bc1,
-                        float bc2, int64_t n) {
+                        float bc2, const float* __restrict__ bc_p,
+                        int64_t n) {
   int64_t i = (int64_t(blockIdx.x) * blockDim.x + threadIdx.x) * 4;
   const int64_t stride = int64_t(gridDim.x) * blockDim.x * 4;
+  if (bc_p) { bc1 = bc_p[0]; bc2 = bc_p[1]; }
   const float decay = 1.0f - lr * wd;
   const float step_size = lr / bc1;
   const float inv_bc2 = 1.0f / bc2;
@@ -61,33 +66,48 @@ __global__ void adamw_k(float* __restrict__ w, const void* __restrict__ gp,
     }
 }
 
+// Device-side step counter tick: t += 1; bc = {1-beta1^t, 1-beta2^t}.
+// Launched (1,1) right before adamw_k inside a captured graph.
+__global__ void adamw_tick_k(int* __restrict__ t, float* __restrict__ bc,
+                             float beta1, float beta2) {
+  const int nt = *t + 1;
+  *t = nt;
+  bc[0] = 1.0f - powf(beta1, float(nt));
+  bc[1] = 1.0f - powf(beta2, float(nt));
+}
+
 }  // namespace
 
+void launch_adamw_tick(int* t, float* bc, float beta1, float beta2,
+                       hipStream_t s) {
+  adamw_tick_k<<<1, 1, 0, s>>>(t, bc, beta1, beta2);
+}
+
 void launch_adamw(float* master, const void* grad, bool grad_is_bf16,
-                  float* m, float* v, bf16_t* out_bf16, int step, float lr,
-                  float beta1, float beta2, float eps, float wd, int64_t n,
-                  hipStream_t s) {
-  const float bc1 = 1.0f - powf(beta1, float(step));
-  const float bc2 = 1.0f - powf(beta2, float(step));
+                  float* m, float* v, bf16_t* out_bf16, int step,
+                  const float* bc_p, float lr, float beta1, float beta2,
+                  float eps, float wd, int64_t n, hipStream_t s) {
+  const float bc1 = bc_p ? 1.0f : 1.0f - powf(beta1, float(step));
+  const float bc2 = bc_p ? 1.0f : 1.0f - powf(beta2, float(step));
   const int block = 256;
   const int grid = elementwise_grid(n, block, 4);
   if (grad_is_bf16) {
     if (out_bf16)
       adamw_k<true, true><<<grid, block, 0, s>>>(master, grad, m, v, out_bf16,
                                                  lr, beta1, beta2, eps, wd,
-                                                 bc1, bc2, n);
+                                                 bc1, bc2, bc_p, n);
     else
       adamw_k<true, false><<<grid, block, 0, s>>>(master, grad, m, v, nullptr,
                                                   lr, beta1, beta2, eps, wd,
-                                                  bc1, bc2, n);
+                                                  bc1, bc2, bc_p, n);
   } else {
     if (out_bf16)
       adamw_k<false, true><<<grid, block, 0, s>>>(master, grad, m, v, out_bf16,
                                                   lr, beta1, beta2, eps, wd,
-                                                  bc1, bc2, n);
+                                                  bc1, bc2, bc_p, n);
     else
       adamw_k<false, false><<<grid, block, 0, s>>>(master, grad, m, v, nullptr,
                                                    lr, beta1, beta2, eps, wd,
-                                                   bc1, bc2, n);
+                                                   bc1, bc2, bc_p, n);
   }
 }

@@ -151,16 +151,37 @@ double l2norm_sq(Tensor x) {
   return out.item<float>();
 }
 
+// bc: empty tensor => host bias-correction from 'step'; else a fp32[2]
+// device buffer maintained by adamw_tick (hipGraph-capturable path).
 void adamw_step(Tensor master, Tensor grad, Tensor m, Tensor v,
                 Tensor out_bf16, int64_t step, double lr, double beta1,
-                double beta2, double eps, double wd) {
+                double beta2, double eps, double wd, Tensor bc) {
   check_f32(master, "master"); check_f32(m, "m"); check_f32(v, "v");
   const bool gb = grad.scalar_type() == torch::kBFloat16;
   bf16_t* outp = out_bf16.numel() ? bfp_mut(out_bf16) : nullptr;
+  const float* bcp = bc.numel() ? bc.data_ptr<float>() : nullptr;
   launch_adamw(master.data_ptr<float>(), grad.data_ptr(), gb,
                m.data_ptr<float>(), v.data_ptr<float>(), outp, int(step),
-               float(lr), float(beta1), float(beta2), float(eps), float(wd),
-               master.numel(), stream());
+               bcp, float(lr), float(beta1), float(beta2), float(eps),
+               float(wd), master.numel(), stream());
+}
+
+void adamw_tick(Tensor t, Tensor bc, double beta1, double beta2) {
+  TORCH_CHECK(t.scalar_type() == torch::kInt32 && t.is_cuda(), "t int32 gpu");
+  check_f32(bc, "bc");
+  launch_adamw_tick(t.data_ptr<int>(), bc.data_ptr<float>(), float(beta1),
+                    float(beta2), stream());
+}
+
+// column sum of a [rows, cols] bf16 matrix -> fp32 [cols] (bias gradient)
+Tensor colsum(Tensor x) {
+  check_bf16(x, "x");
+  TORCH_CHECK(x.dim() == 2 && x.size(1) % 8 == 0,
+              "x must be [rows, cols] with cols % 8 == 0");
+  auto out = torch::zeros({x.size(1)}, x.options().dtype(torch::kFloat32));
+  launch_colsum(bfp(x), out.data_ptr<float>(), x.size(0), int(x.size(1)),
+                stream());
+  return out;
 }
 
 // ---- merge plane -----------------------------------------------------------
@@ -224,13 +245,16 @@ std::vector<Tensor> ce_fwd(Tensor logits, Tensor targets,
   return {loss.squeeze(0), lse, count.squeeze(0)};
 }
 
-Tensor ce_bwd(Tensor logits, Tensor targets, Tensor lse, double scale,
-              int64_t ignore_index) {
+// scale_dev: empty tensor => host 'scale' scalar; else a 0/1-dim fp32
+// device scalar (dloss/count), keeping backward free of host syncs.
+Tensor ce_bwd(Tensor logits, Tensor targets, Tensor lse, Tensor scale_dev,
+              double scale, int64_t ignore_index) {
   check_bf16(logits, "logits");
   const int64_t rows = logits.size(0), vocab = logits.size(1);
   auto dl = torch::empty_like(logits);
+  const float* scp = scale_dev.numel() ? scale_dev.data_ptr<float>() : nullptr;
   launch_ce_bwd(bfp(logits), targets.data_ptr<int64_t>(),
-                lse.data_ptr<float>(), float(scale), ignore_index,
+                lse.data_ptr<float>(), float(scale), scp, ignore_index,
                 bfp_mut(dl), rows, vocab, stream());
   return dl;
 }
@@ -367,6 +391,48 @@ std::vector<Tensor> attn_bwd(Tensor dout, Tensor q, Tensor k, Tensor v,
   return {dq, dk32.to(torch::kBFloat16), dv32.to(torch::kBFloat16)};
 }
 
+// Packed-QKV backward: dq/dk/dv are stored directly into caller-provided
+// [B,S,H,D]/[B,S,Hk,D] strided views of one dqkv buffer (innermost D
+// contiguous) — eliminates the split-backward cat and all transpose copies.
+void attn_bwd_packed(Tensor dout, Tensor q, Tensor k, Tensor v,
+                     Tensor o_bshd, Tensor lse, double scale, Tensor dq_v,
+                     Tensor dk_v, Tensor dv_v) {
+  check_attn_view(dout, "dout");
+  TORCH_CHECK(dq_v.stride(3) == 1 && dk_v.stride(3) == 1 &&
+              dv_v.stride(3) == 1, "dqkv views need contiguous head_dim");
+  const int64_t B = q.size(0), H = q.size(1), S = q.size(2);
+  const int hd = int(q.size(3));
+  const int64_t Hk = k.size(1);
+  AttnGeom geo = make_geom(q, k, v, o_bshd, dout, scale);
+  auto delta = torch::empty({B * H, S}, lse.options());
+  launch_attn_delta(bfp(dout), bfp(o_bshd), delta.data_ptr<float>(), geo,
+                    stream());
+  AttnGeom gq = geo;  // dq_v is [B,S,H,D]: (batch, head, seq) strides
+  gq.ob = dq_v.stride(0); gq.oh = dq_v.stride(2); gq.os_ = dq_v.stride(1);
+  launch_attn_bwd_dq(bfp(dout), bfp(q), bfp(k), bfp(v),
+                     lse.data_ptr<float>(), delta.data_ptr<float>(),
+                     bfp_mut(dq_v), gq, stream());
+  if (geo.grp == 1) {
+    geo.gkb = dk_v.stride(0); geo.gkh = dk_v.stride(2);
+    geo.gks = dk_v.stride(1);
+    launch_attn_bwd_dkv(bfp(dout), bfp(q), bfp(k), bfp(v),
+                        lse.data_ptr<float>(), delta.data_ptr<float>(),
+                        nullptr, nullptr, bfp_mut(dk_v), bfp_mut(dv_v), geo,
+                        stream());
+    return;
+  }
+  // GQA: fp32 accumulation (q-head fold-in), then cast into the views
+  auto dk32 = torch::zeros({B, Hk, S, hd},
+                           q.options().dtype(torch::kFloat32));
+  auto dv32 = torch::zeros_like(dk32);
+  launch_attn_bwd_dkv(bfp(dout), bfp(q), bfp(k), bfp(v),
+                      lse.data_ptr<float>(), delta.data_ptr<float>(),
+                      dk32.data_ptr<float>(), dv32.data_ptr<float>(),
+                      nullptr, nullptr, geo, stream());
+  dk_v.copy_(dk32.permute({0, 2, 1, 3}));
+  dv_v.copy_(dv32.permute({0, 2, 1, 3}));
+}
+
 // ---- mfma self-test --------------------------------------------------------
 Tensor mfma_selftest_16(Tensor A, Tensor B) {
   check_bf16(A, "A"); check_bf16(B, "B");
@@ -397,6 +463,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("has_nan", &has_nan);
   m.def("l2norm_sq", &l2norm_sq);
   m.def("adamw_step", &adamw_step);
+  m.def("adamw_tick", &adamw_tick);
+  m.def("colsum", &colsum);
   m.def("weighted_merge", &weighted_merge);
   m.def("grad_merge_weights", &grad_merge_weights);
   m.def("ce_fwd", &ce_fwd);
@@ -407,6 +475,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("rope_bwd", &rope_bwd);
   m.def("attn_fwd", &attn_fwd);
   m.def("attn_bwd", &attn_bwd);
+  m.def("attn_bwd_packed", &attn_bwd_packed);
   m.def("mfma_selftest_16", &mfma_selftest_16);
   m.def("mfma_selftest_32", &mfma_selftest_32);
 }

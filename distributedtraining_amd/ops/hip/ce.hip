@@ -70,17 +70,22 @@ __global__ void ce_fwd_k(const ushort* __restrict__ logits,
   }
 }
 
+// scale_p: optional device pointer (dloss/count computed on device so the
+// whole backward is hipGraph-capturable with no host sync); if null the
+// host 'scale' scalar is used.
 __global__ void ce_bwd_k(const ushort* __restrict__ logits,
                          const int64_t* __restrict__ targets,
                          const float* __restrict__ lse, float scale,
+                         const float* __restrict__ scale_p,
                          int64_t ignore_index, ushort* __restrict__ dlogits,
                          int64_t rows, int64_t vocab) {
+  const float sc_base = scale_p ? *scale_p : scale;
   for (int64_t row = blockIdx.x; row < rows; row += gridDim.x) {
     const ushort* xr = logits + row * vocab;
     ushort* dxr = dlogits + row * vocab;
     const int64_t tgt = targets[row];
     const float lb = lse[row] * LOG2E;
-    const float sc = (tgt == ignore_index) ? 0.f : scale;
+    const float sc = (tgt == ignore_index) ? 0.f : sc_base;
     int64_t i = int64_t(threadIdx.x) * 8;
     const int64_t stride = int64_t(CE_BLOCK) * 8;
     for (; i + 8 <= vocab; i += stride) {
@@ -112,10 +117,10 @@ void launch_ce_fwd(const bf16_t* logits, const int64_t* targets, int64_t rows,
 }
 
 void launch_ce_bwd(const bf16_t* logits, const int64_t* targets,
-                   const float* lse, float scale, int64_t ignore_index,
-                   bf16_t* dlogits, int64_t rows, int64_t vocab,
-                   hipStream_t s) {
+                   const float* lse, float scale, const float* scale_p,
+                   int64_t ignore_index, bf16_t* dlogits, int64_t rows,
+                   int64_t vocab, hipStream_t s) {
   const int grid = int(rows < 4096 ? (rows > 0 ? rows : 1) : 4096);
-  ce_bwd_k<<<grid, CE_BLOCK, 0, s>>>(logits, targets, lse, scale,
+  ce_bwd_k<<<grid, CE_BLOCK, 0, s>>>(logits, targets, lse, scale, scale_p,
                                      ignore_index, dlogits, rows, vocab);
 }

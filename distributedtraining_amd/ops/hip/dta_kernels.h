@@ -27,9 +27,13 @@ void launch_axpy(float* w, const float* x, float alpha, int64_t n,
 void launch_nan_any(const float* x, int64_t n, int* flag, hipStream_t s);
 void launch_l2norm_sq(const float* x, int64_t n, float* out, hipStream_t s);
 void launch_adamw(float* master, const void* grad, bool grad_is_bf16,
-                  float* m, float* v, bf16_t* out_bf16, int step, float lr,
-                  float beta1, float beta2, float eps, float wd, int64_t n,
-                  hipStream_t s);
+                  float* m, float* v, bf16_t* out_bf16, int step,
+                  const float* bc_p, float lr, float beta1, float beta2,
+                  float eps, float wd, int64_t n, hipStream_t s);
+void launch_adamw_tick(int* t, float* bc, float beta1, float beta2,
+                       hipStream_t s);
+void launch_colsum(const bf16_t* x, float* out, int64_t rows, int cols,
+                   hipStream_t s);
 
 // ---- merge plane ----------------------------------------------------------
 void launch_weighted_merge(const float* base, const float* deltas,  // [N,P]
@@ -64,9 +68,9 @@ void launch_ce_fwd(const bf16_t* logits, const int64_t* targets,
                    int64_t rows, int64_t vocab, int64_t ignore_index,
                    float* lse, float* loss_sum, int* count, hipStream_t s);
 void launch_ce_bwd(const bf16_t* logits, const int64_t* targets,
-                   const float* lse, float scale, int64_t ignore_index,
-                   bf16_t* dlogits, int64_t rows, int64_t vocab,
-                   hipStream_t s);
+                   const float* lse, float scale, const float* scale_p,
+                   int64_t ignore_index, bf16_t* dlogits, int64_t rows,
+                   int64_t vocab, hipStream_t s);
 
 // ---- embedding ------------------------------------------------------------
 void launch_embedding_fwd(const int64_t* ids, const bf16_t* wte,

@@ -214,6 +214,28 @@ __global__ void f32_to_bf16_k(const float* __restrict__ x,
     for (; i < n; ++i) y[i] = f2bf(x[i]);
 }
 
+
+// ---- column sum (bias gradient) -------------------------------------------
+// out[c] = sum_r x[r,c] for x [rows, cols] bf16, out fp32 pre-zeroed.
+// Replaces torch's generic reduce_kernel for dbias (48 calls/step in the
+// GPT-2 backward, ~4x off memory SOL there). Vectorized s16x8 row reads;
+// one atomicAdd per (stripe, column).
+__global__ void colsum_k(const ushort* __restrict__ x,
+                         float* __restrict__ out, int64_t rows, int cols) {
+  const int c8 = (blockIdx.x * blockDim.x + threadIdx.x) * 8;
+  if (c8 >= cols) return;
+  const int64_t r0 = (rows * blockIdx.y) / gridDim.y;
+  const int64_t r1 = (rows * (blockIdx.y + 1)) / gridDim.y;
+  float acc[8] = {0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f};
+  for (int64_t r = r0; r < r1; ++r) {
+    s16x8 vx = *reinterpret_cast<const s16x8*>(x + r * int64_t(cols) + c8);
+#pragma unroll
+    for (int j = 0; j < 8; ++j) acc[j] += bf2f(ushort(vx[j]));
+  }
+#pragma unroll
+  for (int j = 0; j < 8; ++j) atomicAdd(out + c8 + j, acc[j]);
+}
+
 }  // namespace
 
 #define LAUNCH_EW(kernel, n, ...)                                          \
@@ -260,4 +282,15 @@ void launch_l2norm_sq(const float* x, int64_t n, float* out, hipStream_t s) {
 }
 void launch_f32_to_bf16(const float* x, bf16_t* y, int64_t n, hipStream_t s) {
   LAUNCH_EW(f32_to_bf16_k, n, x, y, n);
+}
+void launch_colsum(const bf16_t* x, float* out, int64_t rows, int cols,
+                   hipStream_t s) {
+  const int lanes = cols / 8;                     // cols % 8 == 0 (checked)
+  const int gx = (lanes + 255) / 256;
+  int64_t stripes = 1024 / gx;
+  const int64_t max_stripes = (rows + 31) / 32;
+  if (stripes > max_stripes) stripes = max_stripes;
+  if (stripes < 1) stripes = 1;
+  dim3 grid(gx, unsigned(stripes));
+  colsum_k<<<grid, 256, 0, s>>>(x, out, rows, cols);
 }

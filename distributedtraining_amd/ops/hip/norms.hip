@@ -180,6 +180,41 @@ __global__ void norm_bwd_dwdb_k(const ushort* __restrict__ dy,
   if (!RMS) atomicAdd(db + c, db_acc);
 }
 
+// Vectorized variant (cols % 8 == 0, the production shapes): s16x8 row
+// reads, 8 columns per thread — request-rate bound at 2 B/lane otherwise.
+template <bool RMS>
+__global__ void norm_bwd_dwdb_v8_k(const ushort* __restrict__ dy,
+                                   const ushort* __restrict__ x,
+                                   const float* __restrict__ mean,
+                                   const float* __restrict__ rstd,
+                                   float* __restrict__ dw,
+                                   float* __restrict__ db, int64_t rows,
+                                   int cols) {
+  const int c8 = (blockIdx.x * blockDim.x + threadIdx.x) * 8;
+  if (c8 >= cols) return;
+  const int64_t r0 = (rows * blockIdx.y) / gridDim.y;
+  const int64_t r1 = (rows * (blockIdx.y + 1)) / gridDim.y;
+  float dw_acc[8] = {0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f};
+  float db_acc[8] = {0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f};
+  for (int64_t r = r0; r < r1; ++r) {
+    const float mu = RMS ? 0.f : mean[r];
+    const float rs = rstd[r];
+    s16x8 xv = *reinterpret_cast<const s16x8*>(x + r * int64_t(cols) + c8);
+    s16x8 dv = *reinterpret_cast<const s16x8*>(dy + r * int64_t(cols) + c8);
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      const float dyv = bf2f(ushort(dv[j]));
+      dw_acc[j] = fmaf(dyv, (bf2f(ushort(xv[j])) - mu) * rs, dw_acc[j]);
+      db_acc[j] += dyv;
+    }
+  }
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    atomicAdd(dw + c8 + j, dw_acc[j]);
+    if (!RMS) atomicAdd(db + c8 + j, db_acc[j]);
+  }
+}
+
 template <bool RMS>
 void dispatch_fwd(const ushort* x, const ushort* w, const ushort* b,
                   ushort* y, float* mean, float* rstd, int64_t rows, int cols,
@@ -226,11 +261,22 @@ void dispatch_bwd(const ushort* dy, const ushort* x, const ushort* w,
                                                  rows, cols);
   }
 #undef CASE_B
-  int stripes = int(rows >= 8192 ? 256 : (rows + 31) / 32);
-  if (stripes < 1) stripes = 1;
-  dim3 g2((cols + 255) / 256, stripes);
-  norm_bwd_dwdb_k<RMS><<<g2, 256, 0, s>>>(dy, x, mean, rstd, dw, db, rows,
-                                          cols);
+  if (cols % 8 == 0) {
+    const int gx = (cols / 8 + 255) / 256;
+    int64_t stripes = 1024 / gx;
+    const int64_t max_stripes = (rows + 31) / 32;
+    if (stripes > max_stripes) stripes = max_stripes;
+    if (stripes < 1) stripes = 1;
+    dim3 g2(gx, unsigned(stripes));
+    norm_bwd_dwdb_v8_k<RMS><<<g2, 256, 0, s>>>(dy, x, mean, rstd, dw, db,
+                                               rows, cols);
+  } else {
+    int stripes = int(rows >= 8192 ? 256 : (rows + 31) / 32);
+    if (stripes < 1) stripes = 1;
+    dim3 g2((cols + 255) / 256, stripes);
+    norm_bwd_dwdb_k<RMS><<<g2, 256, 0, s>>>(dy, x, mean, rstd, dw, db, rows,
+                                            cols);
+  }
 }
 
 }  // namespace

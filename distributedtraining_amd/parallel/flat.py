@@ -144,16 +144,36 @@ class FusedAdamW:
         self.m = torch.zeros_like(fp.master)
         self.v = torch.zeros_like(fp.master)
         self.t = 0
+        # device-side step counter + bias-correction buffer: the GPU step is
+        # [tick kernel; adamw kernel], fully hipGraph-capturable (no host
+        # state enters the kernels)
+        if fp.device.type == "cuda":
+            self.t_dev = torch.zeros(1, dtype=torch.int32, device=fp.device)
+            self.bc_dev = torch.zeros(2, dtype=torch.float32,
+                                      device=fp.device)
+        else:
+            self.t_dev = self.bc_dev = None
 
     def reset_state(self) -> None:
         self.m.zero_()
         self.v.zero_()
         self.t = 0
+        if self.t_dev is not None:
+            self.t_dev.zero_()
+            self.bc_dev.zero_()
 
     @torch.no_grad()
     def step(self) -> None:
         self.t += 1
         out_bf16 = self.fp.work if self.fp.work.dtype != torch.float32 else None
+        if self.t_dev is not None:
+            ops.adamw_tick(self.t_dev, self.bc_dev, self.betas[0],
+                           self.betas[1])
+            ops.adamw_step(self.fp.master, self.fp.grad, self.m, self.v,
+                           out_bf16, 0, self.lr, self.betas[0],
+                           self.betas[1], self.eps, self.weight_decay,
+                           bc=self.bc_dev)
+            return
         ops.adamw_step(self.fp.master, self.fp.grad, self.m, self.v,
                        out_bf16, self.t, self.lr, self.betas[0],
                        self.betas[1], self.eps, self.weight_decay)

@@ -55,10 +55,10 @@ class LocalSGDNode:
         self.miner.install_base(self.fp.master)
 
     def train_steps(self, n: int) -> float:
-        last = float("nan")
+        last = None
         for _ in range(n):
             last = self.miner.train_step()
-        return last
+        return float(last) if last is not None else float("nan")
 
     # -- the delta exchange + merge (C1..C5) ---------------------------------
     def merge_round(self, scores: Optional[List[float]] = None) -> None:

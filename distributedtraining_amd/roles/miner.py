@@ -48,7 +48,7 @@ class DeltaLoop:
         self.opt = FusedAdamW(fp, lr=cfg.lr, betas=tuple(cfg.betas),
                               eps=cfg.eps, weight_decay=cfg.weight_decay)
         self.step_count = 0
-        self.total_loss = 0.0
+        self._loss_acc = None   # device-resident Σ(loss·batch)
         self.total_examples = 0
         self.base = fp.snapshot()
         self.base_hash = fp.master_hash()
@@ -81,7 +81,11 @@ class DeltaLoop:
         self.last_base_time = time.time()
 
     # -- training ------------------------------------------------------------
-    def train_step(self, batch=None) -> float:
+    def train_step(self, batch=None) -> torch.Tensor:
+        """One fwd/bwd/AdamW step. Sync-free on GPU: returns the loss as a
+        0-dim device tensor and accumulates it on device; nothing in here
+        forces a host round-trip (the whole body is hipGraph-capturable —
+        see parallel/graphstep.py)."""
         if batch is None:
             batch = next(self.data)
         input_ids = batch["input_ids"].to(self.fp.device, non_blocking=True)
@@ -92,8 +96,11 @@ class DeltaLoop:
         self.opt.step()
         self.opt.zero_grad()
         self.step_count += 1
-        loss = float(out.loss.detach())
-        self.total_loss += loss * input_ids.shape[0]
+        loss = out.loss.detach()
+        if self._loss_acc is None:
+            self._loss_acc = torch.zeros((), dtype=torch.float32,
+                                         device=loss.device)
+        self._loss_acc += loss.float() * input_ids.shape[0]
         self.total_examples += input_ids.shape[0]
         return loss
 
@@ -118,20 +125,21 @@ class DeltaLoop:
         return time.time() - self.last_base_time
 
     def average_loss(self) -> float:
-        if self.total_examples == 0:
+        """Mean loss over all steps so far (syncs once, here)."""
+        if self.total_examples == 0 or self._loss_acc is None:
             return float("nan")
-        return self.total_loss / self.total_examples
+        return float(self._loss_acc) / self.total_examples
 
     def perplexity(self) -> float:
         return math.exp(min(self.average_loss(), 20.0))
 
     def train(self, steps: int) -> float:
         """Run the outer loop: step, poll base, push delta."""
-        last = float("nan")
+        last = None
         for _ in range(steps):
             if (self.cfg.pull_interval_steps and
                     self.step_count % self.cfg.pull_interval_steps == 0):
                 self.maybe_pull_base()
             last = self.train_step()
             self.maybe_push_delta()
-        return last
+        return float(last) if last is not None else float("nan")
