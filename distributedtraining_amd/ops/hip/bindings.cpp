@@ -52,13 +52,18 @@ std::vector<Tensor> layernorm_bwd(Tensor dy, Tensor x, Tensor w, Tensor mean,
   check_bf16(dy, "dy"); check_bf16(x, "x"); check_bf16(w, "w");
   const int64_t rows = x.size(0);
   const int cols = int(x.size(1));
+  TORCH_CHECK(cols % 8 == 0, "cols must be a multiple of 8");
   auto dx = torch::empty_like(x);
-  auto dw32 = torch::zeros({cols}, x.options().dtype(torch::kFloat32));
-  auto db32 = torch::zeros({cols}, x.options().dtype(torch::kFloat32));
+  auto f32 = x.options().dtype(torch::kFloat32);
+  auto dw32 = torch::empty({cols}, f32);
+  auto db32 = torch::empty({cols}, f32);
+  const int stripes = dta_colred_stripes(rows, cols);
+  auto part = torch::empty({2, stripes, cols}, f32);
   launch_layernorm_bwd(bfp(dy), bfp(x), bfp(w), mean.data_ptr<float>(),
                        rstd.data_ptr<float>(), bfp_mut(dx),
-                       dw32.data_ptr<float>(), db32.data_ptr<float>(), rows,
-                       cols, stream());
+                       dw32.data_ptr<float>(), db32.data_ptr<float>(),
+                       part[0].data_ptr<float>(), part[1].data_ptr<float>(),
+                       stripes, rows, cols, stream());
   return {dx, dw32.to(torch::kBFloat16), db32.to(torch::kBFloat16)};
 }
 
@@ -78,11 +83,15 @@ std::vector<Tensor> rmsnorm_bwd(Tensor dy, Tensor x, Tensor w, Tensor rstd) {
   check_bf16(dy, "dy"); check_bf16(x, "x"); check_bf16(w, "w");
   const int64_t rows = x.size(0);
   const int cols = int(x.size(1));
+  TORCH_CHECK(cols % 8 == 0, "cols must be a multiple of 8");
   auto dx = torch::empty_like(x);
-  auto dw32 = torch::zeros({cols}, x.options().dtype(torch::kFloat32));
+  auto f32 = x.options().dtype(torch::kFloat32);
+  auto dw32 = torch::empty({cols}, f32);
+  const int stripes = dta_colred_stripes(rows, cols);
+  auto part = torch::empty({stripes, cols}, f32);
   launch_rmsnorm_bwd(bfp(dy), bfp(x), bfp(w), rstd.data_ptr<float>(),
-                     bfp_mut(dx), dw32.data_ptr<float>(), rows, cols,
-                     stream());
+                     bfp_mut(dx), dw32.data_ptr<float>(),
+                     part.data_ptr<float>(), stripes, rows, cols, stream());
   return {dx, dw32.to(torch::kBFloat16)};
 }
 
@@ -178,9 +187,14 @@ Tensor colsum(Tensor x) {
   check_bf16(x, "x");
   TORCH_CHECK(x.dim() == 2 && x.size(1) % 8 == 0,
               "x must be [rows, cols] with cols % 8 == 0");
-  auto out = torch::zeros({x.size(1)}, x.options().dtype(torch::kFloat32));
-  launch_colsum(bfp(x), out.data_ptr<float>(), x.size(0), int(x.size(1)),
-                stream());
+  const int64_t rows = x.size(0);
+  const int cols = int(x.size(1));
+  const int stripes = dta_colred_stripes(rows, cols);
+  auto f32 = x.options().dtype(torch::kFloat32);
+  auto part = torch::empty({stripes, cols}, f32);
+  auto out = torch::empty({cols}, f32);
+  launch_colsum(bfp(x), part.data_ptr<float>(), out.data_ptr<float>(), rows,
+                cols, stripes, stream());
   return out;
 }
 

@@ -32,8 +32,17 @@ void launch_adamw(float* master, const void* grad, bool grad_is_bf16,
                   float eps, float wd, int64_t n, hipStream_t s);
 void launch_adamw_tick(int* t, float* bc, float beta1, float beta2,
                        hipStream_t s);
-void launch_colsum(const bf16_t* x, float* out, int64_t rows, int cols,
-                   hipStream_t s);
+void launch_colsum(const bf16_t* x, float* part, float* out, int64_t rows,
+                   int cols, int stripes, hipStream_t s);
+// stripe count for the two-phase column reductions (colsum / norm dγdβ)
+inline int dta_colred_stripes(int64_t rows, int cols) {
+  const int gx = int((cols / 8 + 255) / 256);
+  int64_t st = 1024 / (gx > 0 ? gx : 1);
+  const int64_t mx = (rows + 31) / 32;
+  if (st > mx) st = mx;
+  if (st < 1) st = 1;
+  return int(st);
+}
 
 // ---- merge plane ----------------------------------------------------------
 void launch_weighted_merge(const float* base, const float* deltas,  // [N,P]
@@ -52,16 +61,19 @@ void launch_grad_merge_weights(const void* g, bool g_is_bf16,
 void launch_layernorm_fwd(const bf16_t* x, const bf16_t* w, const bf16_t* b,
                           bf16_t* y, float* mean, float* rstd, int64_t rows,
                           int cols, float eps, hipStream_t s);
+// pdw/pdb: [stripes, cols] fp32 workspaces (dta_colred_stripes rows); the
+// two-phase column reduction is atomic-free and deterministic.
 void launch_layernorm_bwd(const bf16_t* dy, const bf16_t* x, const bf16_t* w,
                           const float* mean, const float* rstd, bf16_t* dx,
-                          float* dw, float* db,  // fp32, pre-zeroed
-                          int64_t rows, int cols, hipStream_t s);
+                          float* dw, float* db, float* pdw, float* pdb,
+                          int stripes, int64_t rows, int cols,
+                          hipStream_t s);
 void launch_rmsnorm_fwd(const bf16_t* x, const bf16_t* w, bf16_t* y,
                         float* rstd, int64_t rows, int cols, float eps,
                         hipStream_t s);
 void launch_rmsnorm_bwd(const bf16_t* dy, const bf16_t* x, const bf16_t* w,
-                        const float* rstd, bf16_t* dx, float* dw,
-                        int64_t rows, int cols, hipStream_t s);
+                        const float* rstd, bf16_t* dx, float* dw, float* pdw,
+                        int stripes, int64_t rows, int cols, hipStream_t s);
 
 // ---- fused cross entropy --------------------------------------------------
 void launch_ce_fwd(const bf16_t* logits, const int64_t* targets,

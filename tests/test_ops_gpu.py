@@ -195,7 +195,13 @@ def test_cross_entropy(rows, vocab):
 
     lr = logits.float().detach().requires_grad_(True)
     torch.nn.functional.cross_entropy(lr, targets, ignore_index=-100).backward()
-    dl = m.ce_bwd(logits, targets, lse, 1.0 / n_valid, -100)
+    dl = m.ce_bwd(logits, targets, lse,
+                  torch.empty(0, device=DEV, dtype=torch.float32),
+                  1.0 / n_valid, -100)
+    # device-scale path must agree with the host-scalar path
+    sdev = torch.tensor([1.0 / n_valid], device=DEV)
+    dl2 = m.ce_bwd(logits, targets, lse, sdev, 0.0, -100)
+    torch.testing.assert_close(dl2, dl)
     torch.testing.assert_close(dl.float(), lr.grad, rtol=5e-2,
                                atol=1e-4)
 
