@@ -55,8 +55,8 @@ std::vector<Tensor> layernorm_bwd(Tensor dy, Tensor x, Tensor w, Tensor mean,
   TORCH_CHECK(cols % 8 == 0, "cols must be a multiple of 8");
   auto dx = torch::empty_like(x);
   auto f32 = x.options().dtype(torch::kFloat32);
-  auto dw32 = torch::empty({cols}, f32);
-  auto db32 = torch::empty({cols}, f32);
+  auto dw32 = torch::zeros({cols}, f32);
+  auto db32 = torch::zeros({cols}, f32);
   const int stripes = dta_colred_stripes(rows, cols);
   auto part = torch::empty({2, stripes, cols}, f32);
   launch_layernorm_bwd(bfp(dy), bfp(x), bfp(w), mean.data_ptr<float>(),
@@ -86,7 +86,7 @@ std::vector<Tensor> rmsnorm_bwd(Tensor dy, Tensor x, Tensor w, Tensor rstd) {
   TORCH_CHECK(cols % 8 == 0, "cols must be a multiple of 8");
   auto dx = torch::empty_like(x);
   auto f32 = x.options().dtype(torch::kFloat32);
-  auto dw32 = torch::empty({cols}, f32);
+  auto dw32 = torch::zeros({cols}, f32);
   const int stripes = dta_colred_stripes(rows, cols);
   auto part = torch::empty({stripes, cols}, f32);
   launch_rmsnorm_bwd(bfp(dy), bfp(x), bfp(w), rstd.data_ptr<float>(),
@@ -192,7 +192,7 @@ Tensor colsum(Tensor x) {
   const int stripes = dta_colred_stripes(rows, cols);
   auto f32 = x.options().dtype(torch::kFloat32);
   auto part = torch::empty({stripes, cols}, f32);
-  auto out = torch::empty({cols}, f32);
+  auto out = torch::zeros({cols}, f32);
   launch_colsum(bfp(x), part.data_ptr<float>(), out.data_ptr<float>(), rows,
                 cols, stripes, stream());
   return out;

@@ -152,11 +152,12 @@ __global__ void norm_bwd_dx_k(const ushort* __restrict__ dy,
   }
 }
 
-// ---- backward dγ/dβ (two-phase column reduction, atomic-free) -------------
+// ---- backward dγ/dβ (two-phase column reduction) --------------------------
 // Phase 1: each (col-tile, stripe) block accumulates its stripe's rows into
-// a [stripes, cols] fp32 partial panel (vectorized s16x8 row reads). Phase 2
-// reduces the panel. A same-address atomicAdd version measured 4x slower at
-// [32k, 768] (per-address L2 serialization); this is deterministic too.
+// a [stripes, cols] fp32 partial panel (vectorized s16x8 row reads) — a
+// single-phase version with ~1024 same-address atomicAdds per column
+// measured 4x slower (per-address L2 serialization). Phase 2 reduces the
+// panel 2D with <=32 atomics per output address.
 template <bool RMS>
 __global__ void norm_bwd_dwdb_part_k(const ushort* __restrict__ dy,
                                      const ushort* __restrict__ x,
@@ -193,21 +194,28 @@ __global__ void norm_bwd_dwdb_part_k(const ushort* __restrict__ dy,
   }
 }
 
+// 2D over (col tiles, stripe chunks); <=32 atomics per output address
+// (a 1-block serial stripe loop measured 330 us — pure latency chain).
 __global__ void dwdb_reduce_k(const float* __restrict__ pdw,
                               const float* __restrict__ pdb,
                               float* __restrict__ dw, float* __restrict__ db,
                               int stripes, int cols) {
   const int c4 = (blockIdx.x * blockDim.x + threadIdx.x) * 4;
   if (c4 >= cols) return;
+  const int s0 = (stripes * blockIdx.y) / gridDim.y;
+  const int s1 = (stripes * (blockIdx.y + 1)) / gridDim.y;
   f32x4 aw = {0.f, 0.f, 0.f, 0.f};
   f32x4 ab = {0.f, 0.f, 0.f, 0.f};
-  for (int s_ = 0; s_ < stripes; ++s_) {
+  for (int s_ = s0; s_ < s1; ++s_) {
     aw += *reinterpret_cast<const f32x4*>(pdw + int64_t(s_) * cols + c4);
     if (pdb)
       ab += *reinterpret_cast<const f32x4*>(pdb + int64_t(s_) * cols + c4);
   }
-  *reinterpret_cast<f32x4*>(dw + c4) = aw;
-  if (pdb) *reinterpret_cast<f32x4*>(db + c4) = ab;
+#pragma unroll
+  for (int j = 0; j < 4; ++j) {
+    atomicAdd(dw + c4 + j, aw[j]);
+    if (pdb) atomicAdd(db + c4 + j, ab[j]);
+  }
 }
 
 template <bool RMS>
@@ -263,8 +271,9 @@ void dispatch_bwd(const ushort* dy, const ushort* x, const ushort* w,
                                                RMS ? nullptr : pdb, rows,
                                                cols);
   const int g3 = (cols / 4 + 255) / 256;
-  dwdb_reduce_k<<<g3, 256, 0, s>>>(pdw, RMS ? nullptr : pdb, dw, db, stripes,
-                                   cols);
+  const int ry = stripes < 32 ? stripes : 32;
+  dwdb_reduce_k<<<dim3(g3, ry), 256, 0, s>>>(pdw, RMS ? nullptr : pdb, dw,
+                                             db, stripes, cols);
 }
 
 }  // namespace
