@@ -51,7 +51,7 @@ class DeltaLoop:
         self._loss_acc = None   # device-resident Σ(loss·batch)
         self.total_examples = 0
         self.base = fp.snapshot()
-        self.base_hash = fp.master_hash()
+        self._base_hash = None   # lazy: SHA-256 of 500 MB costs ~300 ms on host
         self.last_push_step = 0
         self.last_base_time = time.time()
         if registry is not None and store is not None:
@@ -77,7 +77,7 @@ class DeltaLoop:
         self.opt.reset_state()
         self.opt.zero_grad()
         self.base = self.fp.snapshot()
-        self.base_hash = self.fp.master_hash()
+        self._base_hash = None
         self.last_base_time = time.time()
 
     # -- training ------------------------------------------------------------
@@ -104,15 +104,25 @@ class DeltaLoop:
         self.total_examples += input_ids.shape[0]
         return loss
 
+    @property
+    def base_hash(self) -> str:
+        """SHA-256 of the current base (reference: calculate_model_hash,
+        training_manager.py:198-203). Computed lazily — the in-node RCCL
+        merge path never needs it; only store pushes do."""
+        if self._base_hash is None:
+            self._base_hash = self.fp.master_hash()
+        return self._base_hash
+
     # -- delta publication (reference :405-427) ------------------------------
-    def make_delta(self) -> DeltaCheckpoint:
+    def make_delta(self, with_hash: bool = False) -> DeltaCheckpoint:
         return self.fp.make_delta(self.base, step=self.step_count,
-                                  base_hash=self.base_hash)
+                                  base_hash=self.base_hash if with_hash
+                                  else "")
 
     def maybe_push_delta(self) -> Optional[DeltaCheckpoint]:
         if self.step_count - self.last_push_step < self.cfg.send_interval_steps:
             return None
-        ckpt = self.make_delta()
+        ckpt = self.make_delta(with_hash=self.store is not None)
         if self.store is not None:
             self.store.push_delta(ckpt)
         if self.on_push is not None:

@@ -272,7 +272,8 @@ def test_adamw_matches_cpu_reference():
         ops.adamw_step(master, grad, mm, vv, None, t, 1e-2, 0.9, 0.999,
                        1e-8, 0.05)  # CPU reference path
         m.adamw_step(master_g, grad.to(DEV, torch.bfloat16), m_g, v_g, wout,
-                     t, 1e-2, 0.9, 0.999, 1e-8, 0.05)
+                     t, 1e-2, 0.9, 0.999, 1e-8, 0.05,
+                     torch.empty(0, device=DEV))
     torch.testing.assert_close(master_g.cpu(), master, rtol=3e-2, atol=1e-3)
     torch.testing.assert_close(wout.float().cpu(), master, rtol=3e-2,
                                atol=2e-2)
