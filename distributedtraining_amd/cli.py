@@ -9,6 +9,11 @@ Usage:
   python -m distributedtraining_amd.cli miner     --comm.root /tmp/ex --hotkey m0 --steps 200
   python -m distributedtraining_amd.cli validator --comm.root /tmp/ex --rounds 1
   python -m distributedtraining_amd.cli averager  --comm.root /tmp/ex --rounds 1
+  python -m distributedtraining_amd.cli bootstrap --port 8500
+
+Every role writes a metrics JSONL under --metrics-dir (utils/metrics.py);
+--resume restores a miner's saved train state (base + step counter) from
+the store.
 """
 
 from __future__ import annotations
@@ -49,19 +54,42 @@ def main(argv=None) -> int:
     argv = argv if argv is not None else sys.argv[1:]
     logging.basicConfig(level=logging.INFO)
     top = argparse.ArgumentParser("distributedtraining_amd")
-    top.add_argument("role", choices=["miner", "validator", "averager"])
+    top.add_argument("role", choices=["miner", "validator", "averager",
+                                      "bootstrap"])
     top.add_argument("--hotkey", default=None)
     top.add_argument("--steps", type=int, default=100)
     top.add_argument("--rounds", type=int, default=1)
     top.add_argument("--tiny", action="store_true",
                      help="use the tiny test-scale model")
+    top.add_argument("--resume", action="store_true",
+                     help="miner: restore saved train state from the store")
+    top.add_argument("--metrics-dir", default="metrics")
+    top.add_argument("--port", type=int, default=8500,
+                     help="bootstrap: HTTP port")
     ns, rest = top.parse_known_args(argv)
     cfg = cfg_mod.from_args(rest)
     if ns.tiny:
         cfg.model = ModelConfig.gpt2_tiny()
     hotkey = ns.hotkey or ns.role
 
+    if ns.role == "bootstrap":
+        import time as _time
+        from .utils.bootstrap_server import BootstrapServer
+        srv = BootstrapServer(port=ns.port)
+        srv.start()
+        print(f"bootstrap server on :{srv.port}")
+        try:
+            while True:
+                _time.sleep(60)
+        except KeyboardInterrupt:
+            srv.stop()
+        return 0
+
+    from .utils.metrics import MetricsRun
     model, fp, store, registry = _setup(cfg, hotkey)
+    metrics = MetricsRun(ns.role, hotkey, log_dir=ns.metrics_dir,
+                         hyperparams={"model": cfg.model.family,
+                                      "lr": cfg.train.lr})
 
     if ns.role == "miner":
         # publish the initial base if none exists yet (first process up)
@@ -73,10 +101,26 @@ def main(argv=None) -> int:
                                  cfg.train.seq_len, seed=cfg.seed + hash(hotkey) % 1000)
         miner = DeltaLoop(model, fp, data, cfg.train, store=store,
                           registry=registry, hotkey=hotkey)
-        miner.maybe_pull_base()
+        if ns.resume:
+            st = store.pull_train_state()
+            if st is not None:
+                fp.load_flat_master(st["flat_master"])
+                miner.base = st["base"].to(fp.device)
+                miner.step_count = int(st.get("step_count", 0))
+                miner.opt.reset_state()  # reference resume contract
+                print(f"miner {hotkey}: resumed at step {miner.step_count}")
+        else:
+            miner.maybe_pull_base()
         miner.train(ns.steps)
         miner.last_push_step = -10**9  # force a final push
         miner.maybe_push_delta()
+        store.push_train_state({"flat_master": fp.master.cpu(),
+                                "base": miner.base.cpu(),
+                                "step_count": miner.step_count})
+        metrics.log(miner.step_count, loss=miner.average_loss(),
+                    perplexity=miner.perplexity(),
+                    staleness_s=miner.gradient_staleness())
+        metrics.close()
         print(f"miner {hotkey}: {ns.steps} steps, "
               f"avg loss {miner.average_loss():.4f}")
         return 0
@@ -90,9 +134,12 @@ def main(argv=None) -> int:
             fp.load_flat_master(sd["flat_master"])
         validator = DeltaValidator(model, fp, ev, cfg.validate, store=store,
                                    registry=registry)
-        for _ in range(ns.rounds):
+        for r in range(ns.rounds):
             scores = validator.validate_and_score()
+            metrics.log(r, base_loss=validator.base_loss,
+                        **{f"score_{k}": v for k, v in scores.items()})
             print("scores:", {k: round(v, 4) for k, v in scores.items()})
+        metrics.close()
         return 0
 
     if ns.role == "averager":
@@ -101,9 +148,11 @@ def main(argv=None) -> int:
             fp.load_flat_master(sd["flat_master"])
         averager = ParameterizedAverager(model, fp, cfg.average, store=store,
                                          registry=registry)
-        for _ in range(ns.rounds):
+        for r in range(ns.rounds):
             averager.run_round(ev)
+            metrics.log(r, merge_rounds=r + 1)
             print("averager: merged + published new base")
+        metrics.close()
         return 0
     return 1
 

@@ -1,9 +1,10 @@
 from ..config import ModelConfig
+from .cnn import SimpleCNN
 from .gpt2 import GPT2LM, CausalLMOutput
 from .llama import LlamaLM
 from .mlp import FeedforwardNN
 
-__all__ = ["build_model", "GPT2LM", "LlamaLM", "FeedforwardNN",
+__all__ = ["build_model", "GPT2LM", "LlamaLM", "FeedforwardNN", "SimpleCNN",
            "CausalLMOutput"]
 
 
@@ -14,4 +15,6 @@ def build_model(cfg: ModelConfig):
         return LlamaLM(cfg)
     if cfg.family == "mlp":
         return FeedforwardNN()
+    if cfg.family == "cnn":
+        return SimpleCNN()
     raise ValueError(f"unknown model family {cfg.family!r}")

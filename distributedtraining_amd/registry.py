@@ -56,18 +56,61 @@ def mad_outlier_scores(metrics: Dict[str, List[float]],
     return out
 
 
+class RateLimiter:
+    """Sliding-window request limiter with blacklist.
+
+    Reference semantics (btt_connector.py:454-480): n=10 requests per 60 s
+    window per hotkey; exceeding the limit rejects the request, repeated
+    abuse blacklists the hotkey permanently for the process lifetime.
+    """
+
+    def __init__(self, max_requests: int = 10, window_s: float = 60.0,
+                 blacklist_after: int = 5):
+        self.max_requests = max_requests
+        self.window_s = window_s
+        self.blacklist_after = blacklist_after
+        self._hits: Dict[str, List[float]] = {}
+        self._violations: Dict[str, int] = {}
+        self.blacklist: set = set()
+        self._lock = threading.Lock()
+
+    def allow(self, hotkey: str, now: Optional[float] = None) -> bool:
+        now = time.monotonic() if now is None else now
+        with self._lock:
+            if hotkey in self.blacklist:
+                return False
+            hits = self._hits.setdefault(hotkey, [])
+            cutoff = now - self.window_s
+            while hits and hits[0] < cutoff:
+                hits.pop(0)
+            if len(hits) >= self.max_requests:
+                v = self._violations.get(hotkey, 0) + 1
+                self._violations[hotkey] = v
+                if v >= self.blacklist_after:
+                    self.blacklist.add(hotkey)
+                return False
+            hits.append(now)
+            return True
+
+
 class Registry:
     """In-process membership + scores registry (thread-safe)."""
+
+    #: stake threshold above which a member counts as a validator
+    #: (reference: get_validator_uids, btt_connector.py:358-380)
+    VALIDATOR_STAKE = 1024.0
 
     def __init__(self, epoch_length: int = 100, ema_alpha: float = 0.333333):
         self._lock = threading.Lock()
         self._addresses: Dict[str, str] = {}     # hotkey -> exchange address
         self._scores: Dict[str, float] = {}      # EMA-folded scores
+        self._stakes: Dict[str, float] = {}      # hotkey -> stake
         self._last_weight_set: int = -10**12      # "block" of last weight set
         self._block0 = time.monotonic()
         self.epoch_length = epoch_length
         self.ema_alpha = ema_alpha
         self._metrics: Dict[str, List[float]] = {}
+        self.rate_limiter = RateLimiter()
 
     # -- membership / address store (reference chain_manager.py:71-115) -----
     def store_address(self, hotkey: str, address: str) -> None:
@@ -87,6 +130,22 @@ class Registry:
     def hotkeys(self) -> List[str]:
         with self._lock:
             return list(self._addresses.keys())
+
+    # -- stake / validator membership (btt_connector.py:358-380) -------------
+    def set_stake(self, hotkey: str, stake: float) -> None:
+        with self._lock:
+            self._stakes[hotkey] = float(stake)
+
+    def get_stake(self, hotkey: str) -> float:
+        with self._lock:
+            return self._stakes.get(hotkey, 0.0)
+
+    def validator_hotkeys(self, threshold: Optional[float] = None) -> List[str]:
+        """Members whose stake clears the validator threshold (the
+        reference's stake >= 1024 TAO rule)."""
+        th = self.VALIDATOR_STAKE if threshold is None else threshold
+        with self._lock:
+            return [h for h, s in self._stakes.items() if s >= th]
 
     # -- scores (reference btt_connector.py:310-356) -------------------------
     def current_block(self) -> int:
@@ -142,12 +201,14 @@ class FileRegistry(Registry):
                     d = json.load(f)
                 self._addresses = d.get("addresses", {})
                 self._scores = d.get("scores", {})
+                self._stakes = d.get("stakes", {})
                 self._last_weight_set = d.get("last_weight_set", -10**12)
             except (json.JSONDecodeError, OSError):
                 pass  # concurrent writer; keep current state
 
     def _save(self) -> None:
         d = {"addresses": self._addresses, "scores": self._scores,
+             "stakes": self._stakes,
              "last_weight_set": self._last_weight_set}
         fd, tmp = tempfile.mkstemp(dir=os.path.dirname(self.path))
         with os.fdopen(fd, "w") as f:
@@ -169,6 +230,8 @@ class FileRegistry(Registry):
                 self._addresses.update(d.get("addresses", {}))
                 for k, v in d.get("scores", {}).items():
                     self._scores.setdefault(k, v)
+                for k, v in d.get("stakes", {}).items():
+                    self._stakes.setdefault(k, v)
             except (json.JSONDecodeError, OSError):
                 pass
 
@@ -188,3 +251,8 @@ class FileRegistry(Registry):
         with self._lock:
             self._save()
         return out
+
+    def set_stake(self, hotkey: str, stake: float) -> None:
+        super().set_stake(hotkey, stake)
+        with self._lock:
+            self._save()

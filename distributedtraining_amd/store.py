@@ -159,6 +159,30 @@ class FileStore:
         self._atomic_save(cpu.state_dict(), path)
         return self.my_address()
 
+    # -- miner-local train state (crash resume) --------------------------
+    # The reference's resume contract is: pull latest base + re-create the
+    # optimizer (training_manager.py:365-378) — optimizer state is
+    # deliberately NOT checkpointed. Train state here is therefore just
+    # {flat_master, base, step_count}; loading it resumes mid-interval
+    # without waiting for the next averager round.
+    def train_state_path(self) -> str:
+        return os.path.join(self.my_address(), "train_state.pt")
+
+    def push_train_state(self, state: dict) -> None:
+        self._atomic_save({"format": "dta-trainstate-v1", **state},
+                          self.train_state_path())
+
+    def pull_train_state(self, map_location="cpu") -> Optional[dict]:
+        path = self.train_state_path()
+        if not os.path.exists(path):
+            return None
+        try:
+            d = torch.load(path, map_location=map_location,
+                           weights_only=False)
+        except Exception:
+            return None
+        return d if d.get("format") == "dta-trainstate-v1" else None
+
     def receive_delta(self, address: str,
                       map_location="cpu") -> Optional[DeltaCheckpoint]:
         """Fetch a miner's delta by its registered address; None on absence
