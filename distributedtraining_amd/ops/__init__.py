@@ -39,9 +39,13 @@ __all__ = [
 # Linear (library GEMM via hipBLASLt) with fast dbias backward
 # --------------------------------------------------------------------------
 class _LinearFn(torch.autograd.Function):
-    """F.linear semantics, but the bias gradient is computed by our
-    vectorized colsum kernel instead of torch's generic reduce_kernel
-    (~4x faster at the GPT-2 shapes; 48 dbias reductions per step)."""
+    """F.linear semantics with two backward optimizations:
+
+    * bias gradient via our vectorized colsum kernel instead of torch's
+      generic reduce_kernel (~4x faster at the GPT-2 shapes),
+    * weight gradient accumulated DIRECTLY into the flat-grad plane
+      (w.main_grad, parallel/flat.py) with one addmm(beta=1) — skipping
+      the fresh-tensor-then-autograd-add pair per weight per step."""
 
     @staticmethod
     def forward(ctx, x, w, b):
@@ -49,6 +53,7 @@ class _LinearFn(torch.autograd.Function):
         ctx.save_for_backward(x2, w)
         ctx.has_b = b is not None
         ctx.xshape = x.shape
+        ctx.wgrad = getattr(w, "main_grad", None)
         y = torch.addmm(b, x2, w.t()) if b is not None else x2.mm(w.t())
         return y.view(*x.shape[:-1], w.shape[0])
 
@@ -59,7 +64,11 @@ class _LinearFn(torch.autograd.Function):
         if dy2.stride(-1) != 1:
             dy2 = dy2.contiguous()
         dx = dy2.mm(w).view(ctx.xshape)
-        dw = dy2.t().mm(x2)
+        if ctx.wgrad is not None:
+            ctx.wgrad.addmm_(dy2.t(), x2)   # flat-plane accumulation
+            dw = None
+        else:
+            dw = dy2.t().mm(x2)
         db = None
         if ctx.has_b:
             db = require_ext().colsum(dy2).to(w.dtype)

@@ -78,6 +78,10 @@ class FlatParams:
             gview = self.grad[o:o + cnt].view(shape)
             newp = nn.Parameter(view)
             newp.grad = gview
+            # main_grad: ops that can accumulate their weight gradient
+            # in place (ops.linear's addmm(beta=1) wgrad) write here and
+            # return None to autograd — skips the alloc+add per weight
+            newp.main_grad = gview
             name_to_view[n] = newp
             o += cnt
         self._rebind(model, name_to_view, seen)
