@@ -222,26 +222,36 @@ __global__ void attn_fwd_k(const ushort* __restrict__ q,
 }
 
 // ---------------- delta = rowsum(dO * O) ----------------
+// 8 lanes per row (head_dim is 32/64/128: s16x8 loads), 8 rows per wave,
+// 32 per block — the one-wave-per-row version left >=half the lanes idle
+// and ran 7x off memory SOL.
 __global__ void attn_delta_k(const ushort* __restrict__ dout,
                              const ushort* __restrict__ o,
                              float* __restrict__ delta, AttnGeom geo) {
   const int lane = threadIdx.x & 63, wid = threadIdx.x >> 6;
+  const int sub = lane & 7;        // lane within row
+  const int rloc = lane >> 3;      // row within wave
   const int hd = geo.hd;
   const int64_t rows = int64_t(geo.B) * geo.H * geo.seq;
-  for (int64_t rr = int64_t(blockIdx.x) * 4 + wid; rr < rows;
-       rr += int64_t(gridDim.x) * 4) {
+  for (int64_t rr = int64_t(blockIdx.x) * 32 + wid * 8 + rloc; rr < rows;
+       rr += int64_t(gridDim.x) * 32) {
     const int s_ = int(rr % geo.seq);
     const int h = int((rr / geo.seq) % geo.H);
     const int b = int(rr / (int64_t(geo.seq) * geo.H));
     const ushort* a = dout + b * geo.db_ + h * geo.dh + int64_t(s_) * geo.ds;
     const ushort* c = o + b * geo.ob + h * geo.oh + int64_t(s_) * geo.os_;
     float acc = 0.f;
-    for (int i = lane * 2; i + 2 <= hd; i += 128) {
-      acc = fmaf(bf2f(a[i]), bf2f(c[i]), acc);
-      acc = fmaf(bf2f(a[i + 1]), bf2f(c[i + 1]), acc);
+    for (int i = sub * 8; i + 8 <= hd; i += 64) {
+      s16x8 va = *reinterpret_cast<const s16x8*>(a + i);
+      s16x8 vc = *reinterpret_cast<const s16x8*>(c + i);
+#pragma unroll
+      for (int j = 0; j < 8; ++j)
+        acc = fmaf(bf2f(ushort(va[j])), bf2f(ushort(vc[j])), acc);
     }
-    acc = wave_sum(acc);
-    if (lane == 0) delta[rr] = acc;
+    acc += __shfl_xor(acc, 4);
+    acc += __shfl_xor(acc, 2);
+    acc += __shfl_xor(acc, 1);
+    if (sub == 0) delta[rr] = acc;
   }
 }
 
@@ -510,7 +520,7 @@ void launch_attn_fwd(const bf16_t* q, const bf16_t* k, const bf16_t* v,
 void launch_attn_delta(const bf16_t* dout, const bf16_t* o, float* delta,
                        const AttnGeom& geo, hipStream_t s) {
   int64_t rows = int64_t(geo.B) * geo.H * geo.seq;
-  int64_t want = (rows + 3) / 4;
+  int64_t want = (rows + 31) / 32;
   const int grid = int(want < 4096 ? (want > 0 ? want : 1) : 4096);
   attn_delta_k<<<grid, 256, 0, s>>>(dout, o, delta, geo);
 }

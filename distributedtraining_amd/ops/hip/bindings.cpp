@@ -300,9 +300,20 @@ std::vector<Tensor> embedding_bwd(Tensor dy, Tensor ids, int64_t vocab,
   auto dwte = torch::zeros({vocab, dim}, f32);
   auto dwpe = npos ? torch::zeros({npos, dim}, f32) : torch::zeros({0}, f32);
   launch_embedding_bwd(bfp(dy), ids.data_ptr<int64_t>(),
-                       dwte.data_ptr<float>(),
-                       npos ? dwpe.data_ptr<float>() : nullptr, n_tok, seq,
-                       dim, npos > 0, stream());
+                       dwte.data_ptr<float>(), nullptr, n_tok, seq, dim,
+                       false, stream());
+  if (npos) {
+    // dwpe[s,:] = sum_b dy[b,s,:] — batch-dim column reduction over the
+    // [B, seq*dim] view (B-way same-address atomics measured dominant in
+    // the scatter version)
+    const int64_t B = n_tok / seq;
+    const int pcols = seq * dim;
+    TORCH_CHECK(pcols % 8 == 0, "seq*dim must be a multiple of 8");
+    const int stripes = dta_colred_stripes(B, pcols);
+    auto part = torch::empty({stripes, pcols}, f32);
+    launch_colsum(bfp(dy), part.data_ptr<float>(), dwpe.data_ptr<float>(),
+                  B, pcols, stripes, stream());
+  }
   return {dwte.to(torch::kBFloat16),
           npos ? dwpe.to(torch::kBFloat16) : dwpe};
 }

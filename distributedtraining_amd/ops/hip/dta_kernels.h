@@ -34,14 +34,23 @@ void launch_adamw_tick(int* t, float* bc, float beta1, float beta2,
                        hipStream_t s);
 void launch_colsum(const bf16_t* x, float* part, float* out, int64_t rows,
                    int cols, int stripes, hipStream_t s);
-// stripe count for the two-phase column reductions (colsum / norm dγdβ)
-inline int dta_colred_stripes(int64_t rows, int cols) {
-  const int gx = int((cols / 8 + 255) / 256);
-  int64_t st = 1024 / (gx > 0 ? gx : 1);
+// launch config for the two-phase column reductions (colsum / norm dγdβ):
+// block sized to the active lane count (cols/8 threads, 64-multiple),
+// stripe count targeting ~512 blocks of 256-thread-equivalents.
+struct ColRedCfg { int threads; int gx; int stripes; };
+inline ColRedCfg dta_colred_cfg(int64_t rows, int cols) {
+  const int lanes = cols / 8;
+  int threads = lanes >= 256 ? 256 : ((lanes + 63) / 64) * 64;
+  if (threads < 64) threads = 64;
+  const int gx = (lanes + threads - 1) / threads;
+  int64_t st = (512 * 256) / (int64_t(gx) * threads);
   const int64_t mx = (rows + 31) / 32;
   if (st > mx) st = mx;
   if (st < 1) st = 1;
-  return int(st);
+  return ColRedCfg{threads, gx, int(st)};
+}
+inline int dta_colred_stripes(int64_t rows, int cols) {
+  return dta_colred_cfg(rows, cols).stripes;
 }
 
 // ---- merge plane ----------------------------------------------------------

@@ -305,9 +305,9 @@ void launch_f32_to_bf16(const float* x, bf16_t* y, int64_t n, hipStream_t s) {
 }
 void launch_colsum(const bf16_t* x, float* part, float* out, int64_t rows,
                    int cols, int stripes, hipStream_t s) {
-  const int gx = (cols / 8 + 255) / 256;
-  dim3 g1(gx, unsigned(stripes));
-  colsum_part_k<<<g1, 256, 0, s>>>(x, part, rows, cols);
+  const ColRedCfg cfg = dta_colred_cfg(rows, cols);
+  dim3 g1(cfg.gx, unsigned(stripes));
+  colsum_part_k<<<g1, cfg.threads, 0, s>>>(x, part, rows, cols);
   const int g2 = (cols / 4 + 255) / 256;
   const int ry = stripes < 32 ? stripes : 32;
   colsum_reduce_k<<<dim3(g2, ry), 256, 0, s>>>(part, out, stripes, cols);

@@ -265,11 +265,12 @@ void dispatch_bwd(const ushort* dy, const ushort* x, const ushort* w,
                                                  rows, cols);
   }
 #undef CASE_B
-  const int gx = (cols / 8 + 255) / 256;
-  dim3 g2(gx, unsigned(stripes));
-  norm_bwd_dwdb_part_k<RMS><<<g2, 256, 0, s>>>(dy, x, mean, rstd, pdw,
-                                               RMS ? nullptr : pdb, rows,
-                                               cols);
+  const ColRedCfg cfg = dta_colred_cfg(rows, cols);
+  dim3 g2(cfg.gx, unsigned(stripes));
+  norm_bwd_dwdb_part_k<RMS><<<g2, cfg.threads, 0, s>>>(dy, x, mean, rstd,
+                                                       pdw,
+                                                       RMS ? nullptr : pdb,
+                                                       rows, cols);
   const int g3 = (cols / 4 + 255) / 256;
   const int ry = stripes < 32 ? stripes : 32;
   dwdb_reduce_k<<<dim3(g3, ry), 256, 0, s>>>(pdw, RMS ? nullptr : pdb, dw,
