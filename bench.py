@@ -42,7 +42,7 @@ def main() -> int:
     p.add_argument("--gpus", type=int, default=1)
     p.add_argument("--steps", type=int, default=30)
     p.add_argument("--warmup", type=int, default=10)
-    p.add_argument("--batch-size", type=int, default=512)
+    p.add_argument("--batch-size", type=int, default=1024)
     p.add_argument("--seq-len", type=int, default=64)
     p.add_argument("--merge-every", type=int, default=25)
     p.add_argument("--model", default="gpt2-small",
