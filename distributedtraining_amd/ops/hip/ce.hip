@@ -3,10 +3,11 @@
 // [tokens, 50257] bf16). Replaces the reference's transformers-internal CE
 // (`labels=` path, training_manager.py:380-385).
 //
-// fwd: one block per row, TWO phases (max pass, then sum-exp pass — the
-//      second pass hits the row in L2/L3); exp/log via the single-
-//      instruction exp2/log2 path (a fused online version was measured
-//      3x slower: the per-element rescale branch serializes on expf).
+// fwd: one block per row, ONE online pass: per s16x8 vector take the local
+//      max (cheap fmax tree), then one branch-free accumulator rescale
+//      s = s*exp2(m_old−m_new) + Σ exp2(x_j−m_new) — 9 exp2 per 8 elements.
+//      (A per-ELEMENT rescale branch measured 3x slower than two-pass; the
+//      per-VECTOR unconditional rescale beats both: single HBM pass.)
 // bwd: dlogits = scale_row * (softmax - onehot), one streaming pass.
 #include "dta_common.h"
 #include "dta_kernels.h"
@@ -23,41 +24,38 @@ __global__ void ce_fwd_k(const ushort* __restrict__ logits,
   __shared__ float lds[16];
   for (int64_t row = blockIdx.x; row < rows; row += gridDim.x) {
     const ushort* xr = logits + row * vocab;
-    // phase 1: row max (branch-free fmax chain, 4 partials for ILP)
-    float m0 = -INFINITY, m1 = -INFINITY, m2 = -INFINITY, m3 = -INFINITY;
+    // online pass: per-thread running (m, s) in exp2 units, one rescale
+    // per 8-wide vector
+    float m_run = -INFINITY, s_run = 0.f;
     int64_t i = int64_t(threadIdx.x) * 8;
     const int64_t stride = int64_t(CE_BLOCK) * 8;
     for (; i + 8 <= vocab; i += stride) {
       s16x8 vx = *reinterpret_cast<const s16x8*>(xr + i);
-      m0 = fmaxf(m0, fmaxf(bf2f(ushort(vx[0])), bf2f(ushort(vx[1]))));
-      m1 = fmaxf(m1, fmaxf(bf2f(ushort(vx[2])), bf2f(ushort(vx[3]))));
-      m2 = fmaxf(m2, fmaxf(bf2f(ushort(vx[4])), bf2f(ushort(vx[5]))));
-      m3 = fmaxf(m3, fmaxf(bf2f(ushort(vx[6])), bf2f(ushort(vx[7]))));
+      float f[8];
+#pragma unroll
+      for (int j = 0; j < 8; ++j) f[j] = bf2f(ushort(vx[j])) * LOG2E;
+      float mx = fmaxf(fmaxf(fmaxf(f[0], f[1]), fmaxf(f[2], f[3])),
+                       fmaxf(fmaxf(f[4], f[5]), fmaxf(f[6], f[7])));
+      const float m_new = fmaxf(m_run, mx);
+      float ps = 0.f;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) ps += __builtin_exp2f(f[j] - m_new);
+      s_run = s_run * __builtin_exp2f(m_run - m_new) + ps;
+      m_run = m_new;
     }
     if (i < vocab && i + 8 > vocab)
-      for (; i < vocab; ++i) m0 = fmaxf(m0, bf2f(xr[i]));
-    float m = fmaxf(fmaxf(m0, m1), fmaxf(m2, m3));
-    const float M = block_max<16>(m, lds);
-    // phase 2: sum exp2((x-M)*log2e) — row now L2-resident
-    const float mb = M * LOG2E;
-    float s0 = 0.f, s1 = 0.f, s2 = 0.f, s3 = 0.f;
-    i = int64_t(threadIdx.x) * 8;
-    for (; i + 8 <= vocab; i += stride) {
-      s16x8 vx = *reinterpret_cast<const s16x8*>(xr + i);
-      s0 += __builtin_exp2f(bf2f(ushort(vx[0])) * LOG2E - mb) +
-            __builtin_exp2f(bf2f(ushort(vx[1])) * LOG2E - mb);
-      s1 += __builtin_exp2f(bf2f(ushort(vx[2])) * LOG2E - mb) +
-            __builtin_exp2f(bf2f(ushort(vx[3])) * LOG2E - mb);
-      s2 += __builtin_exp2f(bf2f(ushort(vx[4])) * LOG2E - mb) +
-            __builtin_exp2f(bf2f(ushort(vx[5])) * LOG2E - mb);
-      s3 += __builtin_exp2f(bf2f(ushort(vx[6])) * LOG2E - mb) +
-            __builtin_exp2f(bf2f(ushort(vx[7])) * LOG2E - mb);
-    }
-    if (i < vocab && i + 8 > vocab)
-      for (; i < vocab; ++i)
-        s0 += __builtin_exp2f(bf2f(xr[i]) * LOG2E - mb);
-    const float S = block_sum<16>(s0 + s1 + s2 + s3, lds);
-    const float l = M + fast_log(S);
+      for (; i < vocab; ++i) {
+        const float f = bf2f(xr[i]) * LOG2E;
+        const float m_new = fmaxf(m_run, f);
+        s_run = s_run * __builtin_exp2f(m_run - m_new) +
+                __builtin_exp2f(f - m_new);
+        m_run = m_new;
+      }
+    // combine the per-thread (m, s) pairs across the block
+    const float M = block_max<16>(m_run, lds);
+    const float s_adj = s_run * __builtin_exp2f(m_run - M);
+    const float S = block_sum<16>(s_adj, lds);
+    const float l = (M + __builtin_log2f(S)) * LN2;  // natural-log lse
     if (threadIdx.x == 0) {
       lse[row] = l;
       const int64_t tgt = targets[row];
@@ -96,7 +94,9 @@ __global__ void ce_bwd_k(const ushort* __restrict__ logits,
         float p = __builtin_exp2f(bf2f(ushort(vx[j])) * LOG2E - lb);
         o[j] = f2bf(sc * (p - ((i + j) == tgt ? 1.f : 0.f)));
       }
-      *reinterpret_cast<s16x8*>(dxr + i) = o;
+      // nontemporal: dlogits is consumed by the big head GEMMs, keep it
+      // out of L2 so the logits read stream stays resident
+      __builtin_nontemporal_store(o, reinterpret_cast<s16x8*>(dxr + i));
     }
     if (i < vocab && i + 8 > vocab)
       for (; i < vocab; ++i) {
