@@ -1,0 +1,70 @@
+"""End-to-end protocol over separate OS processes (the reference's
+deployment shape: independent miner/validator/averager processes sharing
+storage, SURVEY.md §4 'multi-node-without-a-cluster').
+
+Two miners train and push deltas -> validator scores them -> averager
+merges and publishes a new base -> a resumed miner pulls it.
+"""
+
+import json
+import os
+import subprocess
+import sys
+
+import pytest
+import torch
+
+REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+
+
+def _run(args, timeout=240):
+    r = subprocess.run([sys.executable, "-m", "distributedtraining_amd.cli",
+                        *args], cwd=REPO, capture_output=True, text=True,
+                       timeout=timeout)
+    assert r.returncode == 0, f"{args} failed:\n{r.stdout}\n{r.stderr}"
+    return r.stdout
+
+
+@pytest.mark.timeout(600)
+def test_three_role_pipeline(tmp_path):
+    root = str(tmp_path / "ex")
+    common = ["--tiny", "--comm.root", root,
+              "--metrics-dir", str(tmp_path / "metrics"),
+              "--train.batch-size", "2", "--train.seq-len", "16",
+              "--validate.batch-size", "2", "--validate.seq-len", "16",
+              "--validate.n-eval-batches", "2"]
+
+    out1 = _run(["miner", "--hotkey", "m0", "--steps", "6", *common])
+    assert "avg loss" in out1
+    _run(["miner", "--hotkey", "m1", "--steps", "6", *common])
+
+    # both miners registered + deltas present
+    reg = json.load(open(os.path.join(root, "registry.json")))
+    assert set(reg["addresses"]) >= {"m0", "m1"}
+    assert os.path.exists(os.path.join(root, "grads", "m0",
+                                       "weight_diff.pt"))
+
+    out_v = _run(["validator", "--rounds", "1", *common])
+    assert "scores:" in out_v and "m0" in out_v
+
+    base_before = torch.load(os.path.join(root, "model",
+                                          "averaged_model.pt"),
+                             weights_only=False)["flat_master"].clone()
+    out_a = _run(["averager", "--rounds", "1",
+                  "--average.strategy", "mean", *common])
+    assert "merged + published" in out_a
+    base_after = torch.load(os.path.join(root, "model",
+                                         "averaged_model.pt"),
+                            weights_only=False)["flat_master"]
+    assert not torch.equal(base_before, base_after)
+
+    # resumed miner picks up its saved train state
+    out2 = _run(["miner", "--hotkey", "m0", "--steps", "2", "--resume",
+                 *common])
+    assert "resumed at step 6" in out2
+
+    # metrics JSONL written per role
+    mdir = tmp_path / "metrics"
+    names = {p.name for p in mdir.iterdir()}
+    assert {"miner_m0.jsonl", "validator_validator.jsonl",
+            "AVERAGER.jsonl"} <= names

@@ -1,0 +1,88 @@
+"""Numerics parity of the native Llama vs HuggingFace transformers.
+
+The reference has no Llama path; this is the scale-model family (BASELINE
+config #4). A tiny config with GQA (n_kv_head < n_head) is checked against
+transformers.LlamaForCausalLM on identical weights: logits, internal
+shifted-CE loss, and gradient direction.
+"""
+
+import pytest
+import torch
+
+from distributedtraining_amd.config import ModelConfig
+from distributedtraining_amd.models import LlamaLM
+
+transformers = pytest.importorskip("transformers")
+
+
+def _tiny_cfg():
+    return ModelConfig(family="llama", vocab_size=512, n_layer=2, n_head=4,
+                       n_kv_head=2, n_embd=64, n_positions=128,
+                       intermediate_size=176, rope_theta=10000.0,
+                       norm_eps=1e-5, tie_word_embeddings=False)
+
+
+def _hf_and_ours(seed=0):
+    from transformers import LlamaConfig, LlamaForCausalLM
+    cfg = _tiny_cfg()
+    hf_cfg = LlamaConfig(vocab_size=cfg.vocab_size,
+                         hidden_size=cfg.n_embd,
+                         intermediate_size=cfg.intermediate_size,
+                         num_hidden_layers=cfg.n_layer,
+                         num_attention_heads=cfg.n_head,
+                         num_key_value_heads=cfg.n_kv_head,
+                         max_position_embeddings=cfg.n_positions,
+                         rms_norm_eps=cfg.norm_eps,
+                         rope_theta=cfg.rope_theta,
+                         attention_bias=False, tie_word_embeddings=False,
+                         attention_dropout=0.0)
+    torch.manual_seed(seed)
+    hf = LlamaForCausalLM(hf_cfg).eval()
+    ours = LlamaLM(cfg).eval()
+    sd = hf.state_dict()
+    with torch.no_grad():
+        ours.tok_emb.copy_(sd["model.embed_tokens.weight"])
+        ours.lm_head_w.copy_(sd["lm_head.weight"])
+        ours.final_norm_w.copy_(sd["model.norm.weight"])
+        for i, blk in enumerate(ours.blocks):
+            pre = f"model.layers.{i}."
+            blk.attn_norm_w.copy_(sd[pre + "input_layernorm.weight"])
+            blk.q_w.copy_(sd[pre + "self_attn.q_proj.weight"])
+            blk.k_w.copy_(sd[pre + "self_attn.k_proj.weight"])
+            blk.v_w.copy_(sd[pre + "self_attn.v_proj.weight"])
+            blk.o_w.copy_(sd[pre + "self_attn.o_proj.weight"])
+            blk.mlp_norm_w.copy_(sd[pre + "post_attention_layernorm.weight"])
+            blk.gate_w.copy_(sd[pre + "mlp.gate_proj.weight"])
+            blk.up_w.copy_(sd[pre + "mlp.up_proj.weight"])
+            blk.down_w.copy_(sd[pre + "mlp.down_proj.weight"])
+    return hf, ours
+
+
+def _rope_permute(ours):
+    """HF applies RoPE over interleaved-halves the same way we do
+    (rotate_half convention) — no weight permutation needed; helper kept
+    to document the convention choice."""
+    return ours
+
+
+def test_llama_logits_match_transformers():
+    hf, ours = _hf_and_ours()
+    ids = torch.randint(0, 512, (2, 24))
+    with torch.no_grad():
+        ref = hf(input_ids=ids).logits
+        got = ours(input_ids=ids).logits
+    torch.testing.assert_close(got, ref, rtol=2e-4, atol=2e-4)
+
+
+def test_llama_loss_and_grad_match_transformers():
+    hf, ours = _hf_and_ours(seed=1)
+    ids = torch.randint(0, 512, (2, 32))
+    ref_out = hf(input_ids=ids, labels=ids)
+    got_out = ours(input_ids=ids, labels=ids)
+    torch.testing.assert_close(got_out.loss, ref_out.loss, rtol=1e-4,
+                               atol=1e-4)
+    ref_out.loss.backward()
+    got_out.loss.backward()
+    g_ref = hf.model.embed_tokens.weight.grad
+    g_got = ours.tok_emb.grad
+    torch.testing.assert_close(g_got, g_ref, rtol=5e-3, atol=1e-5)
