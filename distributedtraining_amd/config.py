@@ -98,11 +98,16 @@ class AverageConfig:
 
     meta_epochs: int = 7         # reference: neurons/averager.py:106
     meta_lr: float = 0.01        # reference: neurons/averager.py:106
-    strategy: str = "parameterized"  # parameterized | score_weighted | genetic
+    strategy: str = "parameterized"  # parameterized | score_weighted | genetic | mean | nesterov
     # genetic-only (reference: GeneticAverager, averaging_logic.py:830-970)
     population_size: int = 20
     generations: int = 10
     mutation_sigma: float = 0.1
+    # nesterov-only (beyond-parity: DiLoCo-style outer optimizer — the
+    # merged delta treated as an outer pseudo-gradient with Nesterov
+    # momentum; reference merges have no outer optimizer)
+    outer_lr: float = 0.7
+    outer_momentum: float = 0.9
 
 
 @dataclass

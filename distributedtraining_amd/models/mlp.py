@@ -28,8 +28,9 @@ class FeedforwardNN(nn.Module):
         self.fc4 = nn.Linear(hidden, hidden)
         self.fc5 = nn.Linear(hidden, classes)
 
-    def forward(self, x: torch.Tensor,
+    def forward(self, input_ids: torch.Tensor,
                 labels: Optional[torch.Tensor] = None, **_) -> CausalLMOutput:
+        x = input_ids
         h = F.relu(self.fc1(x.flatten(1)))
         h = F.relu(self.fc2(h))
         h = F.relu(self.fc3(h))

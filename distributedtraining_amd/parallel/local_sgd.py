@@ -72,6 +72,8 @@ class LocalSGDNode:
         elif strat == "score_weighted":
             assert scores is not None
             merged = self.averager.score_weighted_merge(base, deltas, scores)
+        elif strat == "nesterov":
+            merged = self.averager.nesterov_merge(base, deltas)
         elif strat == "parameterized":
             if self.comm.rank == 0:
                 merged = self.averager.meta_learning(base, deltas,

@@ -47,6 +47,7 @@ class ParameterizedAverager:
         self.store = store
         self.registry = registry
         self.weights: Optional[torch.Tensor] = None  # W [N, S]
+        self.outer_m: Optional[torch.Tensor] = None  # nesterov momentum
 
     # -- delta collection (reference :365-420) -------------------------------
     def collect_deltas(self) -> Tuple[torch.Tensor, List[str]]:
@@ -105,10 +106,19 @@ class ParameterizedAverager:
                 ids = batch["input_ids"].to(self.fp.device)
                 labels = batch.get("labels", batch["input_ids"]).to(self.fp.device)
                 out = self.model(input_ids=ids, labels=labels)
+                if not bool(torch.isfinite(out.loss)):
+                    # a bad delta can make some merge diverge mid-search;
+                    # skip the update rather than poisoning W (the
+                    # reference has no guard here and NaNs out)
+                    log.warning("meta-learning: non-finite loss, "
+                                "skipping W update")
+                    continue
                 out.loss.backward()
                 gw = ops.grad_merge_weights(self.fp.grad, base, deltas,
                                             merged, self.fp.offsets)
                 W -= lr * gw.to(W.device)
+                if not bool(torch.isfinite(W).all()):
+                    W.copy_(self._uniform_weights(n))   # reset, keep going
                 merged = self.merged_from(base, deltas, W)
         self.weights = W
         return merged
@@ -172,6 +182,28 @@ class ParameterizedAverager:
         W = best.to(self.fp.device).view(n, 1).expand(n, S).contiguous()
         return self.merged_from(base, deltas, W)
 
+    # -- outer-momentum strategy (beyond parity: DiLoCo-style) ---------------
+    def nesterov_merge(self, base: torch.Tensor, deltas: torch.Tensor,
+                       lr: Optional[float] = None,
+                       mu: Optional[float] = None) -> torch.Tensor:
+        """Treat the mean delta as an outer pseudo-gradient and apply
+        Nesterov-momentum SGD to the base:
+            m   <- mu*m + d_avg
+            new <- base + lr*(mu*m + d_avg)
+        With mu=0, lr=1 this reduces to the reference's plain mean merge.
+        Deterministic given identical inputs, so every rank computes it
+        redundantly (no broadcast needed) as long as outer_m stays in
+        lockstep — it does: it's a pure function of the merge history."""
+        if deltas.shape[0] == 0:
+            return base.clone()
+        lr = self.cfg.outer_lr if lr is None else lr
+        mu = self.cfg.outer_momentum if mu is None else mu
+        d_avg = deltas.mean(dim=0)
+        if self.outer_m is None:
+            self.outer_m = torch.zeros_like(d_avg)
+        self.outer_m.mul_(mu).add_(d_avg)
+        return base + lr * (mu * self.outer_m + d_avg)
+
     # -- full round (reference run_periodic_averaging :544-583) ---------------
     def run_round(self, val_batches: List[dict]) -> torch.Tensor:
         """Collect deltas, merge per configured strategy, install + publish
@@ -190,6 +222,8 @@ class ParameterizedAverager:
             merged = (self.merged_from(base, deltas,
                                        self._uniform_weights(deltas.shape[0]))
                       if deltas.shape[0] else base.clone())
+        elif self.cfg.strategy == "nesterov":
+            merged = self.nesterov_merge(base, deltas)
         else:
             raise ValueError(f"unknown strategy {self.cfg.strategy!r}")
         self.fp.load_flat_master(merged)

@@ -114,3 +114,45 @@ def test_train_state_resume(tmp_path):
     assert st is not None and st["step_count"] == 3
     fp2.load_flat_master(st["flat_master"])
     torch.testing.assert_close(fp2.master, saved)
+
+
+def test_fixture_full_protocol_mlp():
+    """The reference's MNIST validator/averager twins: the whole
+    miner->validator->averager round on the classification fixture
+    (validation_logic.py:265-318, averaging_logic.py:586-760)."""
+    from distributedtraining_amd.config import AverageConfig, ValidateConfig
+    from distributedtraining_amd.roles.averager import ParameterizedAverager
+    from distributedtraining_amd.roles.validator import DeltaValidator
+    from distributedtraining_amd.store import DeltaCheckpoint
+
+    cfg = Config()
+    cfg.model = ModelConfig(family="mlp")
+    cfg.train.lr = 1e-3
+    torch.manual_seed(0)
+    model = build_model(cfg.model)
+    fp = FlatParams(model)
+    base = fp.snapshot()
+    ev_iter = mnist_like_batches(n=16, seed=77)
+    ev = [next(ev_iter) for _ in range(3)]
+
+    # two miners from the shared base: one trains, one pushes noise
+    loop = ClassifierLoop(model, fp, mnist_like_batches(n=32, seed=1),
+                          cfg.train)
+    for _ in range(20):
+        loop.train_step()
+    d_good = loop.make_delta()
+    d_noise = DeltaCheckpoint(torch.randn(fp.numel) * 0.02, fp.spec, "")
+
+    fp.load_flat_master(base)
+    validator = DeltaValidator(model, fp, ev, ValidateConfig())
+    scores = validator.validate_and_score({"good": d_good,
+                                           "noise": d_noise})
+    assert scores["good"] > scores["noise"]
+
+    deltas = torch.stack([d_good.flat, d_noise.flat])
+    av = ParameterizedAverager(model, fp, AverageConfig(meta_epochs=3,
+                                                        meta_lr=0.02))
+    merged = av.meta_learning(base, deltas, ev)
+    # meta-learning upweights the trained delta over the noise
+    assert float(av.weights[0].mean()) > float(av.weights[1].mean())
+    assert merged.shape == base.shape

@@ -138,3 +138,25 @@ def test_genetic_merge_runs(tmp_path):
     merged = av.genetic_merge(base, deltas, ev)
     assert merged.shape == base.shape
     assert not torch.isnan(merged).any()
+
+
+def test_nesterov_outer_merge(tmp_path):
+    """DiLoCo-style outer momentum: mu=0/lr=1 reduces to plain mean;
+    momentum accumulates across rounds."""
+    import torch
+    from distributedtraining_amd.config import AverageConfig
+    from distributedtraining_amd.roles.averager import ParameterizedAverager
+    cfg, model, fp, store, registry = _mk(tmp_path, "nv")
+    base = fp.snapshot()
+    deltas = torch.stack([torch.full((fp.numel,), 0.1),
+                          torch.full((fp.numel,), 0.3)])
+    av = ParameterizedAverager(model, fp, AverageConfig(strategy="nesterov"))
+    m0 = av.nesterov_merge(base, deltas, lr=1.0, mu=0.0)
+    torch.testing.assert_close(m0, base + 0.2)
+    # round 2 with momentum: m=0.2 (from round 1) -> m2=0.9*0.2+0.2=0.38;
+    # merged = base + 0.7*(0.9*0.38 + 0.2)
+    av2 = ParameterizedAverager(model, fp, AverageConfig(strategy="nesterov"))
+    av2.nesterov_merge(base, deltas)               # m=0.2
+    m2 = av2.nesterov_merge(base, deltas)          # m=0.38
+    expect = base + 0.7 * (0.9 * 0.38 + 0.2)
+    torch.testing.assert_close(m2, expect)
