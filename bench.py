@@ -125,7 +125,7 @@ def main() -> int:
 
     # warmup (untimed) — includes one merge to warm the collective path
     run_steps(args.warmup, 0)
-    if world > 1 or True:
+    if args.merge_every:
         node.merge_round()
 
     sync()

@@ -72,6 +72,15 @@ class CommPlane:
         dist.all_gather_into_tensor(out, flat.contiguous())
         return out.view(self.world_size, flat.numel())
 
+    def all_reduce_mean(self, flat: torch.Tensor) -> torch.Tensor:
+        """In-place mean over ranks — the O(P)-memory exchange for
+        uniform merges (all_gather_flat is O(world*P): 8 fp32 Llama-3-8B
+        deltas would be 256 GB; the mean never needs them resident)."""
+        if self.is_distributed:
+            dist.all_reduce(flat, op=dist.ReduceOp.SUM)
+            flat.div_(self.world_size)
+        return flat
+
     # -- C2/C5: base publication --------------------------------------------
     def broadcast_flat(self, flat: torch.Tensor, src: int = 0) -> torch.Tensor:
         if self.is_distributed:

@@ -62,19 +62,27 @@ class LocalSGDNode:
 
     # -- the delta exchange + merge (C1..C5) ---------------------------------
     def merge_round(self, scores: Optional[List[float]] = None) -> None:
+        """Exchange + merge. Memory shapes:
+        * mean/nesterov: ONE all-reduce of the flat delta — O(P) resident
+          (required for Llama-3-8B x 8 ranks: a gather would need 256 GB);
+        * score_weighted/parameterized: all-gather, all deltas HBM-resident
+          (8 fp32 GPT-2 deltas ≈ 4 GB — the scoring/meta paths need every
+          delta individually)."""
         base = self.miner.base
         delta = self.fp.make_delta(base)     # fused θ−θ_base
-        deltas = self.comm.all_gather_flat(delta.flat)   # [world, P] resident
         strat = self.merge_strategy
-        if strat in ("mean", "uniform"):
-            W = self.averager._uniform_weights(deltas.shape[0])
-            merged = self.averager.merged_from(base, deltas, W)
+        if strat in ("mean", "uniform", "nesterov"):
+            d = self.comm.all_reduce_mean(delta.flat)
+            if strat == "nesterov":
+                merged = self.averager.nesterov_merge(base, d.unsqueeze(0))
+            else:
+                merged = d.add_(base)        # in place: d becomes merged
         elif strat == "score_weighted":
             assert scores is not None
+            deltas = self.comm.all_gather_flat(delta.flat)
             merged = self.averager.score_weighted_merge(base, deltas, scores)
-        elif strat == "nesterov":
-            merged = self.averager.nesterov_merge(base, deltas)
         elif strat == "parameterized":
+            deltas = self.comm.all_gather_flat(delta.flat)
             if self.comm.rank == 0:
                 merged = self.averager.meta_learning(base, deltas,
                                                      self.val_batches)
