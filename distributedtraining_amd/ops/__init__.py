@@ -8,9 +8,11 @@ Every op has exactly two execution paths:
   ``torch.autograd.Function`` (see hip/*.hip). There is no eager fallback on
   GPU: missing extension ⇒ loud error (backend.require_ext).
 
-Plain projection GEMMs (QKV/out/MLP/LM-head matmuls) intentionally go
-through ``torch.nn.functional.linear`` → hipBLASLt, per the library-GEMM /
-hand-written-fused-op split; everything fused or memory-bound is ours.
+Plain projection GEMMs (QKV/out/MLP/LM-head matmuls) intentionally go to
+hipBLASLt (via ``ops.linear``: addmm with fused bias epilogue forward,
+custom colsum dbias + direct flat-plane wgrad accumulation backward), per
+the library-GEMM / hand-written-fused-op split; everything fused or
+memory-bound is ours.
 
 Reference behavior being reimplemented: the implicit per-step op set of
 GPT-2 training in /root/reference.

@@ -10,8 +10,10 @@
 //     of the projection output [B,S,H*D] (and dq/dk/dv written the same
 //     way) so the model layer does zero transpose/contiguous copies.
 //   * GQA native: kv head = h / group; no repeat_interleave materialization.
-//   * K/V tiles are read directly from global memory — at these sequence
-//     lengths a (b,h)'s K/V fit in L2 (guide §5 common-mistake 7).
+//   * row fragments stream from global (K/V fit in L2 at these seq
+//     lengths); k-major B-fragments are NOT gathered from global (8 scalar
+//     2B loads each, ~32x line waste) — row frags are scatter-stored
+//     transposed into wave-private LDS and re-read as 16B vectors.
 //   * swapped QK^T: mfma(A=K_tile, B=Q^T) puts a query's scores in lanes
 //     sharing (lane&15) so the softmax row-reduce is two shfl_xor ops;
 //     exp via the single-instruction exp2 path.

@@ -79,3 +79,26 @@ def test_two_rank_local_sgd(tmp_path):
     digests = [d for _, _, d in results]
     assert digests[0] == pytest.approx(digests[1], rel=1e-6), \
         "ranks diverged after merge"
+
+
+@pytest.mark.timeout(300)
+def test_four_rank_local_sgd_allreduce(tmp_path):
+    """World 4 through the O(P)-memory all-reduce merge path (the shape the
+    driver's 8-GPU scaling bench exercises)."""
+    world = 4
+    port = 29671
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    procs = [ctx.Process(target=_worker,
+                         args=(r, world, port, str(tmp_path), q))
+             for r in range(world)]
+    for p in procs:
+        p.start()
+    results = [q.get(timeout=240) for _ in range(world)]
+    for p in procs:
+        p.join(timeout=60)
+    assert all(ok for _, ok, _ in results), results
+    digests = [d for _, _, d in results]
+    for d in digests[1:]:
+        assert digests[0] == pytest.approx(d, rel=1e-6), \
+            "ranks diverged after merge"

@@ -458,3 +458,28 @@ def test_graphed_step_matches_eager():
                                atol=1e-4)
     assert g.step_count == eager2.step_count
     del m_e
+
+
+def test_attention_validator_seq512():
+    """The validator's eval shape (seq 512, reference neurons/validator.py:63)
+    through the packed path, vs fp32 sdpa."""
+    from distributedtraining_amd import ops
+    B, H, S, D = 2, 4, 512, 64
+    F = 3 * H * D
+    qkv = _rand_bf16(B, S, F, seed=31)
+    qkv.requires_grad_(True)
+    o = ops.qkv_attention(qkv, H)
+    do = _rand_bf16(B, S, H * D, seed=32)
+    o.backward(do)
+    qf = qkv.detach().float().requires_grad_(True)
+    E = H * D
+    q = qf[..., :E].view(B, S, H, D).transpose(1, 2)
+    k = qf[..., E:2 * E].view(B, S, H, D).transpose(1, 2)
+    v = qf[..., 2 * E:].view(B, S, H, D).transpose(1, 2)
+    of = torch.nn.functional.scaled_dot_product_attention(
+        q, k, v, is_causal=True, scale=1.0 / math.sqrt(D))
+    of = of.transpose(1, 2).reshape(B, S, E)
+    of.backward(do.float())
+    torch.testing.assert_close(o.float(), of.detach(), rtol=3e-2, atol=3e-2)
+    torch.testing.assert_close(qkv.grad.float(), qf.grad, rtol=5e-2,
+                               atol=6e-2)

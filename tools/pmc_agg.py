@@ -10,7 +10,7 @@ with open(out, 'w') as f:
     q = f"select kernel_name, counter_name, sum(value), count(*) from {cc[0]} group by kernel_name, counter_name"
     try:
         for kn, cn, v, n in c.execute(q):
-            short = re.sub(r'\(.*', '', kn)[:70]
+            short = re.sub(r'\(.*', '', kn.replace('(anonymous namespace)::', ''))[:70]
             f.write(f"{short}\t{cn}\t{v}\t{n}\n")
     except Exception as e:
         f.write(f"fallback: {e}\n")
