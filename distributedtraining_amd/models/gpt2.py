@@ -54,8 +54,8 @@ class GPT2Block(nn.Module):
         o = ops.qkv_attention(qkv, self.n_head)
         x = x + ops.linear(o, self.attn_proj_w, self.attn_proj_b)
         h = ops.layer_norm(x, self.ln_2_w, self.ln_2_b)
-        h = ops.gelu(ops.linear(h, self.mlp_fc_w, self.mlp_fc_b))
-        x = x + ops.linear(h, self.mlp_proj_w, self.mlp_proj_b)
+        x = x + ops.mlp_gelu(h, self.mlp_fc_w, self.mlp_fc_b,
+                             self.mlp_proj_w, self.mlp_proj_b)
         return x
 
 

@@ -31,7 +31,8 @@ from .backend import has_ext, require_ext, use_hip
 
 __all__ = [
     "layer_norm", "rms_norm", "gelu", "swiglu", "causal_attention",
-    "qkv_attention", "linear", "cross_entropy_loss", "embedding_fwd",
+    "qkv_attention", "linear", "mlp_gelu", "cross_entropy_loss",
+    "embedding_fwd",
     "rope", "adamw_step", "delta_sub", "axpy_", "weighted_merge",
     "grad_merge_weights", "has_nan", "l2norm",
 ]
@@ -84,6 +85,81 @@ def linear(x: torch.Tensor, w: torch.Tensor,
     if use_hip(x):
         return _LinearFn.apply(x, w, b)
     return F.linear(x, w, b)
+
+
+# --------------------------------------------------------------------------
+# Fused MLP: linear+GELU+linear through hipBLASLt epilogues
+# --------------------------------------------------------------------------
+_lt_fused_ok: Optional[bool] = None
+
+
+class _FusedMLPFn(torch.autograd.Function):
+    """GPT-2 MLP as one node: h = gelu(x W1ᵀ + b1); y = h W2ᵀ + b2.
+
+    Forward fc1 uses HIPBLASLT_EPILOGUE_GELU_AUX_BIAS (gelu fused into the
+    GEMM, pre-gelu aux saved); backward fuses dgelu + db1 into fc2's dgrad
+    GEMM (DGELU_BGRAD) — eliminating the standalone GELU fwd/bwd kernels
+    and the fc1 dbias reduction (≈5 ms/step at batch 1024). Weight grads
+    accumulate into the flat plane (main_grad) like ops.linear."""
+
+    @staticmethod
+    def forward(ctx, x, w1, b1, w2, b2):
+        m = require_ext()
+        x2 = x.reshape(-1, x.shape[-1])
+        h, aux = m.lt_linear_gelu_fwd(x2.contiguous(), w1, b1)
+        y = torch.addmm(b2, h, w2.t())
+        ctx.save_for_backward(x2, w1, w2, h, aux)
+        ctx.xshape = x.shape
+        ctx.w1grad = getattr(w1, "main_grad", None)
+        ctx.w2grad = getattr(w2, "main_grad", None)
+        return y.view(*x.shape[:-1], w2.shape[0])
+
+    @staticmethod
+    def backward(ctx, dy):
+        m = require_ext()
+        x2, w1, w2, h, aux = ctx.saved_tensors
+        dy2 = dy.reshape(-1, dy.shape[-1])
+        if dy2.stride(-1) != 1:
+            dy2 = dy2.contiguous()
+        db2 = m.colsum(dy2).to(w2.dtype)
+        if ctx.w2grad is not None:
+            ctx.w2grad.addmm_(dy2.t(), h)
+            dw2 = None
+        else:
+            dw2 = dy2.t().mm(h)
+        dh_pre, db1 = m.lt_dgrad_dgelu_bgrad(dy2, w2, aux)
+        if ctx.w1grad is not None:
+            ctx.w1grad.addmm_(dh_pre.t(), x2)
+            dw1 = None
+        else:
+            dw1 = dh_pre.t().mm(x2)
+        dx = dh_pre.mm(w1).view(ctx.xshape)
+        return dx, dw1, db1, dw2, db2
+
+
+def mlp_gelu(x: torch.Tensor, w1: torch.Tensor, b1: torch.Tensor,
+             w2: torch.Tensor, b2: torch.Tensor) -> torch.Tensor:
+    """y = gelu(x W1ᵀ + b1) W2ᵀ + b2 (GPT-2 MLP). On GPU, epilogue-fused
+    via hipBLASLt when available; falls back to the composed ops path once
+    if the epilogue heuristics reject the shapes."""
+    global _lt_fused_ok
+    if use_hip(x) and _lt_fused_ok is not False:
+        try:
+            out = _FusedMLPFn.apply(x, w1, b1, w2, b2)
+            _lt_fused_ok = True
+            return out
+        except RuntimeError as e:
+            if _lt_fused_ok is None:
+                import logging
+                logging.getLogger(__name__).warning(
+                    "hipBLASLt epilogue MLP unavailable (%s); composed "
+                    "path", e)
+                _lt_fused_ok = False
+            else:
+                raise
+    if use_hip(x):
+        return linear(gelu(linear(x, w1, b1)), w2, b2)
+    return F.linear(F.gelu(F.linear(x, w1, b1), approximate="tanh"), w2, b2)
 
 
 # --------------------------------------------------------------------------

@@ -474,7 +474,10 @@ Tensor mfma_selftest_32(Tensor A, Tensor B) {
 
 }  // namespace
 
+void register_lt_fused(pybind11::module& m);  // lt_fused.cpp
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  register_lt_fused(m);
   m.def("layernorm_fwd", &layernorm_fwd);
   m.def("layernorm_bwd", &layernorm_bwd);
   m.def("rmsnorm_fwd", &rmsnorm_fwd);

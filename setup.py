@@ -16,6 +16,7 @@ from torch.utils.cpp_extension import BuildExtension, CUDAExtension  # noqa: E40
 
 SRC = [
     "distributedtraining_amd/ops/hip/bindings.cpp",
+    "distributedtraining_amd/ops/hip/lt_fused.cpp",
     "distributedtraining_amd/ops/hip/elementwise.hip",
     "distributedtraining_amd/ops/hip/adamw.hip",
     "distributedtraining_amd/ops/hip/norms.hip",
@@ -34,6 +35,7 @@ setup(
         CUDAExtension(
             name="distributedtraining_amd._dta_hip",
             sources=SRC,
+            libraries=["hipblaslt"],
             extra_compile_args={
                 "cxx": ["-O3", "-std=c++17"],
                 "nvcc": ["-O3", "-std=c++17", "--offload-arch=gfx950"],

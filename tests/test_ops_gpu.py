@@ -483,3 +483,40 @@ def test_attention_validator_seq512():
     torch.testing.assert_close(o.float(), of.detach(), rtol=3e-2, atol=3e-2)
     torch.testing.assert_close(qkv.grad.float(), qf.grad, rtol=5e-2,
                                atol=6e-2)
+
+
+def test_fused_mlp_gelu_matches_composed():
+    """hipBLASLt epilogue MLP (GELU_AUX_BIAS fwd, DGELU_BGRAD bwd) vs the
+    fp32 composed reference; falls back gracefully if heuristics reject."""
+    from distributedtraining_amd import ops
+    T, E, I = 320, 128, 512
+    x = _rand_bf16(T, E, seed=41)
+    w1 = _rand_bf16(I, E, seed=42, scale=0.1)
+    b1 = _rand_bf16(I, seed=43, scale=0.1)
+    w2 = _rand_bf16(E, I, seed=44, scale=0.1)
+    b2 = _rand_bf16(E, seed=45, scale=0.1)
+    for t in (x, w1, b1, w2, b2):
+        t.requires_grad_(True)
+    y = ops.mlp_gelu(x, w1, b1, w2, b2)
+    dy = _rand_bf16(T, E, seed=46)
+    y.backward(dy)
+
+    import torch.nn.functional as F
+    xf = x.detach().float().requires_grad_(True)
+    w1f = w1.detach().float().requires_grad_(True)
+    b1f = b1.detach().float().requires_grad_(True)
+    w2f = w2.detach().float().requires_grad_(True)
+    b2f = b2.detach().float().requires_grad_(True)
+    yf = F.linear(F.gelu(F.linear(xf, w1f, b1f), approximate="tanh"),
+                  w2f, b2f)
+    yf.backward(dy.float())
+    torch.testing.assert_close(y.float(), yf.detach(), rtol=3e-2, atol=3e-2)
+    torch.testing.assert_close(x.grad.float(), xf.grad, rtol=5e-2, atol=5e-2)
+    torch.testing.assert_close(w1.grad.float(), w1f.grad, rtol=5e-2,
+                               atol=3e-1)
+    torch.testing.assert_close(b1.grad.float(), b1f.grad, rtol=5e-2,
+                               atol=3e-1)
+    torch.testing.assert_close(w2.grad.float(), w2f.grad, rtol=5e-2,
+                               atol=3e-1)
+    torch.testing.assert_close(b2.grad.float(), b2f.grad, rtol=5e-2,
+                               atol=3e-1)
