@@ -21,13 +21,12 @@ GPT-2 training in /root/reference.
 from __future__ import annotations
 
 import math
-from typing import Optional, Tuple
+from typing import Optional
 
 import torch
 import torch.nn.functional as F
 
-from . import backend
-from .backend import has_ext, require_ext, use_hip
+from .backend import require_ext, use_hip
 
 __all__ = [
     "layer_norm", "rms_norm", "gelu", "swiglu", "causal_attention",

@@ -18,7 +18,7 @@ is broadcast (C2/C5).
 from __future__ import annotations
 
 import logging
-from typing import Callable, Dict, List, Optional
+from typing import Dict, List, Optional
 
 import torch
 

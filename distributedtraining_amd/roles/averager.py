@@ -23,8 +23,7 @@ SURVEY.md §3.3); we never touch disk in the loop.
 from __future__ import annotations
 
 import logging
-import math
-from typing import Dict, List, Optional, Tuple
+from typing import List, Optional, Tuple
 
 import torch
 
@@ -32,7 +31,7 @@ from .. import ops
 from ..config import AverageConfig
 from ..parallel.flat import FlatParams
 from ..registry import Registry
-from ..store import DeltaCheckpoint, FileStore
+from ..store import FileStore
 
 log = logging.getLogger(__name__)
 

@@ -26,7 +26,7 @@ CNN/MLP fixtures.
 from __future__ import annotations
 
 import logging
-from typing import Dict, List, Optional
+from typing import List, Optional
 
 import torch
 import torch.nn.functional as F

@@ -24,11 +24,9 @@ refresh, matching the reference's design (training_manager.py:371-373).
 from __future__ import annotations
 
 import hashlib
-import io
 import os
 import tempfile
-import time
-from typing import Dict, List, Optional, Tuple
+from typing import List, Optional, Tuple
 
 import torch
 

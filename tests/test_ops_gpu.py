@@ -520,3 +520,53 @@ def test_fused_mlp_gelu_matches_composed():
                                atol=3e-1)
     torch.testing.assert_close(b2.grad.float(), b2f.grad, rtol=5e-2,
                                atol=3e-1)
+
+
+def test_graphed_step_survives_merge_rounds():
+    """The bench interleaves graph replays with eager merge_rounds that
+    rewrite master/work/optimizer state in place — the captured graph must
+    stay valid and match an eager miner driven through the same sequence."""
+    from distributedtraining_amd.config import Config, ModelConfig
+    from distributedtraining_amd.models import build_model
+    from distributedtraining_amd.parallel.flat import FlatParams
+    from distributedtraining_amd.parallel.graphstep import GraphedMinerStep
+    from distributedtraining_amd.roles.miner import DeltaLoop
+    from distributedtraining_amd.utils.data import synthetic_batches
+
+    cfg = Config()
+    cfg.model = ModelConfig.gpt2_tiny()
+
+    def mk():
+        torch.manual_seed(0)
+        model = build_model(cfg.model).to(DEV)
+        fp = FlatParams(model)
+        data = synthetic_batches(cfg.model.vocab_size, 4, 32, seed=1)
+        return DeltaLoop(model, fp, data, cfg.train)
+
+    batches = [{k: t.to(DEV) for k, t in b.items()}
+               for b in [next(synthetic_batches(cfg.model.vocab_size, 4, 32,
+                                                seed=9)) for _ in range(4)]]
+
+    def fake_merge(miner):
+        # what LocalSGDNode.merge_round does at world 1 with mean strategy:
+        # delta -> merged = base + delta -> install (in-place buffer reuse)
+        delta = miner.fp.make_delta(miner.base).flat
+        merged = delta.add_(miner.base)
+        miner.install_base(merged)
+
+    g = mk()
+    eager = mk()
+    for _ in range(3):                      # match GraphedMinerStep warmup
+        eager.train_step(batches[0])
+    gs = GraphedMinerStep(g, batches, warmup=3)
+    for i in range(3):
+        gs.step(batches[i % 4])
+        eager.train_step(batches[i % 4])
+    fake_merge(g)
+    fake_merge(eager)
+    for i in range(3):
+        gs.step(batches[(i + 1) % 4])
+        eager.train_step(batches[(i + 1) % 4])
+    torch.cuda.synchronize()
+    torch.testing.assert_close(g.fp.master, eager.fp.master, rtol=1e-4,
+                               atol=1e-4)
