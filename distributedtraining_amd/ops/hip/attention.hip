@@ -83,6 +83,52 @@ DEV bf16x8 load_frag_col_lds(const ushort* lds_t, int row0, int col) {
   return u.b;
 }
 
+// ---- block-cooperative tile staging ---------------------------------------
+// The 4 waves of a block share one (b,h)'s K/V (fwd, dq) or Q/dO (dkv)
+// tiles; loading them per wave re-read ~590 MB/step from L2/HBM at the
+// flagship shape. All 256 threads cooperatively stage the 32-row tile
+// once per iteration (row-major for A-fragments, transposed for
+// B-fragments); rows past seq are clamped (masked upstream).
+// Row-major pitch: D+8 (2-way max bank conflict on 16-row A reads).
+template <int DTILES>
+DEV void stage_tile_rm(ushort* lds_rm, const ushort* src, int64_t rstride,
+                       int row0, int seq, int tid) {
+  constexpr int CPR = 2 * DTILES;       // 16B chunks per row (D/8)
+  constexpr int RP = 16 * DTILES + 8;
+  for (int i = tid; i < 32 * CPR; i += 256) {
+    const int r = i / CPR, c8 = (i % CPR) * 8;
+    const int rr = row0 + r;
+    const s16x8 v = *reinterpret_cast<const s16x8*>(
+        src + int64_t(rr < seq ? rr : seq - 1) * rstride + c8);
+    *reinterpret_cast<s16x8*>(lds_rm + r * RP + c8) = v;
+  }
+}
+
+template <int DTILES>
+DEV void stage_tile_T2(ushort* lds_t, const ushort* src, int64_t rstride,
+                       int row0, int seq, int tid) {
+  constexpr int CPR = 2 * DTILES;
+  for (int i = tid; i < 32 * CPR; i += 256) {
+    const int r = i / CPR, c8 = (i % CPR) * 8;
+    const int rr = row0 + r;
+    const s16x8 v = *reinterpret_cast<const s16x8*>(
+        src + int64_t(rr < seq ? rr : seq - 1) * rstride + c8);
+#pragma unroll
+    for (int j = 0; j < 8; ++j)
+      lds_t[(c8 + j) * LDS_RP + r] = ushort(v[j]);
+  }
+}
+
+template <int DTILES>
+DEV bf16x8 load_frag_rm(const ushort* lds_rm, int row_rel, int c0) {
+  constexpr int RP = 16 * DTILES + 8;
+  const s16x8 v =
+      *reinterpret_cast<const s16x8*>(lds_rm + row_rel * RP + c0);
+  union { s16x8 s; bf16x8 b; } u;
+  u.s = v;
+  return u.b;
+}
+
 DEV bf16x8 pack_bf16x8(const float* f) {
   bf16x8 o;
 #pragma unroll
@@ -112,20 +158,24 @@ DEV bf16x8 scores_to_afrag(const f32x4& p0, const f32x4& p1, int lane) {
 }
 
 // ---------------- forward ----------------
+// Block = 4 waves = 64 q rows of one (b,h); K/V tiles staged ONCE per
+// block per 32-key iteration (cooperative, barrier-synchronized — every
+// thread reaches both __syncthreads regardless of seq masking).
 template <int DTILES>  // D = 16*DTILES
 __global__ void attn_fwd_k(const ushort* __restrict__ q,
                            const ushort* __restrict__ k,
                            const ushort* __restrict__ v,
                            ushort* __restrict__ o, float* __restrict__ lse,
                            AttnGeom geo) {
-  const int lane = threadIdx.x & 63;
-  const int wid = threadIdx.x >> 6;
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;
   const int64_t bhid = blockIdx.y;
   const int b = int(bhid) / geo.H, h = int(bhid) % geo.H;
   const int hk = h / geo.grp;
   const int seq = geo.seq;
   const int q0 = blockIdx.x * 64 + wid * 16;
-  if (q0 >= seq) return;
+  const bool live = q0 < seq;
   const ushort* qp = q + b * geo.qb + h * geo.qh;
   const ushort* kp = k + b * geo.kb + hk * geo.kh;
   const ushort* vp = v + b * geo.vb + hk * geo.vh;
@@ -143,27 +193,26 @@ __global__ void attn_fwd_k(const ushort* __restrict__ q,
   for (int t = 0; t < DTILES; ++t) acc[t] = f32x4{0, 0, 0, 0};
   float m_run = -INFINITY, l_run = 0.f;  // in exp2 units
 
-  __shared__ ushort ldsv_all[4][16 * DTILES * LDS_RP];
-  ushort* ldsv = ldsv_all[wid];
+  __shared__ ushort k_rm[32 * (16 * DTILES + 8)];
+  __shared__ ushort v_t[16 * DTILES * LDS_RP];
   const int g = lane >> 4;
-  const int kv_end = min(seq, q0 + 16);
-  for (int kv0 = 0; kv0 < kv_end; kv0 += 32) {
+  const int blk_kv_end = min(seq, int(blockIdx.x) * 64 + 64);  // uniform
+  const int my_kv_end = live ? min(seq, q0 + 16) : 0;
+  for (int kv0 = 0; kv0 < blk_kv_end; kv0 += 32) {
+    __syncthreads();   // previous iteration's LDS reads complete
+    stage_tile_rm<DTILES>(k_rm, kp, geo.ks, kv0, seq, tid);
+    stage_tile_T2<DTILES>(v_t, vp, geo.vs, kv0, seq, tid);
+    __syncthreads();
+    if (kv0 >= my_kv_end) continue;   // after barriers: uniform safety
     f32x4 p0 = {0, 0, 0, 0}, p1 = {0, 0, 0, 0};
 #pragma unroll
     for (int sl = 0; sl < DTILES / 2; ++sl) {
       const int c0 = 32 * sl + 8 * (lane >> 4);
-      const int kr0 = kv0 + (lane & 15);
-      const int kr1 = kr0 + 16;
-      bf16x8 ka = load_frag_row(kp, geo.ks, kr0 < seq ? kr0 : seq - 1, c0);
+      bf16x8 ka = load_frag_rm<DTILES>(k_rm, lane & 15, c0);
       p0 = mfma_bf16(ka, qb_[sl], p0);
-      bf16x8 kb2 = load_frag_row(kp, geo.ks, kr1 < seq ? kr1 : seq - 1, c0);
+      bf16x8 kb2 = load_frag_rm<DTILES>(k_rm, 16 + (lane & 15), c0);
       p1 = mfma_bf16(kb2, qb_[sl], p1);
-      bf16x8 va = load_frag_row(vp, geo.vs, kr0 < seq ? kr0 : seq - 1, c0);
-      stage_frag_T(ldsv, va, lane, 0, c0);
-      bf16x8 vb2 = load_frag_row(vp, geo.vs, kr1 < seq ? kr1 : seq - 1, c0);
-      stage_frag_T(ldsv, vb2, lane, 16, c0);
     }
-    __threadfence_block();
     float mx = -INFINITY;
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
@@ -197,11 +246,12 @@ __global__ void attn_fwd_k(const ushort* __restrict__ q,
     for (int t = 0; t < DTILES; ++t) {
       acc[t][0] *= a_r[0]; acc[t][1] *= a_r[1];
       acc[t][2] *= a_r[2]; acc[t][3] *= a_r[3];
-      bf16x8 vb = load_frag_col_lds(ldsv, 8 * g, 16 * t + (lane & 15));
+      bf16x8 vb = load_frag_col_lds(v_t, 8 * g, 16 * t + (lane & 15));
       acc[t] = mfma_bf16(pa, vb, acc[t]);
     }
   }
 
+  if (!live) return;
   // epilogue: O /= l; lse stored in NATURAL-log units = (m + log2 l)*ln2
   if (lane < 16 && qrow < seq)
     lse[bhid * seq + qrow] = (m_run + __builtin_log2f(l_run)) * LN2;
@@ -258,6 +308,8 @@ __global__ void attn_delta_k(const ushort* __restrict__ dout,
 }
 
 // ---------------- backward dQ ----------------
+// Same cooperative structure as forward: K (row-major + transposed) and V
+// (row-major) staged once per block per 32-key iteration.
 template <int DTILES>
 __global__ void attn_bwd_dq_k(const ushort* __restrict__ dout,
                               const ushort* __restrict__ q,
@@ -266,14 +318,15 @@ __global__ void attn_bwd_dq_k(const ushort* __restrict__ dout,
                               const float* __restrict__ lse,
                               const float* __restrict__ delta,
                               ushort* __restrict__ dq, AttnGeom geo) {
-  const int lane = threadIdx.x & 63;
-  const int wid = threadIdx.x >> 6;
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;
   const int64_t bhid = blockIdx.y;
   const int b = int(bhid) / geo.H, h = int(bhid) % geo.H;
   const int hk = h / geo.grp;
   const int seq = geo.seq;
   const int q0 = blockIdx.x * 64 + wid * 16;
-  if (q0 >= seq) return;
+  const bool live = q0 < seq;
   const ushort* qp = q + b * geo.qb + h * geo.qh;
   const ushort* kp = k + b * geo.kb + hk * geo.kh;
   const ushort* vp = v + b * geo.vb + hk * geo.vh;
@@ -296,29 +349,33 @@ __global__ void attn_bwd_dq_k(const ushort* __restrict__ dout,
 #pragma unroll
   for (int t = 0; t < DTILES; ++t) acc[t] = f32x4{0, 0, 0, 0};
 
-  __shared__ ushort ldsk_all[4][16 * DTILES * LDS_RP];
-  ushort* ldsk = ldsk_all[wid];
+  __shared__ ushort k_rm[32 * (16 * DTILES + 8)];
+  __shared__ ushort v_rm[32 * (16 * DTILES + 8)];
+  __shared__ ushort k_t[16 * DTILES * LDS_RP];
   const int g = lane >> 4;
-  const int kv_end = min(seq, q0 + 16);
-  for (int kv0 = 0; kv0 < kv_end; kv0 += 32) {
+  const int blk_kv_end = min(seq, int(blockIdx.x) * 64 + 64);  // uniform
+  const int my_kv_end = live ? min(seq, q0 + 16) : 0;
+  for (int kv0 = 0; kv0 < blk_kv_end; kv0 += 32) {
+    __syncthreads();
+    stage_tile_rm<DTILES>(k_rm, kp, geo.ks, kv0, seq, tid);
+    stage_tile_rm<DTILES>(v_rm, vp, geo.vs, kv0, seq, tid);
+    stage_tile_T2<DTILES>(k_t, kp, geo.ks, kv0, seq, tid);
+    __syncthreads();
+    if (kv0 >= my_kv_end) continue;
     f32x4 s0 = {0, 0, 0, 0}, s1 = {0, 0, 0, 0};
     f32x4 dp0 = {0, 0, 0, 0}, dp1 = {0, 0, 0, 0};
 #pragma unroll
     for (int sl = 0; sl < DTILES / 2; ++sl) {
       const int c0 = 32 * sl + 8 * (lane >> 4);
-      const int kr0 = kv0 + (lane & 15), kr1 = kr0 + 16;
-      bf16x8 ka = load_frag_row(kp, geo.ks, kr0 < seq ? kr0 : seq - 1, c0);
-      bf16x8 kb2 = load_frag_row(kp, geo.ks, kr1 < seq ? kr1 : seq - 1, c0);
-      bf16x8 va = load_frag_row(vp, geo.vs, kr0 < seq ? kr0 : seq - 1, c0);
-      bf16x8 vb2 = load_frag_row(vp, geo.vs, kr1 < seq ? kr1 : seq - 1, c0);
+      bf16x8 ka = load_frag_rm<DTILES>(k_rm, lane & 15, c0);
+      bf16x8 kb2 = load_frag_rm<DTILES>(k_rm, 16 + (lane & 15), c0);
+      bf16x8 va = load_frag_rm<DTILES>(v_rm, lane & 15, c0);
+      bf16x8 vb2 = load_frag_rm<DTILES>(v_rm, 16 + (lane & 15), c0);
       s0 = mfma_bf16(ka, qb_[sl], s0);
       s1 = mfma_bf16(kb2, qb_[sl], s1);
       dp0 = mfma_bf16(va, dob[sl], dp0);
       dp1 = mfma_bf16(vb2, dob[sl], dp1);
-      stage_frag_T(ldsk, ka, lane, 0, c0);
-      stage_frag_T(ldsk, kb2, lane, 16, c0);
     }
-    __threadfence_block();
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
       const int k0a = kv0 + 4 * g + r, k1a = k0a + 16;
@@ -332,10 +389,11 @@ __global__ void attn_bwd_dq_k(const ushort* __restrict__ dout,
     bf16x8 dsa = scores_to_afrag(s0, s1, lane);
 #pragma unroll
     for (int t = 0; t < DTILES; ++t) {
-      bf16x8 kcb = load_frag_col_lds(ldsk, 8 * g, 16 * t + (lane & 15));
+      bf16x8 kcb = load_frag_col_lds(k_t, 8 * g, 16 * t + (lane & 15));
       acc[t] = mfma_bf16(dsa, kcb, acc[t]);
     }
   }
+  if (!live) return;
   ushort* dqp = dq + b * geo.ob + h * geo.oh;  // dq uses o-geometry strides
 #pragma unroll
   for (int t = 0; t < DTILES; ++t)
@@ -349,9 +407,13 @@ __global__ void attn_bwd_dq_k(const ushort* __restrict__ dout,
 }
 
 // ---------------- backward dK/dV ----------------
-// wave owns 16 KV rows; iterates q tiles of 32. With GQA (grp>1) each
-// kv head is walked once per ATTACHED q head (blockIdx.y covers B*H) and
-// results are accumulated with fp32 atomics into dk/dv.
+// Block = 4 waves = 64 KV rows of one (b,h); Q/dO tiles staged once per
+// block per 32-query iteration (row-major for the score mfmas, transposed
+// for the accumulation B-fragments). With GQA (grp>1) each kv head is
+// walked once per ATTACHED q head (blockIdx.y covers B*H) and results are
+// accumulated with fp32 atomics into dk/dv. Early waves see a few
+// below-diagonal query tiles (probabilities masked to 0) — the price of a
+// uniform, barrier-safe loop.
 template <int DTILES, bool ATOMIC>
 __global__ void attn_bwd_dkv_k(const ushort* __restrict__ dout,
                                const ushort* __restrict__ q,
@@ -363,14 +425,15 @@ __global__ void attn_bwd_dkv_k(const ushort* __restrict__ dout,
                                float* __restrict__ dv32,
                                ushort* __restrict__ dk,
                                ushort* __restrict__ dv, AttnGeom geo) {
-  const int lane = threadIdx.x & 63;
-  const int wid = threadIdx.x >> 6;
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;
   const int64_t bhid = blockIdx.y;
   const int b = int(bhid) / geo.H, h = int(bhid) % geo.H;
   const int hk = h / geo.grp;
   const int seq = geo.seq;
   const int kv0 = blockIdx.x * 64 + wid * 16;
-  if (kv0 >= seq) return;
+  const bool live = kv0 < seq;
   const ushort* qp = q + b * geo.qb + h * geo.qh;
   const ushort* kp = k + b * geo.kb + hk * geo.kh;
   const ushort* vp = v + b * geo.vb + hk * geo.vh;
@@ -393,33 +456,34 @@ __global__ void attn_bwd_dkv_k(const ushort* __restrict__ dout,
     accv[t] = f32x4{0, 0, 0, 0};
   }
 
-  __shared__ ushort ldsq_all[4][16 * DTILES * LDS_RP];
-  __shared__ ushort ldsd_all[4][16 * DTILES * LDS_RP];
-  ushort* ldsq = ldsq_all[wid];
-  ushort* ldsd = ldsd_all[wid];
+  __shared__ ushort q_rm[32 * (16 * DTILES + 8)];
+  __shared__ ushort d_rm[32 * (16 * DTILES + 8)];
+  __shared__ ushort q_t[16 * DTILES * LDS_RP];
+  __shared__ ushort d_t[16 * DTILES * LDS_RP];
   const int g = lane >> 4;
-  const int q_start = (kv0 / 32) * 32;
+  const int q_start = int(blockIdx.x) * 64;   // uniform across the block
   for (int q0 = q_start; q0 < seq; q0 += 32) {
+    __syncthreads();
+    stage_tile_rm<DTILES>(q_rm, qp, geo.qs, q0, seq, tid);
+    stage_tile_rm<DTILES>(d_rm, dop, geo.ds, q0, seq, tid);
+    stage_tile_T2<DTILES>(q_t, qp, geo.qs, q0, seq, tid);
+    stage_tile_T2<DTILES>(d_t, dop, geo.ds, q0, seq, tid);
+    __syncthreads();
+    if (!live || q0 + 31 < kv0) continue;   // fully below the diagonal
     f32x4 s0 = {0, 0, 0, 0}, s1 = {0, 0, 0, 0};
     f32x4 dp0 = {0, 0, 0, 0}, dp1 = {0, 0, 0, 0};
 #pragma unroll
     for (int sl = 0; sl < DTILES / 2; ++sl) {
       const int c0 = 32 * sl + 8 * (lane >> 4);
-      const int qr0 = q0 + (lane & 15), qr1 = qr0 + 16;
-      bf16x8 qa = load_frag_row(qp, geo.qs, qr0 < seq ? qr0 : seq - 1, c0);
-      bf16x8 qa1 = load_frag_row(qp, geo.qs, qr1 < seq ? qr1 : seq - 1, c0);
+      bf16x8 qa = load_frag_rm<DTILES>(q_rm, lane & 15, c0);
+      bf16x8 qa1 = load_frag_rm<DTILES>(q_rm, 16 + (lane & 15), c0);
       s0 = mfma_bf16(qa, kb_[sl], s0);
       s1 = mfma_bf16(qa1, kb_[sl], s1);
-      bf16x8 doa = load_frag_row(dop, geo.ds, qr0 < seq ? qr0 : seq - 1, c0);
-      bf16x8 doa1 = load_frag_row(dop, geo.ds, qr1 < seq ? qr1 : seq - 1, c0);
+      bf16x8 doa = load_frag_rm<DTILES>(d_rm, lane & 15, c0);
+      bf16x8 doa1 = load_frag_rm<DTILES>(d_rm, 16 + (lane & 15), c0);
       dp0 = mfma_bf16(doa, vbf[sl], dp0);
       dp1 = mfma_bf16(doa1, vbf[sl], dp1);
-      stage_frag_T(ldsq, qa, lane, 0, c0);
-      stage_frag_T(ldsq, qa1, lane, 16, c0);
-      stage_frag_T(ldsd, doa, lane, 0, c0);
-      stage_frag_T(ldsd, doa1, lane, 16, c0);
     }
-    __threadfence_block();
     f32x4 p0, p1, ds0, ds1;
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
@@ -439,12 +503,13 @@ __global__ void attn_bwd_dkv_k(const ushort* __restrict__ dout,
     bf16x8 dsa = scores_to_afrag(ds0, ds1, lane);
 #pragma unroll
     for (int t = 0; t < DTILES; ++t) {
-      bf16x8 dob2 = load_frag_col_lds(ldsd, 8 * g, 16 * t + (lane & 15));
+      bf16x8 dob2 = load_frag_col_lds(d_t, 8 * g, 16 * t + (lane & 15));
       accv[t] = mfma_bf16(pa, dob2, accv[t]);
-      bf16x8 qcb = load_frag_col_lds(ldsq, 8 * g, 16 * t + (lane & 15));
+      bf16x8 qcb = load_frag_col_lds(q_t, 8 * g, 16 * t + (lane & 15));
       acck[t] = mfma_bf16(dsa, qcb, acck[t]);
     }
   }
+  if (!live) return;
   constexpr int D = 16 * DTILES;
 #pragma unroll
   for (int t = 0; t < DTILES; ++t)
