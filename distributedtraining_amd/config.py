@@ -123,6 +123,11 @@ class CommConfig:
     transport: str = "file"      # file | rccl | gloo
     root: str = "./dt_exchange"  # file transport root directory
     device: str = "cpu"
+    # wire dtype for the gather-based exchanges (score_weighted /
+    # parameterized strategies need all deltas resident): "fp32" is exact;
+    # "bf16" halves xGMI bytes + gather footprint (needed for Llama-scale
+    # models at 8 ranks). mean/nesterov all-reduce stays fp32 always.
+    exchange_dtype: str = "fp32"
 
 
 @dataclass

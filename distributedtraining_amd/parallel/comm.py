@@ -62,15 +62,24 @@ class CommPlane:
                 dist.barrier()
 
     # -- C1/C3/C4: delta exchange -------------------------------------------
-    def all_gather_flat(self, flat: torch.Tensor) -> torch.Tensor:
+    def all_gather_flat(self, flat: torch.Tensor,
+                        wire_dtype: Optional[torch.dtype] = None
+                        ) -> torch.Tensor:
         """Gather each rank's flat tensor -> [world, P] (every rank gets all
-        deltas, HBM-resident)."""
+        deltas, HBM-resident). ``wire_dtype`` (e.g. torch.bfloat16) halves
+        the xGMI bytes AND the resident gather footprint — 8 fp32
+        Llama-3-8B deltas are 256 GB (impossible), 8 bf16 are 128 GB (fits
+        in 288 GB HBM3E); the merge math upcasts per element."""
         if not self.is_distributed:
-            return flat.unsqueeze(0)
-        out = torch.empty(self.world_size * flat.numel(), dtype=flat.dtype,
-                          device=flat.device)
-        dist.all_gather_into_tensor(out, flat.contiguous())
-        return out.view(self.world_size, flat.numel())
+            return (flat if wire_dtype is None
+                    else flat.to(wire_dtype)).unsqueeze(0)
+        send = flat.contiguous()
+        if wire_dtype is not None and wire_dtype != flat.dtype:
+            send = send.to(wire_dtype)
+        out = torch.empty(self.world_size * send.numel(), dtype=send.dtype,
+                          device=send.device)
+        dist.all_gather_into_tensor(out, send)
+        return out.view(self.world_size, send.numel())
 
     def all_reduce_mean(self, flat: torch.Tensor) -> torch.Tensor:
         """In-place mean over ranks — the O(P)-memory exchange for

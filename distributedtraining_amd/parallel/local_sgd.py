@@ -48,6 +48,10 @@ class LocalSGDNode:
         self.averager = ParameterizedAverager(model, fp, cfg.average)
         self.merge_rounds = 0
 
+    def _wire_dtype(self):
+        return (torch.bfloat16 if self.cfg.comm.exchange_dtype == "bf16"
+                else None)
+
     def sync_initial_base(self) -> None:
         """Rank 0's random init becomes the shared base on all ranks."""
         self.comm.broadcast_flat(self.fp.master, src=0)
@@ -79,10 +83,12 @@ class LocalSGDNode:
                 merged = d.add_(base)        # in place: d becomes merged
         elif strat == "score_weighted":
             assert scores is not None
-            deltas = self.comm.all_gather_flat(delta.flat)
-            merged = self.averager.score_weighted_merge(base, deltas, scores)
+            deltas = self.comm.all_gather_flat(delta.flat, self._wire_dtype())
+            merged = self.averager.score_weighted_merge(
+                base, deltas.to(torch.float32), scores)
         elif strat == "parameterized":
-            deltas = self.comm.all_gather_flat(delta.flat)
+            deltas = self.comm.all_gather_flat(delta.flat, self._wire_dtype())
+            deltas = deltas.to(torch.float32)
             if self.comm.rank == 0:
                 merged = self.averager.meta_learning(base, deltas,
                                                      self.val_batches)

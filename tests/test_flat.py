@@ -70,3 +70,16 @@ def test_snapshot_delta_roundtrip():
     torch.testing.assert_close(ck.flat, torch.full_like(ck.flat, 0.5))
     fp.load_flat_master(base)
     torch.testing.assert_close(fp.master, base)
+
+
+def test_all_gather_wire_dtype_single_rank():
+    """bf16 wire-dtype exchange: halves gather bytes; single-rank path."""
+    import torch
+    from distributedtraining_amd.parallel.comm import CommPlane
+    comm = CommPlane(device=torch.device("cpu"))
+    flat = torch.randn(64, dtype=torch.float32)
+    g32 = comm.all_gather_flat(flat)
+    assert g32.dtype == torch.float32 and g32.shape == (1, 64)
+    g16 = comm.all_gather_flat(flat, torch.bfloat16)
+    assert g16.dtype == torch.bfloat16
+    torch.testing.assert_close(g16[0].float(), flat, rtol=1e-2, atol=1e-2)
