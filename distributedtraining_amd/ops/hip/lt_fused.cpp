@@ -242,9 +242,78 @@ std::vector<Tensor> lt_dgrad_dgelu_bgrad(Tensor dy, Tensor w2, Tensor aux) {
   return {dh, db1};
 }
 
+// dW += dy^T @ x with the bias gradient (colsum of dy) emitted by the
+// BGRADB epilogue — replaces the separate colsum pass AND the
+// addmm(beta=1) accumulation when supported.
+// dy [M,N] rm, x [M,E] rm, dw_acc [N,E] rm bf16 (accumulated in place,
+// beta=1). Returns db [N] bf16.
+Tensor lt_wgrad_bgradb(Tensor dy, Tensor x, Tensor dw_acc) {
+  check_in(dy, "dy"); check_in(x, "x"); check_in(dw_acc, "dw_acc");
+  const int64_t M = dy.size(0), N = dy.size(1), E = x.size(1);
+  TORCH_CHECK(dw_acc.size(0) == N && dw_acc.size(1) == E, "dw shape");
+  auto db = torch::empty({N}, dy.options());
+  // cm: D[E,N] = A(x cm [E,M], opN is wrong: x rm [M,E] = cm [E,M]; we
+  // need A=[E,M] yes opN) * B(dy rm [M,N] = cm [N,M], opT -> [M,N])
+  auto key = std::make_tuple(2000000 + int(HIPBLASLT_EPILOGUE_BGRADB), M,
+                             N, E);
+  std::unique_lock<std::mutex> lk(g_mu);
+  auto it = g_plans.find(key);
+  if (it == g_plans.end()) {
+    Plan p;
+    LT_CHECK(hipblasLtMatmulDescCreate(&p.op, HIPBLAS_COMPUTE_32F,
+                                       HIP_R_32F));
+    int32_t ta = HIPBLAS_OP_N, tb = HIPBLAS_OP_T;
+    LT_CHECK(hipblasLtMatmulDescSetAttribute(
+        p.op, HIPBLASLT_MATMUL_DESC_TRANSA, &ta, sizeof(ta)));
+    LT_CHECK(hipblasLtMatmulDescSetAttribute(
+        p.op, HIPBLASLT_MATMUL_DESC_TRANSB, &tb, sizeof(tb)));
+    uint32_t epi = HIPBLASLT_EPILOGUE_BGRADB;
+    LT_CHECK(hipblasLtMatmulDescSetAttribute(
+        p.op, HIPBLASLT_MATMUL_DESC_EPILOGUE, &epi, sizeof(epi)));
+    const void* bp = db.data_ptr();
+    LT_CHECK(hipblasLtMatmulDescSetAttribute(
+        p.op, HIPBLASLT_MATMUL_DESC_BIAS_POINTER, &bp, sizeof(bp)));
+    LT_CHECK(hipblasLtMatrixLayoutCreate(&p.la, HIP_R_16BF, E, M, E));
+    LT_CHECK(hipblasLtMatrixLayoutCreate(&p.lb, HIP_R_16BF, N, M, N));
+    LT_CHECK(hipblasLtMatrixLayoutCreate(&p.lc, HIP_R_16BF, E, N, E));
+    hipblasLtMatmulPreference_t pref;
+    LT_CHECK(hipblasLtMatmulPreferenceCreate(&pref));
+    uint64_t wsz = kWorkspace;
+    LT_CHECK(hipblasLtMatmulPreferenceSetAttribute(
+        pref, HIPBLASLT_MATMUL_PREF_MAX_WORKSPACE_BYTES, &wsz,
+        sizeof(wsz)));
+    hipblasLtMatmulHeuristicResult_t results[4];
+    int nres = 0;
+    LT_CHECK(hipblasLtMatmulAlgoGetHeuristic(handle(), p.op, p.la, p.lb,
+                                             p.lc, p.lc, pref, 4, results,
+                                             &nres));
+    hipblasLtMatmulPreferenceDestroy(pref);
+    TORCH_CHECK(nres > 0, "hipblasLt: no BGRADB algorithm M=", M, " N=",
+                N, " E=", E);
+    p.algo = results[0].algo;
+    p.ready = true;
+    it = g_plans.emplace(key, p).first;
+  }
+  Plan& p = it->second;
+  lk.unlock();
+  const void* bp = db.data_ptr();
+  LT_CHECK(hipblasLtMatmulDescSetAttribute(
+      p.op, HIPBLASLT_MATMUL_DESC_BIAS_POINTER, &bp, sizeof(bp)));
+  auto ws = at::empty({int64_t(kWorkspace)},
+                      dy.options().dtype(torch::kByte));
+  const float alpha = 1.0f, beta = 1.0f;  // accumulate into dw_acc
+  LT_CHECK(hipblasLtMatmul(handle(), p.op, &alpha, x.data_ptr(), p.la,
+                           dy.data_ptr(), p.lb, &beta, dw_acc.data_ptr(),
+                           p.lc, dw_acc.data_ptr(), p.lc, &p.algo,
+                           ws.data_ptr(), kWorkspace,
+                           at::hip::getCurrentHIPStream().stream()));
+  return db;
+}
+
 }  // namespace dta_lt
 
 void register_lt_fused(pybind11::module& m) {
   m.def("lt_linear_gelu_fwd", &dta_lt::lt_linear_gelu_fwd);
   m.def("lt_dgrad_dgelu_bgrad", &dta_lt::lt_dgrad_dgelu_bgrad);
+  m.def("lt_wgrad_bgradb", &dta_lt::lt_wgrad_bgradb);
 }

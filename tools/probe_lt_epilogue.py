@@ -29,6 +29,18 @@ for M in (320, 1024, 4096, 8192, 16384, 32768, 49152, 65536):
         bwd = "ok"
     except RuntimeError as e:
         bwd = "FAIL"
-    print(f"M={M:6d}: gelu_aux_bias={fwd}  dgelu_bgrad={bwd}", flush=True)
+    try:
+        dy2 = torch.randn(M, N, device=dev, dtype=torch.bfloat16)
+        dwacc = torch.zeros(N, K, device=dev, dtype=torch.bfloat16)
+        db2 = m.lt_wgrad_bgradb(dy2, x, dwacc)
+        # numerics spot check vs reference
+        ref_db = dy2.float().sum(0)
+        ok = torch.allclose(db2.float(), ref_db, rtol=3e-2, atol=max(1.0, 3e-3*M))
+        ref_dw = dy2.float().t() @ x.float()
+        okw = torch.allclose(dwacc.float(), ref_dw, rtol=5e-2, atol=max(1.0, 3e-3*M))
+        bg = f"ok(db={ok},dw={okw})"
+    except RuntimeError as e:
+        bg = "FAIL"
+    print(f"M={M:6d}: gelu_aux_bias={fwd}  dgelu_bgrad={bwd}  wgrad_bgradb={bg}", flush=True)
 torch.cuda.synchronize()
 print("done")
