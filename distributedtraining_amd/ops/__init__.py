@@ -66,21 +66,30 @@ class _LinearFn(torch.autograd.Function):
         if dy2.stride(-1) != 1:
             dy2 = dy2.contiguous()
         dx = dy2.mm(w).view(ctx.xshape)
+        m = require_ext()
+        if ctx.has_b and ctx.wgrad is not None:
+            # one GEMM: dW accumulated into the flat plane (beta=1) with
+            # the bias gradient emitted by the BGRADB epilogue — probed
+            # supported at every production shape on gfx950
+            try:
+                db = m.lt_wgrad_bgradb(dy2, x2, ctx.wgrad)
+                return dx, None, db
+            except RuntimeError:   # no epilogue kernel for this shape
+                pass
         if ctx.wgrad is not None:
             ctx.wgrad.addmm_(dy2.t(), x2)   # flat-plane accumulation
             dw = None
         else:
             dw = dy2.t().mm(x2)
-        db = None
-        if ctx.has_b:
-            db = require_ext().colsum(dy2).to(w.dtype)
+        db = m.colsum(dy2).to(w.dtype) if ctx.has_b else None
         return dx, dw, db
 
 
 def linear(x: torch.Tensor, w: torch.Tensor,
            b: Optional[torch.Tensor] = None) -> torch.Tensor:
     """y = x @ w.T (+ b). hipBLASLt GEMM with fused bias epilogue forward;
-    custom colsum dbias backward on GPU."""
+    backward: wgrad+bias-grad in one BGRADB-epilogue GEMM accumulated into
+    the flat-grad plane (colsum fallback)."""
     if use_hip(x):
         return _LinearFn.apply(x, w, b)
     return F.linear(x, w, b)

@@ -570,3 +570,27 @@ def test_graphed_step_survives_merge_rounds():
     torch.cuda.synchronize()
     torch.testing.assert_close(g.fp.master, eager.fp.master, rtol=1e-4,
                                atol=1e-4)
+
+
+def test_linear_bgradb_main_grad_path():
+    """ops.linear with main_grad set: wgrad accumulated via the BGRADB
+    epilogue GEMM (beta=1) with the bias grad emitted by the epilogue."""
+    from distributedtraining_amd import ops
+    T, E, N = 512, 256, 384
+    x = _rand_bf16(T, E, seed=51).requires_grad_(True)
+    w = _rand_bf16(N, E, seed=52, scale=0.1)
+    b = _rand_bf16(N, seed=53, scale=0.1)
+    w.requires_grad_(True)
+    b.requires_grad_(True)
+    pre = _rand_bf16(N, E, seed=54, scale=0.1)   # pre-existing accumulation
+    w.main_grad = pre.clone()
+    y = ops.linear(x, w, b)
+    dy = _rand_bf16(T, N, seed=55)
+    y.backward(dy)
+    ref_dw = pre.float() + dy.float().t() @ x.detach().float()
+    torch.testing.assert_close(w.main_grad.float(), ref_dw, rtol=5e-2,
+                               atol=3e-1)
+    torch.testing.assert_close(b.grad.float(), dy.float().sum(0), rtol=3e-2,
+                               atol=3e-1)
+    # autograd returned None for w -> .grad stays untouched (main_grad owns it)
+    assert w.grad is None
