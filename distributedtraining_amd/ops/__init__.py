@@ -40,6 +40,10 @@ __all__ = [
 # --------------------------------------------------------------------------
 # Linear (library GEMM via hipBLASLt) with fast dbias backward
 # --------------------------------------------------------------------------
+import os as _os
+
+_USE_BGRADB = _os.environ.get("DTA_LT_BGRADB", "0") == "1"
+
 class _LinearFn(torch.autograd.Function):
     """F.linear semantics with two backward optimizations:
 
@@ -67,10 +71,13 @@ class _LinearFn(torch.autograd.Function):
             dy2 = dy2.contiguous()
         dx = dy2.mm(w).view(ctx.xshape)
         m = require_ext()
-        if ctx.has_b and ctx.wgrad is not None:
+        if _USE_BGRADB and ctx.has_b and ctx.wgrad is not None:
             # one GEMM: dW accumulated into the flat plane (beta=1) with
-            # the bias gradient emitted by the BGRADB epilogue — probed
-            # supported at every production shape on gfx950
+            # the bias gradient via the BGRADB epilogue. Works at every
+            # production shape BUT the heuristic algorithm selection loses
+            # to the TunableOp stream-K wgrad kernels at K=65536 (measured
+            # 791k vs 877k tokens/s whole-step) — opt-in until the lt path
+            # gets its own algorithm tuning.
             try:
                 db = m.lt_wgrad_bgradb(dy2, x2, ctx.wgrad)
                 return dx, None, db

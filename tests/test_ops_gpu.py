@@ -574,8 +574,12 @@ def test_graphed_step_survives_merge_rounds():
 
 def test_linear_bgradb_main_grad_path():
     """ops.linear with main_grad set: wgrad accumulated via the BGRADB
-    epilogue GEMM (beta=1) with the bias grad emitted by the epilogue."""
+    epilogue GEMM (beta=1) with the bias grad emitted by the epilogue
+    (opt-in path: heuristic algo selection loses to TunableOp at the
+    flagship wgrad shapes, so production defaults to colsum)."""
     from distributedtraining_amd import ops
+    old = ops._USE_BGRADB
+    ops._USE_BGRADB = True
     T, E, N = 512, 256, 384
     x = _rand_bf16(T, E, seed=51).requires_grad_(True)
     w = _rand_bf16(N, E, seed=52, scale=0.1)
@@ -594,3 +598,4 @@ def test_linear_bgradb_main_grad_path():
                                atol=3e-1)
     # autograd returned None for w -> .grad stays untouched (main_grad owns it)
     assert w.grad is None
+    ops._USE_BGRADB = old
