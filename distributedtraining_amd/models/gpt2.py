@@ -47,15 +47,24 @@ class GPT2Block(nn.Module):
         self.mlp_proj_w = nn.Parameter(torch.empty(E, 4 * E))
         self.mlp_proj_b = nn.Parameter(torch.zeros(E))
 
-    def forward(self, x: torch.Tensor) -> torch.Tensor:
-        h = ops.layer_norm(x, self.ln_1_w, self.ln_1_b)
+    def forward(self, x: torch.Tensor, pending=None):
+        """Residual-join-fused layout: ``pending`` is the previous
+        sub-layer's un-added branch output; every LayerNorm consumes it
+        via ops.add_layer_norm (one kernel produces both the new residual
+        stream and the normalized input — no separate add anywhere in the
+        transformer). Returns (stream, pending)."""
+        if pending is None:
+            s = x
+            h = ops.layer_norm(x, self.ln_1_w, self.ln_1_b)
+        else:
+            s, h = ops.add_layer_norm(x, pending, self.ln_1_w, self.ln_1_b)
         qkv = ops.linear(h, self.attn_qkv_w, self.attn_qkv_b)
         o = ops.qkv_attention(qkv, self.n_head)
-        x = x + ops.linear(o, self.attn_proj_w, self.attn_proj_b)
-        h = ops.layer_norm(x, self.ln_2_w, self.ln_2_b)
-        x = x + ops.mlp_gelu(h, self.mlp_fc_w, self.mlp_fc_b,
-                             self.mlp_proj_w, self.mlp_proj_b)
-        return x
+        a = ops.linear(o, self.attn_proj_w, self.attn_proj_b)
+        s2, h2 = ops.add_layer_norm(s, a, self.ln_2_w, self.ln_2_b)
+        m = ops.mlp_gelu(h2, self.mlp_fc_w, self.mlp_fc_b,
+                         self.mlp_proj_w, self.mlp_proj_b)
+        return s2, m
 
 
 @dataclass
@@ -97,9 +106,13 @@ class GPT2LM(nn.Module):
                 labels: Optional[torch.Tensor] = None,
                 return_logits: Optional[bool] = None) -> CausalLMOutput:
         x = ops.embedding_fwd(input_ids, self.wte, self.wpe)
+        pending = None
         for blk in self.blocks:
-            x = blk(x)
-        x = ops.layer_norm(x, self.ln_f_w, self.ln_f_b)
+            x, pending = blk(x, pending)
+        if pending is None:
+            x = ops.layer_norm(x, self.ln_f_w, self.ln_f_b)
+        else:
+            _, x = ops.add_layer_norm(x, pending, self.ln_f_w, self.ln_f_b)
         if labels is None:
             logits = ops.linear(x, self.wte)
             return CausalLMOutput(loss=None, logits=logits)

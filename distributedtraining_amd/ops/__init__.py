@@ -30,7 +30,8 @@ from .backend import require_ext, use_hip
 
 __all__ = [
     "layer_norm", "rms_norm", "gelu", "swiglu", "causal_attention",
-    "qkv_attention", "linear", "mlp_gelu", "cross_entropy_loss",
+    "qkv_attention", "linear", "mlp_gelu", "add_layer_norm",
+    "cross_entropy_loss",
     "embedding_fwd",
     "rope", "adamw_step", "delta_sub", "axpy_", "weighted_merge",
     "grad_merge_weights", "has_nan", "l2norm",
@@ -180,12 +181,17 @@ def mlp_gelu(x: torch.Tensor, w1: torch.Tensor, b1: torch.Tensor,
 # --------------------------------------------------------------------------
 # LayerNorm
 # --------------------------------------------------------------------------
+def _empty_like0(t):
+    return t.new_empty(0)
+
+
 class _LayerNormFn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, w, b, eps):
         m = require_ext()
         x2 = x.contiguous()
-        y, mean, rstd = m.layernorm_fwd(x2.view(-1, x2.shape[-1]), w, b, eps)
+        y, mean, rstd, _ = m.layernorm_fwd(x2.view(-1, x2.shape[-1]),
+                                           _empty_like0(x2), w, b, eps)
         ctx.save_for_backward(x2, w, mean, rstd)
         return y.view_as(x2)
 
@@ -195,7 +201,8 @@ class _LayerNormFn(torch.autograd.Function):
         x, w, mean, rstd = ctx.saved_tensors
         N = x.shape[-1]
         dx, dw, db = m.layernorm_bwd(dy.contiguous().view(-1, N),
-                                     x.view(-1, N), w, mean, rstd)
+                                     _empty_like0(x), x.view(-1, N), w,
+                                     mean, rstd)
         return dx.view_as(x), dw, db, None
 
 
@@ -204,6 +211,49 @@ def layer_norm(x: torch.Tensor, w: torch.Tensor, b: torch.Tensor,
     if use_hip(x):
         return _LayerNormFn.apply(x, w, b, eps)
     return F.layer_norm(x, (x.shape[-1],), w, b, eps)
+
+
+class _AddLayerNormFn(torch.autograd.Function):
+    """Fused residual-add + LayerNorm: (s, y) = (x+res, LN(x+res)).
+
+    The sum is rounded to bf16 before the statistics so backward (which
+    recomputes the normalized values from the saved bf16 sum) matches
+    exactly; backward folds the sum-stream gradient ds into dx inside the
+    dx kernel. Together this removes the separate residual add kernels
+    forward AND backward (the residual joins were ~48 eager add
+    launches/step)."""
+
+    @staticmethod
+    def forward(ctx, x, res, w, b, eps):
+        m = require_ext()
+        x2 = x.contiguous().view(-1, x.shape[-1])
+        r2 = res.contiguous().view(-1, x.shape[-1])
+        y, mean, rstd, s = m.layernorm_fwd(x2, r2, w, b, eps)
+        ctx.save_for_backward(s, w, mean, rstd)
+        ctx.shape = x.shape
+        return s.view(x.shape), y.view(x.shape)
+
+    @staticmethod
+    def backward(ctx, ds, dy):
+        m = require_ext()
+        s, w, mean, rstd = ctx.saved_tensors
+        N = s.shape[-1]
+        ds2 = (ds.contiguous().view(-1, N) if ds is not None
+               else _empty_like0(s))
+        dx, dw, db = m.layernorm_bwd(dy.contiguous().view(-1, N), ds2, s,
+                                     w, mean, rstd)
+        dx = dx.view(ctx.shape)
+        return dx, dx, dw, db, None
+
+
+def add_layer_norm(x: torch.Tensor, res: torch.Tensor, w: torch.Tensor,
+                   b: torch.Tensor, eps: float = 1e-5):
+    """(s, y) = (x + res, layer_norm(x + res)) — the transformer residual
+    join fused into the norm (fwd add + bwd grad-join add eliminated)."""
+    if use_hip(x):
+        return _AddLayerNormFn.apply(x, res, w, b, eps)
+    s = x + res
+    return s, F.layer_norm(s, (s.shape[-1],), w, b, eps)
 
 
 # --------------------------------------------------------------------------

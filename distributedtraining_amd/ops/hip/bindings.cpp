@@ -33,37 +33,50 @@ void check_f32(const Tensor& t, const char* name) {
 }
 
 // ---- norms ----------------------------------------------------------------
-std::vector<Tensor> layernorm_fwd(Tensor x, Tensor w, Tensor b, double eps) {
+// res: empty tensor => plain LN; else fused residual (returns the bf16
+// sum as an extra output feeding the ongoing residual stream)
+std::vector<Tensor> layernorm_fwd(Tensor x, Tensor res, Tensor w, Tensor b,
+                                  double eps) {
   check_bf16(x, "x"); check_bf16(w, "w"); check_bf16(b, "b");
   const int64_t rows = x.size(0);
   const int cols = int(x.size(1));
   TORCH_CHECK(cols % 8 == 0, "cols must be a multiple of 8");
+  const bool has_res = res.numel() > 0;
+  if (has_res) check_bf16(res, "res");
   auto y = torch::empty_like(x);
+  auto sum = has_res ? torch::empty_like(x) : torch::empty({0}, x.options());
   auto mean = torch::empty({rows}, x.options().dtype(torch::kFloat32));
   auto rstd = torch::empty({rows}, x.options().dtype(torch::kFloat32));
-  launch_layernorm_fwd(bfp(x), bfp(w), bfp(b), bfp_mut(y),
-                       mean.data_ptr<float>(), rstd.data_ptr<float>(), rows,
-                       cols, float(eps), stream());
-  return {y, mean, rstd};
+  launch_layernorm_fwd(bfp(x), has_res ? bfp(res) : nullptr,
+                       has_res ? bfp_mut(sum) : nullptr, bfp(w), bfp(b),
+                       bfp_mut(y), mean.data_ptr<float>(),
+                       rstd.data_ptr<float>(), rows, cols, float(eps),
+                       stream());
+  return {y, mean, rstd, sum};
 }
 
-std::vector<Tensor> layernorm_bwd(Tensor dy, Tensor x, Tensor w, Tensor mean,
-                                  Tensor rstd) {
+// ds: empty tensor => plain; else the gradient arriving on the sum
+// stream, folded into dx (dx = ds + dLN/dx)
+std::vector<Tensor> layernorm_bwd(Tensor dy, Tensor ds, Tensor x, Tensor w,
+                                  Tensor mean, Tensor rstd) {
   check_bf16(dy, "dy"); check_bf16(x, "x"); check_bf16(w, "w");
   const int64_t rows = x.size(0);
   const int cols = int(x.size(1));
   TORCH_CHECK(cols % 8 == 0, "cols must be a multiple of 8");
+  const bool has_ds = ds.numel() > 0;
+  if (has_ds) check_bf16(ds, "ds");
   auto dx = torch::empty_like(x);
   auto f32 = x.options().dtype(torch::kFloat32);
   auto dw32 = torch::zeros({cols}, f32);
   auto db32 = torch::zeros({cols}, f32);
   const int stripes = dta_colred_stripes(rows, cols);
   auto part = torch::empty({2, stripes, cols}, f32);
-  launch_layernorm_bwd(bfp(dy), bfp(x), bfp(w), mean.data_ptr<float>(),
-                       rstd.data_ptr<float>(), bfp_mut(dx),
-                       dw32.data_ptr<float>(), db32.data_ptr<float>(),
-                       part[0].data_ptr<float>(), part[1].data_ptr<float>(),
-                       stripes, rows, cols, stream());
+  launch_layernorm_bwd(bfp(dy), has_ds ? bfp(ds) : nullptr, bfp(x), bfp(w),
+                       mean.data_ptr<float>(), rstd.data_ptr<float>(),
+                       bfp_mut(dx), dw32.data_ptr<float>(),
+                       db32.data_ptr<float>(), part[0].data_ptr<float>(),
+                       part[1].data_ptr<float>(), stripes, rows, cols,
+                       stream());
   return {dx, dw32.to(torch::kBFloat16), db32.to(torch::kBFloat16)};
 }
 

@@ -69,12 +69,17 @@ void launch_grad_merge_weights(const void* g, bool g_is_bf16,
                                int n_segs, hipStream_t s);
 
 // ---- norms ----------------------------------------------------------------
-void launch_layernorm_fwd(const bf16_t* x, const bf16_t* w, const bf16_t* b,
+// res/sum_out: optional fused residual (sum = bf16(x+res) feeds both the
+// statistics and the ongoing stream); ds: optional additive gradient on
+// the sum stream folded into dx.
+void launch_layernorm_fwd(const bf16_t* x, const bf16_t* res,
+                          bf16_t* sum_out, const bf16_t* w, const bf16_t* b,
                           bf16_t* y, float* mean, float* rstd, int64_t rows,
                           int cols, float eps, hipStream_t s);
 // pdw/pdb: [stripes, cols] fp32 workspaces (dta_colred_stripes rows); the
 // two-phase column reduction is atomic-free and deterministic.
-void launch_layernorm_bwd(const bf16_t* dy, const bf16_t* x, const bf16_t* w,
+void launch_layernorm_bwd(const bf16_t* dy, const bf16_t* ds,
+                          const bf16_t* x, const bf16_t* w,
                           const float* mean, const float* rstd, bf16_t* dx,
                           float* dw, float* db, float* pdw, float* pdb,
                           int stripes, int64_t rows, int cols,

@@ -18,8 +18,13 @@ namespace {
 constexpr int ROW_WAVES = 4;
 
 // ---- forward --------------------------------------------------------------
-template <int ITERS, bool RMS>
+// RES: residual fusion — vals = round_bf16(x + res) (rounded BEFORE the
+// statistics so the saved sum tensor reproduces them exactly in backward),
+// and the sum is written out for the ongoing residual stream.
+template <int ITERS, bool RMS, bool RES>
 __global__ void norm_fwd_k(const ushort* __restrict__ x,
+                           const ushort* __restrict__ res,
+                           ushort* __restrict__ sum_out,
                            const ushort* __restrict__ w,
                            const ushort* __restrict__ b,
                            ushort* __restrict__ y, float* __restrict__ mean,
@@ -30,6 +35,8 @@ __global__ void norm_fwd_k(const ushort* __restrict__ x,
   for (int64_t row = int64_t(blockIdx.x) * ROW_WAVES + wid; row < rows;
        row += int64_t(gridDim.x) * ROW_WAVES) {
     const ushort* xr = x + row * cols;
+    const ushort* rr = RES ? res + row * cols : nullptr;
+    ushort* so = RES ? sum_out + row * cols : nullptr;
     float vals[ITERS][8];
     float sum = 0.f, sumsq = 0.f;
 #pragma unroll
@@ -37,13 +44,22 @@ __global__ void norm_fwd_k(const ushort* __restrict__ x,
       const int c = lane + it * 64;
       if (c < nchunk) {
         s16x8 vx = *reinterpret_cast<const s16x8*>(xr + c * 8);
+        s16x8 vr;
+        if (RES) vr = *reinterpret_cast<const s16x8*>(rr + c * 8);
+        s16x8 vs;
 #pragma unroll
         for (int j = 0; j < 8; ++j) {
           float f = bf2f(ushort(vx[j]));
+          if (RES) {
+            const ushort sb = f2bf(f + bf2f(ushort(vr[j])));
+            vs[j] = sb;
+            f = bf2f(sb);
+          }
           vals[it][j] = f;
           sum += f;
           sumsq = fmaf(f, f, sumsq);
         }
+        if (RES) *reinterpret_cast<s16x8*>(so + c * 8) = vs;
       } else {
 #pragma unroll
         for (int j = 0; j < 8; ++j) vals[it][j] = 0.f;
@@ -84,8 +100,11 @@ __global__ void norm_fwd_k(const ushort* __restrict__ x,
 // ---- backward dx (streaming, no cross-row state) --------------------------
 // LN:  dx = rs*(dyw - mean(dyw) - xh*mean(dyw*xh)),  dyw = dy*w
 // RMS: dx = rs*(dyw - xh*mean(dyw*xh))
-template <int ITERS, bool RMS>
+// DS: residual fusion — dx += ds (the gradient arriving on the sum
+// stream from its downstream consumer), saving the separate add kernel.
+template <int ITERS, bool RMS, bool DS>
 __global__ void norm_bwd_dx_k(const ushort* __restrict__ dy,
+                              const ushort* __restrict__ ds,
                               const ushort* __restrict__ x,
                               const ushort* __restrict__ w,
                               const float* __restrict__ mean,
@@ -135,16 +154,21 @@ __global__ void norm_bwd_dx_k(const ushort* __restrict__ dy,
     s1 = wave_sum(s1) / cols;
     s2 = wave_sum(s2) / cols;
     ushort* dxr = dx + row * cols;
+    const ushort* dsr = DS ? ds + row * cols : nullptr;
 #pragma unroll
     for (int it = 0; it < ITERS; ++it) {
       const int c = lane + it * 64;
       if (c < nchunk) {
+        s16x8 vds;
+        if (DS) vds = *reinterpret_cast<const s16x8*>(dsr + c * 8);
         s16x8 o;
 #pragma unroll
         for (int j = 0; j < 8; ++j) {
           float v = RMS ? (dyw[it][j] - xh[it][j] * s2)
                         : (dyw[it][j] - s1 - xh[it][j] * s2);
-          o[j] = f2bf(rs * v);
+          float g = rs * v;
+          if (DS) g += bf2f(ushort(vds[j]));
+          o[j] = f2bf(g);
         }
         *reinterpret_cast<s16x8*>(dxr + c * 8) = o;
       }
@@ -218,10 +242,11 @@ __global__ void dwdb_reduce_k(const float* __restrict__ pdw,
   }
 }
 
-template <bool RMS>
-void dispatch_fwd(const ushort* x, const ushort* w, const ushort* b,
-                  ushort* y, float* mean, float* rstd, int64_t rows, int cols,
-                  float eps, hipStream_t s) {
+template <bool RMS, bool RES>
+void dispatch_fwd(const ushort* x, const ushort* res, ushort* sum_out,
+                  const ushort* w, const ushort* b, ushort* y, float* mean,
+                  float* rstd, int64_t rows, int cols, float eps,
+                  hipStream_t s) {
   const int nchunk = cols >> 3;
   const int iters = (nchunk + 63) / 64;
   int64_t want = (rows + ROW_WAVES - 1) / ROW_WAVES;
@@ -229,25 +254,26 @@ void dispatch_fwd(const ushort* x, const ushort* w, const ushort* b,
   const dim3 blk(64 * ROW_WAVES);
 #define CASE_F(I)                                                         \
   case I:                                                                 \
-    norm_fwd_k<I, RMS><<<grid, blk, 0, s>>>(x, w, b, y, mean, rstd, rows, \
-                                            cols, eps);                   \
+    norm_fwd_k<I, RMS, RES><<<grid, blk, 0, s>>>(x, res, sum_out, w, b,   \
+                                                 y, mean, rstd, rows,     \
+                                                 cols, eps);              \
     break;
   switch (iters) {
     CASE_F(1) CASE_F(2) CASE_F(3) CASE_F(4) CASE_F(6) CASE_F(8) CASE_F(16)
     default: {
-      if (iters <= 6) { norm_fwd_k<6, RMS><<<grid, blk, 0, s>>>(x, w, b, y, mean, rstd, rows, cols, eps); }
-      else if (iters <= 8) { norm_fwd_k<8, RMS><<<grid, blk, 0, s>>>(x, w, b, y, mean, rstd, rows, cols, eps); }
-      else { norm_fwd_k<16, RMS><<<grid, blk, 0, s>>>(x, w, b, y, mean, rstd, rows, cols, eps); }
+      if (iters <= 6) { norm_fwd_k<6, RMS, RES><<<grid, blk, 0, s>>>(x, res, sum_out, w, b, y, mean, rstd, rows, cols, eps); }
+      else if (iters <= 8) { norm_fwd_k<8, RMS, RES><<<grid, blk, 0, s>>>(x, res, sum_out, w, b, y, mean, rstd, rows, cols, eps); }
+      else { norm_fwd_k<16, RMS, RES><<<grid, blk, 0, s>>>(x, res, sum_out, w, b, y, mean, rstd, rows, cols, eps); }
     }
   }
 #undef CASE_F
 }
 
-template <bool RMS>
-void dispatch_bwd(const ushort* dy, const ushort* x, const ushort* w,
-                  const float* mean, const float* rstd, ushort* dx, float* dw,
-                  float* db, float* pdw, float* pdb, int stripes,
-                  int64_t rows, int cols, hipStream_t s) {
+template <bool RMS, bool DS>
+void dispatch_bwd(const ushort* dy, const ushort* ds, const ushort* x,
+                  const ushort* w, const float* mean, const float* rstd,
+                  ushort* dx, float* dw, float* db, float* pdw, float* pdb,
+                  int stripes, int64_t rows, int cols, hipStream_t s) {
   const int nchunk = cols >> 3;
   const int iters = (nchunk + 63) / 64;
   int64_t want = (rows + ROW_WAVES - 1) / ROW_WAVES;
@@ -255,14 +281,14 @@ void dispatch_bwd(const ushort* dy, const ushort* x, const ushort* w,
   const dim3 blk(64 * ROW_WAVES);
 #define CASE_B(I)                                                            \
   case I:                                                                    \
-    norm_bwd_dx_k<I, RMS><<<grid, blk, 0, s>>>(dy, x, w, mean, rstd, dx,     \
-                                               rows, cols);                  \
+    norm_bwd_dx_k<I, RMS, DS><<<grid, blk, 0, s>>>(dy, ds, x, w, mean,       \
+                                                   rstd, dx, rows, cols);    \
     break;
   switch (iters) {
     CASE_B(1) CASE_B(2) CASE_B(3) CASE_B(4) CASE_B(6) CASE_B(8)
     default:
-      norm_bwd_dx_k<8, RMS><<<grid, blk, 0, s>>>(dy, x, w, mean, rstd, dx,
-                                                 rows, cols);
+      norm_bwd_dx_k<8, RMS, DS><<<grid, blk, 0, s>>>(dy, ds, x, w, mean,
+                                                     rstd, dx, rows, cols);
   }
 #undef CASE_B
   const ColRedCfg cfg = dta_colred_cfg(rows, cols);
@@ -279,27 +305,39 @@ void dispatch_bwd(const ushort* dy, const ushort* x, const ushort* w,
 
 }  // namespace
 
-void launch_layernorm_fwd(const bf16_t* x, const bf16_t* w, const bf16_t* b,
+void launch_layernorm_fwd(const bf16_t* x, const bf16_t* res,
+                          bf16_t* sum_out, const bf16_t* w, const bf16_t* b,
                           bf16_t* y, float* mean, float* rstd, int64_t rows,
                           int cols, float eps, hipStream_t s) {
-  dispatch_fwd<false>(x, w, b, y, mean, rstd, rows, cols, eps, s);
+  if (res)
+    dispatch_fwd<false, true>(x, res, sum_out, w, b, y, mean, rstd, rows,
+                              cols, eps, s);
+  else
+    dispatch_fwd<false, false>(x, nullptr, nullptr, w, b, y, mean, rstd,
+                               rows, cols, eps, s);
 }
-void launch_layernorm_bwd(const bf16_t* dy, const bf16_t* x, const bf16_t* w,
+void launch_layernorm_bwd(const bf16_t* dy, const bf16_t* ds,
+                          const bf16_t* x, const bf16_t* w,
                           const float* mean, const float* rstd, bf16_t* dx,
                           float* dw, float* db, float* pdw, float* pdb,
                           int stripes, int64_t rows, int cols,
                           hipStream_t s) {
-  dispatch_bwd<false>(dy, x, w, mean, rstd, dx, dw, db, pdw, pdb, stripes,
-                      rows, cols, s);
+  if (ds)
+    dispatch_bwd<false, true>(dy, ds, x, w, mean, rstd, dx, dw, db, pdw,
+                              pdb, stripes, rows, cols, s);
+  else
+    dispatch_bwd<false, false>(dy, nullptr, x, w, mean, rstd, dx, dw, db,
+                               pdw, pdb, stripes, rows, cols, s);
 }
 void launch_rmsnorm_fwd(const bf16_t* x, const bf16_t* w, bf16_t* y,
                         float* rstd, int64_t rows, int cols, float eps,
                         hipStream_t s) {
-  dispatch_fwd<true>(x, w, nullptr, y, nullptr, rstd, rows, cols, eps, s);
+  dispatch_fwd<true, false>(x, nullptr, nullptr, w, nullptr, y, nullptr,
+                            rstd, rows, cols, eps, s);
 }
 void launch_rmsnorm_bwd(const bf16_t* dy, const bf16_t* x, const bf16_t* w,
                         const float* rstd, bf16_t* dx, float* dw, float* pdw,
                         int stripes, int64_t rows, int cols, hipStream_t s) {
-  dispatch_bwd<true>(dy, x, w, nullptr, rstd, dx, dw, nullptr, pdw, nullptr,
-                     stripes, rows, cols, s);
+  dispatch_bwd<true, false>(dy, nullptr, x, w, nullptr, rstd, dx, dw,
+                            nullptr, pdw, nullptr, stripes, rows, cols, s);
 }

@@ -65,13 +65,13 @@ def test_layernorm_fwd_bwd(rows, cols):
     x = _rand_bf16(rows, cols, seed=1)
     w = _rand_bf16(cols, seed=2, scale=0.5)
     b = _rand_bf16(cols, seed=3, scale=0.5)
-    y, mean, rstd = m.layernorm_fwd(x, w, b, 1e-5)
+    y, mean, rstd, _ = m.layernorm_fwd(x, torch.empty(0, device=DEV, dtype=torch.bfloat16), w, b, 1e-5)
     ref = torch.nn.functional.layer_norm(
         x.float(), (cols,), w.float(), b.float(), 1e-5)
     torch.testing.assert_close(y.float(), ref, rtol=2e-2, atol=2e-2)
 
     dy = _rand_bf16(rows, cols, seed=4)
-    dx, dw, db = m.layernorm_bwd(dy, x, w, mean, rstd)
+    dx, dw, db = m.layernorm_bwd(dy, torch.empty(0, device=DEV, dtype=torch.bfloat16), x, w, mean, rstd)
     xr = x.float().detach().requires_grad_(True)
     wr = w.float().detach().requires_grad_(True)
     br = b.float().detach().requires_grad_(True)
@@ -599,3 +599,32 @@ def test_linear_bgradb_main_grad_path():
     # autograd returned None for w -> .grad stays untouched (main_grad owns it)
     assert w.grad is None
     ops._USE_BGRADB = old
+
+
+def test_add_layer_norm_fused():
+    """Fused residual-add+LN vs fp32 composition, incl. the ds-folded
+    backward (both outputs consumed)."""
+    from distributedtraining_amd import ops
+    T, E = 300, 256
+    x = _rand_bf16(T, E, seed=61).requires_grad_(True)
+    r = _rand_bf16(T, E, seed=62).requires_grad_(True)
+    w = _rand_bf16(E, seed=63, scale=0.5).requires_grad_(True)
+    b = _rand_bf16(E, seed=64, scale=0.5).requires_grad_(True)
+    s, y = ops.add_layer_norm(x, r, w, b)
+    ds = _rand_bf16(T, E, seed=65)
+    dy = _rand_bf16(T, E, seed=66)
+    (s.float() * ds.float() + y.float() * dy.float()).sum().backward()
+
+    xf = x.detach().float().requires_grad_(True)
+    rf = r.detach().float().requires_grad_(True)
+    wf = w.detach().float().requires_grad_(True)
+    bf = b.detach().float().requires_grad_(True)
+    sf = xf + rf
+    yf = torch.nn.functional.layer_norm(sf, (E,), wf, bf, 1e-5)
+    (sf * ds.float() + yf * dy.float()).sum().backward()
+    torch.testing.assert_close(s.float(), sf.detach(), rtol=2e-2, atol=2e-2)
+    torch.testing.assert_close(y.float(), yf.detach(), rtol=3e-2, atol=3e-2)
+    torch.testing.assert_close(x.grad.float(), xf.grad, rtol=5e-2, atol=8e-2)
+    torch.testing.assert_close(r.grad.float(), rf.grad, rtol=5e-2, atol=8e-2)
+    torch.testing.assert_close(w.grad.float(), wf.grad, rtol=5e-2, atol=5e-1)
+    torch.testing.assert_close(b.grad.float(), bf.grad, rtol=5e-2, atol=5e-1)
