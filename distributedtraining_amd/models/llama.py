@@ -50,10 +50,17 @@ class LlamaBlock(nn.Module):
         self.down_w = nn.Parameter(torch.empty(E, I))
 
     def forward(self, x: torch.Tensor, cos: torch.Tensor,
-                sin: torch.Tensor) -> torch.Tensor:
+                sin: torch.Tensor, pending=None):
+        """Residual-join-fused layout (see gpt2.GPT2Block.forward):
+        returns (stream, pending)."""
         B, S, E = x.shape
         D = self.head_dim
-        h = ops.rms_norm(x, self.attn_norm_w, self.norm_eps)
+        if pending is None:
+            s = x
+            h = ops.rms_norm(x, self.attn_norm_w, self.norm_eps)
+        else:
+            s, h = ops.add_rms_norm(x, pending, self.attn_norm_w,
+                                    self.norm_eps)
         q = ops.linear(h, self.q_w).view(B, S, self.n_head, D).transpose(1, 2)
         k = ops.linear(h, self.k_w).view(B, S, self.n_kv, D).transpose(1, 2)
         v = ops.linear(h, self.v_w).view(B, S, self.n_kv, D).transpose(1, 2)
@@ -61,11 +68,10 @@ class LlamaBlock(nn.Module):
         k = ops.rope(k, cos, sin)
         o = ops.causal_attention(q, k, v)
         o = o.transpose(1, 2).reshape(B, S, E)
-        x = x + ops.linear(o, self.o_w)
-        h = ops.rms_norm(x, self.mlp_norm_w, self.norm_eps)
-        h = ops.swiglu(ops.linear(h, self.gate_w), ops.linear(h, self.up_w))
-        x = x + ops.linear(h, self.down_w)
-        return x
+        a = ops.linear(o, self.o_w)
+        s2, h2 = ops.add_rms_norm(s, a, self.mlp_norm_w, self.norm_eps)
+        h2 = ops.swiglu(ops.linear(h2, self.gate_w), ops.linear(h2, self.up_w))
+        return s2, ops.linear(h2, self.down_w)
 
 
 class LlamaLM(nn.Module):
@@ -109,9 +115,14 @@ class LlamaLM(nn.Module):
         cos = self.rope_cos[:S]
         sin = self.rope_sin[:S]
         x = ops.embedding_fwd(input_ids, self.tok_emb, None)
+        pending = None
         for blk in self.blocks:
-            x = blk(x, cos, sin)
-        x = ops.rms_norm(x, self.final_norm_w, self.cfg.norm_eps)
+            x, pending = blk(x, cos, sin, pending)
+        if pending is None:
+            x = ops.rms_norm(x, self.final_norm_w, self.cfg.norm_eps)
+        else:
+            _, x = ops.add_rms_norm(x, pending, self.final_norm_w,
+                                    self.cfg.norm_eps)
         if labels is None:
             return CausalLMOutput(loss=None, logits=ops.linear(x, self._head()))
         logits = ops.linear(x[:, :-1, :].contiguous(), self._head())

@@ -31,6 +31,7 @@ from .backend import require_ext, use_hip
 __all__ = [
     "layer_norm", "rms_norm", "gelu", "swiglu", "causal_attention",
     "qkv_attention", "linear", "mlp_gelu", "add_layer_norm",
+    "add_rms_norm",
     "cross_entropy_loss",
     "embedding_fwd",
     "rope", "adamw_step", "delta_sub", "axpy_", "weighted_merge",
@@ -264,7 +265,8 @@ class _RMSNormFn(torch.autograd.Function):
     def forward(ctx, x, w, eps):
         m = require_ext()
         x2 = x.contiguous()
-        y, rstd = m.rmsnorm_fwd(x2.view(-1, x2.shape[-1]), w, eps)
+        y, rstd, _ = m.rmsnorm_fwd(x2.view(-1, x2.shape[-1]),
+                                   _empty_like0(x2), w, eps)
         ctx.save_for_backward(x2, w, rstd)
         return y.view_as(x2)
 
@@ -273,9 +275,47 @@ class _RMSNormFn(torch.autograd.Function):
         m = require_ext()
         x, w, rstd = ctx.saved_tensors
         N = x.shape[-1]
-        dx, dw = m.rmsnorm_bwd(dy.contiguous().view(-1, N), x.view(-1, N),
-                               w, rstd)
+        dx, dw = m.rmsnorm_bwd(dy.contiguous().view(-1, N), _empty_like0(x),
+                               x.view(-1, N), w, rstd)
         return dx.view_as(x), dw, None
+
+
+class _AddRMSNormFn(torch.autograd.Function):
+    """Fused residual-add + RMSNorm (the Llama-family twin of
+    _AddLayerNormFn): (s, y) = (x+res, RMSNorm(x+res)); backward folds the
+    sum-stream gradient into dx."""
+
+    @staticmethod
+    def forward(ctx, x, res, w, eps):
+        m = require_ext()
+        x2 = x.contiguous().view(-1, x.shape[-1])
+        r2 = res.contiguous().view(-1, x.shape[-1])
+        y, rstd, s = m.rmsnorm_fwd(x2, r2, w, eps)
+        ctx.save_for_backward(s, w, rstd)
+        ctx.shape = x.shape
+        return s.view(x.shape), y.view(x.shape)
+
+    @staticmethod
+    def backward(ctx, ds, dy):
+        m = require_ext()
+        s, w, rstd = ctx.saved_tensors
+        N = s.shape[-1]
+        ds2 = (ds.contiguous().view(-1, N) if ds is not None
+               else _empty_like0(s))
+        dx, dw = m.rmsnorm_bwd(dy.contiguous().view(-1, N), ds2, s, w, rstd)
+        dx = dx.view(ctx.shape)
+        return dx, dx, dw, None
+
+
+def add_rms_norm(x: torch.Tensor, res: torch.Tensor, w: torch.Tensor,
+                 eps: float = 1e-5):
+    """(s, y) = (x + res, rms_norm(x + res))."""
+    if use_hip(x):
+        return _AddRMSNormFn.apply(x, res, w, eps)
+    s = x + res
+    sf = s.float()
+    y = sf * torch.rsqrt(sf.pow(2).mean(-1, keepdim=True) + eps)
+    return s, (y * w.float()).to(s.dtype)
 
 
 def rms_norm(x: torch.Tensor, w: torch.Tensor, eps: float = 1e-5) -> torch.Tensor:

@@ -80,31 +80,41 @@ std::vector<Tensor> layernorm_bwd(Tensor dy, Tensor ds, Tensor x, Tensor w,
   return {dx, dw32.to(torch::kBFloat16), db32.to(torch::kBFloat16)};
 }
 
-std::vector<Tensor> rmsnorm_fwd(Tensor x, Tensor w, double eps) {
+std::vector<Tensor> rmsnorm_fwd(Tensor x, Tensor res, Tensor w,
+                                double eps) {
   check_bf16(x, "x"); check_bf16(w, "w");
   const int64_t rows = x.size(0);
   const int cols = int(x.size(1));
   TORCH_CHECK(cols % 8 == 0, "cols must be a multiple of 8");
+  const bool has_res = res.numel() > 0;
+  if (has_res) check_bf16(res, "res");
   auto y = torch::empty_like(x);
+  auto sum = has_res ? torch::empty_like(x) : torch::empty({0}, x.options());
   auto rstd = torch::empty({rows}, x.options().dtype(torch::kFloat32));
-  launch_rmsnorm_fwd(bfp(x), bfp(w), bfp_mut(y), rstd.data_ptr<float>(),
-                     rows, cols, float(eps), stream());
-  return {y, rstd};
+  launch_rmsnorm_fwd(bfp(x), has_res ? bfp(res) : nullptr,
+                     has_res ? bfp_mut(sum) : nullptr, bfp(w), bfp_mut(y),
+                     rstd.data_ptr<float>(), rows, cols, float(eps),
+                     stream());
+  return {y, rstd, sum};
 }
 
-std::vector<Tensor> rmsnorm_bwd(Tensor dy, Tensor x, Tensor w, Tensor rstd) {
+std::vector<Tensor> rmsnorm_bwd(Tensor dy, Tensor ds, Tensor x, Tensor w,
+                                Tensor rstd) {
   check_bf16(dy, "dy"); check_bf16(x, "x"); check_bf16(w, "w");
   const int64_t rows = x.size(0);
   const int cols = int(x.size(1));
   TORCH_CHECK(cols % 8 == 0, "cols must be a multiple of 8");
+  const bool has_ds = ds.numel() > 0;
+  if (has_ds) check_bf16(ds, "ds");
   auto dx = torch::empty_like(x);
   auto f32 = x.options().dtype(torch::kFloat32);
   auto dw32 = torch::zeros({cols}, f32);
   const int stripes = dta_colred_stripes(rows, cols);
   auto part = torch::empty({stripes, cols}, f32);
-  launch_rmsnorm_bwd(bfp(dy), bfp(x), bfp(w), rstd.data_ptr<float>(),
-                     bfp_mut(dx), dw32.data_ptr<float>(),
-                     part.data_ptr<float>(), stripes, rows, cols, stream());
+  launch_rmsnorm_bwd(bfp(dy), has_ds ? bfp(ds) : nullptr, bfp(x), bfp(w),
+                     rstd.data_ptr<float>(), bfp_mut(dx),
+                     dw32.data_ptr<float>(), part.data_ptr<float>(),
+                     stripes, rows, cols, stream());
   return {dx, dw32.to(torch::kBFloat16)};
 }
 

@@ -84,10 +84,12 @@ void launch_layernorm_bwd(const bf16_t* dy, const bf16_t* ds,
                           float* dw, float* db, float* pdw, float* pdb,
                           int stripes, int64_t rows, int cols,
                           hipStream_t s);
-void launch_rmsnorm_fwd(const bf16_t* x, const bf16_t* w, bf16_t* y,
+void launch_rmsnorm_fwd(const bf16_t* x, const bf16_t* res,
+                        bf16_t* sum_out, const bf16_t* w, bf16_t* y,
                         float* rstd, int64_t rows, int cols, float eps,
                         hipStream_t s);
-void launch_rmsnorm_bwd(const bf16_t* dy, const bf16_t* x, const bf16_t* w,
+void launch_rmsnorm_bwd(const bf16_t* dy, const bf16_t* ds,
+                        const bf16_t* x, const bf16_t* w,
                         const float* rstd, bf16_t* dx, float* dw, float* pdw,
                         int stripes, int64_t rows, int cols, hipStream_t s);
 

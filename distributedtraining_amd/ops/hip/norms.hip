@@ -329,15 +329,25 @@ void launch_layernorm_bwd(const bf16_t* dy, const bf16_t* ds,
     dispatch_bwd<false, false>(dy, nullptr, x, w, mean, rstd, dx, dw, db,
                                pdw, pdb, stripes, rows, cols, s);
 }
-void launch_rmsnorm_fwd(const bf16_t* x, const bf16_t* w, bf16_t* y,
+void launch_rmsnorm_fwd(const bf16_t* x, const bf16_t* res,
+                        bf16_t* sum_out, const bf16_t* w, bf16_t* y,
                         float* rstd, int64_t rows, int cols, float eps,
                         hipStream_t s) {
-  dispatch_fwd<true, false>(x, nullptr, nullptr, w, nullptr, y, nullptr,
-                            rstd, rows, cols, eps, s);
+  if (res)
+    dispatch_fwd<true, true>(x, res, sum_out, w, nullptr, y, nullptr,
+                             rstd, rows, cols, eps, s);
+  else
+    dispatch_fwd<true, false>(x, nullptr, nullptr, w, nullptr, y, nullptr,
+                              rstd, rows, cols, eps, s);
 }
-void launch_rmsnorm_bwd(const bf16_t* dy, const bf16_t* x, const bf16_t* w,
+void launch_rmsnorm_bwd(const bf16_t* dy, const bf16_t* ds,
+                        const bf16_t* x, const bf16_t* w,
                         const float* rstd, bf16_t* dx, float* dw, float* pdw,
                         int stripes, int64_t rows, int cols, hipStream_t s) {
-  dispatch_bwd<true, false>(dy, nullptr, x, w, nullptr, rstd, dx, dw,
-                            nullptr, pdw, nullptr, stripes, rows, cols, s);
+  if (ds)
+    dispatch_bwd<true, true>(dy, ds, x, w, nullptr, rstd, dx, dw, nullptr,
+                             pdw, nullptr, stripes, rows, cols, s);
+  else
+    dispatch_bwd<true, false>(dy, nullptr, x, w, nullptr, rstd, dx, dw,
+                              nullptr, pdw, nullptr, stripes, rows, cols, s);
 }

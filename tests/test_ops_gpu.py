@@ -89,7 +89,7 @@ def test_rmsnorm_fwd_bwd(rows, cols):
     m = _ext()
     x = _rand_bf16(rows, cols, seed=1)
     w = _rand_bf16(cols, seed=2, scale=0.5)
-    y, rstd = m.rmsnorm_fwd(x, w, 1e-5)
+    y, rstd, _ = m.rmsnorm_fwd(x, torch.empty(0, device=DEV, dtype=torch.bfloat16), w, 1e-5)
     xf = x.float()
     ref = xf * torch.rsqrt(xf.pow(2).mean(-1, keepdim=True) + 1e-5) * w.float()
     torch.testing.assert_close(y.float(), ref, rtol=2e-2, atol=2e-2)
@@ -99,7 +99,7 @@ def test_rmsnorm_fwd_bwd(rows, cols):
     wr = w.float().detach().requires_grad_(True)
     (xr * torch.rsqrt(xr.pow(2).mean(-1, keepdim=True) + 1e-5) * wr).backward(
         dy.float())
-    dx, dw = m.rmsnorm_bwd(dy, x, w, rstd)
+    dx, dw = m.rmsnorm_bwd(dy, torch.empty(0, device=DEV, dtype=torch.bfloat16), x, w, rstd)
     torch.testing.assert_close(dx.float(), xr.grad, rtol=5e-2, atol=5e-2)
     torch.testing.assert_close(dw.float(), wr.grad, rtol=5e-2, atol=0.3)
 
