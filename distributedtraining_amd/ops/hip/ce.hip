@@ -84,21 +84,27 @@ __global__ void ce_bwd_k(const ushort* __restrict__ logits,
     const int64_t tgt = targets[row];
     const float lb = lse[row] * LOG2E;
     const float sc = (tgt == ignore_index) ? 0.f : sc_base;
-    int64_t i = int64_t(threadIdx.x) * 8;
-    const int64_t stride = int64_t(CE_BLOCK) * 8;
-    for (; i + 8 <= vocab; i += stride) {
-      s16x8 vx = *reinterpret_cast<const s16x8*>(xr + i);
-      s16x8 o;
+    // two 16 B vectors in flight per iteration (load/store ILP)
+    int64_t i = int64_t(threadIdx.x) * 16;
+    const int64_t stride = int64_t(CE_BLOCK) * 16;
+    for (; i + 16 <= vocab; i += stride) {
+      s16x8 va = *reinterpret_cast<const s16x8*>(xr + i);
+      s16x8 vb = *reinterpret_cast<const s16x8*>(xr + i + 8);
+      s16x8 oa, ob;
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
-        float p = __builtin_exp2f(bf2f(ushort(vx[j])) * LOG2E - lb);
-        o[j] = f2bf(sc * (p - ((i + j) == tgt ? 1.f : 0.f)));
+        float pa = __builtin_exp2f(bf2f(ushort(va[j])) * LOG2E - lb);
+        float pb = __builtin_exp2f(bf2f(ushort(vb[j])) * LOG2E - lb);
+        oa[j] = f2bf(sc * (pa - ((i + j) == tgt ? 1.f : 0.f)));
+        ob[j] = f2bf(sc * (pb - ((i + 8 + j) == tgt ? 1.f : 0.f)));
       }
       // nontemporal: dlogits is consumed by the big head GEMMs, keep it
       // out of L2 so the logits read stream stays resident
-      __builtin_nontemporal_store(o, reinterpret_cast<s16x8*>(dxr + i));
+      __builtin_nontemporal_store(oa, reinterpret_cast<s16x8*>(dxr + i));
+      __builtin_nontemporal_store(ob,
+                                  reinterpret_cast<s16x8*>(dxr + i + 8));
     }
-    if (i < vocab && i + 8 > vocab)
+    if (i < vocab)
       for (; i < vocab; ++i) {
         float p = __builtin_exp2f(bf2f(xr[i]) * LOG2E - lb);
         dxr[i] = f2bf(sc * (p - (i == tgt ? 1.f : 0.f)));
