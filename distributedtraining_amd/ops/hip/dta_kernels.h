@@ -40,9 +40,11 @@ void launch_colsum(const bf16_t* x, float* part, float* out, int64_t rows,
 struct ColRedCfg { int threads; int gx; int stripes; };
 inline ColRedCfg dta_colred_cfg(int64_t rows, int cols) {
   const int lanes = cols / 8;
-  int threads = lanes >= 256 ? 256 : ((lanes + 63) / 64) * 64;
+  // balance lanes across gx blocks (a 256-thread split left trailing
+  // blocks ~12% occupied at lanes=288, e.g. the qkv dbias shape)
+  const int gx = (lanes + 255) / 256;
+  int threads = ((lanes + gx - 1) / gx + 63) / 64 * 64;
   if (threads < 64) threads = 64;
-  const int gx = (lanes + threads - 1) / threads;
   // target ~1024 blocks of 256-thread-equivalents (measured best; halving
   // it cost colsum_part 24us -> 33us avg)
   int64_t st = (1024 * 256) / (int64_t(gx) * threads);
