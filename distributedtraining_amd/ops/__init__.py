@@ -329,9 +329,6 @@ def rms_norm(x: torch.Tensor, w: torch.Tensor, eps: float = 1e-5) -> torch.Tenso
 # --------------------------------------------------------------------------
 # GELU (tanh approximation — GPT-2 uses gelu_new)
 # --------------------------------------------------------------------------
-_GELU_C = math.sqrt(2.0 / math.pi)
-
-
 class _GeluFn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x):

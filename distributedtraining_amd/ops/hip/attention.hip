@@ -40,40 +40,16 @@ DEV bf16x8 load_frag_row(const ushort* base, int64_t row_stride, int row,
   return u.b;
 }
 
-DEV bf16x8 load_frag_col(const ushort* base, int64_t row_stride, int row0,
-                         int col, int row_max) {
-  bf16x8 out;
-#pragma unroll
-  for (int j = 0; j < 8; ++j) {
-    int r = row0 + j;
-    ushort u = (r < row_max) ? base[int64_t(r) * row_stride + col] : 0;
-    union { ushort s; __bf16 b; } c;
-    c.s = u;
-    out[j] = c.b;
-  }
-  return out;
-}
-
-// ---- per-wave LDS tile staging (transposed) --------------------------------
+// ---- LDS tile staging ------------------------------------------------------
 // B-fragments (k-major columns) gathered straight from global memory cost 8
 // scalar 2 B loads each, every one touching its own 64 B line (~32x wasted
-// HBM traffic — measured 204 us on the dkv kernel). Instead the row
-// fragments we ALREADY load coalesced are scatter-stored transposed into a
-// wave-private LDS tile [cols][LDS_RP] and B-fragments become single 16 B
-// LDS reads. Rows beyond seq hold clamped garbage — safe everywhere because
-// the matching score/probability lanes are already masked to 0 upstream.
+// HBM traffic — measured 204 us on the dkv kernel). Tiles are staged
+// block-cooperatively (the 4 waves share one (b,h)'s K/V or Q/dO) as
+// row-major panels for A-fragments and transposed [cols][LDS_RP] panels
+// for B-fragments (single 16 B LDS reads). Rows beyond seq hold clamped
+// garbage — safe everywhere because the matching score/probability lanes
+// are already masked to 0 upstream.
 constexpr int LDS_RP = 40;  // 32 rows + 8 pad (multiple of 8: aligned reads)
-
-DEV void stage_frag_T(ushort* lds_t, bf16x8 v, int lane, int row_base,
-                      int c0) {
-  const int row = row_base + (lane & 15);
-#pragma unroll
-  for (int j = 0; j < 8; ++j) {
-    union { __bf16 b; ushort s; } u;
-    u.b = v[j];
-    lds_t[(c0 + j) * LDS_RP + row] = u.s;
-  }
-}
 
 DEV bf16x8 load_frag_col_lds(const ushort* lds_t, int row0, int col) {
   const s16x8 v =
