@@ -8,7 +8,7 @@ the repo snapshot to GPU boxes).
 
 import os
 
-from setuptools import setup
+from setuptools import find_packages, setup
 
 os.environ.setdefault("PYTORCH_ROCM_ARCH", "gfx950")
 
@@ -30,7 +30,7 @@ SRC = [
 setup(
     name="distributedtraining_amd",
     version="0.1.0",
-    packages=["distributedtraining_amd"],
+    packages=find_packages(include=["distributedtraining_amd*"]),
     ext_modules=[
         CUDAExtension(
             name="distributedtraining_amd._dta_hip",
