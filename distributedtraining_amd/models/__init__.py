@@ -1,11 +1,12 @@
 from ..config import ModelConfig
 from .cnn import SimpleCNN
+from .generate import generate
 from .gpt2 import GPT2LM, CausalLMOutput
 from .llama import LlamaLM
 from .mlp import FeedforwardNN
 
-__all__ = ["build_model", "GPT2LM", "LlamaLM", "FeedforwardNN", "SimpleCNN",
-           "CausalLMOutput"]
+__all__ = ["build_model", "generate", "GPT2LM", "LlamaLM", "FeedforwardNN",
+           "SimpleCNN", "CausalLMOutput"]
 
 
 def build_model(cfg: ModelConfig):

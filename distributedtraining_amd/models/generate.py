@@ -1,0 +1,58 @@
+"""Autoregressive generation over the native models (serving path).
+
+The reference is training-only; this provides the serving-side surface a
+deployed base model needs: batched greedy / temperature / top-k sampling
+driven through the same CDNA4 kernel stack (full-context re-forward per
+step — at the framework's seq lengths of 64-512 the per-token forward is
+GEMM-dominated and a KV cache is a later optimization, not a correctness
+feature).
+"""
+
+from __future__ import annotations
+
+from typing import Optional
+
+import torch
+
+
+@torch.no_grad()
+def generate(model, input_ids: torch.Tensor, max_new_tokens: int,
+             temperature: float = 0.0, top_k: int = 0,
+             eos_token_id: Optional[int] = None,
+             generator: Optional[torch.Generator] = None) -> torch.Tensor:
+    """Extend ``input_ids`` [B, S] by up to ``max_new_tokens``.
+
+    temperature 0 = greedy; top_k > 0 restricts sampling to the k highest
+    logits. Stops early when every sequence has produced eos_token_id.
+    """
+    model_was_training = model.training
+    model.eval()
+    n_pos = getattr(model, "cfg").n_positions
+    ids = input_ids
+    done = torch.zeros(ids.shape[0], dtype=torch.bool, device=ids.device)
+    try:
+        for _ in range(max_new_tokens):
+            ctx = ids[:, -n_pos:]
+            logits = model(input_ids=ctx).logits[:, -1].float()
+            if temperature <= 0:
+                nxt = logits.argmax(dim=-1)
+            else:
+                logits = logits / temperature
+                if top_k > 0 and top_k < logits.shape[-1]:
+                    kth = torch.topk(logits, top_k, dim=-1).values[:, -1:]
+                    logits = logits.masked_fill(logits < kth,
+                                                float("-inf"))
+                probs = torch.softmax(logits, dim=-1)
+                nxt = torch.multinomial(probs, 1, generator=generator
+                                        ).squeeze(-1)
+            if eos_token_id is not None:
+                nxt = torch.where(done, torch.full_like(nxt, eos_token_id),
+                                  nxt)
+                done |= nxt == eos_token_id
+            ids = torch.cat([ids, nxt.unsqueeze(1)], dim=1)
+            if eos_token_id is not None and bool(done.all()):
+                break
+    finally:
+        if model_was_training:
+            model.train()
+    return ids

@@ -628,3 +628,16 @@ def test_add_layer_norm_fused():
     torch.testing.assert_close(r.grad.float(), rf.grad, rtol=5e-2, atol=8e-2)
     torch.testing.assert_close(w.grad.float(), wf.grad, rtol=5e-2, atol=5e-1)
     torch.testing.assert_close(b.grad.float(), bf.grad, rtol=5e-2, atol=5e-1)
+
+
+def test_generate_on_gpu():
+    """Serving path through the HIP kernel stack (labels=None forward)."""
+    from distributedtraining_amd.config import ModelConfig
+    from distributedtraining_amd.models import build_model, generate
+    cfg = ModelConfig.gpt2_tiny()
+    torch.manual_seed(0)
+    model = build_model(cfg).to(DEV)
+    ids = torch.randint(0, cfg.vocab_size, (2, 6), device=DEV)
+    out = generate(model, ids, max_new_tokens=5)
+    assert out.shape == (2, 11)
+    assert torch.equal(out, generate(model, ids, max_new_tokens=5))
