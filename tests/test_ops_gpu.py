@@ -636,7 +636,9 @@ def test_generate_on_gpu():
     from distributedtraining_amd.models import build_model, generate
     cfg = ModelConfig.gpt2_tiny()
     torch.manual_seed(0)
-    model = build_model(cfg).to(DEV)
+    # serving deployment: cast to the kernels' bf16 compute dtype (training
+    # goes through FlatParams which does this as part of the flat plane)
+    model = build_model(cfg).to(DEV, torch.bfloat16)
     ids = torch.randint(0, cfg.vocab_size, (2, 6), device=DEV)
     out = generate(model, ids, max_new_tokens=5)
     assert out.shape == (2, 11)
