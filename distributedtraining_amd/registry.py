@@ -256,3 +256,13 @@ class FileRegistry(Registry):
         super().set_stake(hotkey, stake)
         with self._lock:
             self._save()
+
+    def deregister(self, hotkey: str) -> None:
+        # must persist the removal BEFORE any reload: _load_unlocked merges
+        # the file back in, which would resurrect the hotkey otherwise
+        with self._lock:
+            self._load_unlocked()
+            self._addresses.pop(hotkey, None)
+            self._scores.pop(hotkey, None)
+            self._stakes.pop(hotkey, None)
+            self._save()

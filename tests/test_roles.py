@@ -160,3 +160,44 @@ def test_nesterov_outer_merge(tmp_path):
     m2 = av2.nesterov_merge(base, deltas)          # m=0.38
     expect = base + 0.7 * (0.9 * 0.38 + 0.2)
     torch.testing.assert_close(m2, expect)
+
+
+def test_empty_rounds_are_safe(tmp_path):
+    """Zero registered miners / zero deltas: validator normalizes to empty,
+    averager publishes the unchanged base (the reference merges 'whatever
+    exists'; elasticity contract SURVEY §2.3)."""
+    import torch
+    from distributedtraining_amd.config import AverageConfig
+    from distributedtraining_amd.roles.averager import ParameterizedAverager
+    from distributedtraining_amd.roles.validator import DeltaValidator
+    cfg, model, fp, store, registry = _mk(tmp_path, "e0")
+    ev = synthetic_eval_set(cfg.model.vocab_size, 1, 2, 16)
+    base = fp.snapshot()
+
+    validator = DeltaValidator(model, fp, ev, cfg.validate, store=store,
+                               registry=registry)
+    assert validator.validate_and_score() == {}          # no miners at all
+
+    av = ParameterizedAverager(model, fp, AverageConfig(strategy="mean"),
+                               store=store, registry=registry)
+    merged = av.run_round(ev)
+    torch.testing.assert_close(merged, base)             # base unchanged
+    # meta-learning with an empty stack is also a no-op
+    empty = torch.zeros(0, fp.numel)
+    torch.testing.assert_close(av.meta_learning(base, empty, ev), base)
+    torch.testing.assert_close(av.nesterov_merge(base, empty), base)
+
+
+def test_registry_deregister_drops_miner(tmp_path):
+    """Miner leaves: deregistered hotkey vanishes from membership and its
+    stale delta is no longer collected (elastic membership)."""
+    cfg, model, fp, store, registry = _mk(tmp_path, "d0")
+    data = synthetic_batches(cfg.model.vocab_size, 2, 16, seed=1)
+    miner = DeltaLoop(model, fp, data, cfg.train, store=store,
+                      registry=registry, hotkey="gone")
+    miner.train(2)
+    miner.last_push_step = -10**9
+    miner.maybe_push_delta()
+    assert "gone" in registry.hotkeys
+    registry.deregister("gone")
+    assert "gone" not in registry.hotkeys
