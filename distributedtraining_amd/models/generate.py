@@ -2,10 +2,13 @@
 
 The reference is training-only; this provides the serving-side surface a
 deployed base model needs: batched greedy / temperature / top-k sampling
-driven through the same CDNA4 kernel stack (full-context re-forward per
-step — at the framework's seq lengths of 64-512 the per-token forward is
-GEMM-dominated and a KV cache is a later optimization, not a correctness
-feature).
+driven through the same CDNA4 kernel stack. Two execution modes:
+
+* ``use_cache=True`` (models exposing new_cache/prefill/decode_step, i.e.
+  GPT-2): prompt prefill through the causal-attention kernel, then one
+  token per step through the flash-decoding KV-cache kernel
+  (ops.decode_attention) — O(S) per token;
+* ``use_cache=False``: full-context re-forward per token (any model).
 """
 
 from __future__ import annotations
@@ -15,11 +18,23 @@ from typing import Optional
 import torch
 
 
+def _sample(logits, temperature, top_k, generator):
+    if temperature <= 0:
+        return logits.argmax(dim=-1)
+    logits = logits / temperature
+    if top_k > 0 and top_k < logits.shape[-1]:
+        kth = torch.topk(logits, top_k, dim=-1).values[:, -1:]
+        logits = logits.masked_fill(logits < kth, float("-inf"))
+    probs = torch.softmax(logits, dim=-1)
+    return torch.multinomial(probs, 1, generator=generator).squeeze(-1)
+
+
 @torch.no_grad()
 def generate(model, input_ids: torch.Tensor, max_new_tokens: int,
              temperature: float = 0.0, top_k: int = 0,
              eos_token_id: Optional[int] = None,
-             generator: Optional[torch.Generator] = None) -> torch.Tensor:
+             generator: Optional[torch.Generator] = None,
+             use_cache: bool = False) -> torch.Tensor:
     """Extend ``input_ids`` [B, S] by up to ``max_new_tokens``.
 
     temperature 0 = greedy; top_k > 0 restricts sampling to the k highest
@@ -31,20 +46,27 @@ def generate(model, input_ids: torch.Tensor, max_new_tokens: int,
     ids = input_ids
     done = torch.zeros(ids.shape[0], dtype=torch.bool, device=ids.device)
     try:
+        if use_cache and hasattr(model, "prefill"):
+            total = min(ids.shape[1] + max_new_tokens, n_pos)
+            cache = model.new_cache(ids.shape[0], total, ids.device)
+            logits = model.prefill(ids, cache).float()
+            for _ in range(max_new_tokens):
+                nxt = _sample(logits, temperature, top_k, generator)
+                if eos_token_id is not None:
+                    nxt = torch.where(done,
+                                      torch.full_like(nxt, eos_token_id),
+                                      nxt)
+                    done |= nxt == eos_token_id
+                ids = torch.cat([ids, nxt.unsqueeze(1)], dim=1)
+                if ((eos_token_id is not None and bool(done.all()))
+                        or cache.len >= cache.max_len):
+                    break
+                logits = model.decode_step(ids[:, -1:], cache).float()
+            return ids
         for _ in range(max_new_tokens):
             ctx = ids[:, -n_pos:]
             logits = model(input_ids=ctx).logits[:, -1].float()
-            if temperature <= 0:
-                nxt = logits.argmax(dim=-1)
-            else:
-                logits = logits / temperature
-                if top_k > 0 and top_k < logits.shape[-1]:
-                    kth = torch.topk(logits, top_k, dim=-1).values[:, -1:]
-                    logits = logits.masked_fill(logits < kth,
-                                                float("-inf"))
-                probs = torch.softmax(logits, dim=-1)
-                nxt = torch.multinomial(probs, 1, generator=generator
-                                        ).squeeze(-1)
+            nxt = _sample(logits, temperature, top_k, generator)
             if eos_token_id is not None:
                 nxt = torch.where(done, torch.full_like(nxt, eos_token_id),
                                   nxt)

@@ -124,3 +124,100 @@ class GPT2LM(nn.Module):
         return CausalLMOutput(
             loss=loss,
             logits=logits if (return_logits or not self.training) else None)
+
+
+class KVCache:
+    """Preallocated per-layer KV cache for serving ([B, H, max_len, D];
+    the decode kernel reads the first ``len`` positions)."""
+
+    def __init__(self, n_layer: int, batch: int, n_kv_head: int,
+                 max_len: int, head_dim: int, device, dtype=None):
+        dtype = dtype or torch.bfloat16
+        self.k = [torch.zeros(batch, n_kv_head, max_len, head_dim,
+                              device=device, dtype=dtype)
+                  for _ in range(n_layer)]
+        self.v = [torch.zeros_like(self.k[0]) for _ in range(n_layer)]
+        self.max_len = max_len
+        self.len = 0
+
+    def append(self, layer: int, k_new: torch.Tensor,
+               v_new: torch.Tensor) -> None:
+        """k_new/v_new [B, S_new, Hk, D] written at position self.len
+        (advance once per model step via ``advance``)."""
+        S = k_new.shape[1]
+        assert self.len + S <= self.max_len, "KV cache overflow"
+        self.k[layer][:, :, self.len:self.len + S].copy_(
+            k_new.transpose(1, 2))
+        self.v[layer][:, :, self.len:self.len + S].copy_(
+            v_new.transpose(1, 2))
+
+    def advance(self, n: int) -> None:
+        self.len += n
+
+
+def _gpt2_block_attn_cached(blk: GPT2Block, x: torch.Tensor,
+                            cache: KVCache, i: int) -> torch.Tensor:
+    """One block with the KV cache: prefill (S>1, causal over itself) or
+    decode (S==1, decode_attention over the cache)."""
+    B, S, E = x.shape
+    H = blk.n_head
+    D = E // H
+    h = ops.layer_norm(x, blk.ln_1_w, blk.ln_1_b)
+    qkv = ops.linear(h, blk.attn_qkv_w, blk.attn_qkv_b)
+    q, k, v = qkv.split(E, dim=-1)
+    cache.append(i, k.view(B, S, H, D), v.view(B, S, H, D))
+    if S == 1:
+        o = ops.decode_attention(q.reshape(B, H, D), cache.k[i],
+                                 cache.v[i], cache.len + 1)
+        o = o.view(B, 1, E)
+    else:
+        assert cache.len == 0, "prefill must start an empty cache"
+        qv = q.view(B, S, H, D).transpose(1, 2)
+        kv_ = k.view(B, S, H, D).transpose(1, 2)
+        vv = v.view(B, S, H, D).transpose(1, 2)
+        o = ops.causal_attention(qv, kv_, vv).transpose(1, 2).reshape(B, S, E)
+    x = x + ops.linear(o, blk.attn_proj_w, blk.attn_proj_b)
+    h = ops.layer_norm(x, blk.ln_2_w, blk.ln_2_b)
+    x = x + ops.mlp_gelu(h, blk.mlp_fc_w, blk.mlp_fc_b, blk.mlp_proj_w,
+                         blk.mlp_proj_b)
+    return x
+
+
+def _gpt2_cached_forward(model: "GPT2LM", input_ids: torch.Tensor,
+                         cache: KVCache) -> torch.Tensor:
+    """Shared prefill/decode body: returns last-position logits [B, V]."""
+    S = input_ids.shape[1]
+    off = cache.len
+    wpe_slice = model.wpe[off:off + S].contiguous()
+    x = ops.embedding_fwd(input_ids, model.wte, wpe_slice)
+    for i, blk in enumerate(model.blocks):
+        x = _gpt2_block_attn_cached(blk, x, cache, i)
+    cache.advance(S)
+    x = ops.layer_norm(x[:, -1:], model.ln_f_w, model.ln_f_b)
+    return ops.linear(x, model.wte).squeeze(1)
+
+
+def _gpt2_new_cache(self: "GPT2LM", batch: int, max_len: int, device,
+                    dtype=None) -> KVCache:
+    cfg = self.cfg
+    return KVCache(cfg.n_layer, batch, cfg.n_head, max_len,
+                   cfg.n_embd // cfg.n_head, device,
+                   dtype or next(self.parameters()).dtype)
+
+
+def _gpt2_prefill(self: "GPT2LM", input_ids: torch.Tensor,
+                  cache: KVCache) -> torch.Tensor:
+    """Fill the cache with the prompt; returns last-position logits."""
+    return _gpt2_cached_forward(self, input_ids, cache)
+
+
+def _gpt2_decode_step(self: "GPT2LM", input_ids: torch.Tensor,
+                      cache: KVCache) -> torch.Tensor:
+    """One-token step ([B,1]) through the decode-attention kernel."""
+    assert input_ids.shape[1] == 1
+    return _gpt2_cached_forward(self, input_ids, cache)
+
+
+GPT2LM.new_cache = _gpt2_new_cache
+GPT2LM.prefill = _gpt2_prefill
+GPT2LM.decode_step = _gpt2_decode_step

@@ -34,7 +34,8 @@ __all__ = [
     "add_rms_norm",
     "cross_entropy_loss",
     "embedding_fwd",
-    "rope", "adamw_step", "delta_sub", "axpy_", "weighted_merge",
+    "decode_attention", "rope", "adamw_step", "delta_sub", "axpy_",
+    "weighted_merge",
     "grad_merge_weights", "has_nan", "l2norm",
 ]
 
@@ -489,6 +490,30 @@ def qkv_attention(qkv: torch.Tensor, n_head: int,
     o = F.scaled_dot_product_attention(q, k, v, is_causal=True, scale=scale,
                                        enable_gqa=True)
     return o.transpose(1, 2).reshape(B, S, E)
+
+
+def decode_attention(q: torch.Tensor, k_cache: torch.Tensor,
+                     v_cache: torch.Tensor, kv_len: int,
+                     scale: Optional[float] = None) -> torch.Tensor:
+    """Serving decode step: one new query per head attends the whole KV
+    cache. q [B,H,D]; caches [B,Hk,Lmax,D] with the first kv_len rows
+    valid. Inference-only (no autograd)."""
+    if scale is None:
+        scale = 1.0 / math.sqrt(q.shape[-1])
+    if use_hip(q):
+        return require_ext().attn_decode(q.contiguous(), k_cache, v_cache,
+                                         kv_len, scale)
+    qf = q.float().unsqueeze(2)                       # [B,H,1,D]
+    B, Hk, _, D = k_cache.shape
+    H = q.shape[1]
+    kf = k_cache[:, :, :kv_len].float()
+    vf = v_cache[:, :, :kv_len].float()
+    if H != Hk:
+        rep = H // Hk
+        kf = kf.repeat_interleave(rep, dim=1)
+        vf = vf.repeat_interleave(rep, dim=1)
+    att = torch.softmax((qf @ kf.transpose(-1, -2)) * scale, dim=-1)
+    return (att @ vf).squeeze(2).to(q.dtype)
 
 
 # --------------------------------------------------------------------------

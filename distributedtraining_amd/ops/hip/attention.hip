@@ -508,6 +508,98 @@ __global__ void attn_bwd_dkv_k(const ushort* __restrict__ dout,
     }
 }
 
+// ---------------- decode (KV-cache serving path) ----------------
+// One new query per (b,h) against a cached K/V of length kvlen (no mask:
+// every cached key is attended). One wave per (b,h), flash-decoding
+// shape: per 64-key chunk, phase 1 computes the chunk's scores (lane =
+// key; full q register-resident), phase 2 rescales the per-lane output
+// columns online (lane = head-dim column, CPL columns each). Wave-private
+// LDS carries the chunk probabilities between phases.
+template <int DTILES>  // D = 16*DTILES; CPL = D/64 columns per lane
+__global__ void attn_decode_k(const ushort* __restrict__ q,   // [B,H,D]
+                              const ushort* __restrict__ kc,  // [B,Hk,L,D]
+                              const ushort* __restrict__ vc,
+                              ushort* __restrict__ o,         // [B,H,D]
+                              int B, int H, int grp, int kvlen,
+                              int64_t cb, int64_t ch,  // cache strides
+                              float scale) {
+  constexpr int D = 16 * DTILES;
+  constexpr int CPL = D / 64;
+  const int lane = threadIdx.x & 63, wid = threadIdx.x >> 6;
+  const int bh = blockIdx.x * 4 + wid;
+  if (bh >= B * H) return;
+  const int b = bh / H, h = bh % H;
+  const int hk = h / grp;
+  const ushort* qp = q + (int64_t(b) * H + h) * D;
+  const ushort* kp = kc + b * cb + hk * ch;
+  const ushort* vp = vc + b * cb + hk * ch;
+  const float sc2 = scale * LOG2E;
+
+  // full query register-resident (bf16 pairs -> fp32 on use)
+  float qr[D];
+#pragma unroll
+  for (int i = 0; i < D; i += 8) {
+    s16x8 vq = *reinterpret_cast<const s16x8*>(qp + i);
+#pragma unroll
+    for (int j = 0; j < 8; ++j) qr[i + j] = bf2f(ushort(vq[j]));
+  }
+
+  __shared__ float p_all[4][64];
+  float* p = p_all[wid];
+  float m_run = -INFINITY, l_run = 0.f;
+  float acc[CPL];
+#pragma unroll
+  for (int c = 0; c < CPL; ++c) acc[c] = 0.f;
+
+  for (int s0 = 0; s0 < kvlen; s0 += 64) {
+    // phase 1: this lane's key
+    const int srow = s0 + lane;
+    float sc = -INFINITY;
+    if (srow < kvlen) {
+      const ushort* kr = kp + int64_t(srow) * D;
+      float dot = 0.f;
+#pragma unroll
+      for (int i = 0; i < D; i += 8) {
+        s16x8 vk = *reinterpret_cast<const s16x8*>(kr + i);
+#pragma unroll
+        for (int j = 0; j < 8; ++j)
+          dot = fmaf(qr[i + j], bf2f(ushort(vk[j])), dot);
+      }
+      sc = dot * sc2;
+    }
+    float mx = sc;
+#pragma unroll
+    for (int off = 32; off; off >>= 1) mx = fmaxf(mx, __shfl_xor(mx, off));
+    const float m_new = fmaxf(m_run, mx);
+    const float pj = (sc == -INFINITY) ? 0.f : __builtin_exp2f(sc - m_new);
+    float ps = pj;
+#pragma unroll
+    for (int off = 32; off; off >>= 1) ps += __shfl_xor(ps, off);
+    const float alpha =
+        (m_run == -INFINITY) ? 0.f : __builtin_exp2f(m_run - m_new);
+    l_run = l_run * alpha + ps;
+    m_run = m_new;
+    p[lane] = pj;
+    __threadfence_block();   // wave-local LDS visibility
+    // phase 2: lane owns CPL output columns
+    const int n = min(64, kvlen - s0);
+#pragma unroll
+    for (int c = 0; c < CPL; ++c) acc[c] *= alpha;
+    for (int s_ = 0; s_ < n; ++s_) {
+      const ushort* vr = vp + int64_t(s0 + s_) * D;
+      const float ps_ = p[s_];
+#pragma unroll
+      for (int c = 0; c < CPL; ++c)
+        acc[c] = fmaf(ps_, bf2f(vr[64 * c + lane]), acc[c]);
+    }
+    __threadfence_block();   // p reads done before next chunk overwrites
+  }
+  const float inv = l_run > 0.f ? 1.0f / l_run : 0.f;
+  ushort* op = o + (int64_t(b) * H + h) * D;
+#pragma unroll
+  for (int c = 0; c < CPL; ++c) op[64 * c + lane] = f2bf(acc[c] * inv);
+}
+
 // ---------------- mfma layout probes ----------------
 __global__ void mfma_probe16_k(const ushort* A, const ushort* B, float* D) {
   const int lane = threadIdx.x & 63;
@@ -611,4 +703,16 @@ void launch_mfma_probe_16(const bf16_t* A, const bf16_t* B, float* D,
 void launch_mfma_probe_32(const bf16_t* A, const bf16_t* B, float* D,
                           hipStream_t s) {
   mfma_probe32_k<<<1, 64, 0, s>>>(A, B, D);
+}
+
+void launch_attn_decode(const bf16_t* q, const bf16_t* kc, const bf16_t* vc,
+                        bf16_t* o, int B, int H, int grp, int kvlen, int hd,
+                        int64_t cb, int64_t ch, float scale, hipStream_t s) {
+  const int grid = (B * H + 3) / 4;
+  if (hd == 64)
+    attn_decode_k<4><<<grid, 256, 0, s>>>(q, kc, vc, o, B, H, grp, kvlen,
+                                          cb, ch, scale);
+  else if (hd == 128)
+    attn_decode_k<8><<<grid, 256, 0, s>>>(q, kc, vc, o, B, H, grp, kvlen,
+                                          cb, ch, scale);
 }

@@ -481,6 +481,23 @@ void attn_bwd_packed(Tensor dout, Tensor q, Tensor k, Tensor v,
   dv_v.copy_(dv32.permute({0, 2, 1, 3}));
 }
 
+// Decode attention over a KV cache: q [B,H,D], k/v caches [B,Hk,Lmax,D]
+// contiguous, first kv_len positions valid -> o [B,H,D].
+Tensor attn_decode(Tensor q, Tensor kc, Tensor vc, int64_t kv_len,
+                   double scale) {
+  check_bf16(q, "q"); check_bf16(kc, "kcache"); check_bf16(vc, "vcache");
+  TORCH_CHECK(q.dim() == 3 && kc.dim() == 4, "q [B,H,D], cache [B,Hk,L,D]");
+  const int B = int(q.size(0)), H = int(q.size(1)), hd = int(q.size(2));
+  const int Hk = int(kc.size(1));
+  TORCH_CHECK(hd == 64 || hd == 128, "decode head dim must be 64/128");
+  TORCH_CHECK(H % Hk == 0 && kv_len <= kc.size(2), "bad cache geometry");
+  auto o = torch::empty_like(q);
+  launch_attn_decode(bfp(q), bfp(kc), bfp(vc), bfp_mut(o), B, H, H / Hk,
+                     int(kv_len), hd, kc.stride(0), kc.stride(1),
+                     float(scale), stream());
+  return o;
+}
+
 // ---- mfma self-test --------------------------------------------------------
 Tensor mfma_selftest_16(Tensor A, Tensor B) {
   check_bf16(A, "A"); check_bf16(B, "B");
@@ -527,6 +544,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("attn_fwd", &attn_fwd);
   m.def("attn_bwd", &attn_bwd);
   m.def("attn_bwd_packed", &attn_bwd_packed);
+  m.def("attn_decode", &attn_decode);
   m.def("mfma_selftest_16", &mfma_selftest_16);
   m.def("mfma_selftest_32", &mfma_selftest_32);
 }

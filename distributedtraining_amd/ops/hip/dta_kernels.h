@@ -146,6 +146,10 @@ void launch_attn_bwd_dkv(const bf16_t* dout, const bf16_t* q,
                          bf16_t* dk, bf16_t* dv, const AttnGeom& geo,
                          hipStream_t s);
 
+void launch_attn_decode(const bf16_t* q, const bf16_t* kc, const bf16_t* vc,
+                        bf16_t* o, int B, int H, int grp, int kvlen, int hd,
+                        int64_t cb, int64_t ch, float scale, hipStream_t s);
+
 // ---- mfma layout self-test ------------------------------------------------
 // D[32,32] = A[32,16] x B[16,32] and D[16,16] = A[16,32] x B[32,16]
 void launch_mfma_probe_32(const bf16_t* A, const bf16_t* B, float* D,

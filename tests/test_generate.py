@@ -59,3 +59,31 @@ def test_trained_model_completes_the_corpus():
     out = generate(model, prompt, max_new_tokens=6)
     completion = tok.decode(out[0, prompt.shape[1]:].tolist())
     assert completion == "defgh "
+
+
+def test_cached_generation_matches_uncached():
+    """KV-cache path (prefill + decode_step) must produce the same greedy
+    tokens as the full-context re-forward path."""
+    cfg = ModelConfig.gpt2_tiny()
+    torch.manual_seed(0)
+    model = build_model(cfg)
+    ids = torch.randint(0, cfg.vocab_size, (2, 7))
+    ref = generate(model, ids, max_new_tokens=8, use_cache=False)
+    got = generate(model, ids, max_new_tokens=8, use_cache=True)
+    assert torch.equal(got, ref)
+
+
+def test_decode_attention_cpu_fallback_matches_full():
+    """ops.decode_attention (CPU composition) == last row of full causal
+    attention, incl. GQA."""
+    import math
+    from distributedtraining_amd import ops
+    B, H, Hk, L, D = 2, 4, 2, 9, 16
+    torch.manual_seed(1)
+    k = torch.randn(B, Hk, L, D)
+    v = torch.randn(B, Hk, L, D)
+    q = torch.randn(B, H, D)
+    out = ops.decode_attention(q, k, v, L)
+    full = torch.nn.functional.scaled_dot_product_attention(
+        q.unsqueeze(2), k, v, scale=1.0 / math.sqrt(D), enable_gqa=True)
+    torch.testing.assert_close(out, full.squeeze(2), rtol=1e-5, atol=1e-5)

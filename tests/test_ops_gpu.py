@@ -643,3 +643,13 @@ def test_generate_on_gpu():
     out = generate(model, ids, max_new_tokens=5)
     assert out.shape == (2, 11)
     assert torch.equal(out, generate(model, ids, max_new_tokens=5))
+    # KV-cache decode kernel path: greedy tokens must match uncached.
+    # gpt2-tiny head_dim must be >=64 for the decode kernel -> build one
+    cfg2 = ModelConfig(family="gpt2", vocab_size=512, n_layer=2, n_head=2,
+                       n_embd=128, n_positions=128)
+    torch.manual_seed(1)
+    m2 = build_model(cfg2).to(DEV, torch.bfloat16)
+    ids2 = torch.randint(0, 512, (2, 9), device=DEV)
+    ref = generate(m2, ids2, max_new_tokens=6, use_cache=False)
+    got = generate(m2, ids2, max_new_tokens=6, use_cache=True)
+    assert torch.equal(got, ref)
