@@ -132,3 +132,67 @@ class LlamaLM(nn.Module):
         return CausalLMOutput(
             loss=loss,
             logits=logits if (return_logits or not self.training) else None)
+
+
+def _llama_block_attn_cached(blk: LlamaBlock, x: torch.Tensor, cos, sin,
+                             cache, i: int) -> torch.Tensor:
+    """One Llama block with the KV cache (GQA-aware): prefill (S>1) or
+    decode (S==1 through ops.decode_attention). cos/sin are already
+    position-offset slices."""
+    B, S, E = x.shape
+    D = blk.head_dim
+    h = ops.rms_norm(x, blk.attn_norm_w, blk.norm_eps)
+    q = ops.linear(h, blk.q_w).view(B, S, blk.n_head, D).transpose(1, 2)
+    k = ops.linear(h, blk.k_w).view(B, S, blk.n_kv, D).transpose(1, 2)
+    v = ops.linear(h, blk.v_w).view(B, S, blk.n_kv, D).transpose(1, 2)
+    q = ops.rope(q, cos, sin)
+    k = ops.rope(k, cos, sin)
+    cache.append(i, k.transpose(1, 2), v.transpose(1, 2))
+    if S == 1:
+        o = ops.decode_attention(q.reshape(B, blk.n_head, D), cache.k[i],
+                                 cache.v[i], cache.len + 1)
+        o = o.view(B, 1, E)
+    else:
+        assert cache.len == 0, "prefill must start an empty cache"
+        o = ops.causal_attention(q, k, v).transpose(1, 2).reshape(B, S, E)
+    x = x + ops.linear(o, blk.o_w)
+    h = ops.rms_norm(x, blk.mlp_norm_w, blk.norm_eps)
+    h = ops.swiglu(ops.linear(h, blk.gate_w), ops.linear(h, blk.up_w))
+    return x + ops.linear(h, blk.down_w)
+
+
+def _llama_cached_forward(model: "LlamaLM", input_ids: torch.Tensor,
+                          cache) -> torch.Tensor:
+    S = input_ids.shape[1]
+    off = cache.len
+    cos = model.rope_cos[off:off + S].contiguous()
+    sin = model.rope_sin[off:off + S].contiguous()
+    x = ops.embedding_fwd(input_ids, model.tok_emb, None)
+    for i, blk in enumerate(model.blocks):
+        x = _llama_block_attn_cached(blk, x, cos, sin, cache, i)
+    cache.advance(S)
+    x = ops.rms_norm(x[:, -1:], model.final_norm_w, model.cfg.norm_eps)
+    return ops.linear(x, model._head()).squeeze(1)
+
+
+def _llama_new_cache(self: "LlamaLM", batch: int, max_len: int, device,
+                     dtype=None):
+    from .gpt2 import KVCache
+    cfg = self.cfg
+    return KVCache(cfg.n_layer, batch, cfg.n_kv_head or cfg.n_head,
+                   max_len, cfg.n_embd // cfg.n_head, device,
+                   dtype or next(self.parameters()).dtype)
+
+
+def _llama_prefill(self: "LlamaLM", input_ids: torch.Tensor, cache):
+    return _llama_cached_forward(self, input_ids, cache)
+
+
+def _llama_decode_step(self: "LlamaLM", input_ids: torch.Tensor, cache):
+    assert input_ids.shape[1] == 1
+    return _llama_cached_forward(self, input_ids, cache)
+
+
+LlamaLM.new_cache = _llama_new_cache
+LlamaLM.prefill = _llama_prefill
+LlamaLM.decode_step = _llama_decode_step

@@ -87,3 +87,17 @@ def test_decode_attention_cpu_fallback_matches_full():
     full = torch.nn.functional.scaled_dot_product_attention(
         q.unsqueeze(2), k, v, scale=1.0 / math.sqrt(D), enable_gqa=True)
     torch.testing.assert_close(out, full.squeeze(2), rtol=1e-5, atol=1e-5)
+
+
+def test_cached_generation_matches_uncached_llama():
+    """Llama KV-cache path (GQA, rope offsets) == full re-forward."""
+    cfg = ModelConfig(family="llama", vocab_size=256, n_layer=2, n_head=4,
+                      n_kv_head=2, n_embd=64, n_positions=64,
+                      intermediate_size=128, rope_theta=10000.0,
+                      tie_word_embeddings=False)
+    torch.manual_seed(2)
+    model = build_model(cfg)
+    ids = torch.randint(0, 256, (2, 6))
+    ref = generate(model, ids, max_new_tokens=7, use_cache=False)
+    got = generate(model, ids, max_new_tokens=7, use_cache=True)
+    assert torch.equal(got, ref)
