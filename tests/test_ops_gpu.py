@@ -653,3 +653,14 @@ def test_generate_on_gpu():
     ref = generate(m2, ids2, max_new_tokens=6, use_cache=False)
     got = generate(m2, ids2, max_new_tokens=6, use_cache=True)
     assert torch.equal(got, ref)
+    # llama GQA decode kernel path
+    cfg3 = ModelConfig(family="llama", vocab_size=512, n_layer=2, n_head=4,
+                       n_kv_head=2, n_embd=256, n_positions=128,
+                       intermediate_size=512, rope_theta=10000.0,
+                       tie_word_embeddings=False)
+    torch.manual_seed(2)
+    m3 = build_model(cfg3).to(DEV, torch.bfloat16)
+    ids3 = torch.randint(0, 512, (2, 8), device=DEV)
+    ref3 = generate(m3, ids3, max_new_tokens=5, use_cache=False)
+    got3 = generate(m3, ids3, max_new_tokens=5, use_cache=True)
+    assert torch.equal(got3, ref3)
