@@ -603,7 +603,11 @@ class _RopeFn(torch.autograd.Function):
 
 
 def rope(x: torch.Tensor, cos: torch.Tensor, sin: torch.Tensor) -> torch.Tensor:
-    """Apply rotary embedding to x [B,H,S,D] with cos/sin [S,D/2] (fp32)."""
+    """Apply rotary embedding to x [B,H,S,D] with cos/sin [S,D/2] (fp32;
+    other dtypes are upcast — Module.to(bf16) casts registered buffers)."""
+    if cos.dtype != torch.float32:
+        cos = cos.float()
+        sin = sin.float()
     if use_hip(x):
         return _RopeFn.apply(x, cos, sin)
     xf = x.float()
