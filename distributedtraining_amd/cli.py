@@ -55,7 +55,7 @@ def main(argv=None) -> int:
     logging.basicConfig(level=logging.INFO)
     top = argparse.ArgumentParser("distributedtraining_amd")
     top.add_argument("role", choices=["miner", "validator", "averager",
-                                      "bootstrap"])
+                                      "bootstrap", "serve"])
     top.add_argument("--hotkey", default=None)
     top.add_argument("--steps", type=int, default=100)
     top.add_argument("--rounds", type=int, default=1)
@@ -71,6 +71,33 @@ def main(argv=None) -> int:
     if ns.tiny:
         cfg.model = ModelConfig.gpt2_tiny()
     hotkey = ns.hotkey or ns.role
+
+    if ns.role == "serve":
+        import time as _time
+        from .models import build_model as _bm
+        from .parallel.flat import FlatParams as _FP
+        from .store import FileStore as _FS
+        from .utils.serve import InferenceServer
+        device = torch.device(cfg.comm.device if not torch.cuda.is_available()
+                              else "cuda:0")
+        torch.manual_seed(cfg.seed)
+        model = _bm(cfg.model).to(device)
+        if device.type == "cuda":
+            model = model.to(torch.bfloat16)
+        st = _FS(cfg.comm.root, hotkey="serve").pull_model()
+        if st is not None and "flat_master" in st:
+            fp = _FP(model, device=device)
+            fp.load_flat_master(st["flat_master"])
+            print("serving the store's current base model")
+        srv = InferenceServer(model, device, port=ns.port)
+        srv.start()
+        print(f"inference server on :{srv.port} (POST /generate)")
+        try:
+            while True:
+                _time.sleep(60)
+        except KeyboardInterrupt:
+            srv.stop()
+        return 0
 
     if ns.role == "bootstrap":
         import time as _time

@@ -101,3 +101,38 @@ def test_cached_generation_matches_uncached_llama():
     ref = generate(model, ids, max_new_tokens=7, use_cache=False)
     got = generate(model, ids, max_new_tokens=7, use_cache=True)
     assert torch.equal(got, ref)
+
+
+def test_inference_server_roundtrip():
+    """HTTP serving endpoint: /generate completes ids, /health reports."""
+    import json
+    import urllib.request
+    from distributedtraining_amd.utils.serve import (InferenceServer,
+                                                     post_generate)
+    cfg = ModelConfig.gpt2_tiny()
+    torch.manual_seed(0)
+    model = build_model(cfg)
+    srv = InferenceServer(model, torch.device("cpu"))
+    srv.start()
+    url = f"http://127.0.0.1:{srv.port}"
+    try:
+        ids = [[1, 2, 3], [4, 5, 6]]
+        out = post_generate(url, ids, max_new_tokens=4)
+        assert out is not None and len(out) == 2 and len(out[0]) == 7
+        # matches direct generation (greedy, cached)
+        ref = generate(model, torch.tensor(ids), 4, use_cache=True)
+        assert out == ref.tolist()
+        # bad request -> 400, not a crash
+        req = urllib.request.Request(url + "/generate", data=b"{}",
+                                     headers={"Content-Type":
+                                              "application/json"})
+        try:
+            urllib.request.urlopen(req, timeout=10)
+            assert False, "expected 400"
+        except urllib.error.HTTPError as e:
+            assert e.code == 400
+        with urllib.request.urlopen(url + "/health", timeout=10) as r:
+            h = json.loads(r.read())
+        assert h["status"] == "ok" and h["served"] == 1
+    finally:
+        srv.stop()
