@@ -3,16 +3,16 @@
 This is the framework's first-class parallelism: each of
 the node's GPUs runs one miner doing communication-sparse local SGD on a
 shared base; every merge interval the per-rank weight deltas are exchanged
-with ONE RCCL all-gather over the xGMI clique, merged (uniform,
-score-weighted or meta-learned weights), and the merged base is installed
-on every rank — replacing the reference's miner→HF-hub→averager→HF-hub→miner
-round trip with two collectives.
+with ONE RCCL collective over the xGMI clique, merged, and the merged base
+installed on every rank — replacing the reference's
+miner→HF-hub→averager→HF-hub→miner round trip.
 
-Merge placement: the all-gather leaves ALL deltas resident on EVERY rank,
-so the uniform/score-weighted merges are computed redundantly by each rank
-(deterministic kernels ⇒ identical bases, no broadcast needed); the
-meta-learned merge runs on rank 0 (it needs val-loss backward passes) and
-is broadcast (C2/C5).
+Collective choice per strategy (see merge_round): mean/nesterov use an
+all-reduce (O(P) resident — mandatory at Llama scale); score-weighted and
+meta-learned use an all-gather (every delta HBM-resident, optionally bf16
+on the wire). Deterministic merges are computed redundantly by every rank
+(identical bases, no broadcast); the meta-learned merge runs on rank 0
+(it needs val-loss backward passes) and is broadcast (C2/C5).
 """
 
 from __future__ import annotations
