@@ -116,7 +116,11 @@ class DeltaValidator:
         self.maybe_pull_base()
         total = 0.0
         for hotkey, ckpt in deltas.items():
-            loss, ppl, loss_score, ppl_score = self.score_delta(ckpt)
+            try:
+                loss, ppl, loss_score, ppl_score = self.score_delta(ckpt)
+            except Exception as e:   # reference: any per-miner failure
+                log.warning("%s: scoring failed (%s), score 0", hotkey, e)
+                loss, ppl, loss_score, ppl_score = 1e8, 1e8, 0.0, 0.0
             self.scores[hotkey] = ppl_score   # reference scores by ppl (:136-140)
             total += ppl_score
             log.info("%s: loss=%.4f ppl=%.2f score=%.4f", hotkey, loss, ppl,

@@ -201,3 +201,18 @@ def test_registry_deregister_drops_miner(tmp_path):
     assert "gone" in registry.hotkeys
     registry.deregister("gone")
     assert "gone" not in registry.hotkeys
+
+
+def test_validator_survives_corrupt_delta(tmp_path):
+    """Any unexpected per-miner failure scores 0 instead of crashing the
+    round (reference: per-miner try/except, validation_logic.py:152-166)."""
+    cfg, model, fp, store, registry = _mk(tmp_path, "c0")
+    ev = synthetic_eval_set(cfg.model.vocab_size, 1, 2, 16)
+    from distributedtraining_amd.roles.validator import DeltaValidator
+    validator = DeltaValidator(model, fp, ev, cfg.validate)
+    # dtype-corrupt checkpoint: spec validates but axpy_ will raise on CPU
+    bad = DeltaCheckpoint(torch.zeros(fp.numel, dtype=torch.int64).float()[:3],
+                          fp.spec, "")   # wrong numel despite matching spec
+    bad.flat = "not a tensor"            # hard corruption
+    scores = validator.validate_and_score({"bad": bad})
+    assert scores == {"bad": 0.0}
