@@ -43,6 +43,49 @@ static hipblasLtHandle_t handle() {
 }
 
 constexpr size_t kWorkspace = 32u << 20;  // 32 MiB
+constexpr int kMaxAlgos = 64;
+
+// Time each heuristic candidate (3 reps, beta=0 into scratch) and return
+// the fastest — hipBLASLt's first heuristic pick measured 8 ms/step slower
+// than TunableOp's stream-K selection on the wgrad shapes, so every lt
+// plan here gets its own mini-TunableOp pass at creation (one-time, in
+// eager warmup).
+static int pick_algo(hipblasLtMatmulDesc_t op, hipblasLtMatrixLayout_t la,
+                     hipblasLtMatrixLayout_t lb, hipblasLtMatrixLayout_t lc,
+                     const void* A, const void* B, void* Dscratch,
+                     void* ws, hipblasLtMatmulHeuristicResult_t* results,
+                     int nres, hipStream_t stream) {
+  if (nres <= 1) return 0;
+  const float alpha = 1.0f, beta = 0.0f;
+  hipEvent_t t0, t1;
+  (void)hipEventCreate(&t0);
+  (void)hipEventCreate(&t1);
+  int best = 0;
+  float best_ms = 1e30f;
+  for (int i = 0; i < nres; ++i) {
+    // warm once; skip candidates that fail at run time
+    if (hipblasLtMatmul(handle(), op, &alpha, A, la, B, lb, &beta, Dscratch,
+                        lc, Dscratch, lc, &results[i].algo, ws, kWorkspace,
+                        stream) != HIPBLAS_STATUS_SUCCESS)
+      continue;
+    (void)hipEventRecord(t0, stream);
+    for (int r = 0; r < 3; ++r)
+      (void)hipblasLtMatmul(handle(), op, &alpha, A, la, B, lb, &beta,
+                            Dscratch, lc, Dscratch, lc, &results[i].algo,
+                            ws, kWorkspace, stream);
+    (void)hipEventRecord(t1, stream);
+    (void)hipEventSynchronize(t1);
+    float ms = 1e30f;
+    (void)hipEventElapsedTime(&ms, t0, t1);
+    if (ms < best_ms) {
+      best_ms = ms;
+      best = i;
+    }
+  }
+  (void)hipEventDestroy(t0);
+  (void)hipEventDestroy(t1);
+  return best;
+}
 
 struct Plan {
   hipblasLtMatmulDesc_t op{};
@@ -282,15 +325,24 @@ Tensor lt_wgrad_bgradb(Tensor dy, Tensor x, Tensor dw_acc) {
     LT_CHECK(hipblasLtMatmulPreferenceSetAttribute(
         pref, HIPBLASLT_MATMUL_PREF_MAX_WORKSPACE_BYTES, &wsz,
         sizeof(wsz)));
-    hipblasLtMatmulHeuristicResult_t results[4];
+    hipblasLtMatmulHeuristicResult_t results[kMaxAlgos];
     int nres = 0;
     LT_CHECK(hipblasLtMatmulAlgoGetHeuristic(handle(), p.op, p.la, p.lb,
-                                             p.lc, p.lc, pref, 4, results,
-                                             &nres));
+                                             p.lc, p.lc, pref, kMaxAlgos,
+                                             results, &nres));
     hipblasLtMatmulPreferenceDestroy(pref);
     TORCH_CHECK(nres > 0, "hipblasLt: no BGRADB algorithm M=", M, " N=",
                 N, " E=", E);
-    p.algo = results[0].algo;
+    {
+      auto dscr = torch::empty({N, E}, dy.options());
+      auto wst = at::empty({int64_t(kWorkspace)},
+                           dy.options().dtype(torch::kByte));
+      const int bi = pick_algo(p.op, p.la, p.lb, p.lc, x.data_ptr(),
+                               dy.data_ptr(), dscr.data_ptr(),
+                               wst.data_ptr(), results, nres,
+                               at::hip::getCurrentHIPStream().stream());
+      p.algo = results[bi].algo;
+    }
     p.ready = true;
     it = g_plans.emplace(key, p).first;
   }
