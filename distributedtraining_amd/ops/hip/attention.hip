@@ -249,50 +249,21 @@ __global__ void attn_fwd_k(const ushort* __restrict__ q,
     }
 }
 
-// ---------------- delta = rowsum(dO * O) ----------------
-// 8 lanes per row (head_dim is 32/64/128: s16x8 loads), 8 rows per wave,
-// 32 per block — the one-wave-per-row version left >=half the lanes idle
-// and ran 7x off memory SOL.
-__global__ void attn_delta_k(const ushort* __restrict__ dout,
-                             const ushort* __restrict__ o,
-                             float* __restrict__ delta, AttnGeom geo) {
-  const int lane = threadIdx.x & 63, wid = threadIdx.x >> 6;
-  const int sub = lane & 7;        // lane within row
-  const int rloc = lane >> 3;      // row within wave
-  const int hd = geo.hd;
-  const int64_t rows = int64_t(geo.B) * geo.H * geo.seq;
-  for (int64_t rr = int64_t(blockIdx.x) * 32 + wid * 8 + rloc; rr < rows;
-       rr += int64_t(gridDim.x) * 32) {
-    const int s_ = int(rr % geo.seq);
-    const int h = int((rr / geo.seq) % geo.H);
-    const int b = int(rr / (int64_t(geo.seq) * geo.H));
-    const ushort* a = dout + b * geo.db_ + h * geo.dh + int64_t(s_) * geo.ds;
-    const ushort* c = o + b * geo.ob + h * geo.oh + int64_t(s_) * geo.os_;
-    float acc = 0.f;
-    for (int i = sub * 8; i + 8 <= hd; i += 64) {
-      s16x8 va = *reinterpret_cast<const s16x8*>(a + i);
-      s16x8 vc = *reinterpret_cast<const s16x8*>(c + i);
-#pragma unroll
-      for (int j = 0; j < 8; ++j)
-        acc = fmaf(bf2f(ushort(va[j])), bf2f(ushort(vc[j])), acc);
-    }
-    acc += __shfl_xor(acc, 4);
-    acc += __shfl_xor(acc, 2);
-    acc += __shfl_xor(acc, 1);
-    if (sub == 0) delta[rr] = acc;
-  }
-}
-
 // ---------------- backward dQ ----------------
 // Same cooperative structure as forward: K (row-major + transposed) and V
 // (row-major) staged once per block per 32-key iteration.
+// Also COMPUTES delta[q] = rowsum(dO[q]*O[q]) from its own rows (it loads
+// dO anyway; O rows are one extra coalesced read) and publishes it for the
+// dkv kernel — the standalone delta kernel is gone.
 template <int DTILES>
 __global__ void attn_bwd_dq_k(const ushort* __restrict__ dout,
                               const ushort* __restrict__ q,
                               const ushort* __restrict__ k,
                               const ushort* __restrict__ v,
+                              const ushort* __restrict__ o,
+                              int64_t ob2, int64_t oh2, int64_t os2,
                               const float* __restrict__ lse,
-                              const float* __restrict__ delta,
+                              float* __restrict__ delta,
                               ushort* __restrict__ dq, AttnGeom geo) {
   const int tid = threadIdx.x;
   const int lane = tid & 63;
@@ -307,18 +278,29 @@ __global__ void attn_bwd_dq_k(const ushort* __restrict__ dout,
   const ushort* kp = k + b * geo.kb + hk * geo.kh;
   const ushort* vp = v + b * geo.vb + hk * geo.vh;
   const ushort* dop = dout + b * geo.db_ + h * geo.dh;
+  const ushort* op2 = o + b * ob2 + h * oh2;
 
   const int qrow = q0 + (lane & 15);
   const int qr_ld = qrow < seq ? qrow : seq - 1;
   bf16x8 qb_[DTILES / 2], dob[DTILES / 2];
+  float dpart = 0.f;
 #pragma unroll
   for (int sl = 0; sl < DTILES / 2; ++sl) {
     const int c0 = 32 * sl + 8 * (lane >> 4);
     qb_[sl] = load_frag_row(qp, geo.qs, qr_ld, c0);
     dob[sl] = load_frag_row(dop, geo.ds, qr_ld, c0);
+    bf16x8 orow = load_frag_row(op2, os2, qr_ld, c0);
+#pragma unroll
+    for (int j = 0; j < 8; ++j)
+      dpart = fmaf(float(dob[sl][j]), float(orow[j]), dpart);
   }
+  // delta[qrow]: each of the 4 lane-groups holds 16 of the 64 columns of
+  // row (lane&15); fold across groups
+  dpart += __shfl_xor(dpart, 16);
+  dpart += __shfl_xor(dpart, 32);
+  const float dlt_q = dpart;
+  if (lane < 16 && qrow < seq) delta[bhid * seq + qrow] = dlt_q;
   const float lse_q = lse[bhid * seq + qr_ld] * LOG2E;  // exp2 units
-  const float dlt_q = delta[bhid * seq + qr_ld];
   const float sc2 = geo.scale * LOG2E;
 
   f32x4 acc[DTILES];
@@ -652,22 +634,16 @@ void launch_attn_fwd(const bf16_t* q, const bf16_t* k, const bf16_t* v,
   else if (hd == 32) attn_fwd_k<2><<<grid, 256, 0, s>>>(q, k, v, o, lse, geo);
 }
 
-void launch_attn_delta(const bf16_t* dout, const bf16_t* o, float* delta,
-                       const AttnGeom& geo, hipStream_t s) {
-  int64_t rows = int64_t(geo.B) * geo.H * geo.seq;
-  int64_t want = (rows + 31) / 32;
-  const int grid = int(want < 4096 ? (want > 0 ? want : 1) : 4096);
-  attn_delta_k<<<grid, 256, 0, s>>>(dout, o, delta, geo);
-}
-
 void launch_attn_bwd_dq(const bf16_t* dout, const bf16_t* q, const bf16_t* k,
-                        const bf16_t* v, const float* lse, const float* delta,
-                        bf16_t* dq, const AttnGeom& geo, hipStream_t s) {
+                        const bf16_t* v, const bf16_t* o, int64_t ob2,
+                        int64_t oh2, int64_t os2, const float* lse,
+                        float* delta, bf16_t* dq, const AttnGeom& geo,
+                        hipStream_t s) {
   dim3 grid((geo.seq + 63) / 64, int64_t(geo.B) * geo.H);
   const int hd = geo.hd;
-  if (hd == 64) attn_bwd_dq_k<4><<<grid, 256, 0, s>>>(dout, q, k, v, lse, delta, dq, geo);
-  else if (hd == 128) attn_bwd_dq_k<8><<<grid, 256, 0, s>>>(dout, q, k, v, lse, delta, dq, geo);
-  else if (hd == 32) attn_bwd_dq_k<2><<<grid, 256, 0, s>>>(dout, q, k, v, lse, delta, dq, geo);
+  if (hd == 64) attn_bwd_dq_k<4><<<grid, 256, 0, s>>>(dout, q, k, v, o, ob2, oh2, os2, lse, delta, dq, geo);
+  else if (hd == 128) attn_bwd_dq_k<8><<<grid, 256, 0, s>>>(dout, q, k, v, o, ob2, oh2, os2, lse, delta, dq, geo);
+  else if (hd == 32) attn_bwd_dq_k<2><<<grid, 256, 0, s>>>(dout, q, k, v, o, ob2, oh2, os2, lse, delta, dq, geo);
 }
 
 void launch_attn_bwd_dkv(const bf16_t* dout, const bf16_t* q,

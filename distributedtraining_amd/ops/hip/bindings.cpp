@@ -411,10 +411,9 @@ std::vector<Tensor> attn_bwd(Tensor dout, Tensor q, Tensor k, Tensor v,
   const int64_t Hk = k.size(1);
   AttnGeom geo = make_geom(q, k, v, o_bshd, dout, scale);
   auto delta = torch::empty({B * H, S}, lse.options());
-  launch_attn_delta(bfp(dout), bfp(o_bshd), delta.data_ptr<float>(), geo,
-                    stream());
   auto dq = torch::empty({B, S, H, hd}, q.options());
-  launch_attn_bwd_dq(bfp(dout), bfp(q), bfp(k), bfp(v),
+  launch_attn_bwd_dq(bfp(dout), bfp(q), bfp(k), bfp(v), bfp(o_bshd),
+                     o_bshd.stride(0), o_bshd.stride(2), o_bshd.stride(1),
                      lse.data_ptr<float>(), delta.data_ptr<float>(),
                      bfp_mut(dq), geo, stream());
   if (geo.grp == 1) {
@@ -453,11 +452,10 @@ void attn_bwd_packed(Tensor dout, Tensor q, Tensor k, Tensor v,
   const int64_t Hk = k.size(1);
   AttnGeom geo = make_geom(q, k, v, o_bshd, dout, scale);
   auto delta = torch::empty({B * H, S}, lse.options());
-  launch_attn_delta(bfp(dout), bfp(o_bshd), delta.data_ptr<float>(), geo,
-                    stream());
   AttnGeom gq = geo;  // dq_v is [B,S,H,D]: (batch, head, seq) strides
   gq.ob = dq_v.stride(0); gq.oh = dq_v.stride(2); gq.os_ = dq_v.stride(1);
-  launch_attn_bwd_dq(bfp(dout), bfp(q), bfp(k), bfp(v),
+  launch_attn_bwd_dq(bfp(dout), bfp(q), bfp(k), bfp(v), bfp(o_bshd),
+                     o_bshd.stride(0), o_bshd.stride(2), o_bshd.stride(1),
                      lse.data_ptr<float>(), delta.data_ptr<float>(),
                      bfp_mut(dq_v), gq, stream());
   if (geo.grp == 1) {

@@ -135,11 +135,13 @@ struct AttnGeom {
 void launch_attn_fwd(const bf16_t* q, const bf16_t* k, const bf16_t* v,
                      bf16_t* o, float* lse, const AttnGeom& geo,
                      hipStream_t s);
-void launch_attn_delta(const bf16_t* dout, const bf16_t* o, float* delta,
-                       const AttnGeom& geo, hipStream_t s);
+// computes delta[q]=rowsum(dO*O) itself (o + its strides passed
+// explicitly: geo.ob may be overridden for the packed dq output)
 void launch_attn_bwd_dq(const bf16_t* dout, const bf16_t* q, const bf16_t* k,
-                        const bf16_t* v, const float* lse, const float* delta,
-                        bf16_t* dq, const AttnGeom& geo, hipStream_t s);
+                        const bf16_t* v, const bf16_t* o, int64_t ob2,
+                        int64_t oh2, int64_t os2, const float* lse,
+                        float* delta, bf16_t* dq, const AttnGeom& geo,
+                        hipStream_t s);
 void launch_attn_bwd_dkv(const bf16_t* dout, const bf16_t* q,
                          const bf16_t* k, const bf16_t* v, const float* lse,
                          const float* delta, float* dk32, float* dv32,
