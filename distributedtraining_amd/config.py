@@ -140,6 +140,15 @@ class Config:
     seed: int = 0
 
 
+def _str2bool(v: str) -> bool:
+    # argparse type=bool is a trap: bool("False") is True
+    if v.lower() in ("1", "true", "yes", "on"):
+        return True
+    if v.lower() in ("0", "false", "no", "off"):
+        return False
+    raise argparse.ArgumentTypeError(f"expected a boolean, got {v!r}")
+
+
 def _add_dataclass_args(parser: argparse.ArgumentParser, prefix: str, dc) -> None:
     for f in dataclasses.fields(dc):
         if dataclasses.is_dataclass(f.type) or isinstance(f.default, (tuple,)):
@@ -148,7 +157,8 @@ def _add_dataclass_args(parser: argparse.ArgumentParser, prefix: str, dc) -> Non
         typ = f.type if callable(f.type) and f.type in (int, float, str) else None
         if typ is None:
             # dataclasses store types as strings under future annotations
-            typ = {"int": int, "float": float, "str": str, "bool": bool}.get(
+            typ = {"int": int, "float": float, "str": str,
+                   "bool": _str2bool}.get(
                 str(f.type).replace("typing.Optional[", "").rstrip("]"), str)
         parser.add_argument(name, type=typ, default=None, dest=f"{prefix}__{f.name}")
 
