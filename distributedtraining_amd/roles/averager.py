@@ -231,3 +231,19 @@ class ParameterizedAverager:
                                    "flat_master": merged.cpu(),
                                    "spec": self.fp.spec})
         return merged
+
+    def run_periodic_averaging(self, val_batches: List[dict],
+                               interval_s: float = 1200.0,
+                               max_rounds: Optional[int] = None) -> None:
+        """The reference's outer loop (averaging_logic.py:544-583):
+        collect + merge + publish, then sleep the REMAINDER of the
+        interval (its sleep(max(0, t - elapsed)) contract)."""
+        import time as _time
+        n = 0
+        while max_rounds is None or n < max_rounds:
+            t0 = _time.monotonic()
+            self.run_round(val_batches)
+            n += 1
+            if max_rounds is not None and n >= max_rounds:
+                break
+            _time.sleep(max(0.0, interval_s - (_time.monotonic() - t0)))

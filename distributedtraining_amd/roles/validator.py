@@ -133,3 +133,17 @@ class DeltaValidator:
         if self.registry is not None and self.registry.should_set_weights():
             self.registry.set_weights(self.normalized_scores)
         return dict(self.normalized_scores)
+
+    def start_periodic_validation(self, interval_s: float = 1800.0,
+                                  max_rounds: Optional[int] = None) -> None:
+        """The reference's outer loop (validation_logic.py:191-196):
+        validate, sleep ``interval_s`` (its default 1800 s), repeat.
+        ``max_rounds`` bounds the loop for tests/finite runs."""
+        import time as _time
+        n = 0
+        while max_rounds is None or n < max_rounds:
+            self.validate_and_score()
+            n += 1
+            if max_rounds is not None and n >= max_rounds:
+                break
+            _time.sleep(interval_s)

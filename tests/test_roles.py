@@ -216,3 +216,19 @@ def test_validator_survives_corrupt_delta(tmp_path):
     bad.flat = "not a tensor"            # hard corruption
     scores = validator.validate_and_score({"bad": bad})
     assert scores == {"bad": 0.0}
+
+
+def test_periodic_wrappers(tmp_path):
+    """The reference's outer loops, bounded for tests."""
+    cfg, model, fp, store, registry = _mk(tmp_path, "p0")
+    ev = synthetic_eval_set(cfg.model.vocab_size, 1, 2, 16)
+    from distributedtraining_amd.config import AverageConfig
+    from distributedtraining_amd.roles.averager import ParameterizedAverager
+    from distributedtraining_amd.roles.validator import DeltaValidator
+    v = DeltaValidator(model, fp, ev, cfg.validate, store=store,
+                       registry=registry)
+    v.start_periodic_validation(interval_s=0.0, max_rounds=2)
+    av = ParameterizedAverager(model, fp, AverageConfig(strategy="mean"),
+                               store=store, registry=registry)
+    av.run_periodic_averaging(ev, interval_s=0.0, max_rounds=2)
+    assert store.pull_model() is not None
