@@ -664,3 +664,26 @@ def test_generate_on_gpu():
     ref3 = generate(m3, ids3, max_new_tokens=5, use_cache=False)
     got3 = generate(m3, ids3, max_new_tokens=5, use_cache=True)
     assert torch.equal(got3, ref3)
+
+
+@pytest.mark.parametrize("B,H,Hk,L,D", [(2, 4, 4, 64, 64),
+                                        (2, 8, 2, 500, 64),
+                                        (1, 4, 4, 130, 128)])
+def test_decode_attention_kernel(B, H, Hk, L, D):
+    """Flash-decoding kernel vs fp32 sdpa: long caches cross the 64-key
+    chunk boundary (online rescale), GQA head folding, D=128 column split."""
+    from distributedtraining_amd import ops
+    Lmax = L + 7                      # cache longer than the valid prefix
+    q = _rand_bf16(B, H, D, seed=71)
+    k = _rand_bf16(B, Hk, Lmax, D, seed=72)
+    v = _rand_bf16(B, Hk, Lmax, D, seed=73)
+    out = ops.decode_attention(q, k, v, L)
+    scale = 1.0 / math.sqrt(D)
+    kf = k[:, :, :L].float()
+    vf = v[:, :, :L].float()
+    if H != Hk:
+        kf = kf.repeat_interleave(H // Hk, dim=1)
+        vf = vf.repeat_interleave(H // Hk, dim=1)
+    ref = torch.nn.functional.scaled_dot_product_attention(
+        q.float().unsqueeze(2), kf, vf, scale=scale).squeeze(2)
+    torch.testing.assert_close(out.float(), ref, rtol=3e-2, atol=3e-2)
