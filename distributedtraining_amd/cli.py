@@ -63,6 +63,12 @@ def main(argv=None) -> int:
                      help="use the tiny test-scale model")
     top.add_argument("--resume", action="store_true",
                      help="miner: restore saved train state from the store")
+    top.add_argument("--gradient-mode", action="store_true",
+                     help="legacy protocol: miner publishes aggregated "
+                          "normalized gradients (gradients.pt) instead of "
+                          "weight deltas; averager applies their "
+                          "score-weighted mean with --average.meta-lr as "
+                          "alpha (reference TrainingLoop + Averager)")
     top.add_argument("--metrics-dir", default="metrics")
     top.add_argument("--port", type=int, default=8500,
                      help="bootstrap: HTTP port")
@@ -126,8 +132,13 @@ def main(argv=None) -> int:
                               "spec": fp.spec})
         data = synthetic_batches(cfg.model.vocab_size, cfg.train.batch_size,
                                  cfg.train.seq_len, seed=cfg.seed + hash(hotkey) % 1000)
-        miner = DeltaLoop(model, fp, data, cfg.train, store=store,
-                          registry=registry, hotkey=hotkey)
+        if ns.gradient_mode:
+            from .roles.gradient_loop import GradientLoop
+            miner = GradientLoop(model, fp, data, cfg.train, store=store,
+                                 registry=registry, hotkey=hotkey)
+        else:
+            miner = DeltaLoop(model, fp, data, cfg.train, store=store,
+                              registry=registry, hotkey=hotkey)
         if ns.resume:
             st = store.pull_train_state()
             if st is not None:
@@ -139,8 +150,11 @@ def main(argv=None) -> int:
         else:
             miner.maybe_pull_base()
         miner.train(ns.steps)
-        miner.last_push_step = -10**9  # force a final push
-        miner.maybe_push_delta()
+        if ns.gradient_mode:
+            miner.push_gradients()
+        else:
+            miner.last_push_step = -10**9  # force a final push
+            miner.maybe_push_delta()
         store.push_train_state({"flat_master": fp.master.cpu(),
                                 "base": miner.base.cpu(),
                                 "step_count": miner.step_count})
@@ -175,6 +189,25 @@ def main(argv=None) -> int:
             fp.load_flat_master(sd["flat_master"])
         averager = ParameterizedAverager(model, fp, cfg.average, store=store,
                                          registry=registry)
+        if ns.gradient_mode:
+            from .roles.gradient_loop import apply_gradient_average
+            for r in range(ns.rounds):
+                cks = [store.receive_delta(registry.retrieve_address(h))
+                       for h in registry.hotkeys]
+                cks = [c for c in cks
+                       if c is not None and c.meta.get("kind") == "gradients"]
+                sc = registry.get_weights()
+                scores = ([sc.get(h, 1.0) for h in registry.hotkeys][:len(cks)]
+                          if sc else None)
+                n = apply_gradient_average(fp, cks, scores,
+                                           alpha=cfg.average.meta_lr)
+                store.push_model({"format": "dta-base-v1",
+                                  "flat_master": fp.master.cpu(),
+                                  "spec": fp.spec})
+                metrics.log(r, merged_gradients=n)
+                print(f"averager: applied {n} gradient aggregates")
+            metrics.close()
+            return 0
         for r in range(ns.rounds):
             averager.run_round(ev)
             metrics.log(r, merge_rounds=r + 1)

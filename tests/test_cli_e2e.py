@@ -68,3 +68,26 @@ def test_three_role_pipeline(tmp_path):
     names = {p.name for p in mdir.iterdir()}
     assert {"miner_m0.jsonl", "validator_validator.jsonl",
             "AVERAGER.jsonl"} <= names
+
+
+@pytest.mark.timeout(600)
+def test_gradient_mode_pipeline(tmp_path):
+    """Legacy gradient-publication protocol over the CLI (reference
+    TrainingLoop gradients.pt + Averager alpha-apply)."""
+    root = str(tmp_path / "gx")
+    common = ["--tiny", "--gradient-mode", "--comm.root", root,
+              "--metrics-dir", str(tmp_path / "m"),
+              "--train.batch-size", "2", "--train.seq-len", "16",
+              "--validate.batch-size", "2", "--validate.seq-len", "16",
+              "--validate.n-eval-batches", "1"]
+    _run(["miner", "--hotkey", "g0", "--steps", "4", *common])
+    base_before = torch.load(os.path.join(root, "model",
+                                          "averaged_model.pt"),
+                             weights_only=False)["flat_master"].clone()
+    out = _run(["averager", "--rounds", "1", "--average.meta-lr", "0.001",
+                *common])
+    assert "applied 1 gradient aggregates" in out
+    base_after = torch.load(os.path.join(root, "model",
+                                         "averaged_model.pt"),
+                            weights_only=False)["flat_master"]
+    assert not torch.equal(base_before, base_after)
