@@ -10,10 +10,12 @@ Usage:
   python -m distributedtraining_amd.cli validator --comm.root /tmp/ex --rounds 1
   python -m distributedtraining_amd.cli averager  --comm.root /tmp/ex --rounds 1
   python -m distributedtraining_amd.cli bootstrap --port 8500
+  python -m distributedtraining_amd.cli serve     --comm.root /tmp/ex --port 8600
 
 Every role writes a metrics JSONL under --metrics-dir (utils/metrics.py);
 --resume restores a miner's saved train state (base + step counter) from
-the store.
+the store; --gradient-mode switches miner/averager to the legacy
+gradient-publication protocol.
 """
 
 from __future__ import annotations
