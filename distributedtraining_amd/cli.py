@@ -1,4 +1,4 @@
-"""CLI entry points: miner | validator | averager | bench.
+"""CLI entry points: miner | validator | averager | bootstrap | serve.
 
 Replaces the reference's neurons/{miner,validator,averager}.py process
 entry points with explicit subcommands over the file
