@@ -59,28 +59,11 @@ class GradientLoop(DeltaLoop):
         self.apply_every = apply_every      # 0 = never apply locally
         self.apply_alpha = apply_alpha
 
-    def train_step(self, batch=None) -> torch.Tensor:
-        """DeltaLoop's step with the gradient captured before zeroing, then
-        L2-normalized and folded into the aggregate."""
-        if batch is None:
-            batch = next(self.data)
-        input_ids = batch["input_ids"].to(self.fp.device, non_blocking=True)
-        labels = batch.get("labels", batch["input_ids"]).to(
-            self.fp.device, non_blocking=True)
-        out = self.model(input_ids=input_ids, labels=labels)
-        out.loss.backward()
-        self.opt.step()
+    def _on_grad_available(self) -> None:
+        """DeltaLoop hook: capture the step's gradient (before zeroing),
+        L2-normalize and fold it into the aggregate."""
         g32 = self.fp.grad.detach().to(torch.float32).clone()
-        self.opt.zero_grad()
-        self.step_count += 1
-        loss = out.loss.detach()
-        if self._loss_acc is None:
-            self._loss_acc = torch.zeros((), dtype=torch.float32,
-                                         device=loss.device)
-        self._loss_acc += loss.float() * input_ids.shape[0]
-        self.total_examples += input_ids.shape[0]
         self._fold_gradient(g32)
-        return loss
 
     def _fold_gradient(self, g32: torch.Tensor) -> None:
         normalize_flat_(g32)
@@ -164,7 +147,7 @@ class ClassifierLoop(GradientLoop):
         loss = F.cross_entropy(logits.float(), y)
         loss.backward()
         self.opt.step()
-        g32 = self.fp.grad.detach().to(torch.float32).clone()
+        self._on_grad_available()
         self.opt.zero_grad()
         self.step_count += 1
         dloss = loss.detach()
@@ -173,5 +156,4 @@ class ClassifierLoop(GradientLoop):
                                          device=dloss.device)
         self._loss_acc += dloss.float() * x.shape[0]
         self.total_examples += x.shape[0]
-        self._fold_gradient(g32)
         return dloss

@@ -94,6 +94,7 @@ class DeltaLoop:
         out = self.model(input_ids=input_ids, labels=labels)
         out.loss.backward()
         self.opt.step()
+        self._on_grad_available()   # subclass hook (gradient protocol)
         self.opt.zero_grad()
         self.step_count += 1
         loss = out.loss.detach()
@@ -112,6 +113,10 @@ class DeltaLoop:
         if self._base_hash is None:
             self._base_hash = self.fp.master_hash()
         return self._base_hash
+
+    def _on_grad_available(self) -> None:
+        """Called after the optimizer step, BEFORE grads are zeroed —
+        GradientLoop captures and folds the step's gradient here."""
 
     # -- delta publication (reference :405-427) ------------------------------
     def make_delta(self, with_hash: bool = False) -> DeltaCheckpoint:
