@@ -52,6 +52,8 @@ def main() -> int:
                    help="disable hipGraph step capture (eager launches)")
     p.add_argument("--cpu", action="store_true",
                    help="CPU smoke mode (tiny model, tests only)")
+    p.add_argument("--no-preflight", action="store_true",
+                   help="skip the RCCL collective-shape pre-flight")
     args = p.parse_args()
 
     from distributedtraining_amd.config import Config, ModelConfig
@@ -91,6 +93,25 @@ def main() -> int:
                              args.seq_len, seed=1000 + rank)
     node = LocalSGDNode(model, fp, data, cfg, comm,
                         merge_strategy=args.merge_strategy)
+
+    if comm.is_distributed and not args.no_preflight:
+        # RCCL pre-flight: exercise the exact collective shapes the merge
+        # path needs, BEFORE any training state exists — failures surface
+        # early and attributably (round-1 verdict item #2).
+        t = torch.ones(1, device=device)
+        torch.distributed.all_reduce(t)
+        probe = torch.empty_like(fp.master)
+        d = comm.all_reduce_mean(probe.zero_())
+        if args.merge_strategy in ("score_weighted", "parameterized"):
+            comm.all_gather_flat(probe, torch.bfloat16
+                                 if cfg.comm.exchange_dtype == "bf16"
+                                 else None)
+        comm.broadcast_flat(probe, src=0)
+        del probe, d
+        if rank == 0:
+            print(f"preflight ok: world={world} flat={fp.numel}",
+                  file=sys.stderr)
+
     node.sync_initial_base()
 
     def sync():
@@ -128,11 +149,20 @@ def main() -> int:
     if args.merge_every:
         node.merge_round()
 
+    # The step counter CARRIES ACROSS warmup (round-1 fix): with the
+    # driver flags (steps=20, warmup=10, merge-every=25) the merge at
+    # global step 25 now lands INSIDE the timed region; merges_timed
+    # reports how many exchanges the measurement actually includes.
+    merges_before = node.merge_rounds
     sync()
     t0 = time.perf_counter()
-    run_steps(args.steps, 0)
+    run_steps(args.steps, args.warmup)
     sync()
     elapsed = time.perf_counter() - t0
+    merges_timed = node.merge_rounds - merges_before
+    if args.merge_every and merges_timed == 0 and rank == 0:
+        print(f"WARN: merges_timed=0 (steps={args.steps} never crossed a "
+              f"merge-every={args.merge_every} boundary)", file=sys.stderr)
 
     # MAX over ranks
     t = torch.tensor([elapsed], dtype=torch.float64, device=device
@@ -163,6 +193,7 @@ def main() -> int:
                 "seq_len": args.seq_len,
                 "parallelism": f"local_sgd_dp{world}",
                 "merge_every": args.merge_every,
+                "merges_timed": merges_timed,
                 "merge_strategy": args.merge_strategy,
                 "optimizer": "fused AdamW (lr 5e-4, reference config)",
                 "final_train_loss": node.miner.average_loss(),

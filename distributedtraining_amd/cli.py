@@ -69,8 +69,8 @@ def main(argv=None) -> int:
                      help="legacy protocol: miner publishes aggregated "
                           "normalized gradients (gradients.pt) instead of "
                           "weight deltas; averager applies their "
-                          "score-weighted mean with --average.meta-lr as "
-                          "alpha (reference TrainingLoop + Averager)")
+                          "score-weighted mean with --average.gradient-alpha "
+                          "(default 1e-5, reference TrainingLoop + Averager)")
     top.add_argument("--metrics-dir", default="metrics")
     top.add_argument("--port", type=int, default=8500,
                      help="bootstrap: HTTP port")
@@ -194,15 +194,18 @@ def main(argv=None) -> int:
         if ns.gradient_mode:
             from .roles.gradient_loop import apply_gradient_average
             for r in range(ns.rounds):
-                cks = [store.receive_delta(registry.retrieve_address(h))
-                       for h in registry.hotkeys]
-                cks = [c for c in cks
-                       if c is not None and c.meta.get("kind") == "gradients"]
+                # one snapshot of the hotkey list; filter (hotkey, ckpt)
+                # pairs together so scores stay aligned to their miners
+                hks = registry.hotkeys
+                pairs = [(h, store.receive_delta(registry.retrieve_address(h)))
+                         for h in hks]
+                pairs = [(h, c) for h, c in pairs
+                         if c is not None and c.meta.get("kind") == "gradients"]
+                cks = [c for _, c in pairs]
                 sc = registry.get_weights()
-                scores = ([sc.get(h, 1.0) for h in registry.hotkeys][:len(cks)]
-                          if sc else None)
+                scores = [sc.get(h, 1.0) for h, _ in pairs] if sc else None
                 n = apply_gradient_average(fp, cks, scores,
-                                           alpha=cfg.average.meta_lr)
+                                           alpha=cfg.average.gradient_alpha)
                 store.push_model({"format": "dta-base-v1",
                                   "flat_master": fp.master.cpu(),
                                   "spec": fp.spec})

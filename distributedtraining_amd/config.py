@@ -98,6 +98,10 @@ class AverageConfig:
 
     meta_epochs: int = 7         # reference: neurons/averager.py:106
     meta_lr: float = 0.01        # reference: neurons/averager.py:106
+    # legacy gradient protocol: step size for applying the score-weighted
+    # gradient average, θ −= α·ḡ (reference: apply_averaged_gradients,
+    # averaging_logic.py:149-153 — default 1e-5, NOT the meta lr)
+    gradient_alpha: float = 1e-5
     strategy: str = "parameterized"  # parameterized | score_weighted | genetic | mean | nesterov
     # genetic-only (reference: GeneticAverager, averaging_logic.py:830-970)
     population_size: int = 20

@@ -145,7 +145,9 @@ class KVCache:
         """k_new/v_new [B, S_new, Hk, D] written at position self.len
         (advance once per model step via ``advance``)."""
         S = k_new.shape[1]
-        assert self.len + S <= self.max_len, "KV cache overflow"
+        if self.len + S > self.max_len:
+            raise ValueError(
+                f"KV cache overflow: {self.len}+{S} > {self.max_len}")
         self.k[layer][:, :, self.len:self.len + S].copy_(
             k_new.transpose(1, 2))
         self.v[layer][:, :, self.len:self.len + S].copy_(

@@ -108,7 +108,11 @@ def linear(x: torch.Tensor, w: torch.Tensor,
 # --------------------------------------------------------------------------
 # Fused MLP: linear+GELU+linear through hipBLASLt epilogues
 # --------------------------------------------------------------------------
-_lt_fused_ok: Optional[bool] = None
+# hipBLASLt epilogue availability is SHAPE-dependent (the heuristics reject
+# e.g. GELU_AUX_BIAS at M=512 N=3072 K=768) — cache the verdict per
+# (M, N, K) so a serving decode shape rejected after a successful prefill
+# shape falls back instead of crashing.
+_lt_fused_shape_ok: dict = {}
 
 
 class _FusedMLPFn(torch.autograd.Function):
@@ -158,24 +162,22 @@ class _FusedMLPFn(torch.autograd.Function):
 def mlp_gelu(x: torch.Tensor, w1: torch.Tensor, b1: torch.Tensor,
              w2: torch.Tensor, b2: torch.Tensor) -> torch.Tensor:
     """y = gelu(x W1ᵀ + b1) W2ᵀ + b2 (GPT-2 MLP). On GPU, epilogue-fused
-    via hipBLASLt when available; falls back to the composed ops path once
-    if the epilogue heuristics reject the shapes."""
-    global _lt_fused_ok
-    if use_hip(x) and _lt_fused_ok is not False:
-        try:
-            out = _FusedMLPFn.apply(x, w1, b1, w2, b2)
-            _lt_fused_ok = True
-            return out
-        except RuntimeError as e:
-            if _lt_fused_ok is None:
+    via hipBLASLt when available for this (M, N, K); shapes the epilogue
+    heuristics reject fall back to the composed ops path (cached, logged
+    once per shape)."""
+    if use_hip(x):
+        key = (x.numel() // x.shape[-1], w1.shape[0], w1.shape[1])
+        if _lt_fused_shape_ok.get(key, True):
+            try:
+                out = _FusedMLPFn.apply(x, w1, b1, w2, b2)
+                _lt_fused_shape_ok[key] = True
+                return out
+            except RuntimeError as e:
                 import logging
                 logging.getLogger(__name__).warning(
-                    "hipBLASLt epilogue MLP unavailable (%s); composed "
-                    "path", e)
-                _lt_fused_ok = False
-            else:
-                raise
-    if use_hip(x):
+                    "hipBLASLt epilogue MLP unavailable at M,N,K=%s (%s); "
+                    "composed path", key, e)
+                _lt_fused_shape_ok[key] = False
         return linear(gelu(linear(x, w1, b1)), w2, b2)
     return F.linear(F.gelu(F.linear(x, w1, b1), approximate="tanh"), w2, b2)
 

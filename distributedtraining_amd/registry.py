@@ -222,16 +222,20 @@ class FileRegistry(Registry):
             self._save()
 
     def _load_unlocked(self):
-        # helper used while already holding the lock
+        # helper used while already holding the lock. The file is the
+        # source of truth for concurrent writers: every mutator persists
+        # under this lock before releasing it, so plain update (last
+        # writer wins) is correct — setdefault would pin the FIRST value
+        # ever seen and keep stale scores forever (round-1 advisory).
         if os.path.exists(self.path):
             try:
                 with open(self.path) as f:
                     d = json.load(f)
                 self._addresses.update(d.get("addresses", {}))
-                for k, v in d.get("scores", {}).items():
-                    self._scores.setdefault(k, v)
-                for k, v in d.get("stakes", {}).items():
-                    self._stakes.setdefault(k, v)
+                self._scores.update(d.get("scores", {}))
+                self._stakes.update(d.get("stakes", {}))
+                self._last_weight_set = max(self._last_weight_set,
+                                            d.get("last_weight_set", -10**12))
             except (json.JSONDecodeError, OSError):
                 pass
 
@@ -247,6 +251,8 @@ class FileRegistry(Registry):
             return list(self._addresses.keys())
 
     def set_weights(self, scores: Dict[str, float]) -> Dict[str, float]:
+        with self._lock:          # fold the EMA into the LATEST persisted
+            self._load_unlocked()  # scores, not this process's stale copy
         out = super().set_weights(scores)
         with self._lock:
             self._save()

@@ -107,11 +107,14 @@ class DeltaLoop:
 
     @property
     def base_hash(self) -> str:
-        """SHA-256 of the current base (reference: calculate_model_hash,
-        training_manager.py:198-203). Computed lazily — the in-node RCCL
-        merge path never needs it; only store pushes do."""
+        """SHA-256 of the BASE snapshot the delta is relative to
+        (reference: calculate_model_hash, training_manager.py:198-203).
+        Computed lazily — the in-node RCCL merge path never needs it; only
+        store pushes do. Hashes ``self.base`` (not the drifted master), so
+        the value is stable no matter when it is first read."""
         if self._base_hash is None:
-            self._base_hash = self.fp.master_hash()
+            from ..store import tensor_sha256
+            self._base_hash = tensor_sha256(self.base)
         return self._base_hash
 
     def _on_grad_available(self) -> None:

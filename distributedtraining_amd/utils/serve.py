@@ -76,6 +76,11 @@ class InferenceServer:
         if int(ids.min()) < 0 or int(ids.max()) >= vocab:
             raise ValueError("token id out of range")
         mnt = min(int(req.get("max_new_tokens", 32)), self.cap)
+        npos = getattr(self.model.cfg, "n_positions", None)
+        if npos is not None and ids.shape[1] + mnt > npos:
+            raise ValueError(
+                f"prompt ({ids.shape[1]}) + max_new_tokens ({mnt}) exceeds "
+                f"the model context length {npos}")
         temperature = float(req.get("temperature", 0.0))
         top_k = int(req.get("top_k", 0))
         eos = req.get("eos_token_id")

@@ -189,3 +189,13 @@ def test_text_dataset_miner_training(tmp_path):
     for _ in range(25):
         last = float(loop.train_step())
     assert last < first          # byte-level LM learns the repetitive corpus
+
+
+def test_gradient_alpha_default_is_reference_1e5():
+    """The legacy gradient-apply step size defaults to the reference's
+    alpha=1e-5 (averaging_logic.py:149-153), with its OWN config knob —
+    not the meta-learning lr (round-1 verdict item #8)."""
+    from distributedtraining_amd.config import AverageConfig
+    cfg = AverageConfig()
+    assert cfg.gradient_alpha == 1e-5
+    assert cfg.meta_lr == 0.01          # untouched, separate knob

@@ -49,3 +49,27 @@ def test_file_registry_cross_instance(tmp_path):
     assert b.retrieve_address("hk1") == "addr1"
     b.store_address("hk2", "addr2")
     assert set(a.hotkeys) == {"hk1", "hk2"}
+
+
+def test_file_registry_score_update_not_stale(tmp_path):
+    """Concurrent-writer score merge: a reload must pick up NEWER persisted
+    scores (round-1 advisory: setdefault pinned the first value forever)."""
+    a = FileRegistry(str(tmp_path), ema_alpha=1.0)
+    b = FileRegistry(str(tmp_path), ema_alpha=1.0)
+    a.set_weights({"hk": 1.0})
+    assert b.get_weights() == {}          # b hasn't reloaded yet
+    b.store_address("hk", "addr")          # any reload path
+    assert b.get_weights().get("hk") == 1.0
+    a.set_weights({"hk": 0.25})
+    b.retrieve_address("hk")               # reload again
+    assert b.get_weights().get("hk") == 0.25   # newer value wins
+
+
+def test_file_registry_set_weights_folds_latest(tmp_path):
+    """set_weights EMA must fold into the latest persisted scores, not this
+    process's stale copy."""
+    a = FileRegistry(str(tmp_path), ema_alpha=0.5)
+    b = FileRegistry(str(tmp_path), ema_alpha=0.5)
+    a.set_weights({"hk": 1.0})             # file: 0.5
+    out = b.set_weights({"hk": 1.0})       # folds into 0.5 -> 0.75
+    assert abs(out["hk"] - 0.75) < 1e-9
