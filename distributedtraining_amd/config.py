@@ -38,6 +38,12 @@ class ModelConfig:
     rope_theta: float = 500000.0
     norm_eps: float = 1e-5
     tie_word_embeddings: bool = True
+    # training dropout — the reference trains transformers GPT-2 with its
+    # defaults (resid/embd/attn pdrop 0.1, from_pretrained at
+    # neurons/miner.py:60-62); Llama-3 trains dropout-free
+    resid_pdrop: float = 0.1
+    embd_pdrop: float = 0.1
+    attn_pdrop: float = 0.1
 
     @staticmethod
     def gpt2_small() -> "ModelConfig":
@@ -45,23 +51,27 @@ class ModelConfig:
 
     @staticmethod
     def gpt2_tiny() -> "ModelConfig":
-        """CPU-test scale: same code path, toy size."""
+        """CPU-test scale: same code path, toy size (dropout off so test
+        numerics are deterministic; dropout tests opt in explicitly)."""
         return ModelConfig(vocab_size=512, n_layer=2, n_head=2, n_embd=64,
-                           n_positions=128)
+                           n_positions=128, resid_pdrop=0.0, embd_pdrop=0.0,
+                           attn_pdrop=0.0)
 
     @staticmethod
     def llama3_8b() -> "ModelConfig":
         return ModelConfig(family="llama", vocab_size=128256, n_layer=32,
                            n_head=32, n_kv_head=8, n_embd=4096,
                            intermediate_size=14336, n_positions=8192,
-                           tie_word_embeddings=False)
+                           tie_word_embeddings=False, resid_pdrop=0.0,
+                           embd_pdrop=0.0, attn_pdrop=0.0)
 
     @staticmethod
     def llama_tiny() -> "ModelConfig":
         return ModelConfig(family="llama", vocab_size=512, n_layer=2,
                            n_head=4, n_kv_head=2, n_embd=64,
                            intermediate_size=176, n_positions=128,
-                           tie_word_embeddings=True)
+                           tie_word_embeddings=True, resid_pdrop=0.0,
+                           embd_pdrop=0.0, attn_pdrop=0.0)
 
 
 @dataclass

@@ -30,10 +30,13 @@ from ..config import ModelConfig
 
 
 class GPT2Block(nn.Module):
-    def __init__(self, cfg: ModelConfig):
+    def __init__(self, cfg: ModelConfig, layer_idx: int = 0):
         super().__init__()
         E = cfg.n_embd
         self.n_head = cfg.n_head
+        self.layer_idx = layer_idx
+        self.attn_pdrop = cfg.attn_pdrop
+        self.resid_pdrop = cfg.resid_pdrop
         self.ln_1_w = nn.Parameter(torch.ones(E))
         self.ln_1_b = nn.Parameter(torch.zeros(E))
         self.attn_qkv_w = nn.Parameter(torch.empty(3 * E, E))
@@ -47,23 +50,35 @@ class GPT2Block(nn.Module):
         self.mlp_proj_w = nn.Parameter(torch.empty(E, 4 * E))
         self.mlp_proj_b = nn.Parameter(torch.zeros(E))
 
-    def forward(self, x: torch.Tensor, pending=None):
+    def forward(self, x: torch.Tensor, pending=None, kvlen=None):
         """Residual-join-fused layout: ``pending`` is the previous
         sub-layer's un-added branch output; every LayerNorm consumes it
         via ops.add_layer_norm (one kernel produces both the new residual
         stream and the normalized input — no separate add anywhere in the
-        transformer). Returns (stream, pending)."""
+        transformer). Returns (stream, pending).
+
+        ``kvlen`` (int32 [B], optional): right-padding mask — the
+        reference's attention_mask path (training_manager.py:380-385).
+        Residual branches get counter-RNG dropout in train mode
+        (transformers resid_pdrop/attn_pdrop semantics); dropout sites
+        3i+1..3i+3 keep fwd/bwd masks aligned per layer."""
+        t = self.training
+        i3 = 3 * self.layer_idx
         if pending is None:
             s = x
             h = ops.layer_norm(x, self.ln_1_w, self.ln_1_b)
         else:
             s, h = ops.add_layer_norm(x, pending, self.ln_1_w, self.ln_1_b)
         qkv = ops.linear(h, self.attn_qkv_w, self.attn_qkv_b)
-        o = ops.qkv_attention(qkv, self.n_head)
+        o = ops.qkv_attention(qkv, self.n_head, kvlen=kvlen,
+                              p_drop=self.attn_pdrop if t else 0.0,
+                              site=i3 + 1)
         a = ops.linear(o, self.attn_proj_w, self.attn_proj_b)
+        a = ops.dropout(a, self.resid_pdrop, site=i3 + 2, training=t)
         s2, h2 = ops.add_layer_norm(s, a, self.ln_2_w, self.ln_2_b)
         m = ops.mlp_gelu(h2, self.mlp_fc_w, self.mlp_fc_b,
                          self.mlp_proj_w, self.mlp_proj_b)
+        m = ops.dropout(m, self.resid_pdrop, site=i3 + 3, training=t)
         return s2, m
 
 
@@ -84,7 +99,8 @@ class GPT2LM(nn.Module):
         self.cfg = cfg
         self.wte = nn.Parameter(torch.empty(cfg.vocab_size, cfg.n_embd))
         self.wpe = nn.Parameter(torch.empty(cfg.n_positions, cfg.n_embd))
-        self.blocks = nn.ModuleList(GPT2Block(cfg) for _ in range(cfg.n_layer))
+        self.blocks = nn.ModuleList(GPT2Block(cfg, i)
+                                    for i in range(cfg.n_layer))
         self.ln_f_w = nn.Parameter(torch.ones(cfg.n_embd))
         self.ln_f_b = nn.Parameter(torch.zeros(cfg.n_embd))
         self.reset_parameters()
@@ -105,10 +121,21 @@ class GPT2LM(nn.Module):
                 attention_mask: Optional[torch.Tensor] = None,
                 labels: Optional[torch.Tensor] = None,
                 return_logits: Optional[bool] = None) -> CausalLMOutput:
+        """``attention_mask`` ([B,S] of 1/0, right-padded — the reference's
+        collate, neurons/miner.py:95-99): padded keys are masked out of
+        attention AND padded label positions are ignored in the loss
+        (set to -100 → the CE kernel's ignore_index path). The reference
+        passes the mask but lets pad targets into the loss; ignoring them
+        is the corrected semantics (round-1 verdict item #1)."""
+        kvlen = None
+        if attention_mask is not None:
+            kvlen = attention_mask.to(torch.int32).sum(dim=1).clamp_(min=1)
         x = ops.embedding_fwd(input_ids, self.wte, self.wpe)
+        x = ops.dropout(x, self.cfg.embd_pdrop, site=0,
+                        training=self.training)
         pending = None
         for blk in self.blocks:
-            x, pending = blk(x, pending)
+            x, pending = blk(x, pending, kvlen=kvlen)
         if pending is None:
             x = ops.layer_norm(x, self.ln_f_w, self.ln_f_b)
         else:
@@ -118,7 +145,10 @@ class GPT2LM(nn.Module):
             return CausalLMOutput(loss=None, logits=logits)
         # shifted CE: predict token t+1 from position t (HF semantics)
         logits = ops.linear(x[:, :-1, :].contiguous(), self.wte)
-        tgt = labels[:, 1:].contiguous().view(-1)
+        tgt = labels[:, 1:]
+        if attention_mask is not None:
+            tgt = tgt.masked_fill(attention_mask[:, 1:] == 0, -100)
+        tgt = tgt.contiguous().view(-1)
         loss = ops.cross_entropy_loss(
             logits.reshape(-1, self.cfg.vocab_size), tgt)
         return CausalLMOutput(

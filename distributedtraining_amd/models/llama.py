@@ -50,9 +50,9 @@ class LlamaBlock(nn.Module):
         self.down_w = nn.Parameter(torch.empty(E, I))
 
     def forward(self, x: torch.Tensor, cos: torch.Tensor,
-                sin: torch.Tensor, pending=None):
+                sin: torch.Tensor, pending=None, kvlen=None):
         """Residual-join-fused layout (see gpt2.GPT2Block.forward):
-        returns (stream, pending)."""
+        returns (stream, pending). kvlen: right-padding mask (int32 [B])."""
         B, S, E = x.shape
         D = self.head_dim
         if pending is None:
@@ -66,7 +66,7 @@ class LlamaBlock(nn.Module):
         v = ops.linear(h, self.v_w).view(B, S, self.n_kv, D).transpose(1, 2)
         q = ops.rope(q, cos, sin)
         k = ops.rope(k, cos, sin)
-        o = ops.causal_attention(q, k, v)
+        o = ops.causal_attention(q, k, v, kvlen=kvlen)
         o = o.transpose(1, 2).reshape(B, S, E)
         a = ops.linear(o, self.o_w)
         s2, h2 = ops.add_rms_norm(s, a, self.mlp_norm_w, self.norm_eps)
@@ -114,10 +114,13 @@ class LlamaLM(nn.Module):
         S = input_ids.shape[1]
         cos = self.rope_cos[:S]
         sin = self.rope_sin[:S]
+        kvlen = None
+        if attention_mask is not None:   # right-padded mask → key prefix
+            kvlen = attention_mask.to(torch.int32).sum(dim=1).clamp_(min=1)
         x = ops.embedding_fwd(input_ids, self.tok_emb, None)
         pending = None
         for blk in self.blocks:
-            x, pending = blk(x, cos, sin, pending)
+            x, pending = blk(x, cos, sin, pending, kvlen=kvlen)
         if pending is None:
             x = ops.rms_norm(x, self.final_norm_w, self.cfg.norm_eps)
         else:
@@ -126,7 +129,10 @@ class LlamaLM(nn.Module):
         if labels is None:
             return CausalLMOutput(loss=None, logits=ops.linear(x, self._head()))
         logits = ops.linear(x[:, :-1, :].contiguous(), self._head())
-        tgt = labels[:, 1:].contiguous().view(-1)
+        tgt = labels[:, 1:]
+        if attention_mask is not None:   # pad targets ignored (-100)
+            tgt = tgt.masked_fill(attention_mask[:, 1:] == 0, -100)
+        tgt = tgt.contiguous().view(-1)
         loss = ops.cross_entropy_loss(
             logits.reshape(-1, self.cfg.vocab_size), tgt)
         return CausalLMOutput(

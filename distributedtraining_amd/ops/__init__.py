@@ -26,6 +26,7 @@ from typing import Optional
 import torch
 import torch.nn.functional as F
 
+from . import droprng
 from .backend import require_ext, use_hip
 
 __all__ = [
@@ -37,6 +38,7 @@ __all__ = [
     "decode_attention", "rope", "adamw_step", "delta_sub", "axpy_",
     "weighted_merge",
     "grad_merge_weights", "has_nan", "l2norm",
+    "dropout", "rng_tick",
 ]
 
 
@@ -330,6 +332,51 @@ def rms_norm(x: torch.Tensor, w: torch.Tensor, eps: float = 1e-5) -> torch.Tenso
 
 
 # --------------------------------------------------------------------------
+# Counter-based dropout (transformers GPT-2 trains with pdrop 0.1 —
+# resid/embd/attn; the reference inherits those defaults via
+# from_pretrained, neurons/miner.py:60-62). Masks regenerate from
+# (device counter, site, index) — storage-free, hipGraph-capturable
+# (ops/droprng.py documents the chain). The caller advances the stream
+# once per step via rng_tick().
+# --------------------------------------------------------------------------
+class _DropoutFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, site, p):
+        m = require_ext()
+        c = droprng.counter(x.device)
+        ctx.drop_args = (c, site, p)
+        return m.dropout_apply(x.contiguous(), c, site, p)
+
+    @staticmethod
+    def backward(ctx, dy):
+        # same masked scale; valid while the counter hasn't ticked again
+        # (backward runs within the producing step in every training loop)
+        m = require_ext()
+        c, site, p = ctx.drop_args
+        return m.dropout_apply(dy.contiguous(), c, site, p), None, None
+
+
+def dropout(x: torch.Tensor, p: float, site: int,
+            training: bool = True) -> torch.Tensor:
+    """y = x ⊙ mask / (1-p) with a counter-derived mask; identity when
+    p == 0 or not training. ``site`` decorrelates call sites within one
+    step (fwd and bwd of one call share it)."""
+    if p <= 0.0 or not training:
+        return x
+    if use_hip(x):
+        return _DropoutFn.apply(x, site, p)
+    keep = droprng.elem_keep_mask(x.numel(), droprng.value(x.device),
+                                  site, p)
+    mask = torch.from_numpy(keep.astype("float32")).reshape(x.shape)
+    return x * (mask * droprng.inv_keep(p)).to(x.dtype)
+
+
+def rng_tick(device) -> None:
+    """Advance the dropout RNG one step (capturable on GPU)."""
+    droprng.tick(device)
+
+
+# --------------------------------------------------------------------------
 # GELU (tanh approximation — GPT-2 uses gelu_new)
 # --------------------------------------------------------------------------
 class _GeluFn(torch.autograd.Function):
@@ -387,39 +434,83 @@ def _dense_last(t: torch.Tensor) -> torch.Tensor:
     return t if t.stride(-1) == 1 else t.contiguous()
 
 
+def _empty_kvlen(t: torch.Tensor) -> torch.Tensor:
+    return torch.empty(0, dtype=torch.int32, device=t.device)
+
+
 class _AttnFn(torch.autograd.Function):
     @staticmethod
-    def forward(ctx, q, k, v, scale):
+    def forward(ctx, q, k, v, scale, kvlen, p_drop, site):
         m = require_ext()
         q, k, v = _dense_last(q), _dense_last(k), _dense_last(v)
-        o_bshd, lse = m.attn_fwd(q, k, v, scale)
-        ctx.save_for_backward(q, k, v, o_bshd, lse)
-        ctx.scale = scale
+        kv = kvlen if kvlen is not None else _empty_kvlen(q)
+        rng = droprng.counter(q.device)
+        o_bshd, lse = m.attn_fwd(q, k, v, scale, kv, rng, site, p_drop)
+        ctx.save_for_backward(q, k, v, o_bshd, lse, kv, rng)
+        ctx.attn_args = (scale, site, p_drop)
         return o_bshd.permute(0, 2, 1, 3)  # [B,H,S,D] view, zero-copy
 
     @staticmethod
     def backward(ctx, do):
         m = require_ext()
-        q, k, v, o_bshd, lse = ctx.saved_tensors
+        q, k, v, o_bshd, lse, kv, rng = ctx.saved_tensors
+        scale, site, p_drop = ctx.attn_args
         dq_bshd, dk, dv = m.attn_bwd(_dense_last(do), q, k, v, o_bshd, lse,
-                                     ctx.scale)
-        return dq_bshd.permute(0, 2, 1, 3), dk, dv, None
+                                     scale, kv, rng, site, p_drop)
+        return (dq_bshd.permute(0, 2, 1, 3), dk, dv, None, None, None,
+                None)
+
+
+def _attn_ref(q, k, v, scale, kvlen, p_drop, site):
+    """CPU gold: explicit masked softmax attention with the SAME padding
+    and dropout semantics the HIP kernels implement (mask from the shared
+    RNG chain, so GPU tests compare bit-identical masks). fp32 math;
+    differentiable through plain torch ops."""
+    B, H, S, D = q.shape
+    Hk = k.shape[1]
+    kf, vf = k.float(), v.float()
+    if Hk != H:
+        rep = H // Hk
+        kf = kf.repeat_interleave(rep, dim=1)
+        vf = vf.repeat_interleave(rep, dim=1)
+    scores = torch.einsum("bhqd,bhkd->bhqk", q.float(), kf) * scale
+    idx = torch.arange(S)
+    allowed = (idx[None, :] <= idx[:, None])[None, None]   # causal [1,1,q,k]
+    if kvlen is not None:
+        allowed = allowed & (idx[None, None, None, :]
+                             < kvlen.view(B, 1, 1, 1).to(torch.long))
+    A = torch.softmax(scores.masked_fill(~allowed, float("-inf")), dim=-1)
+    if p_drop > 0.0:
+        keep = droprng.attn_keep_mask(B * H, S, S,
+                                      droprng.value(q.device), site, p_drop)
+        A = A * (torch.from_numpy(keep.astype("float32")).view(B, H, S, S)
+                 * droprng.inv_keep(p_drop))
+    return torch.einsum("bhqk,bhkd->bhqd", A, vf).to(q.dtype)
 
 
 def causal_attention(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
-                     scale: Optional[float] = None) -> torch.Tensor:
+                     scale: Optional[float] = None,
+                     kvlen: Optional[torch.Tensor] = None,
+                     p_drop: float = 0.0, site: int = 0) -> torch.Tensor:
     """softmax(q kᵀ · scale + causal_mask) v over [B, H, S, D] tensors.
 
     GQA: k/v may have fewer heads (H_kv dividing H) — handled natively by
     the kernel (no repeat_interleave). Inputs may be permuted views of
     [B,S,H*D] projections; only the head_dim must be contiguous.
+
+    ``kvlen`` (int32 [B]): right-padding mask — key j of row b is attended
+    iff j < kvlen[b] (the reference's attention_mask semantics,
+    training_manager.py:380-385). ``p_drop``/``site``: attention-prob
+    dropout from the counter RNG (transformers attn_pdrop).
     """
     if scale is None:
         scale = 1.0 / math.sqrt(q.shape[-1])
     if use_hip(q):
-        return _AttnFn.apply(q, k, v, scale)
-    return F.scaled_dot_product_attention(q, k, v, is_causal=True,
-                                          scale=scale, enable_gqa=True)
+        return _AttnFn.apply(q, k, v, scale, kvlen, p_drop, site)
+    if kvlen is None and p_drop <= 0.0:
+        return F.scaled_dot_product_attention(q, k, v, is_causal=True,
+                                              scale=scale, enable_gqa=True)
+    return _attn_ref(q, k, v, scale, kvlen, p_drop, site)
 
 
 # --------------------------------------------------------------------------
@@ -441,21 +532,24 @@ def _qkv_views(t: torch.Tensor, H: int, Hk: int, D: int):
 
 class _QKVAttnFn(torch.autograd.Function):
     @staticmethod
-    def forward(ctx, qkv, H, Hk, scale):
+    def forward(ctx, qkv, H, Hk, scale, kvlen, p_drop, site):
         m = require_ext()
         B, S, Fdim = qkv.shape
         D = Fdim // (H + 2 * Hk)
         q, k, v = _qkv_views(qkv, H, Hk, D)
-        o_bshd, lse = m.attn_fwd(q, k, v, scale)   # [B,S,H,D] contiguous
-        ctx.save_for_backward(qkv, o_bshd, lse)
-        ctx.geom = (H, Hk, D, scale)
+        kv = kvlen if kvlen is not None else _empty_kvlen(qkv)
+        rng = droprng.counter(qkv.device)
+        o_bshd, lse = m.attn_fwd(q, k, v, scale, kv, rng, site,
+                                 p_drop)   # [B,S,H,D] contiguous
+        ctx.save_for_backward(qkv, o_bshd, lse, kv, rng)
+        ctx.geom = (H, Hk, D, scale, site, p_drop)
         return o_bshd.view(B, S, H * D)
 
     @staticmethod
     def backward(ctx, do):
         m = require_ext()
-        qkv, o_bshd, lse = ctx.saved_tensors
-        H, Hk, D, scale = ctx.geom
+        qkv, o_bshd, lse, kv, rng = ctx.saved_tensors
+        H, Hk, D, scale, site, p_drop = ctx.geom
         B, S, Fdim = qkv.shape
         q, k, v = _qkv_views(qkv, H, Hk, D)
         do4 = do.view(B, S, H, D).permute(0, 2, 1, 3)
@@ -468,29 +562,36 @@ class _QKVAttnFn(torch.autograd.Function):
         dk_v = dqkv.as_strided((B, S, Hk, D), (sb, ss, D, 1), o0 + H * D)
         dv_v = dqkv.as_strided((B, S, Hk, D), (sb, ss, D, 1),
                                o0 + (H + Hk) * D)
-        m.attn_bwd_packed(do4, q, k, v, o_bshd, lse, scale, dq_v, dk_v, dv_v)
-        return dqkv, None, None, None
+        m.attn_bwd_packed(do4, q, k, v, o_bshd, lse, scale, dq_v, dk_v,
+                          dv_v, kv, rng, site, p_drop)
+        return dqkv, None, None, None, None, None, None
 
 
 def qkv_attention(qkv: torch.Tensor, n_head: int,
                   n_kv_head: Optional[int] = None,
-                  scale: Optional[float] = None) -> torch.Tensor:
+                  scale: Optional[float] = None,
+                  kvlen: Optional[torch.Tensor] = None,
+                  p_drop: float = 0.0, site: int = 0) -> torch.Tensor:
     """Causal attention over a packed qkv projection [B,S,(H+2Hk)·D] →
-    [B,S,H·D]. GQA when n_kv_head < n_head."""
+    [B,S,H·D]. GQA when n_kv_head < n_head. kvlen/p_drop/site as in
+    :func:`causal_attention`."""
     Hk = n_kv_head or n_head
     B, S, Fdim = qkv.shape
     D = Fdim // (n_head + 2 * Hk)
     if scale is None:
         scale = 1.0 / math.sqrt(D)
     if use_hip(qkv):
-        return _QKVAttnFn.apply(qkv, n_head, Hk, scale)
+        return _QKVAttnFn.apply(qkv, n_head, Hk, scale, kvlen, p_drop, site)
     E = n_head * D
     kvd = Hk * D
     q = qkv[..., :E].view(B, S, n_head, D).transpose(1, 2)
     k = qkv[..., E:E + kvd].view(B, S, Hk, D).transpose(1, 2)
     v = qkv[..., E + kvd:].view(B, S, Hk, D).transpose(1, 2)
-    o = F.scaled_dot_product_attention(q, k, v, is_causal=True, scale=scale,
-                                       enable_gqa=True)
+    if kvlen is None and p_drop <= 0.0:
+        o = F.scaled_dot_product_attention(q, k, v, is_causal=True,
+                                           scale=scale, enable_gqa=True)
+    else:
+        o = _attn_ref(q, k, v, scale, kvlen, p_drop, site)
     return o.transpose(1, 2).reshape(B, S, E)
 
 

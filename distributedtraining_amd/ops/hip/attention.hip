@@ -156,6 +156,13 @@ __global__ void attn_fwd_k(const ushort* __restrict__ q,
   const ushort* kp = k + b * geo.kb + hk * geo.kh;
   const ushort* vp = v + b * geo.vb + hk * geo.vh;
   const float sc2 = geo.scale * LOG2E;  // fold scale into the exp2 argument
+  // padded batches: keys >= kvl are masked for every query (block-uniform)
+  const int kvl = geo.kvlen ? min(seq, geo.kvlen[b]) : seq;
+  const uint32_t thr = geo.thr16;
+  uint64_t drop_s2 = 0;
+  if (thr)
+    drop_s2 = sm64(sm64(*geo.rng + geo.site * DTA_RNG_SITE_K) ^
+                   (uint64_t(bhid) * DTA_RNG_HEAD_K));
 
   bf16x8 qb_[DTILES / 2];
   const int qrow = q0 + (lane & 15);
@@ -172,7 +179,9 @@ __global__ void attn_fwd_k(const ushort* __restrict__ q,
   __shared__ ushort k_rm[32 * (16 * DTILES + 8)];
   __shared__ ushort v_t[16 * DTILES * LDS_RP];
   const int g = lane >> 4;
-  const int blk_kv_end = min(seq, int(blockIdx.x) * 64 + 64);  // uniform
+  // loop + mask bounds honor the pad length (kvl == seq when unmasked)
+  const int blk_kv_end =
+      min(min(seq, int(blockIdx.x) * 64 + 64), (kvl + 31) & ~31);  // uniform
   const int my_kv_end = live ? min(seq, q0 + 16) : 0;
   for (int kv0 = 0; kv0 < blk_kv_end; kv0 += 32) {
     __syncthreads();   // previous iteration's LDS reads complete
@@ -193,8 +202,8 @@ __global__ void attn_fwd_k(const ushort* __restrict__ q,
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
       const int k0a = kv0 + 4 * g + r, k1a = k0a + 16;
-      p0[r] = (k0a <= qrow && k0a < seq) ? p0[r] * sc2 : -INFINITY;
-      p1[r] = (k1a <= qrow && k1a < seq) ? p1[r] * sc2 : -INFINITY;
+      p0[r] = (k0a <= qrow && k0a < kvl) ? p0[r] * sc2 : -INFINITY;
+      p1[r] = (k1a <= qrow && k1a < kvl) ? p1[r] * sc2 : -INFINITY;
       mx = fmaxf(mx, fmaxf(p0[r], p1[r]));
     }
     mx = fmaxf(mx, __shfl_xor(mx, 16));
@@ -214,6 +223,24 @@ __global__ void attn_fwd_k(const ushort* __restrict__ q,
     l_run = l_run * alpha + psum;
     m_run = m_new;
 
+    if (thr) {
+      // dropout on the PV accumulation only: l keeps the undropped sum,
+      // so O = dropout(P) V with P the true softmax. One hash covers the
+      // 4 consecutive keys this lane holds per half (k & 3 == r).
+      const uint64_t hq = drop_s2 + (uint64_t(uint32_t(qrow)) << 24) *
+                                        DTA_RNG_IDX_K;
+      const uint64_t h0 =
+          sm64(hq + uint64_t(uint32_t((kv0 + 4 * g) >> 2)) * DTA_RNG_IDX_K);
+      const uint64_t h1 = sm64(
+          hq + uint64_t(uint32_t((kv0 + 16 + 4 * g) >> 2)) * DTA_RNG_IDX_K);
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        p0[r] = (uint32_t(h0 >> (16 * r)) & 0xFFFF) >= thr
+                    ? p0[r] * geo.inv_keep : 0.f;
+        p1[r] = (uint32_t(h1 >> (16 * r)) & 0xFFFF) >= thr
+                    ? p1[r] * geo.inv_keep : 0.f;
+      }
+    }
     bf16x8 pa = scores_to_afrag(p0, p1, lane);
     float a_r[4];
 #pragma unroll
@@ -279,6 +306,12 @@ __global__ void attn_bwd_dq_k(const ushort* __restrict__ dout,
   const ushort* vp = v + b * geo.vb + hk * geo.vh;
   const ushort* dop = dout + b * geo.db_ + h * geo.dh;
   const ushort* op2 = o + b * ob2 + h * oh2;
+  const int kvl = geo.kvlen ? min(seq, geo.kvlen[b]) : seq;
+  const uint32_t thr = geo.thr16;
+  uint64_t drop_s2 = 0;
+  if (thr)
+    drop_s2 = sm64(sm64(*geo.rng + geo.site * DTA_RNG_SITE_K) ^
+                   (uint64_t(bhid) * DTA_RNG_HEAD_K));
 
   const int qrow = q0 + (lane & 15);
   const int qr_ld = qrow < seq ? qrow : seq - 1;
@@ -311,7 +344,8 @@ __global__ void attn_bwd_dq_k(const ushort* __restrict__ dout,
   __shared__ ushort v_rm[32 * (16 * DTILES + 8)];
   __shared__ ushort k_t[16 * DTILES * LDS_RP];
   const int g = lane >> 4;
-  const int blk_kv_end = min(seq, int(blockIdx.x) * 64 + 64);  // uniform
+  const int blk_kv_end =
+      min(min(seq, int(blockIdx.x) * 64 + 64), (kvl + 31) & ~31);  // uniform
   const int my_kv_end = live ? min(seq, q0 + 16) : 0;
   for (int kv0 = 0; kv0 < blk_kv_end; kv0 += 32) {
     __syncthreads();
@@ -334,15 +368,32 @@ __global__ void attn_bwd_dq_k(const ushort* __restrict__ dout,
       dp0 = mfma_bf16(va, dob[sl], dp0);
       dp1 = mfma_bf16(vb2, dob[sl], dp1);
     }
+    uint64_t h0 = 0, h1 = 0;
+    if (thr) {
+      const uint64_t hq =
+          drop_s2 + (uint64_t(uint32_t(qrow)) << 24) * DTA_RNG_IDX_K;
+      h0 = sm64(hq + uint64_t(uint32_t((kv0 + 4 * g) >> 2)) * DTA_RNG_IDX_K);
+      h1 = sm64(hq + uint64_t(uint32_t((kv0 + 16 + 4 * g) >> 2)) *
+                         DTA_RNG_IDX_K);
+    }
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
       const int k0a = kv0 + 4 * g + r, k1a = k0a + 16;
-      const bool v0 = (k0a <= qrow && k0a < seq);
-      const bool v1 = (k1a <= qrow && k1a < seq);
+      const bool v0 = (k0a <= qrow && k0a < kvl);
+      const bool v1 = (k1a <= qrow && k1a < kvl);
       const float P0 = v0 ? __builtin_exp2f(s0[r] * sc2 - lse_q) : 0.f;
       const float P1 = v1 ? __builtin_exp2f(s1[r] * sc2 - lse_q) : 0.f;
-      s0[r] = P0 * (dp0[r] - dlt_q);
-      s1[r] = P1 * (dp1[r] - dlt_q);
+      // dS = P ⊙ (M/(1-p) ⊙ (dO Vᵀ) − delta): the dropout mask gates the
+      // dO·V term only; delta = rowsum(dO ⊙ O) already absorbs the mask
+      float g0 = dp0[r], g1 = dp1[r];
+      if (thr) {
+        g0 = (uint32_t(h0 >> (16 * r)) & 0xFFFF) >= thr
+                 ? g0 * geo.inv_keep : 0.f;
+        g1 = (uint32_t(h1 >> (16 * r)) & 0xFFFF) >= thr
+                 ? g1 * geo.inv_keep : 0.f;
+      }
+      s0[r] = P0 * (g0 - dlt_q);
+      s1[r] = P1 * (g1 - dlt_q);
     }
     bf16x8 dsa = scores_to_afrag(s0, s1, lane);
 #pragma unroll
@@ -397,6 +448,12 @@ __global__ void attn_bwd_dkv_k(const ushort* __restrict__ dout,
   const ushort* vp = v + b * geo.vb + hk * geo.vh;
   const ushort* dop = dout + b * geo.db_ + h * geo.dh;
   const float sc2 = geo.scale * LOG2E;
+  const int kvl = geo.kvlen ? min(seq, geo.kvlen[b]) : seq;
+  const uint32_t thr = geo.thr16;
+  uint64_t drop_s2 = 0;
+  if (thr)
+    drop_s2 = sm64(sm64(*geo.rng + geo.site * DTA_RNG_SITE_K) ^
+                   (uint64_t(bhid) * DTA_RNG_HEAD_K));
 
   const int krow = kv0 + (lane & 15);
   const int kr_ld = krow < seq ? krow : seq - 1;
@@ -419,7 +476,11 @@ __global__ void attn_bwd_dkv_k(const ushort* __restrict__ dout,
   __shared__ ushort q_t[16 * DTILES * LDS_RP];
   __shared__ ushort d_t[16 * DTILES * LDS_RP];
   const int g = lane >> 4;
-  const int q_start = int(blockIdx.x) * 64;   // uniform across the block
+  // a block whose 64 key rows are all padding has zero dk/dv: skip the
+  // whole walk (uniform) and fall through to the zero-initialized stores
+  const int q_start =
+      (int(blockIdx.x) * 64 >= kvl) ? seq : int(blockIdx.x) * 64;
+  const bool kv_ok = krow < kvl;
   for (int q0 = q_start; q0 < seq; q0 += 32) {
     __syncthreads();
     stage_tile_rm<DTILES>(q_rm, qp, geo.qs, q0, seq, tid);
@@ -446,16 +507,38 @@ __global__ void attn_bwd_dkv_k(const ushort* __restrict__ dout,
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
       const int qa0 = q0 + 4 * g + r, qa1 = qa0 + 16;
-      const bool v0 = (qa0 >= krow && qa0 < seq);
-      const bool v1 = (qa1 >= krow && qa1 < seq);
+      const bool v0 = (qa0 >= krow && qa0 < seq && kv_ok);
+      const bool v1 = (qa1 >= krow && qa1 < seq && kv_ok);
       const float l0 = lse[bhid * seq + (v0 ? qa0 : 0)] * LOG2E;
       const float l1 = lse[bhid * seq + (v1 ? qa1 : 0)] * LOG2E;
       p0[r] = v0 ? __builtin_exp2f(s0[r] * sc2 - l0) : 0.f;
       p1[r] = v1 ? __builtin_exp2f(s1[r] * sc2 - l1) : 0.f;
       const float d0 = delta[bhid * seq + ((qa0 < seq) ? qa0 : 0)];
       const float d1 = delta[bhid * seq + ((qa1 < seq) ? qa1 : 0)];
-      ds0[r] = p0[r] * (dp0[r] - d0);
-      ds1[r] = p1[r] * (dp1[r] - d1);
+      if (thr) {
+        // element (q, k=krow): slice krow&3 of hash(q, krow>>2) — the
+        // transposed view of the SAME per-(b,h,q,k) draw the fwd/dq
+        // kernels take; dV uses the dropped P, dK the dropout-gated dS
+        const uint64_t hk0 =
+            sm64(drop_s2 + ((uint64_t(uint32_t(qa0)) << 24) |
+                            uint64_t(uint32_t(krow) >> 2)) * DTA_RNG_IDX_K);
+        const uint64_t hk1 =
+            sm64(drop_s2 + ((uint64_t(uint32_t(qa1)) << 24) |
+                            uint64_t(uint32_t(krow) >> 2)) * DTA_RNG_IDX_K);
+        const int sl = 16 * (krow & 3);
+        const float m0 = (uint32_t(hk0 >> sl) & 0xFFFF) >= thr
+                             ? geo.inv_keep : 0.f;
+        const float m1 = (uint32_t(hk1 >> sl) & 0xFFFF) >= thr
+                             ? geo.inv_keep : 0.f;
+        // dS = A ⊙ (M/(1-p)·dp − delta); dV gets the dropped P = A·M/(1-p)
+        ds0[r] = p0[r] * (m0 * dp0[r] - d0);
+        ds1[r] = p1[r] * (m1 * dp1[r] - d1);
+        p0[r] *= m0;
+        p1[r] *= m1;
+      } else {
+        ds0[r] = p0[r] * (dp0[r] - d0);
+        ds1[r] = p1[r] * (dp1[r] - d1);
+      }
     }
     bf16x8 pa = scores_to_afrag(p0, p1, lane);
     bf16x8 dsa = scores_to_afrag(ds0, ds1, lane);

@@ -3,6 +3,7 @@
 #include <torch/extension.h>
 #include <ATen/hip/HIPContext.h>
 
+#include <cmath>
 #include <vector>
 
 #include "dta_kernels.h"
@@ -148,6 +149,32 @@ Tensor swiglu_fwd(Tensor g, Tensor u) {
   launch_swiglu_fwd(bfp(g), bfp(u), bfp_mut(y), g.numel(), stream());
   return y;
 }
+// counter-based dropout: y = x ⊙ mask / keep. The same call with the same
+// (rng value, site) regenerates the identical mask — backward is this very
+// function applied to dy (no stored mask).
+Tensor dropout_apply(Tensor x, Tensor rng, int64_t site, double p) {
+  check_bf16(x, "x");
+  TORCH_CHECK(p > 0.0 && p < 1.0, "dropout p must be in (0,1)");
+  TORCH_CHECK(rng.numel() == 1 && rng.scalar_type() == torch::kInt64 &&
+                  rng.is_cuda(), "rng must be an int64 [1] GPU counter");
+  unsigned int thr = (unsigned int)std::ceil(p * 65536.0);
+  if (thr > 65535u) thr = 65535u;
+  const float inv_keep = float(65536.0 / (65536.0 - double(thr)));
+  auto y = torch::empty_like(x);
+  launch_dropout(bfp(x), bfp_mut(y), x.numel(),
+                 reinterpret_cast<const unsigned long long*>(
+                     rng.data_ptr<int64_t>()),
+                 (unsigned long long)site, thr, inv_keep, stream());
+  return y;
+}
+
+void rng_tick(Tensor ctr) {
+  TORCH_CHECK(ctr.numel() == 1 && ctr.scalar_type() == torch::kInt64 &&
+                  ctr.is_cuda(), "ctr must be an int64 [1] GPU counter");
+  launch_rng_tick(reinterpret_cast<unsigned long long*>(
+                      ctr.data_ptr<int64_t>()), stream());
+}
+
 std::vector<Tensor> swiglu_bwd(Tensor dy, Tensor g, Tensor u) {
   check_bf16(dy, "dy");
   auto dg = torch::empty_like(g);
@@ -369,6 +396,33 @@ static void check_attn_view(const Tensor& t, const char* n) {
               " must be a [B,H,S,D] view with contiguous D");
 }
 
+// kvlen: int32 [B] valid-key prefix per batch row (empty = no mask);
+// rng: int64 [1] device step counter + site/p: attention-prob dropout
+// (p == 0 disables). thr16 quantizes p to 1/65536 steps; inv_keep uses the
+// REALIZED keep probability so the expectation stays unbiased.
+static void set_mask_drop(AttnGeom& g, const Tensor& kvlen, const Tensor& rng,
+                          int64_t site, double p, int64_t B) {
+  if (kvlen.numel() > 0) {
+    TORCH_CHECK(kvlen.scalar_type() == torch::kInt32 && kvlen.is_cuda() &&
+                    kvlen.is_contiguous() && kvlen.numel() == B,
+                "kvlen must be a contiguous int32 [B] GPU tensor");
+    g.kvlen = kvlen.data_ptr<int>();
+  }
+  if (p > 0.0) {
+    TORCH_CHECK(p < 1.0, "dropout p must be < 1");
+    TORCH_CHECK(rng.numel() == 1 && rng.scalar_type() == torch::kInt64 &&
+                    rng.is_cuda(),
+                "rng must be an int64 [1] GPU counter");
+    g.rng = reinterpret_cast<const unsigned long long*>(
+        rng.data_ptr<int64_t>());
+    g.site = (unsigned long long)site;
+    unsigned int thr = (unsigned int)std::ceil(p * 65536.0);
+    if (thr > 65535u) thr = 65535u;
+    g.thr16 = thr;
+    g.inv_keep = float(65536.0 / (65536.0 - double(thr)));
+  }
+}
+
 static AttnGeom make_geom(const Tensor& q, const Tensor& k, const Tensor& v,
                           const Tensor& o_bshd, const Tensor& dout,
                           double scale) {
@@ -390,7 +444,9 @@ static AttnGeom make_geom(const Tensor& q, const Tensor& k, const Tensor& v,
   return g;
 }
 
-std::vector<Tensor> attn_fwd(Tensor q, Tensor k, Tensor v, double scale) {
+std::vector<Tensor> attn_fwd(Tensor q, Tensor k, Tensor v, double scale,
+                             Tensor kvlen, Tensor rng, int64_t site,
+                             double p) {
   check_attn_view(q, "q"); check_attn_view(k, "k"); check_attn_view(v, "v");
   const int hd = int(q.size(3));
   TORCH_CHECK(hd == 32 || hd == 64 || hd == 128, "head dim must be 32/64/128");
@@ -398,18 +454,22 @@ std::vector<Tensor> attn_fwd(Tensor q, Tensor k, Tensor v, double scale) {
   auto o = torch::empty({B, S, H, hd}, q.options());
   auto lse = torch::empty({B * H, S}, q.options().dtype(torch::kFloat32));
   AttnGeom geo = make_geom(q, k, v, o, Tensor(), scale);
+  set_mask_drop(geo, kvlen, rng, site, p, B);
   launch_attn_fwd(bfp(q), bfp(k), bfp(v), bfp_mut(o),
                   lse.data_ptr<float>(), geo, stream());
   return {o, lse};  // caller views o as [B,H,S,D] via permute(0,2,1,3)
 }
 
 std::vector<Tensor> attn_bwd(Tensor dout, Tensor q, Tensor k, Tensor v,
-                             Tensor o_bshd, Tensor lse, double scale) {
+                             Tensor o_bshd, Tensor lse, double scale,
+                             Tensor kvlen, Tensor rng, int64_t site,
+                             double p) {
   check_attn_view(dout, "dout");
   const int64_t B = q.size(0), H = q.size(1), S = q.size(2);
   const int hd = int(q.size(3));
   const int64_t Hk = k.size(1);
   AttnGeom geo = make_geom(q, k, v, o_bshd, dout, scale);
+  set_mask_drop(geo, kvlen, rng, site, p, B);
   auto delta = torch::empty({B * H, S}, lse.options());
   auto dq = torch::empty({B, S, H, hd}, q.options());
   launch_attn_bwd_dq(bfp(dout), bfp(q), bfp(k), bfp(v), bfp(o_bshd),
@@ -443,7 +503,8 @@ std::vector<Tensor> attn_bwd(Tensor dout, Tensor q, Tensor k, Tensor v,
 // contiguous) — eliminates the split-backward cat and all transpose copies.
 void attn_bwd_packed(Tensor dout, Tensor q, Tensor k, Tensor v,
                      Tensor o_bshd, Tensor lse, double scale, Tensor dq_v,
-                     Tensor dk_v, Tensor dv_v) {
+                     Tensor dk_v, Tensor dv_v, Tensor kvlen, Tensor rng,
+                     int64_t site, double p) {
   check_attn_view(dout, "dout");
   TORCH_CHECK(dq_v.stride(3) == 1 && dk_v.stride(3) == 1 &&
               dv_v.stride(3) == 1, "dqkv views need contiguous head_dim");
@@ -451,6 +512,7 @@ void attn_bwd_packed(Tensor dout, Tensor q, Tensor k, Tensor v,
   const int hd = int(q.size(3));
   const int64_t Hk = k.size(1);
   AttnGeom geo = make_geom(q, k, v, o_bshd, dout, scale);
+  set_mask_drop(geo, kvlen, rng, site, p, B);
   auto delta = torch::empty({B * H, S}, lse.options());
   AttnGeom gq = geo;  // dq_v is [B,S,H,D]: (batch, head, seq) strides
   gq.ob = dq_v.stride(0); gq.oh = dq_v.stride(2); gq.os_ = dq_v.stride(1);
@@ -524,6 +586,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("gelu_bwd", &gelu_bwd);
   m.def("swiglu_fwd", &swiglu_fwd);
   m.def("swiglu_bwd", &swiglu_bwd);
+  m.def("dropout_apply", &dropout_apply);
+  m.def("rng_tick", &rng_tick);
   m.def("delta_sub", &delta_sub);
   m.def("axpy", &axpy);
   m.def("has_nan", &has_nan);

@@ -96,6 +96,32 @@ DEV float block_max(float v, float* lds_scratch) {
   return r;
 }
 
+// ---- counter-based dropout RNG (splitmix64 finalizer) ---------------------
+// Graph-capturable, storage-free dropout: masks are a pure function of a
+// device-resident step counter, a per-call-site salt and the element
+// coordinates, so forward and both backward kernels regenerate identical
+// masks with zero mask memory, and a hipGraph replay gets fresh masks from
+// the ticked counter. One 64-bit hash yields FOUR 16-bit keep-draws
+// (elements 4j..4j+3 of the innermost axis). The exact chain (mirrored
+// bit-for-bit by the host gold in ops/droprng.py):
+//   s1 = sm64(*ctr + site * 0xA24BAED4963EE407)
+//   attention (per b,h head): s2 = sm64(s1 ^ (bh * 0x9E3779B97F4A7C15))
+//       h  = sm64(s2 + ((q << 24) | (k >> 2)) * 0xD1B54A32D192ED03)
+//       draw = (h >> (16 * (k & 3))) & 0xFFFF
+//   elementwise: h = sm64(s1 + (i >> 2) * 0xD1B54A32D192ED03)
+//       draw = (h >> (16 * (i & 3))) & 0xFFFF
+// keep iff draw >= thr16, thr16 = ceil(p * 65536) => the realized keep
+// probability is exactly 1 - thr16/65536 and scale = 65536/(65536-thr16)
+// keeps the expectation unbiased.
+DEV uint64_t sm64(uint64_t z) {
+  z = (z ^ (z >> 30)) * 0xBF58476D1CE4E5B9ULL;
+  z = (z ^ (z >> 27)) * 0x94D049BB133111EBULL;
+  return z ^ (z >> 31);
+}
+#define DTA_RNG_SITE_K 0xA24BAED4963EE407ULL
+#define DTA_RNG_HEAD_K 0x9E3779B97F4A7C15ULL
+#define DTA_RNG_IDX_K 0xD1B54A32D192ED03ULL
+
 // ---- grid sizing (Guideline 11: cap + grid-stride for memory-bound) -------
 constexpr int MAX_RESIDENT_BLOCKS = 2048;  // 256 CU x 8 blocks
 inline int elementwise_grid(int64_t n, int block, int vec) {

@@ -18,6 +18,12 @@ void launch_swiglu_fwd(const bf16_t* g, const bf16_t* u, bf16_t* y, int64_t n,
                        hipStream_t s);
 void launch_swiglu_bwd(const bf16_t* dy, const bf16_t* g, const bf16_t* u,
                        bf16_t* dg, bf16_t* du, int64_t n, hipStream_t s);
+// counter-based dropout (fwd and bwd are the same masked scale; the mask
+// is regenerated, never stored). rng: device uint64 step counter.
+void launch_dropout(const bf16_t* x, bf16_t* y, int64_t n,
+                    const unsigned long long* rng, unsigned long long site,
+                    unsigned int thr16, float inv_keep, hipStream_t s);
+void launch_rng_tick(unsigned long long* ctr, hipStream_t s);
 
 // ---- flat-plane ops -------------------------------------------------------
 void launch_delta_sub(const float* w, const float* base, float* out,
@@ -131,6 +137,16 @@ struct AttnGeom {
   int64_t ob, oh, os_;     // o (and dq) strides
   int64_t db_, dh, ds;     // dout strides
   int64_t gkb, gkh, gks;   // dk/dv output strides (grp==1 direct-store path)
+  // padding mask (reference attention_mask, right-padded batches:
+  // training_manager.py:380-385): key j of batch row b is valid iff
+  // j < kvlen[b]. null = no mask (all keys valid).
+  const int* kvlen = nullptr;
+  // attention-probability dropout (transformers GPT-2 attn_pdrop):
+  // thr16 == 0 disables; rng = device step counter (dta_common.h RNG).
+  const unsigned long long* rng = nullptr;
+  unsigned long long site = 0;
+  unsigned int thr16 = 0;
+  float inv_keep = 1.0f;
 };
 void launch_attn_fwd(const bf16_t* q, const bf16_t* k, const bf16_t* v,
                      bf16_t* o, float* lse, const AttnGeom& geo,

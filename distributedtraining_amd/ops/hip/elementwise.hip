@@ -256,6 +256,42 @@ __global__ void colsum_reduce_k(const float* __restrict__ part,
   for (int j = 0; j < 4; ++j) atomicAdd(out + c4 + j, acc[j]);
 }
 
+// -- counter-based dropout (residual/embedding paths; transformers GPT-2
+// resid_pdrop/embd_pdrop). One kernel serves fwd (x) and bwd (dy): both
+// are y = x * mask * inv_keep with the mask regenerated from
+// (counter, site, index) — dta_common.h RNG chain. 8 elements per 16 B
+// vector = two sm64 hashes (4 draws each).
+__global__ void dropout_k(const ushort* __restrict__ x,
+                          ushort* __restrict__ y, int64_t n,
+                          const unsigned long long* __restrict__ rng,
+                          unsigned long long site, unsigned int thr16,
+                          float inv_keep) {
+  const uint64_t s1 = sm64(*rng + site * DTA_RNG_SITE_K);
+  int64_t i = (int64_t(blockIdx.x) * blockDim.x + threadIdx.x) * 8;
+  const int64_t stride = int64_t(gridDim.x) * blockDim.x * 8;
+  for (; i + 8 <= n; i += stride) {
+    s16x8 vx = *reinterpret_cast<const s16x8*>(x + i);
+    const uint64_t h0 = sm64(s1 + uint64_t(i >> 2) * DTA_RNG_IDX_K);
+    const uint64_t h1 = sm64(s1 + uint64_t((i >> 2) + 1) * DTA_RNG_IDX_K);
+    s16x8 o;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      const uint64_t h = (j < 4) ? h0 : h1;
+      const bool keep = (uint32_t(h >> (16 * (j & 3))) & 0xFFFF) >= thr16;
+      o[j] = keep ? f2bf(bf2f(ushort(vx[j])) * inv_keep) : ushort(0);
+    }
+    *reinterpret_cast<s16x8*>(y + i) = o;
+  }
+  if (i < n && i + 8 > n)
+    for (; i < n; ++i) {
+      const uint64_t h = sm64(s1 + uint64_t(i >> 2) * DTA_RNG_IDX_K);
+      const bool keep = (uint32_t(h >> (16 * (i & 3))) & 0xFFFF) >= thr16;
+      y[i] = keep ? f2bf(bf2f(x[i]) * inv_keep) : ushort(0);
+    }
+}
+
+__global__ void rng_tick_k(unsigned long long* ctr) { ++(*ctr); }
+
 }  // namespace
 
 #define LAUNCH_EW(kernel, n, ...)                                          \
@@ -311,4 +347,12 @@ void launch_colsum(const bf16_t* x, float* part, float* out, int64_t rows,
   const int g2 = (cols / 4 + 255) / 256;
   const int ry = stripes < 32 ? stripes : 32;
   colsum_reduce_k<<<dim3(g2, ry), 256, 0, s>>>(part, out, stripes, cols);
+}
+void launch_dropout(const bf16_t* x, bf16_t* y, int64_t n,
+                    const unsigned long long* rng, unsigned long long site,
+                    unsigned int thr16, float inv_keep, hipStream_t s) {
+  LAUNCH_EW(dropout_k, n, x, y, n, rng, site, thr16, inv_keep);
+}
+void launch_rng_tick(unsigned long long* ctr, hipStream_t s) {
+  rng_tick_k<<<1, 1, 0, s>>>(ctr);
 }

@@ -47,6 +47,10 @@ class GraphedMinerStep:
         self.labels = (self.ids if lbl is b0["input_ids"]
                        else lbl.to(dev).clone())
         self._static = {"input_ids": self.ids, "labels": self.labels}
+        self.mask = None
+        if b0.get("attention_mask") is not None:
+            self.mask = b0["attention_mask"].to(dev).clone()
+            self._static["attention_mask"] = self.mask
 
         # warmup on a side stream (torch requirement before capture)
         s = torch.cuda.Stream()
@@ -71,6 +75,8 @@ class GraphedMinerStep:
         lbl = batch.get("labels", batch["input_ids"])
         if self.labels is not self.ids:
             self.labels.copy_(lbl, non_blocking=True)
+        if self.mask is not None:
+            self.mask.copy_(batch["attention_mask"], non_blocking=True)
         self.graph.replay()
         self.miner.step_count += 1
         self.miner.total_examples += self.batch_size

@@ -88,10 +88,16 @@ class DeltaLoop:
         see parallel/graphstep.py)."""
         if batch is None:
             batch = next(self.data)
+        from .. import ops
+        ops.rng_tick(self.fp.device)   # advance the dropout stream
         input_ids = batch["input_ids"].to(self.fp.device, non_blocking=True)
         labels = batch.get("labels", batch["input_ids"]).to(
             self.fp.device, non_blocking=True)
-        out = self.model(input_ids=input_ids, labels=labels)
+        am = batch.get("attention_mask")
+        if am is not None:
+            am = am.to(self.fp.device, non_blocking=True)
+        out = self.model(input_ids=input_ids, attention_mask=am,
+                         labels=labels)
         out.loss.backward()
         self.opt.step()
         self._on_grad_available()   # subclass hook (gradient protocol)

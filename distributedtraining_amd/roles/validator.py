@@ -57,7 +57,10 @@ class DeltaValidator:
         for batch in self.eval_batches:
             ids = batch["input_ids"].to(self.fp.device)
             labels = batch.get("labels", batch["input_ids"]).to(self.fp.device)
-            out = self.model(input_ids=ids, labels=labels)
+            am = batch.get("attention_mask")
+            if am is not None:
+                am = am.to(self.fp.device)
+            out = self.model(input_ids=ids, attention_mask=am, labels=labels)
             total_loss += float(out.loss) * ids.shape[0]
             total_samples += ids.shape[0]
         self.model.train()
