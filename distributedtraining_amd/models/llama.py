@@ -116,7 +116,7 @@ class LlamaLM(nn.Module):
         sin = self.rope_sin[:S]
         kvlen = None
         if attention_mask is not None:   # right-padded mask → key prefix
-            kvlen = attention_mask.to(torch.int32).sum(dim=1).clamp_(min=1)
+            kvlen = attention_mask.sum(dim=1, dtype=torch.int32).clamp_(min=1)
         x = ops.embedding_fwd(input_ids, self.tok_emb, None)
         pending = None
         for blk in self.blocks:
