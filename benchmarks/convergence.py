@@ -31,6 +31,11 @@ def main() -> int:
     ap.add_argument("--strategy", default="mean",
                     choices=["mean", "nesterov", "parameterized",
                              "score_weighted"])
+    ap.add_argument("--tokenizer", default="byte", choices=["byte", "bpe"],
+                    help="byte: 258-vocab byte tokenizer; bpe: GPT-2-style "
+                         "byte-level BPE trained offline on the corpus "
+                         "(ragged lengths -> the padding/attention-mask "
+                         "path is exercised end to end)")
     ap.add_argument("--table", action="store_true")
     args = ap.parse_args()
 
@@ -47,7 +52,6 @@ def main() -> int:
 
     use_gpu = torch.cuda.is_available()
     dev = torch.device("cuda:0" if use_gpu else "cpu")
-    tok = ByteTokenizer()
 
     # synthetic byte corpus with real sequential structure, sharded
     # disjointly across miners (each miner sees a different slice — the
@@ -57,8 +61,20 @@ def main() -> int:
                 f"the lazy dog while counting {i * shard + i} apples"
                 for i in range(n)]
 
+    if args.tokenizer == "bpe":
+        # offline-trained GPT-2-style BPE (utils/bpe.py): variable-length
+        # rows => padding + attention_mask flow through the kernels
+        from distributedtraining_amd.utils.bpe import train_bpe
+        train_corpus = [t for s in range(1, args.miners + 1)
+                        for t in corpus(s)]
+        tok = train_bpe(train_corpus, vocab_size=512)
+    else:
+        tok = ByteTokenizer()
+
     cfg = Config()
-    cfg.model = ModelConfig(family="gpt2", vocab_size=tok.vocab_size,
+    # vocab rounded up to a multiple of 8 (norm/CE kernel width contract)
+    vs = (tok.vocab_size + 7) // 8 * 8
+    cfg.model = ModelConfig(family="gpt2", vocab_size=vs,
                             n_layer=2, n_head=4, n_embd=128, n_positions=64)
     cfg.train.lr = 5e-4
     seq = 48
@@ -108,6 +124,7 @@ def main() -> int:
 
     out = {"metric": "held-out eval loss after K averaging rounds",
            "device": "MI355X" if use_gpu else "cpu",
+           "tokenizer": args.tokenizer, "vocab_size": vs,
            "strategy": args.strategy, "miners": args.miners,
            "steps_per_round": args.steps_per_round,
            "rounds": args.rounds, "initial_loss": round(losses[0], 4),
