@@ -47,6 +47,8 @@ class LocalSGDNode:
                                hotkey=f"rank{comm.rank}")
         self.averager = ParameterizedAverager(model, fp, cfg.average)
         self.merge_rounds = 0
+        self._validator: Optional[DeltaValidator] = None
+        self._validator_base = None   # base tensor identity at last refresh
 
     def _wire_dtype(self):
         return (torch.bfloat16 if self.cfg.comm.exchange_dtype == "bf16"
@@ -101,22 +103,55 @@ class LocalSGDNode:
         self.merge_rounds += 1
 
     # -- optional distributed validation (BASELINE config #3) ----------------
+    def _refresh_validator(self) -> DeltaValidator:
+        """Build the validator once and re-evaluate its cached base loss
+        only when the base actually changed (round-1 verdict: a fresh
+        validator per round recomputed the base loss every time).
+        Caller must have the BASE loaded into fp.master."""
+        if self._validator is None:
+            self._validator = DeltaValidator(self.model, self.fp,
+                                             self.val_batches,
+                                             self.cfg.validate)
+            self._validator_base = self.miner.base
+        elif self._validator_base is not self.miner.base:
+            v = self._validator
+            v.base_loss, v.base_perplexity = v.evaluate_model()
+            self._validator_base = self.miner.base
+        return self._validator
+
     def validation_round(self) -> Dict[str, float]:
-        """Every rank scores every delta on its own eval set shard; rank 0's
-        normalized scores win (kept simple: identical eval data ⇒ identical
-        scores)."""
+        """Distributed scoring (round-1 verdict #9): the deltas are
+        all-gathered once; each rank scores only its shard
+        (i ≡ rank mod world) against the shared BASE model, and the raw
+        per-delta scores are all-gathered — per-rank eval cost is
+        ~1/world of the reference's every-validator-scores-everything
+        loop (validation_logic.py:126-139). Normalization happens
+        identically on every rank from the gathered totals."""
         base = self.miner.base
         delta = self.fp.make_delta(base)
         deltas = self.comm.all_gather_flat(delta.flat)
         saved = self.fp.master.clone()
-        validator = DeltaValidator(self.model, self.fp, self.val_batches,
-                                   self.cfg.validate)
-        ckpts = {f"rank{i}": DeltaCheckpoint(deltas[i], self.fp.spec, "")
-                 for i in range(deltas.shape[0])}
-        scores = validator.validate_and_score(ckpts)
+        self.fp.load_flat_master(base)   # score against the base, not the
+        validator = self._refresh_validator()   # locally-drifted weights
+        my: Dict[str, float] = {}
+        N = deltas.shape[0]
+        for i in range(self.comm.rank, N, self.comm.world_size):
+            ck = DeltaCheckpoint(deltas[i].to(torch.float32), self.fp.spec,
+                                 "")
+            try:
+                _, _, _, ppl_score = validator.score_delta(ck)
+            except Exception as e:   # reference: per-miner failure → 0
+                log.warning("rank%d scoring failed (%s), score 0", i, e)
+                ppl_score = 0.0
+            my[f"rank{i}"] = ppl_score
         self.fp.master.copy_(saved)
         self.fp.sync_work_from_master()
-        return scores
+        raw: Dict[str, float] = {}
+        for d in self.comm.all_gather_object(my):
+            raw.update(d)
+        total = sum(raw.values())
+        return {h: (max(0.0, s / total) if total > 0 else 0.0)
+                for h, s in sorted(raw.items())}
 
     def run(self, total_steps: int, merge_every: int) -> None:
         self.sync_initial_base()
