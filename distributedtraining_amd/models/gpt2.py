@@ -60,25 +60,29 @@ class GPT2Block(nn.Module):
         ``kvlen`` (int32 [B], optional): right-padding mask — the
         reference's attention_mask path (training_manager.py:380-385).
         Residual branches get counter-RNG dropout in train mode
-        (transformers resid_pdrop/attn_pdrop semantics); dropout sites
-        3i+1..3i+3 keep fwd/bwd masks aligned per layer."""
+        (transformers resid_pdrop/attn_pdrop semantics), FUSED into the
+        add_layer_norm join that consumes each branch (zero extra HBM
+        passes); per-layer sites 3i..3i+2 keep fwd/bwd masks aligned."""
         t = self.training
+        rp = self.resid_pdrop if t else 0.0
         i3 = 3 * self.layer_idx
         if pending is None:
             s = x
             h = ops.layer_norm(x, self.ln_1_w, self.ln_1_b)
         else:
-            s, h = ops.add_layer_norm(x, pending, self.ln_1_w, self.ln_1_b)
+            # consumes the PREVIOUS block's mlp branch: its resid dropout
+            # happens here (site 3i = 3(i-1)+3)
+            s, h = ops.add_layer_norm(x, pending, self.ln_1_w, self.ln_1_b,
+                                      p_drop=rp, site=i3)
         qkv = ops.linear(h, self.attn_qkv_w, self.attn_qkv_b)
         o = ops.qkv_attention(qkv, self.n_head, kvlen=kvlen,
                               p_drop=self.attn_pdrop if t else 0.0,
                               site=i3 + 1)
         a = ops.linear(o, self.attn_proj_w, self.attn_proj_b)
-        a = ops.dropout(a, self.resid_pdrop, site=i3 + 2, training=t)
-        s2, h2 = ops.add_layer_norm(s, a, self.ln_2_w, self.ln_2_b)
+        s2, h2 = ops.add_layer_norm(s, a, self.ln_2_w, self.ln_2_b,
+                                    p_drop=rp, site=i3 + 2)
         m = ops.mlp_gelu(h2, self.mlp_fc_w, self.mlp_fc_b,
                          self.mlp_proj_w, self.mlp_proj_b)
-        m = ops.dropout(m, self.resid_pdrop, site=i3 + 3, training=t)
         return s2, m
 
 
@@ -139,7 +143,11 @@ class GPT2LM(nn.Module):
         if pending is None:
             x = ops.layer_norm(x, self.ln_f_w, self.ln_f_b)
         else:
-            _, x = ops.add_layer_norm(x, pending, self.ln_f_w, self.ln_f_b)
+            # final join consumes the last block's mlp branch (site 3L)
+            rp = self.cfg.resid_pdrop if self.training else 0.0
+            _, x = ops.add_layer_norm(x, pending, self.ln_f_w, self.ln_f_b,
+                                      p_drop=rp,
+                                      site=3 * self.cfg.n_layer)
         if labels is None:
             logits = ops.linear(x, self.wte)
             return CausalLMOutput(loss=None, logits=logits)

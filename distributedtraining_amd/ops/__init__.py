@@ -191,13 +191,18 @@ def _empty_like0(t):
     return t.new_empty(0)
 
 
+def _rng0(t: torch.Tensor) -> torch.Tensor:
+    return droprng.counter(t.device)
+
+
 class _LayerNormFn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, w, b, eps):
         m = require_ext()
         x2 = x.contiguous()
         y, mean, rstd, _ = m.layernorm_fwd(x2.view(-1, x2.shape[-1]),
-                                           _empty_like0(x2), w, b, eps)
+                                           _empty_like0(x2), w, b, eps,
+                                           _rng0(x), 0, 0.0)
         ctx.save_for_backward(x2, w, mean, rstd)
         return y.view_as(x2)
 
@@ -206,9 +211,9 @@ class _LayerNormFn(torch.autograd.Function):
         m = require_ext()
         x, w, mean, rstd = ctx.saved_tensors
         N = x.shape[-1]
-        dx, dw, db = m.layernorm_bwd(dy.contiguous().view(-1, N),
-                                     _empty_like0(x), x.view(-1, N), w,
-                                     mean, rstd)
+        dx, dw, db, _ = m.layernorm_bwd(dy.contiguous().view(-1, N),
+                                        _empty_like0(x), x.view(-1, N), w,
+                                        mean, rstd, _rng0(x), 0, 0.0)
         return dx.view_as(x), dw, db, None
 
 
@@ -227,37 +232,54 @@ class _AddLayerNormFn(torch.autograd.Function):
     exactly; backward folds the sum-stream gradient ds into dx inside the
     dx kernel. Together this removes the separate residual add kernels
     forward AND backward (the residual joins were ~48 eager add
-    launches/step)."""
+    launches/step). With p_drop > 0 the INCOMING branch is dropout-ed
+    inside the same kernels (counter RNG): s = x + drop(res), and
+    backward regenerates the mask to emit dres — the standalone dropout
+    fwd+bwd HBM passes disappear entirely."""
 
     @staticmethod
-    def forward(ctx, x, res, w, b, eps):
+    def forward(ctx, x, res, w, b, eps, p_drop, site):
         m = require_ext()
         x2 = x.contiguous().view(-1, x.shape[-1])
         r2 = res.contiguous().view(-1, x.shape[-1])
-        y, mean, rstd, s = m.layernorm_fwd(x2, r2, w, b, eps)
+        y, mean, rstd, s = m.layernorm_fwd(x2, r2, w, b, eps, _rng0(x),
+                                           site, p_drop)
         ctx.save_for_backward(s, w, mean, rstd)
         ctx.shape = x.shape
+        ctx.drop = (p_drop, site)
         return s.view(x.shape), y.view(x.shape)
 
     @staticmethod
     def backward(ctx, ds, dy):
         m = require_ext()
         s, w, mean, rstd = ctx.saved_tensors
+        p_drop, site = ctx.drop
         N = s.shape[-1]
         ds2 = (ds.contiguous().view(-1, N) if ds is not None
                else _empty_like0(s))
-        dx, dw, db = m.layernorm_bwd(dy.contiguous().view(-1, N), ds2, s,
-                                     w, mean, rstd)
+        dx, dw, db, dres = m.layernorm_bwd(dy.contiguous().view(-1, N),
+                                           ds2, s, w, mean, rstd, _rng0(s),
+                                           site, p_drop)
         dx = dx.view(ctx.shape)
-        return dx, dx, dw, db, None
+        dr = dres.view(ctx.shape) if p_drop > 0.0 else dx
+        return dx, dr, dw, db, None, None, None
 
 
 def add_layer_norm(x: torch.Tensor, res: torch.Tensor, w: torch.Tensor,
-                   b: torch.Tensor, eps: float = 1e-5):
-    """(s, y) = (x + res, layer_norm(x + res)) — the transformer residual
-    join fused into the norm (fwd add + bwd grad-join add eliminated)."""
+                   b: torch.Tensor, eps: float = 1e-5, p_drop: float = 0.0,
+                   site: int = 0):
+    """(s, y) = (x + dropout(res), layer_norm(...)) — the transformer
+    residual join fused into the norm; with p_drop > 0 the incoming
+    branch additionally gets counter-RNG dropout inside the same kernel
+    (transformers resid_pdrop applied at the join that consumes the
+    branch)."""
     if use_hip(x):
-        return _AddLayerNormFn.apply(x, res, w, b, eps)
+        return _AddLayerNormFn.apply(x, res, w, b, eps, p_drop, site)
+    if p_drop > 0.0:
+        keep = droprng.elem_keep_mask(res.numel(),
+                                      droprng.value(x.device), site, p_drop)
+        mask = torch.from_numpy(keep.astype("float32")).reshape(res.shape)
+        res = res * (mask * droprng.inv_keep(p_drop)).to(res.dtype)
     s = x + res
     return s, F.layer_norm(s, (s.shape[-1],), w, b, eps)
 
@@ -271,7 +293,8 @@ class _RMSNormFn(torch.autograd.Function):
         m = require_ext()
         x2 = x.contiguous()
         y, rstd, _ = m.rmsnorm_fwd(x2.view(-1, x2.shape[-1]),
-                                   _empty_like0(x2), w, eps)
+                                   _empty_like0(x2), w, eps, _rng0(x), 0,
+                                   0.0)
         ctx.save_for_backward(x2, w, rstd)
         return y.view_as(x2)
 
@@ -280,43 +303,54 @@ class _RMSNormFn(torch.autograd.Function):
         m = require_ext()
         x, w, rstd = ctx.saved_tensors
         N = x.shape[-1]
-        dx, dw = m.rmsnorm_bwd(dy.contiguous().view(-1, N), _empty_like0(x),
-                               x.view(-1, N), w, rstd)
+        dx, dw, _ = m.rmsnorm_bwd(dy.contiguous().view(-1, N),
+                                  _empty_like0(x), x.view(-1, N), w, rstd,
+                                  _rng0(x), 0, 0.0)
         return dx.view_as(x), dw, None
 
 
 class _AddRMSNormFn(torch.autograd.Function):
     """Fused residual-add + RMSNorm (the Llama-family twin of
-    _AddLayerNormFn): (s, y) = (x+res, RMSNorm(x+res)); backward folds the
-    sum-stream gradient into dx."""
+    _AddLayerNormFn): (s, y) = (x+drop(res), RMSNorm(...)); backward folds
+    the sum-stream gradient into dx and regenerates the dropout mask for
+    dres when p_drop > 0."""
 
     @staticmethod
-    def forward(ctx, x, res, w, eps):
+    def forward(ctx, x, res, w, eps, p_drop, site):
         m = require_ext()
         x2 = x.contiguous().view(-1, x.shape[-1])
         r2 = res.contiguous().view(-1, x.shape[-1])
-        y, rstd, s = m.rmsnorm_fwd(x2, r2, w, eps)
+        y, rstd, s = m.rmsnorm_fwd(x2, r2, w, eps, _rng0(x), site, p_drop)
         ctx.save_for_backward(s, w, rstd)
         ctx.shape = x.shape
+        ctx.drop = (p_drop, site)
         return s.view(x.shape), y.view(x.shape)
 
     @staticmethod
     def backward(ctx, ds, dy):
         m = require_ext()
         s, w, rstd = ctx.saved_tensors
+        p_drop, site = ctx.drop
         N = s.shape[-1]
         ds2 = (ds.contiguous().view(-1, N) if ds is not None
                else _empty_like0(s))
-        dx, dw = m.rmsnorm_bwd(dy.contiguous().view(-1, N), ds2, s, w, rstd)
+        dx, dw, dres = m.rmsnorm_bwd(dy.contiguous().view(-1, N), ds2, s,
+                                     w, rstd, _rng0(s), site, p_drop)
         dx = dx.view(ctx.shape)
-        return dx, dx, dw, None
+        dr = dres.view(ctx.shape) if p_drop > 0.0 else dx
+        return dx, dr, dw, None, None, None
 
 
 def add_rms_norm(x: torch.Tensor, res: torch.Tensor, w: torch.Tensor,
-                 eps: float = 1e-5):
-    """(s, y) = (x + res, rms_norm(x + res))."""
+                 eps: float = 1e-5, p_drop: float = 0.0, site: int = 0):
+    """(s, y) = (x + dropout(res), rms_norm(...))."""
     if use_hip(x):
-        return _AddRMSNormFn.apply(x, res, w, eps)
+        return _AddRMSNormFn.apply(x, res, w, eps, p_drop, site)
+    if p_drop > 0.0:
+        keep = droprng.elem_keep_mask(res.numel(),
+                                      droprng.value(x.device), site, p_drop)
+        mask = torch.from_numpy(keep.astype("float32")).reshape(res.shape)
+        res = res * (mask * droprng.inv_keep(p_drop)).to(res.dtype)
     s = x + res
     sf = s.float()
     y = sf * torch.rsqrt(sf.pow(2).mean(-1, keepdim=True) + eps)

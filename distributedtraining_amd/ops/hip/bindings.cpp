@@ -34,16 +34,45 @@ void check_f32(const Tensor& t, const char* name) {
 }
 
 // ---- norms ----------------------------------------------------------------
+// drop params shared by the norm bindings: p > 0 enables the fused
+// residual-branch dropout (counter RNG; thr16 quantization as elsewhere)
+struct DropArgs {
+  const unsigned long long* rng = nullptr;
+  unsigned long long site = 0;
+  unsigned int thr16 = 0;
+  float ik = 1.0f;
+};
+static DropArgs drop_args(const Tensor& rng, int64_t site, double p) {
+  DropArgs d;
+  if (p > 0.0) {
+    TORCH_CHECK(p < 1.0, "dropout p must be < 1");
+    TORCH_CHECK(rng.numel() == 1 && rng.scalar_type() == torch::kInt64 &&
+                    rng.is_cuda(), "rng must be an int64 [1] GPU counter");
+    d.rng = reinterpret_cast<const unsigned long long*>(
+        rng.data_ptr<int64_t>());
+    d.site = (unsigned long long)site;
+    unsigned int thr = (unsigned int)std::ceil(p * 65536.0);
+    if (thr > 65535u) thr = 65535u;
+    d.thr16 = thr;
+    d.ik = float(65536.0 / (65536.0 - double(thr)));
+  }
+  return d;
+}
+
 // res: empty tensor => plain LN; else fused residual (returns the bf16
-// sum as an extra output feeding the ongoing residual stream)
+// sum as an extra output feeding the ongoing residual stream). p > 0:
+// the branch is dropout-ed BEFORE the add (sum = x + drop(res)).
 std::vector<Tensor> layernorm_fwd(Tensor x, Tensor res, Tensor w, Tensor b,
-                                  double eps) {
+                                  double eps, Tensor rng, int64_t site,
+                                  double p) {
   check_bf16(x, "x"); check_bf16(w, "w"); check_bf16(b, "b");
   const int64_t rows = x.size(0);
   const int cols = int(x.size(1));
   TORCH_CHECK(cols % 8 == 0, "cols must be a multiple of 8");
   const bool has_res = res.numel() > 0;
   if (has_res) check_bf16(res, "res");
+  TORCH_CHECK(p == 0.0 || has_res, "dropout needs the residual branch");
+  const DropArgs da = drop_args(rng, site, p);
   auto y = torch::empty_like(x);
   auto sum = has_res ? torch::empty_like(x) : torch::empty({0}, x.options());
   auto mean = torch::empty({rows}, x.options().dtype(torch::kFloat32));
@@ -52,21 +81,26 @@ std::vector<Tensor> layernorm_fwd(Tensor x, Tensor res, Tensor w, Tensor b,
                        has_res ? bfp_mut(sum) : nullptr, bfp(w), bfp(b),
                        bfp_mut(y), mean.data_ptr<float>(),
                        rstd.data_ptr<float>(), rows, cols, float(eps),
-                       stream());
+                       da.rng, da.site, da.thr16, da.ik, stream());
   return {y, mean, rstd, sum};
 }
 
 // ds: empty tensor => plain; else the gradient arriving on the sum
-// stream, folded into dx (dx = ds + dLN/dx)
+// stream, folded into dx (dx = ds + dLN/dx). p > 0: also emits
+// dres = dx ⊙ mask/(1-p) (4th output) for the dropped branch.
 std::vector<Tensor> layernorm_bwd(Tensor dy, Tensor ds, Tensor x, Tensor w,
-                                  Tensor mean, Tensor rstd) {
+                                  Tensor mean, Tensor rstd, Tensor rng,
+                                  int64_t site, double p) {
   check_bf16(dy, "dy"); check_bf16(x, "x"); check_bf16(w, "w");
   const int64_t rows = x.size(0);
   const int cols = int(x.size(1));
   TORCH_CHECK(cols % 8 == 0, "cols must be a multiple of 8");
   const bool has_ds = ds.numel() > 0;
   if (has_ds) check_bf16(ds, "ds");
+  const DropArgs da = drop_args(rng, site, p);
   auto dx = torch::empty_like(x);
+  auto dres = da.thr16 ? torch::empty_like(x)
+                       : torch::empty({0}, x.options());
   auto f32 = x.options().dtype(torch::kFloat32);
   auto dw32 = torch::zeros({cols}, f32);
   auto db32 = torch::zeros({cols}, f32);
@@ -74,49 +108,59 @@ std::vector<Tensor> layernorm_bwd(Tensor dy, Tensor ds, Tensor x, Tensor w,
   auto part = torch::empty({2, stripes, cols}, f32);
   launch_layernorm_bwd(bfp(dy), has_ds ? bfp(ds) : nullptr, bfp(x), bfp(w),
                        mean.data_ptr<float>(), rstd.data_ptr<float>(),
-                       bfp_mut(dx), dw32.data_ptr<float>(),
-                       db32.data_ptr<float>(), part[0].data_ptr<float>(),
-                       part[1].data_ptr<float>(), stripes, rows, cols,
-                       stream());
-  return {dx, dw32.to(torch::kBFloat16), db32.to(torch::kBFloat16)};
+                       bfp_mut(dx), da.thr16 ? bfp_mut(dres) : nullptr,
+                       dw32.data_ptr<float>(), db32.data_ptr<float>(),
+                       part[0].data_ptr<float>(), part[1].data_ptr<float>(),
+                       stripes, rows, cols, da.rng, da.site, da.thr16,
+                       da.ik, stream());
+  return {dx, dw32.to(torch::kBFloat16), db32.to(torch::kBFloat16), dres};
 }
 
 std::vector<Tensor> rmsnorm_fwd(Tensor x, Tensor res, Tensor w,
-                                double eps) {
+                                double eps, Tensor rng, int64_t site,
+                                double p) {
   check_bf16(x, "x"); check_bf16(w, "w");
   const int64_t rows = x.size(0);
   const int cols = int(x.size(1));
   TORCH_CHECK(cols % 8 == 0, "cols must be a multiple of 8");
   const bool has_res = res.numel() > 0;
   if (has_res) check_bf16(res, "res");
+  TORCH_CHECK(p == 0.0 || has_res, "dropout needs the residual branch");
+  const DropArgs da = drop_args(rng, site, p);
   auto y = torch::empty_like(x);
   auto sum = has_res ? torch::empty_like(x) : torch::empty({0}, x.options());
   auto rstd = torch::empty({rows}, x.options().dtype(torch::kFloat32));
   launch_rmsnorm_fwd(bfp(x), has_res ? bfp(res) : nullptr,
                      has_res ? bfp_mut(sum) : nullptr, bfp(w), bfp_mut(y),
                      rstd.data_ptr<float>(), rows, cols, float(eps),
-                     stream());
+                     da.rng, da.site, da.thr16, da.ik, stream());
   return {y, rstd, sum};
 }
 
 std::vector<Tensor> rmsnorm_bwd(Tensor dy, Tensor ds, Tensor x, Tensor w,
-                                Tensor rstd) {
+                                Tensor rstd, Tensor rng, int64_t site,
+                                double p) {
   check_bf16(dy, "dy"); check_bf16(x, "x"); check_bf16(w, "w");
   const int64_t rows = x.size(0);
   const int cols = int(x.size(1));
   TORCH_CHECK(cols % 8 == 0, "cols must be a multiple of 8");
   const bool has_ds = ds.numel() > 0;
   if (has_ds) check_bf16(ds, "ds");
+  const DropArgs da = drop_args(rng, site, p);
   auto dx = torch::empty_like(x);
+  auto dres = da.thr16 ? torch::empty_like(x)
+                       : torch::empty({0}, x.options());
   auto f32 = x.options().dtype(torch::kFloat32);
   auto dw32 = torch::zeros({cols}, f32);
   const int stripes = dta_colred_stripes(rows, cols);
   auto part = torch::empty({stripes, cols}, f32);
   launch_rmsnorm_bwd(bfp(dy), has_ds ? bfp(ds) : nullptr, bfp(x), bfp(w),
                      rstd.data_ptr<float>(), bfp_mut(dx),
+                     da.thr16 ? bfp_mut(dres) : nullptr,
                      dw32.data_ptr<float>(), part.data_ptr<float>(),
-                     stripes, rows, cols, stream());
-  return {dx, dw32.to(torch::kBFloat16)};
+                     stripes, rows, cols, da.rng, da.site, da.thr16, da.ik,
+                     stream());
+  return {dx, dw32.to(torch::kBFloat16), dres};
 }
 
 // ---- elementwise ----------------------------------------------------------

@@ -80,26 +80,38 @@ void launch_grad_merge_weights(const void* g, bool g_is_bf16,
 // res/sum_out: optional fused residual (sum = bf16(x+res) feeds both the
 // statistics and the ongoing stream); ds: optional additive gradient on
 // the sum stream folded into dx.
+// rng/site/thr16/ik: counter-RNG dropout fused onto the incoming residual
+// branch (thr16 == 0 disables; requires res). Backward emits the branch
+// gradient dres = dx ⊙ mask/(1-p) when enabled.
 void launch_layernorm_fwd(const bf16_t* x, const bf16_t* res,
                           bf16_t* sum_out, const bf16_t* w, const bf16_t* b,
                           bf16_t* y, float* mean, float* rstd, int64_t rows,
-                          int cols, float eps, hipStream_t s);
+                          int cols, float eps, const unsigned long long* rng,
+                          unsigned long long site, unsigned int thr16,
+                          float ik, hipStream_t s);
 // pdw/pdb: [stripes, cols] fp32 workspaces (dta_colred_stripes rows); the
 // two-phase column reduction is atomic-free and deterministic.
 void launch_layernorm_bwd(const bf16_t* dy, const bf16_t* ds,
                           const bf16_t* x, const bf16_t* w,
                           const float* mean, const float* rstd, bf16_t* dx,
-                          float* dw, float* db, float* pdw, float* pdb,
-                          int stripes, int64_t rows, int cols,
-                          hipStream_t s);
+                          bf16_t* dres, float* dw, float* db, float* pdw,
+                          float* pdb, int stripes, int64_t rows, int cols,
+                          const unsigned long long* rng,
+                          unsigned long long site, unsigned int thr16,
+                          float ik, hipStream_t s);
 void launch_rmsnorm_fwd(const bf16_t* x, const bf16_t* res,
                         bf16_t* sum_out, const bf16_t* w, bf16_t* y,
                         float* rstd, int64_t rows, int cols, float eps,
-                        hipStream_t s);
+                        const unsigned long long* rng,
+                        unsigned long long site, unsigned int thr16,
+                        float ik, hipStream_t s);
 void launch_rmsnorm_bwd(const bf16_t* dy, const bf16_t* ds,
                         const bf16_t* x, const bf16_t* w,
-                        const float* rstd, bf16_t* dx, float* dw, float* pdw,
-                        int stripes, int64_t rows, int cols, hipStream_t s);
+                        const float* rstd, bf16_t* dx, bf16_t* dres,
+                        float* dw, float* pdw, int stripes, int64_t rows,
+                        int cols, const unsigned long long* rng,
+                        unsigned long long site, unsigned int thr16,
+                        float ik, hipStream_t s);
 
 // ---- fused cross entropy --------------------------------------------------
 void launch_ce_fwd(const bf16_t* logits, const int64_t* targets,

@@ -16,6 +16,7 @@
 #include <torch/extension.h>
 #include <ATen/hip/HIPContext.h>
 #include <hipblaslt/hipblaslt.h>
+#include <hipblaslt/hipblaslt-ext.hpp>
 
 #include <map>
 #include <mutex>
@@ -42,8 +43,41 @@ static hipblasLtHandle_t handle() {
   return h;
 }
 
-constexpr size_t kWorkspace = 32u << 20;  // 32 MiB
+constexpr size_t kWorkspace = 128u << 20;  // 128 MiB (stream-K slabs)
 constexpr int kMaxAlgos = 64;
+
+// Full-catalog candidate enumeration (round-1 parked plan): the heuristic
+// top-N for BGRADB never offered the stream-K wgrad kernels TunableOp
+// finds for the plain GEMM, so pull EVERY algorithm of the dtype/op combo
+// from hipblaslt_ext::getAllAlgos and keep the ones that accept this
+// problem; pick_algo then times them.
+static int all_supported_algos(hipblasLtMatmulDesc_t op,
+                               hipblasOperation_t ta, hipblasOperation_t tb,
+                               hipblasLtMatrixLayout_t la,
+                               hipblasLtMatrixLayout_t lb,
+                               hipblasLtMatrixLayout_t lc,
+                               hipblasLtMatmulHeuristicResult_t* out,
+                               int cap) {
+  std::vector<hipblasLtMatmulHeuristicResult_t> all;
+  if (hipblaslt_ext::getAllAlgos(handle(), hipblaslt_ext::GemmType::HIPBLASLT_GEMM,
+                                 ta, tb, HIP_R_16BF, HIP_R_16BF, HIP_R_16BF,
+                                 HIP_R_16BF, HIPBLAS_COMPUTE_32F,
+                                 all) != HIPBLAS_STATUS_SUCCESS)
+    return 0;
+  const float alpha = 1.0f, beta = 1.0f;
+  int n = 0;
+  for (auto& r : all) {
+    size_t ws = 0;
+    if (hipblaslt_ext::matmulIsAlgoSupported(handle(), op, &alpha, la, lb,
+                                             &beta, lc, lc, r.algo,
+                                             ws) == HIPBLAS_STATUS_SUCCESS &&
+        ws <= kWorkspace) {
+      out[n++] = r;
+      if (n >= cap) break;
+    }
+  }
+  return n;
+}
 
 // Time each heuristic candidate (3 reps, beta=0 into scratch) and return
 // the fastest — hipBLASLt's first heuristic pick measured 8 ms/step slower
@@ -325,11 +359,15 @@ Tensor lt_wgrad_bgradb(Tensor dy, Tensor x, Tensor dw_acc) {
     LT_CHECK(hipblasLtMatmulPreferenceSetAttribute(
         pref, HIPBLASLT_MATMUL_PREF_MAX_WORKSPACE_BYTES, &wsz,
         sizeof(wsz)));
-    hipblasLtMatmulHeuristicResult_t results[kMaxAlgos];
-    int nres = 0;
-    LT_CHECK(hipblasLtMatmulAlgoGetHeuristic(handle(), p.op, p.la, p.lb,
-                                             p.lc, p.lc, pref, kMaxAlgos,
-                                             results, &nres));
+    // full catalog first (stream-K lives outside the heuristic top-N),
+    // heuristic as the fallback
+    static hipblasLtMatmulHeuristicResult_t results[512];
+    int nres = all_supported_algos(p.op, HIPBLAS_OP_N, HIPBLAS_OP_T,
+                                   p.la, p.lb, p.lc, results, 512);
+    if (nres == 0)
+      LT_CHECK(hipblasLtMatmulAlgoGetHeuristic(handle(), p.op, p.la, p.lb,
+                                               p.lc, p.lc, pref, kMaxAlgos,
+                                               results, &nres));
     hipblasLtMatmulPreferenceDestroy(pref);
     TORCH_CHECK(nres > 0, "hipblasLt: no BGRADB algorithm M=", M, " N=",
                 N, " E=", E);

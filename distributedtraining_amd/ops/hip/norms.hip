@@ -21,7 +21,11 @@ constexpr int ROW_WAVES = 4;
 // RES: residual fusion — vals = round_bf16(x + res) (rounded BEFORE the
 // statistics so the saved sum tensor reproduces them exactly in backward),
 // and the sum is written out for the ongoing residual stream.
-template <int ITERS, bool RMS, bool RES>
+// DROP: counter-RNG dropout fused onto the INCOMING residual branch
+// (transformers resid_pdrop: x = x + dropout(branch)) — elementwise
+// chain over the branch's flat index, same draws as the standalone
+// dropout kernel, so the host gold is ops.droprng.elem_keep_mask.
+template <int ITERS, bool RMS, bool RES, bool DROP>
 __global__ void norm_fwd_k(const ushort* __restrict__ x,
                            const ushort* __restrict__ res,
                            ushort* __restrict__ sum_out,
@@ -29,9 +33,13 @@ __global__ void norm_fwd_k(const ushort* __restrict__ x,
                            const ushort* __restrict__ b,
                            ushort* __restrict__ y, float* __restrict__ mean,
                            float* __restrict__ rstd, int64_t rows, int cols,
-                           float eps) {
+                           float eps,
+                           const unsigned long long* __restrict__ rng,
+                           unsigned long long site, unsigned int thr16,
+                           float inv_keep) {
   const int lane = threadIdx.x & 63, wid = threadIdx.x >> 6;
   const int nchunk = cols >> 3;
+  const uint64_t s1d = DROP ? sm64(*rng + site * DTA_RNG_SITE_K) : 0;
   for (int64_t row = int64_t(blockIdx.x) * ROW_WAVES + wid; row < rows;
        row += int64_t(gridDim.x) * ROW_WAVES) {
     const ushort* xr = x + row * cols;
@@ -46,12 +54,24 @@ __global__ void norm_fwd_k(const ushort* __restrict__ x,
         s16x8 vx = *reinterpret_cast<const s16x8*>(xr + c * 8);
         s16x8 vr;
         if (RES) vr = *reinterpret_cast<const s16x8*>(rr + c * 8);
+        uint64_t h0 = 0, h1 = 0;
+        if (DROP) {
+          const int64_t i0 = row * cols + int64_t(c) * 8;
+          h0 = sm64(s1d + uint64_t(i0 >> 2) * DTA_RNG_IDX_K);
+          h1 = sm64(s1d + uint64_t((i0 >> 2) + 1) * DTA_RNG_IDX_K);
+        }
         s16x8 vs;
 #pragma unroll
         for (int j = 0; j < 8; ++j) {
           float f = bf2f(ushort(vx[j]));
           if (RES) {
-            const ushort sb = f2bf(f + bf2f(ushort(vr[j])));
+            float rf = bf2f(ushort(vr[j]));
+            if (DROP) {
+              const uint64_t h = (j < 4) ? h0 : h1;
+              rf = ((uint32_t(h >> (16 * (j & 3))) & 0xFFFF) >= thr16)
+                       ? rf * inv_keep : 0.f;
+            }
+            const ushort sb = f2bf(f + rf);
             vs[j] = sb;
             f = bf2f(sb);
           }
@@ -102,17 +122,26 @@ __global__ void norm_fwd_k(const ushort* __restrict__ x,
 // RMS: dx = rs*(dyw - xh*mean(dyw*xh))
 // DS: residual fusion — dx += ds (the gradient arriving on the sum
 // stream from its downstream consumer), saving the separate add kernel.
-template <int ITERS, bool RMS, bool DS>
+// DROP: the forward dropped the residual branch, so its gradient is the
+// total sum-gradient gated by the regenerated mask: dres = dx ⊙ M/(1-p)
+// — one extra streaming write here instead of a separate dropout-bwd
+// pass over dx.
+template <int ITERS, bool RMS, bool DS, bool DROP>
 __global__ void norm_bwd_dx_k(const ushort* __restrict__ dy,
                               const ushort* __restrict__ ds,
                               const ushort* __restrict__ x,
                               const ushort* __restrict__ w,
                               const float* __restrict__ mean,
                               const float* __restrict__ rstd,
-                              ushort* __restrict__ dx, int64_t rows,
-                              int cols) {
+                              ushort* __restrict__ dx,
+                              ushort* __restrict__ dres, int64_t rows,
+                              int cols,
+                              const unsigned long long* __restrict__ rng,
+                              unsigned long long site, unsigned int thr16,
+                              float inv_keep) {
   const int lane = threadIdx.x & 63, wid = threadIdx.x >> 6;
   const int nchunk = cols >> 3;
+  const uint64_t s1d = DROP ? sm64(*rng + site * DTA_RNG_SITE_K) : 0;
   float wv[ITERS][8];
 #pragma unroll
   for (int it = 0; it < ITERS; ++it) {
@@ -154,6 +183,7 @@ __global__ void norm_bwd_dx_k(const ushort* __restrict__ dy,
     s1 = wave_sum(s1) / cols;
     s2 = wave_sum(s2) / cols;
     ushort* dxr = dx + row * cols;
+    ushort* drr = DROP ? dres + row * cols : nullptr;
     const ushort* dsr = DS ? ds + row * cols : nullptr;
 #pragma unroll
     for (int it = 0; it < ITERS; ++it) {
@@ -161,7 +191,13 @@ __global__ void norm_bwd_dx_k(const ushort* __restrict__ dy,
       if (c < nchunk) {
         s16x8 vds;
         if (DS) vds = *reinterpret_cast<const s16x8*>(dsr + c * 8);
-        s16x8 o;
+        uint64_t h0 = 0, h1 = 0;
+        if (DROP) {
+          const int64_t i0 = row * cols + int64_t(c) * 8;
+          h0 = sm64(s1d + uint64_t(i0 >> 2) * DTA_RNG_IDX_K);
+          h1 = sm64(s1d + uint64_t((i0 >> 2) + 1) * DTA_RNG_IDX_K);
+        }
+        s16x8 o, od;
 #pragma unroll
         for (int j = 0; j < 8; ++j) {
           float v = RMS ? (dyw[it][j] - xh[it][j] * s2)
@@ -169,8 +205,14 @@ __global__ void norm_bwd_dx_k(const ushort* __restrict__ dy,
           float g = rs * v;
           if (DS) g += bf2f(ushort(vds[j]));
           o[j] = f2bf(g);
+          if (DROP) {
+            const uint64_t h = (j < 4) ? h0 : h1;
+            od[j] = ((uint32_t(h >> (16 * (j & 3))) & 0xFFFF) >= thr16)
+                        ? f2bf(g * inv_keep) : ushort(0);
+          }
         }
         *reinterpret_cast<s16x8*>(dxr + c * 8) = o;
+        if (DROP) *reinterpret_cast<s16x8*>(drr + c * 8) = od;
       }
     }
   }
@@ -242,11 +284,12 @@ __global__ void dwdb_reduce_k(const float* __restrict__ pdw,
   }
 }
 
-template <bool RMS, bool RES>
+template <bool RMS, bool RES, bool DROP>
 void dispatch_fwd(const ushort* x, const ushort* res, ushort* sum_out,
                   const ushort* w, const ushort* b, ushort* y, float* mean,
                   float* rstd, int64_t rows, int cols, float eps,
-                  hipStream_t s) {
+                  const unsigned long long* rng, unsigned long long site,
+                  unsigned int thr16, float ik, hipStream_t s) {
   const int nchunk = cols >> 3;
   const int iters = (nchunk + 63) / 64;
   int64_t want = (rows + ROW_WAVES - 1) / ROW_WAVES;
@@ -254,26 +297,28 @@ void dispatch_fwd(const ushort* x, const ushort* res, ushort* sum_out,
   const dim3 blk(64 * ROW_WAVES);
 #define CASE_F(I)                                                         \
   case I:                                                                 \
-    norm_fwd_k<I, RMS, RES><<<grid, blk, 0, s>>>(x, res, sum_out, w, b,   \
-                                                 y, mean, rstd, rows,     \
-                                                 cols, eps);              \
+    norm_fwd_k<I, RMS, RES, DROP><<<grid, blk, 0, s>>>(                   \
+        x, res, sum_out, w, b, y, mean, rstd, rows, cols, eps, rng, site, \
+        thr16, ik);                                                       \
     break;
   switch (iters) {
     CASE_F(1) CASE_F(2) CASE_F(3) CASE_F(4) CASE_F(6) CASE_F(8) CASE_F(16)
     default: {
-      if (iters <= 6) { norm_fwd_k<6, RMS, RES><<<grid, blk, 0, s>>>(x, res, sum_out, w, b, y, mean, rstd, rows, cols, eps); }
-      else if (iters <= 8) { norm_fwd_k<8, RMS, RES><<<grid, blk, 0, s>>>(x, res, sum_out, w, b, y, mean, rstd, rows, cols, eps); }
-      else { norm_fwd_k<16, RMS, RES><<<grid, blk, 0, s>>>(x, res, sum_out, w, b, y, mean, rstd, rows, cols, eps); }
+      if (iters <= 6) { norm_fwd_k<6, RMS, RES, DROP><<<grid, blk, 0, s>>>(x, res, sum_out, w, b, y, mean, rstd, rows, cols, eps, rng, site, thr16, ik); }
+      else if (iters <= 8) { norm_fwd_k<8, RMS, RES, DROP><<<grid, blk, 0, s>>>(x, res, sum_out, w, b, y, mean, rstd, rows, cols, eps, rng, site, thr16, ik); }
+      else { norm_fwd_k<16, RMS, RES, DROP><<<grid, blk, 0, s>>>(x, res, sum_out, w, b, y, mean, rstd, rows, cols, eps, rng, site, thr16, ik); }
     }
   }
 #undef CASE_F
 }
 
-template <bool RMS, bool DS>
+template <bool RMS, bool DS, bool DROP>
 void dispatch_bwd(const ushort* dy, const ushort* ds, const ushort* x,
                   const ushort* w, const float* mean, const float* rstd,
-                  ushort* dx, float* dw, float* db, float* pdw, float* pdb,
-                  int stripes, int64_t rows, int cols, hipStream_t s) {
+                  ushort* dx, ushort* dres, float* dw, float* db, float* pdw,
+                  float* pdb, int stripes, int64_t rows, int cols,
+                  const unsigned long long* rng, unsigned long long site,
+                  unsigned int thr16, float ik, hipStream_t s) {
   const int nchunk = cols >> 3;
   const int iters = (nchunk + 63) / 64;
   int64_t want = (rows + ROW_WAVES - 1) / ROW_WAVES;
@@ -281,14 +326,16 @@ void dispatch_bwd(const ushort* dy, const ushort* ds, const ushort* x,
   const dim3 blk(64 * ROW_WAVES);
 #define CASE_B(I)                                                            \
   case I:                                                                    \
-    norm_bwd_dx_k<I, RMS, DS><<<grid, blk, 0, s>>>(dy, ds, x, w, mean,       \
-                                                   rstd, dx, rows, cols);    \
+    norm_bwd_dx_k<I, RMS, DS, DROP><<<grid, blk, 0, s>>>(                    \
+        dy, ds, x, w, mean, rstd, dx, dres, rows, cols, rng, site, thr16,    \
+        ik);                                                                 \
     break;
   switch (iters) {
     CASE_B(1) CASE_B(2) CASE_B(3) CASE_B(4) CASE_B(6) CASE_B(8)
     default:
-      norm_bwd_dx_k<8, RMS, DS><<<grid, blk, 0, s>>>(dy, ds, x, w, mean,
-                                                     rstd, dx, rows, cols);
+      norm_bwd_dx_k<8, RMS, DS, DROP><<<grid, blk, 0, s>>>(
+          dy, ds, x, w, mean, rstd, dx, dres, rows, cols, rng, site, thr16,
+          ik);
   }
 #undef CASE_B
   const ColRedCfg cfg = dta_colred_cfg(rows, cols);
@@ -308,46 +355,92 @@ void dispatch_bwd(const ushort* dy, const ushort* ds, const ushort* x,
 void launch_layernorm_fwd(const bf16_t* x, const bf16_t* res,
                           bf16_t* sum_out, const bf16_t* w, const bf16_t* b,
                           bf16_t* y, float* mean, float* rstd, int64_t rows,
-                          int cols, float eps, hipStream_t s) {
-  if (res)
-    dispatch_fwd<false, true>(x, res, sum_out, w, b, y, mean, rstd, rows,
-                              cols, eps, s);
+                          int cols, float eps, const unsigned long long* rng,
+                          unsigned long long site, unsigned int thr16,
+                          float ik, hipStream_t s) {
+  if (res && thr16)
+    dispatch_fwd<false, true, true>(x, res, sum_out, w, b, y, mean, rstd,
+                                    rows, cols, eps, rng, site, thr16, ik,
+                                    s);
+  else if (res)
+    dispatch_fwd<false, true, false>(x, res, sum_out, w, b, y, mean, rstd,
+                                     rows, cols, eps, nullptr, 0, 0, 1.f,
+                                     s);
   else
-    dispatch_fwd<false, false>(x, nullptr, nullptr, w, b, y, mean, rstd,
-                               rows, cols, eps, s);
+    dispatch_fwd<false, false, false>(x, nullptr, nullptr, w, b, y, mean,
+                                      rstd, rows, cols, eps, nullptr, 0, 0,
+                                      1.f, s);
 }
 void launch_layernorm_bwd(const bf16_t* dy, const bf16_t* ds,
                           const bf16_t* x, const bf16_t* w,
                           const float* mean, const float* rstd, bf16_t* dx,
-                          float* dw, float* db, float* pdw, float* pdb,
-                          int stripes, int64_t rows, int cols,
-                          hipStream_t s) {
-  if (ds)
-    dispatch_bwd<false, true>(dy, ds, x, w, mean, rstd, dx, dw, db, pdw,
-                              pdb, stripes, rows, cols, s);
+                          bf16_t* dres, float* dw, float* db, float* pdw,
+                          float* pdb, int stripes, int64_t rows, int cols,
+                          const unsigned long long* rng,
+                          unsigned long long site, unsigned int thr16,
+                          float ik, hipStream_t s) {
+  if (thr16) {
+    if (ds)
+      dispatch_bwd<false, true, true>(dy, ds, x, w, mean, rstd, dx, dres,
+                                      dw, db, pdw, pdb, stripes, rows,
+                                      cols, rng, site, thr16, ik, s);
+    else
+      dispatch_bwd<false, false, true>(dy, nullptr, x, w, mean, rstd, dx,
+                                       dres, dw, db, pdw, pdb, stripes,
+                                       rows, cols, rng, site, thr16, ik, s);
+  } else if (ds)
+    dispatch_bwd<false, true, false>(dy, ds, x, w, mean, rstd, dx, nullptr,
+                                     dw, db, pdw, pdb, stripes, rows, cols,
+                                     nullptr, 0, 0, 1.f, s);
   else
-    dispatch_bwd<false, false>(dy, nullptr, x, w, mean, rstd, dx, dw, db,
-                               pdw, pdb, stripes, rows, cols, s);
+    dispatch_bwd<false, false, false>(dy, nullptr, x, w, mean, rstd, dx,
+                                      nullptr, dw, db, pdw, pdb, stripes,
+                                      rows, cols, nullptr, 0, 0, 1.f, s);
 }
 void launch_rmsnorm_fwd(const bf16_t* x, const bf16_t* res,
                         bf16_t* sum_out, const bf16_t* w, bf16_t* y,
                         float* rstd, int64_t rows, int cols, float eps,
-                        hipStream_t s) {
-  if (res)
-    dispatch_fwd<true, true>(x, res, sum_out, w, nullptr, y, nullptr,
-                             rstd, rows, cols, eps, s);
+                        const unsigned long long* rng,
+                        unsigned long long site, unsigned int thr16,
+                        float ik, hipStream_t s) {
+  if (res && thr16)
+    dispatch_fwd<true, true, true>(x, res, sum_out, w, nullptr, y, nullptr,
+                                   rstd, rows, cols, eps, rng, site, thr16,
+                                   ik, s);
+  else if (res)
+    dispatch_fwd<true, true, false>(x, res, sum_out, w, nullptr, y,
+                                    nullptr, rstd, rows, cols, eps,
+                                    nullptr, 0, 0, 1.f, s);
   else
-    dispatch_fwd<true, false>(x, nullptr, nullptr, w, nullptr, y, nullptr,
-                              rstd, rows, cols, eps, s);
+    dispatch_fwd<true, false, false>(x, nullptr, nullptr, w, nullptr, y,
+                                     nullptr, rstd, rows, cols, eps,
+                                     nullptr, 0, 0, 1.f, s);
 }
 void launch_rmsnorm_bwd(const bf16_t* dy, const bf16_t* ds,
                         const bf16_t* x, const bf16_t* w,
-                        const float* rstd, bf16_t* dx, float* dw, float* pdw,
-                        int stripes, int64_t rows, int cols, hipStream_t s) {
-  if (ds)
-    dispatch_bwd<true, true>(dy, ds, x, w, nullptr, rstd, dx, dw, nullptr,
-                             pdw, nullptr, stripes, rows, cols, s);
+                        const float* rstd, bf16_t* dx, bf16_t* dres,
+                        float* dw, float* pdw, int stripes, int64_t rows,
+                        int cols, const unsigned long long* rng,
+                        unsigned long long site, unsigned int thr16,
+                        float ik, hipStream_t s) {
+  if (thr16) {
+    if (ds)
+      dispatch_bwd<true, true, true>(dy, ds, x, w, nullptr, rstd, dx, dres,
+                                     dw, nullptr, pdw, nullptr, stripes,
+                                     rows, cols, rng, site, thr16, ik, s);
+    else
+      dispatch_bwd<true, false, true>(dy, nullptr, x, w, nullptr, rstd, dx,
+                                      dres, dw, nullptr, pdw, nullptr,
+                                      stripes, rows, cols, rng, site,
+                                      thr16, ik, s);
+  } else if (ds)
+    dispatch_bwd<true, true, false>(dy, ds, x, w, nullptr, rstd, dx,
+                                    nullptr, dw, nullptr, pdw, nullptr,
+                                    stripes, rows, cols, nullptr, 0, 0,
+                                    1.f, s);
   else
-    dispatch_bwd<true, false>(dy, nullptr, x, w, nullptr, rstd, dx, dw,
-                              nullptr, pdw, nullptr, stripes, rows, cols, s);
+    dispatch_bwd<true, false, false>(dy, nullptr, x, w, nullptr, rstd, dx,
+                                     nullptr, dw, nullptr, pdw, nullptr,
+                                     stripes, rows, cols, nullptr, 0, 0,
+                                     1.f, s);
 }

@@ -226,3 +226,22 @@ def test_dropout_convergence_tiny():
     for _ in range(30):
         last = float(miner.train_step(batch))
     assert last < first - 0.5, (first, last)
+
+
+def test_add_layer_norm_dropout_cpu_semantics():
+    """CPU path of the fused-join dropout: masked branch, exact zeros in
+    the branch gradient, stream gradient unmasked."""
+    droprng.counter("cpu").fill_(400)
+    R, C, p, site = 64, 32, 0.25, 11
+    x = torch.randn(R, C, requires_grad=True)
+    res = torch.randn(R, C, requires_grad=True)
+    w, b = torch.ones(C, requires_grad=True), torch.zeros(C,
+                                                          requires_grad=True)
+    s, y = ops.add_layer_norm(x, res, w, b, p_drop=p, site=site)
+    keep = droprng.elem_keep_mask(R * C, 400, site, p)
+    mask = torch.from_numpy(keep.astype("float32")).view(R, C)
+    torch.testing.assert_close(
+        s, x + res * mask * droprng.inv_keep(p))
+    (s.sum() + y.sum()).backward()
+    assert torch.all(res.grad[mask == 0] == 0)
+    assert torch.all(x.grad != 0)   # stream gradient not masked

@@ -869,3 +869,73 @@ def test_gpt2_train_dropout_changes_loss_gpu():
         droprng.tick(DEV)
         e2 = model(input_ids=ids, labels=ids).loss
     torch.testing.assert_close(e1, e2)    # eval mode: dropout off
+
+
+def test_add_layer_norm_fused_dropout():
+    """Fused residual-branch dropout in add_layer_norm: GPU kernels vs the
+    host-gold mask (elementwise chain), fwd and bwd."""
+    from distributedtraining_amd import ops
+    from distributedtraining_amd.ops import droprng
+    droprng.counter(DEV).fill_(31)
+    droprng.counter("cpu").fill_(31)
+    R, C, p, site = 512, 256, 0.2, 9
+    x = _rand_bf16(R, C, seed=50).requires_grad_(True)
+    res = _rand_bf16(R, C, seed=51).requires_grad_(True)
+    w = _rand_bf16(C, seed=52).requires_grad_(True)
+    b = _rand_bf16(C, seed=53).requires_grad_(True)
+    s, y = ops.add_layer_norm(x, res, w, b, p_drop=p, site=site)
+    # CPU gold with the same mask
+    xr = x.detach().cpu().float().requires_grad_(True)
+    rr = res.detach().cpu().float().requires_grad_(True)
+    wr = w.detach().cpu().float().requires_grad_(True)
+    br = b.detach().cpu().float().requires_grad_(True)
+    keep = droprng.elem_keep_mask(R * C, 31, site, p)
+    mask = torch.from_numpy(keep.astype("float32")).view(R, C) \
+        * droprng.inv_keep(p)
+    sr = xr + rr * mask
+    yr = torch.nn.functional.layer_norm(sr, (C,), wr, br, 1e-5)
+    torch.testing.assert_close(s.float().cpu(), sr, rtol=2e-2, atol=2e-2)
+    torch.testing.assert_close(y.float().cpu(), yr, rtol=3e-2, atol=3e-2)
+    # exact mask agreement on the sum: where mask==0, s == x exactly
+    zeros = (mask.cpu() == 0)
+    torch.testing.assert_close(s.cpu()[zeros].float(),
+                               x.detach().cpu()[zeros].float())
+    ds = _rand_bf16(R, C, seed=54)
+    dy = _rand_bf16(R, C, seed=55)
+    torch.autograd.backward([s, y], [ds, dy])
+    torch.autograd.backward([sr, yr], [ds.float().cpu(), dy.float().cpu()])
+    torch.testing.assert_close(x.grad.float().cpu(), xr.grad,
+                               rtol=5e-2, atol=5e-2)
+    torch.testing.assert_close(res.grad.float().cpu(), rr.grad,
+                               rtol=5e-2, atol=5e-2)
+    # dropped branch positions get exactly zero gradient
+    assert torch.all(res.grad.cpu()[zeros] == 0)
+    torch.testing.assert_close(w.grad.float().cpu(), wr.grad,
+                               rtol=5e-2, atol=0.3)
+
+
+def test_add_rms_norm_fused_dropout():
+    from distributedtraining_amd import ops
+    from distributedtraining_amd.ops import droprng
+    droprng.counter(DEV).fill_(77)
+    R, C, p, site = 256, 128, 0.3, 4
+    x = _rand_bf16(R, C, seed=60).requires_grad_(True)
+    res = _rand_bf16(R, C, seed=61).requires_grad_(True)
+    w = _rand_bf16(C, seed=62).requires_grad_(True)
+    s, y = ops.add_rms_norm(x, res, w, p_drop=p, site=site)
+    keep = droprng.elem_keep_mask(R * C, 77, site, p)
+    mask = torch.from_numpy(keep.astype("float32")).view(R, C) \
+        * droprng.inv_keep(p)
+    xr = x.detach().cpu().float().requires_grad_(True)
+    rr = res.detach().cpu().float().requires_grad_(True)
+    wr = w.detach().cpu().float().requires_grad_(True)
+    sr = xr + rr * mask
+    yr = sr * torch.rsqrt(sr.pow(2).mean(-1, keepdim=True) + 1e-5) * wr
+    torch.testing.assert_close(s.float().cpu(), sr, rtol=2e-2, atol=2e-2)
+    torch.testing.assert_close(y.float().cpu(), yr, rtol=3e-2, atol=3e-2)
+    dy = _rand_bf16(R, C, seed=63)
+    y.backward(dy)
+    yr.backward(dy.float().cpu())
+    torch.testing.assert_close(res.grad.float().cpu(), rr.grad,
+                               rtol=5e-2, atol=5e-2)
+    assert torch.all(res.grad.cpu()[mask.cpu() == 0] == 0)
