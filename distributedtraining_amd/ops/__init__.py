@@ -79,10 +79,13 @@ class _LinearFn(torch.autograd.Function):
         if _USE_BGRADB and ctx.has_b and ctx.wgrad is not None:
             # one GEMM: dW accumulated into the flat plane (beta=1) with
             # the bias gradient via the BGRADB epilogue. Works at every
-            # production shape BUT the heuristic algorithm selection loses
-            # to the TunableOp stream-K wgrad kernels at K=65536 (measured
-            # 791k vs 877k tokens/s whole-step) — opt-in until the lt path
-            # gets its own algorithm tuning.
+            # production shape BUT loses to TunableOp stream-K wgrad +
+            # colsum — FINAL negative result r02: even with the FULL
+            # algorithm catalog (hipblaslt_ext::getAllAlgos, 128 MiB
+            # workspace) timed per shape, the best BGRADB plan measured
+            # 815k vs 882k tokens/s whole-step (80.4 vs 74.3 ms). The
+            # epilogue's bias reduction forces non-stream-K kernels at
+            # K=65536. Kept opt-in for other shapes/batches.
             try:
                 db = m.lt_wgrad_bgradb(dy2, x2, ctx.wgrad)
                 return dx, None, db

@@ -13,6 +13,11 @@ pytestmark = pytest.mark.gpu
 DEV = "cuda:0"
 
 
+def _rng():
+    from distributedtraining_amd.ops import droprng
+    return droprng.counter(DEV)
+
+
 def _ext():
     from distributedtraining_amd.ops.backend import require_ext
     return require_ext()
@@ -65,13 +70,13 @@ def test_layernorm_fwd_bwd(rows, cols):
     x = _rand_bf16(rows, cols, seed=1)
     w = _rand_bf16(cols, seed=2, scale=0.5)
     b = _rand_bf16(cols, seed=3, scale=0.5)
-    y, mean, rstd, _ = m.layernorm_fwd(x, torch.empty(0, device=DEV, dtype=torch.bfloat16), w, b, 1e-5)
+    y, mean, rstd, _ = m.layernorm_fwd(x, torch.empty(0, device=DEV, dtype=torch.bfloat16), w, b, 1e-5, _rng(), 0, 0.0)
     ref = torch.nn.functional.layer_norm(
         x.float(), (cols,), w.float(), b.float(), 1e-5)
     torch.testing.assert_close(y.float(), ref, rtol=2e-2, atol=2e-2)
 
     dy = _rand_bf16(rows, cols, seed=4)
-    dx, dw, db = m.layernorm_bwd(dy, torch.empty(0, device=DEV, dtype=torch.bfloat16), x, w, mean, rstd)
+    dx, dw, db, _ = m.layernorm_bwd(dy, torch.empty(0, device=DEV, dtype=torch.bfloat16), x, w, mean, rstd, _rng(), 0, 0.0)
     xr = x.float().detach().requires_grad_(True)
     wr = w.float().detach().requires_grad_(True)
     br = b.float().detach().requires_grad_(True)
@@ -89,7 +94,7 @@ def test_rmsnorm_fwd_bwd(rows, cols):
     m = _ext()
     x = _rand_bf16(rows, cols, seed=1)
     w = _rand_bf16(cols, seed=2, scale=0.5)
-    y, rstd, _ = m.rmsnorm_fwd(x, torch.empty(0, device=DEV, dtype=torch.bfloat16), w, 1e-5)
+    y, rstd, _ = m.rmsnorm_fwd(x, torch.empty(0, device=DEV, dtype=torch.bfloat16), w, 1e-5, _rng(), 0, 0.0)
     xf = x.float()
     ref = xf * torch.rsqrt(xf.pow(2).mean(-1, keepdim=True) + 1e-5) * w.float()
     torch.testing.assert_close(y.float(), ref, rtol=2e-2, atol=2e-2)
@@ -99,7 +104,7 @@ def test_rmsnorm_fwd_bwd(rows, cols):
     wr = w.float().detach().requires_grad_(True)
     (xr * torch.rsqrt(xr.pow(2).mean(-1, keepdim=True) + 1e-5) * wr).backward(
         dy.float())
-    dx, dw = m.rmsnorm_bwd(dy, torch.empty(0, device=DEV, dtype=torch.bfloat16), x, w, rstd)
+    dx, dw, _ = m.rmsnorm_bwd(dy, torch.empty(0, device=DEV, dtype=torch.bfloat16), x, w, rstd, _rng(), 0, 0.0)
     torch.testing.assert_close(dx.float(), xr.grad, rtol=5e-2, atol=5e-2)
     torch.testing.assert_close(dw.float(), wr.grad, rtol=5e-2, atol=0.3)
 
