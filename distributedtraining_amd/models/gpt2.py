@@ -151,17 +151,18 @@ class GPT2LM(nn.Module):
         if labels is None:
             logits = ops.linear(x, self.wte)
             return CausalLMOutput(loss=None, logits=logits)
-        # shifted CE: predict token t+1 from position t (HF semantics)
-        logits = ops.linear(x[:, :-1, :].contiguous(), self.wte)
+        # shifted CE: predict token t+1 from position t (HF semantics);
+        # head GEMM + CE fused into one pipelined node (ops.lm_head_ce)
         tgt = labels[:, 1:]
         if attention_mask is not None:
             tgt = tgt.masked_fill(attention_mask[:, 1:] == 0, -100)
-        tgt = tgt.contiguous().view(-1)
-        loss = ops.cross_entropy_loss(
-            logits.reshape(-1, self.cfg.vocab_size), tgt)
+        B, S = input_ids.shape
+        loss, logits = ops.lm_head_ce(x[:, :-1, :].contiguous(), self.wte,
+                                      tgt.contiguous().view(-1))
         return CausalLMOutput(
             loss=loss,
-            logits=logits if (return_logits or not self.training) else None)
+            logits=(logits.view(B, S - 1, self.cfg.vocab_size)
+                    if (return_logits or not self.training) else None))
 
 
 class KVCache:

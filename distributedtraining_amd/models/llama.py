@@ -128,16 +128,17 @@ class LlamaLM(nn.Module):
                                     self.cfg.norm_eps)
         if labels is None:
             return CausalLMOutput(loss=None, logits=ops.linear(x, self._head()))
-        logits = ops.linear(x[:, :-1, :].contiguous(), self._head())
         tgt = labels[:, 1:]
         if attention_mask is not None:   # pad targets ignored (-100)
             tgt = tgt.masked_fill(attention_mask[:, 1:] == 0, -100)
-        tgt = tgt.contiguous().view(-1)
-        loss = ops.cross_entropy_loss(
-            logits.reshape(-1, self.cfg.vocab_size), tgt)
+        B = input_ids.shape[0]
+        loss, logits = ops.lm_head_ce(x[:, :-1, :].contiguous(),
+                                      self._head(),
+                                      tgt.contiguous().view(-1))
         return CausalLMOutput(
             loss=loss,
-            logits=logits if (return_logits or not self.training) else None)
+            logits=(logits.view(B, S - 1, self.cfg.vocab_size)
+                    if (return_logits or not self.training) else None))
 
 
 def _llama_block_attn_cached(blk: LlamaBlock, x: torch.Tensor, cos, sin,

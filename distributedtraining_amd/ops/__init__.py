@@ -33,7 +33,7 @@ __all__ = [
     "layer_norm", "rms_norm", "gelu", "swiglu", "causal_attention",
     "qkv_attention", "linear", "mlp_gelu", "add_layer_norm",
     "add_rms_norm",
-    "cross_entropy_loss",
+    "cross_entropy_loss", "lm_head_ce",
     "embedding_fwd",
     "decode_attention", "rope", "adamw_step", "delta_sub", "axpy_",
     "weighted_merge",
@@ -690,6 +690,96 @@ def cross_entropy_loss(logits: torch.Tensor, targets: torch.Tensor,
     if use_hip(logits):
         return _CrossEntropyFn.apply(logits, targets, ignore_index)
     return F.cross_entropy(logits.float(), targets, ignore_index=ignore_index)
+
+
+# --------------------------------------------------------------------------
+# LM head + CE as one node, with a PIPELINED forward: the head GEMM runs
+# tile-by-tile and each tile's online-softmax contribution is consumed on
+# a side stream while the tile is still L2/Infinity-Cache resident —
+# removing the 6.5 GB logits HBM re-read of the one-shot ce_fwd at the
+# flagship shape AND overlapping the CE math with the next GEMM tile.
+# Logits stay fully materialized (backward's ce_bwd + head GEMMs need
+# them). Tile sizes keep two tiles inside the 256 MiB Infinity Cache.
+# --------------------------------------------------------------------------
+_CE_PIPE = _os.environ.get("DTA_CE_PIPELINE", "1") == "1"
+_CE_TT = int(_os.environ.get("DTA_CE_TT", "8192"))
+_CE_VC = int(_os.environ.get("DTA_CE_VC", "4224"))
+_ce_stream: Optional[torch.cuda.Stream] = None
+
+
+def _ce_side_stream() -> torch.cuda.Stream:
+    global _ce_stream
+    if _ce_stream is None:
+        _ce_stream = torch.cuda.Stream()
+    return _ce_stream
+
+
+class _LMHeadCEFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x2, w, targets, ignore_index):
+        m = require_ext()
+        T, _K = x2.shape
+        V = w.shape[0]
+        logits = torch.empty(T, V, dtype=x2.dtype, device=x2.device)
+        if _CE_PIPE and T >= 2 * _CE_TT:
+            mr = torch.full((T,), float("-inf"), dtype=torch.float32,
+                            device=x2.device)
+            sr = torch.zeros(T, dtype=torch.float32, device=x2.device)
+            main = torch.cuda.current_stream()
+            side = _ce_side_stream()
+            for t0 in range(0, T, _CE_TT):
+                te = min(t0 + _CE_TT, T)
+                xt = x2[t0:te]
+                for v0 in range(0, V, _CE_VC):
+                    ve = min(v0 + _CE_VC, V)
+                    torch.mm(xt, w[v0:ve].t(), out=logits[t0:te, v0:ve])
+                    side.wait_stream(main)
+                    with torch.cuda.stream(side):
+                        m.ce_chunk(logits[t0:te, v0:ve], mr[t0:te],
+                                   sr[t0:te])
+            main.wait_stream(side)
+            loss_sum, lse, count = m.ce_finalize(logits, targets, mr, sr,
+                                                 ignore_index)
+        else:
+            torch.mm(x2, w.t(), out=logits)
+            loss_sum, lse, count = m.ce_fwd(logits, targets, ignore_index)
+        countf = count.clamp(min=1).to(torch.float32)
+        ctx.save_for_backward(x2, w, logits, targets, lse, countf)
+        ctx.ignore_index = ignore_index
+        ctx.wgrad = getattr(w, "main_grad", None)
+        ctx.mark_non_differentiable(logits)
+        return loss_sum / countf, logits
+
+    @staticmethod
+    def backward(ctx, dloss, _dlogits):
+        m = require_ext()
+        x2, w, logits, targets, lse, countf = ctx.saved_tensors
+        scale_dev = (dloss.to(torch.float32) / countf).reshape(1)
+        dlog = m.ce_bwd(logits, targets, lse, scale_dev, 0.0,
+                        ctx.ignore_index)
+        dx = dlog.mm(w)
+        if ctx.wgrad is not None:
+            ctx.wgrad.addmm_(dlog.t(), x2)   # flat-plane accumulation
+            dw = None
+        else:
+            dw = dlog.t().mm(x2)
+        return dx, dw, None, None
+
+
+def lm_head_ce(x: torch.Tensor, w: torch.Tensor, targets: torch.Tensor,
+               ignore_index: int = -100):
+    """(loss, logits) = mean-CE(x @ wᵀ, targets) as one node. x [..., E]
+    is flattened to [T, E]; logits come back [T, V] (non-differentiable
+    reference — backward runs through the fused node)."""
+    x2 = x.reshape(-1, x.shape[-1])
+    if use_hip(x):
+        if not x2.is_contiguous():
+            x2 = x2.contiguous()
+        return _LMHeadCEFn.apply(x2, w, targets, ignore_index)
+    logits = F.linear(x2, w)
+    loss = F.cross_entropy(logits.float(), targets,
+                           ignore_index=ignore_index)
+    return loss, logits
 
 
 # --------------------------------------------------------------------------

@@ -353,6 +353,34 @@ std::vector<Tensor> ce_fwd(Tensor logits, Tensor targets,
   return {loss.squeeze(0), lse, count.squeeze(0)};
 }
 
+// pipelined head-GEMM+CE forward helpers: ce_chunk folds one logits tile
+// (a strided [rows, cols] view with leading dim ld) into per-row running
+// (m, s); ce_finalize emits lse / loss_sum / count.
+void ce_chunk(Tensor chunk, Tensor m_run, Tensor s_run) {
+  TORCH_CHECK(chunk.is_cuda() && chunk.scalar_type() == torch::kBFloat16 &&
+                  chunk.dim() == 2 && chunk.stride(1) == 1,
+              "chunk must be a bf16 [rows, cols] view, contiguous cols");
+  check_f32(m_run, "m_run"); check_f32(s_run, "s_run");
+  launch_ce_chunk(bfp(chunk), chunk.stride(0), chunk.size(0), chunk.size(1),
+                  m_run.data_ptr<float>(), s_run.data_ptr<float>(),
+                  stream());
+}
+
+std::vector<Tensor> ce_finalize(Tensor logits, Tensor targets, Tensor m_run,
+                                Tensor s_run, int64_t ignore_index) {
+  check_bf16(logits, "logits");
+  const int64_t rows = logits.size(0), vocab = logits.size(1);
+  auto lse = torch::empty({rows}, logits.options().dtype(torch::kFloat32));
+  auto loss = torch::zeros({1}, logits.options().dtype(torch::kFloat32));
+  auto count = torch::zeros({1}, logits.options().dtype(torch::kInt32));
+  launch_ce_finalize(bfp(logits), targets.data_ptr<int64_t>(),
+                     m_run.data_ptr<float>(), s_run.data_ptr<float>(), rows,
+                     vocab, ignore_index, lse.data_ptr<float>(),
+                     loss.data_ptr<float>(), count.data_ptr<int>(),
+                     stream());
+  return {loss.squeeze(0), lse, count.squeeze(0)};
+}
+
 // scale_dev: empty tensor => host 'scale' scalar; else a 0/1-dim fp32
 // device scalar (dloss/count), keeping backward free of host syncs.
 Tensor ce_bwd(Tensor logits, Tensor targets, Tensor lse, Tensor scale_dev,
@@ -643,6 +671,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("grad_merge_weights", &grad_merge_weights);
   m.def("ce_fwd", &ce_fwd);
   m.def("ce_bwd", &ce_bwd);
+  m.def("ce_chunk", &ce_chunk);
+  m.def("ce_finalize", &ce_finalize);
   m.def("embedding_fwd", &embedding_fwd);
   m.def("embedding_bwd", &embedding_bwd);
   m.def("rope_fwd", &rope_fwd);

@@ -112,7 +112,107 @@ __global__ void ce_bwd_k(const ushort* __restrict__ logits,
   }
 }
 
+// ---- pipelined head-GEMM + CE forward -------------------------------------
+// The LM-head logits ([tokens, 50k] bf16, ~6.5 GB at the flagship shape)
+// are produced tile-by-tile by the head GEMM; consuming each tile's
+// online-softmax contribution IMMEDIATELY after its GEMM (while the tile
+// is L2/Infinity-Cache resident) removes the full-logits HBM re-read of
+// the one-shot ce_fwd. State: per-row running (m, s) in exp2 units.
+__global__ void ce_chunk_k(const ushort* __restrict__ chunk, int64_t ld,
+                           int64_t rows, int64_t cols,
+                           float* __restrict__ m_run,
+                           float* __restrict__ s_run) {
+  __shared__ float lds[16];
+  for (int64_t row = blockIdx.x; row < rows; row += gridDim.x) {
+    const ushort* xr = chunk + row * ld;
+    float m_l = -INFINITY, s_l = 0.f;
+    int64_t i = int64_t(threadIdx.x) * 8;
+    const int64_t stride = int64_t(CE_BLOCK) * 8;
+    for (; i + 8 <= cols; i += stride) {
+      s16x8 vx = *reinterpret_cast<const s16x8*>(xr + i);
+      float f[8];
+#pragma unroll
+      for (int j = 0; j < 8; ++j) f[j] = bf2f(ushort(vx[j])) * LOG2E;
+      float mx = fmaxf(fmaxf(fmaxf(f[0], f[1]), fmaxf(f[2], f[3])),
+                       fmaxf(fmaxf(f[4], f[5]), fmaxf(f[6], f[7])));
+      const float m_new = fmaxf(m_l, mx);
+      float ps = 0.f;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) ps += __builtin_exp2f(f[j] - m_new);
+      s_l = s_l * __builtin_exp2f(m_l - m_new) + ps;
+      m_l = m_new;
+    }
+    if (i < cols && i + 8 > cols)
+      for (; i < cols; ++i) {
+        const float f = bf2f(xr[i]) * LOG2E;
+        const float m_new = fmaxf(m_l, f);
+        s_l = s_l * __builtin_exp2f(m_l - m_new) +
+              __builtin_exp2f(f - m_new);
+        m_l = m_new;
+      }
+    const float M = block_max<16>(m_l, lds);
+    const float S = block_sum<16>(s_l * __builtin_exp2f(m_l - M), lds);
+    if (threadIdx.x == 0) {
+      const float m0 = m_run[row], s0 = s_run[row];
+      const float mn = fmaxf(m0, M);
+      // first chunk: s0 == 0 (the -inf m0 exp2 underflows to 0 cleanly)
+      s_run[row] = s0 * __builtin_exp2f(m0 - mn) +
+                   S * __builtin_exp2f(M - mn);
+      m_run[row] = mn;
+    }
+    __syncthreads();
+  }
+}
+
+// lse[t] = ln-units lse from (m, s); loss_sum/count accumulated over
+// non-ignored rows with one target-logit gather per row.
+__global__ void ce_finalize_k(const ushort* __restrict__ logits,
+                              const int64_t* __restrict__ targets,
+                              const float* __restrict__ m_run,
+                              const float* __restrict__ s_run, int64_t rows,
+                              int64_t vocab, int64_t ignore_index,
+                              float* __restrict__ lse,
+                              float* __restrict__ loss_sum,
+                              int* __restrict__ count) {
+  float part = 0.f;
+  int c = 0;
+  for (int64_t t = int64_t(blockIdx.x) * blockDim.x + threadIdx.x; t < rows;
+       t += int64_t(gridDim.x) * blockDim.x) {
+    const float l = (m_run[t] + __builtin_log2f(s_run[t])) * LN2;
+    lse[t] = l;
+    const int64_t tg = targets[t];
+    if (tg != ignore_index) {
+      part += l - bf2f(logits[t * vocab + tg]);
+      ++c;
+    }
+  }
+  part = wave_sum(part);
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) c += __shfl_xor(c, off);
+  if ((threadIdx.x & 63) == 0) {
+    atomicAdd(loss_sum, part);
+    atomicAdd(count, c);
+  }
+}
+
 }  // namespace
+
+void launch_ce_chunk(const bf16_t* chunk, int64_t ld, int64_t rows,
+                     int64_t cols, float* m_run, float* s_run,
+                     hipStream_t s) {
+  const int grid = int(rows < 4096 ? (rows > 0 ? rows : 1) : 4096);
+  ce_chunk_k<<<grid, CE_BLOCK, 0, s>>>(chunk, ld, rows, cols, m_run, s_run);
+}
+
+void launch_ce_finalize(const bf16_t* logits, const int64_t* targets,
+                        const float* m_run, const float* s_run, int64_t rows,
+                        int64_t vocab, int64_t ignore_index, float* lse,
+                        float* loss_sum, int* count, hipStream_t s) {
+  const int grid = elementwise_grid(rows, 256, 1);
+  ce_finalize_k<<<grid, 256, 0, s>>>(logits, targets, m_run, s_run, rows,
+                                     vocab, ignore_index, lse, loss_sum,
+                                     count);
+}
 
 void launch_ce_fwd(const bf16_t* logits, const int64_t* targets, int64_t rows,
                    int64_t vocab, int64_t ignore_index, float* lse,

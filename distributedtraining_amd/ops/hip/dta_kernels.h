@@ -121,6 +121,14 @@ void launch_ce_bwd(const bf16_t* logits, const int64_t* targets,
                    const float* lse, float scale, const float* scale_p,
                    int64_t ignore_index, bf16_t* dlogits, int64_t rows,
                    int64_t vocab, hipStream_t s);
+// pipelined head-GEMM+CE: per-tile online-softmax update + finalize
+void launch_ce_chunk(const bf16_t* chunk, int64_t ld, int64_t rows,
+                     int64_t cols, float* m_run, float* s_run,
+                     hipStream_t s);
+void launch_ce_finalize(const bf16_t* logits, const int64_t* targets,
+                        const float* m_run, const float* s_run, int64_t rows,
+                        int64_t vocab, int64_t ignore_index, float* lse,
+                        float* loss_sum, int* count, hipStream_t s);
 
 // ---- embedding ------------------------------------------------------------
 void launch_embedding_fwd(const int64_t* ids, const bf16_t* wte,

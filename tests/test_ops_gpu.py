@@ -944,3 +944,42 @@ def test_add_rms_norm_fused_dropout():
     torch.testing.assert_close(res.grad.float().cpu(), rr.grad,
                                rtol=5e-2, atol=5e-2)
     assert torch.all(res.grad.cpu()[mask.cpu() == 0] == 0)
+
+
+def test_lm_head_ce_pipelined_matches_oneshot():
+    """The pipelined (tiled GEMM + side-stream online-softmax) head+CE
+    forward must equal the one-shot path, and backward must match the
+    fp32 torch reference."""
+    from distributedtraining_amd import ops
+    T, E, V = 4096, 256, 3000
+    x = _rand_bf16(T, E, seed=70, scale=0.5).requires_grad_(True)
+    w = _rand_bf16(V, E, seed=71, scale=0.02).requires_grad_(True)
+    tgt = torch.randint(0, V, (T,), device=DEV)
+    tgt[::17] = -100   # exercise ignore_index
+    # pipelined (shrink tiles so the path engages at test scale)
+    tt, vc, pipe = ops._CE_TT, ops._CE_VC, ops._CE_PIPE
+    try:
+        ops._CE_TT, ops._CE_VC, ops._CE_PIPE = 1024, 640, True
+        loss_p, logits_p = ops.lm_head_ce(x, w, tgt)
+        loss_p.backward()
+        gx_p, gw_p = x.grad.clone(), w.grad.clone()
+        x.grad = w.grad = None
+        ops._CE_PIPE = False
+        loss_o, logits_o = ops.lm_head_ce(x, w, tgt)
+        loss_o.backward()
+    finally:
+        ops._CE_TT, ops._CE_VC, ops._CE_PIPE = tt, vc, pipe
+    torch.testing.assert_close(loss_p, loss_o, rtol=1e-3, atol=1e-3)
+    torch.testing.assert_close(logits_p, logits_o, rtol=0, atol=0)
+    torch.testing.assert_close(gx_p, x.grad, rtol=2e-2, atol=2e-2)
+    torch.testing.assert_close(gw_p, w.grad, rtol=2e-2, atol=2e-2)
+    # fp32 torch reference
+    xr = x.detach().cpu().float().requires_grad_(True)
+    wr = w.detach().cpu().float().requires_grad_(True)
+    lr = torch.nn.functional.cross_entropy(xr.mm(wr.t()), tgt.cpu(),
+                                           ignore_index=-100)
+    torch.testing.assert_close(loss_p.float().cpu(), lr.detach(),
+                               rtol=2e-2, atol=2e-2)
+    lr.backward()
+    torch.testing.assert_close(gx_p.float().cpu(), xr.grad,
+                               rtol=5e-2, atol=5e-2)
