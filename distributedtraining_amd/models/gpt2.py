@@ -157,12 +157,14 @@ class GPT2LM(nn.Module):
         if attention_mask is not None:
             tgt = tgt.masked_fill(attention_mask[:, 1:] == 0, -100)
         B, S = input_ids.shape
+        want_logits = bool(return_logits) or not self.training
         loss, logits = ops.lm_head_ce(x[:, :-1, :].contiguous(), self.wte,
-                                      tgt.contiguous().view(-1))
+                                      tgt.contiguous().view(-1),
+                                      need_logits=want_logits)
         return CausalLMOutput(
             loss=loss,
             logits=(logits.view(B, S - 1, self.cfg.vocab_size)
-                    if (return_logits or not self.training) else None))
+                    if (want_logits and logits is not None) else None))
 
 
 class KVCache:

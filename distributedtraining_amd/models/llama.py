@@ -132,13 +132,15 @@ class LlamaLM(nn.Module):
         if attention_mask is not None:   # pad targets ignored (-100)
             tgt = tgt.masked_fill(attention_mask[:, 1:] == 0, -100)
         B = input_ids.shape[0]
+        want_logits = bool(return_logits) or not self.training
         loss, logits = ops.lm_head_ce(x[:, :-1, :].contiguous(),
                                       self._head(),
-                                      tgt.contiguous().view(-1))
+                                      tgt.contiguous().view(-1),
+                                      need_logits=want_logits)
         return CausalLMOutput(
             loss=loss,
             logits=(logits.view(B, S - 1, self.cfg.vocab_size)
-                    if (return_logits or not self.training) else None))
+                    if (want_logits and logits is not None) else None))
 
 
 def _llama_block_attn_cached(blk: LlamaBlock, x: torch.Tensor, cos, sin,

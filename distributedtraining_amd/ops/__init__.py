@@ -680,7 +680,7 @@ class _CrossEntropyFn(torch.autograd.Function):
         logits, targets, lse, countf = ctx.saved_tensors
         scale_dev = (dloss.to(torch.float32) / countf).reshape(1)
         dlogits = m.ce_bwd(logits, targets, lse, scale_dev, 0.0,
-                           ctx.ignore_index)
+                           ctx.ignore_index, 0, True)
         return dlogits, None, None
 
 
@@ -716,15 +716,23 @@ def _ce_side_stream() -> torch.cuda.Stream:
 
 class _LMHeadCEFn(torch.autograd.Function):
     @staticmethod
-    def forward(ctx, x2, w, targets, ignore_index):
+    def forward(ctx, x2, w, targets, ignore_index, need_logits):
         m = require_ext()
         T, _K = x2.shape
         V = w.shape[0]
-        logits = torch.empty(T, V, dtype=x2.dtype, device=x2.device)
-        if _CE_PIPE and T >= 2 * _CE_TT:
+        pipe = (_CE_PIPE and not need_logits and T >= 2 * _CE_TT
+                and torch.is_grad_enabled())
+        if pipe:
+            # tiled: per-(token, vocab) tile CONTIGUOUS buffers — the GEMM
+            # writes a tile, the side stream folds its online-softmax
+            # contribution while the next GEMM runs, and backward re-walks
+            # the tiles so each dlogits tile feeds its two head GEMMs
+            # straight out of cache.
             mr = torch.full((T,), float("-inf"), dtype=torch.float32,
                             device=x2.device)
             sr = torch.zeros(T, dtype=torch.float32, device=x2.device)
+            tl = torch.zeros(T, dtype=torch.float32, device=x2.device)
+            tiles = []
             main = torch.cuda.current_stream()
             side = _ce_side_stream()
             for t0 in range(0, T, _CE_TT):
@@ -732,20 +740,27 @@ class _LMHeadCEFn(torch.autograd.Function):
                 xt = x2[t0:te]
                 for v0 in range(0, V, _CE_VC):
                     ve = min(v0 + _CE_VC, V)
-                    torch.mm(xt, w[v0:ve].t(), out=logits[t0:te, v0:ve])
+                    tile = torch.empty(te - t0, ve - v0, dtype=x2.dtype,
+                                       device=x2.device)
+                    torch.mm(xt, w[v0:ve].t(), out=tile)
                     side.wait_stream(main)
                     with torch.cuda.stream(side):
-                        m.ce_chunk(logits[t0:te, v0:ve], mr[t0:te],
-                                   sr[t0:te])
+                        m.ce_chunk(tile, targets[t0:te], v0, tl[t0:te],
+                                   mr[t0:te], sr[t0:te])
+                    tiles.append(tile)
             main.wait_stream(side)
-            loss_sum, lse, count = m.ce_finalize(logits, targets, mr, sr,
+            loss_sum, lse, count = m.ce_finalize(targets, mr, sr, tl,
                                                  ignore_index)
+            logits = x2.new_empty(0)
         else:
-            torch.mm(x2, w.t(), out=logits)
+            logits = x2.mm(w.t())
             loss_sum, lse, count = m.ce_fwd(logits, targets, ignore_index)
+            tiles = [logits]
         countf = count.clamp(min=1).to(torch.float32)
-        ctx.save_for_backward(x2, w, logits, targets, lse, countf)
+        ctx.save_for_backward(x2, w, targets, lse, countf, *tiles)
         ctx.ignore_index = ignore_index
+        ctx.pipe = pipe
+        ctx.dims = (T, V)
         ctx.wgrad = getattr(w, "main_grad", None)
         ctx.mark_non_differentiable(logits)
         return loss_sum / countf, logits
@@ -753,29 +768,56 @@ class _LMHeadCEFn(torch.autograd.Function):
     @staticmethod
     def backward(ctx, dloss, _dlogits):
         m = require_ext()
-        x2, w, logits, targets, lse, countf = ctx.saved_tensors
+        x2, w, targets, lse, countf, *tiles = ctx.saved_tensors
         scale_dev = (dloss.to(torch.float32) / countf).reshape(1)
-        dlog = m.ce_bwd(logits, targets, lse, scale_dev, 0.0,
-                        ctx.ignore_index)
-        dx = dlog.mm(w)
-        if ctx.wgrad is not None:
-            ctx.wgrad.addmm_(dlog.t(), x2)   # flat-plane accumulation
-            dw = None
-        else:
-            dw = dlog.t().mm(x2)
-        return dx, dw, None, None
+        ii = ctx.ignore_index
+        if not ctx.pipe:
+            dlog = m.ce_bwd(tiles[0], targets, lse, scale_dev, 0.0, ii,
+                            0, True)
+            dx = dlog.mm(w)
+            if ctx.wgrad is not None:
+                ctx.wgrad.addmm_(dlog.t(), x2)   # flat-plane accumulation
+                dw = None
+            else:
+                dw = dlog.t().mm(x2)
+            return dx, dw, None, None, None
+        T, V = ctx.dims
+        dx = torch.empty_like(x2)
+        dw = None if ctx.wgrad is not None else torch.zeros_like(w)
+        i = 0
+        for t0 in range(0, T, _CE_TT):
+            te = min(t0 + _CE_TT, T)
+            first = True
+            for v0 in range(0, V, _CE_VC):
+                ve = min(v0 + _CE_VC, V)
+                # no nontemporal: dlog is re-read by both GEMMs right away
+                dlog = m.ce_bwd(tiles[i], targets[t0:te], lse[t0:te],
+                                scale_dev, 0.0, ii, v0, False)
+                i += 1
+                if first:
+                    torch.mm(dlog, w[v0:ve], out=dx[t0:te])
+                    first = False
+                else:
+                    dx[t0:te].addmm_(dlog, w[v0:ve])
+                if ctx.wgrad is not None:
+                    ctx.wgrad[v0:ve].addmm_(dlog.t(), x2[t0:te])
+                else:
+                    dw[v0:ve].addmm_(dlog.t(), x2[t0:te])
+        return dx, dw, None, None, None
 
 
 def lm_head_ce(x: torch.Tensor, w: torch.Tensor, targets: torch.Tensor,
-               ignore_index: int = -100):
+               ignore_index: int = -100, need_logits: bool = True):
     """(loss, logits) = mean-CE(x @ wᵀ, targets) as one node. x [..., E]
-    is flattened to [T, E]; logits come back [T, V] (non-differentiable
-    reference — backward runs through the fused node)."""
+    is flattened to [T, E]. With ``need_logits=False`` (training) the
+    pipelined tiled path may engage and logits come back as ``None``."""
     x2 = x.reshape(-1, x.shape[-1])
     if use_hip(x):
         if not x2.is_contiguous():
             x2 = x2.contiguous()
-        return _LMHeadCEFn.apply(x2, w, targets, ignore_index)
+        loss, logits = _LMHeadCEFn.apply(x2, w, targets, ignore_index,
+                                         need_logits)
+        return loss, (logits if logits.numel() else None)
     logits = F.linear(x2, w)
     loss = F.cross_entropy(logits.float(), targets,
                            ignore_index=ignore_index)

@@ -354,28 +354,31 @@ std::vector<Tensor> ce_fwd(Tensor logits, Tensor targets,
 }
 
 // pipelined head-GEMM+CE forward helpers: ce_chunk folds one logits tile
-// (a strided [rows, cols] view with leading dim ld) into per-row running
-// (m, s); ce_finalize emits lse / loss_sum / count.
-void ce_chunk(Tensor chunk, Tensor m_run, Tensor s_run) {
+// (a [rows, cols] view with leading dim ld, vocab offset v0) into per-row
+// running (m, s) and records the target logit; ce_finalize emits
+// lse / loss_sum / count from the state alone.
+void ce_chunk(Tensor chunk, Tensor targets, int64_t v0, Tensor tlogit,
+              Tensor m_run, Tensor s_run) {
   TORCH_CHECK(chunk.is_cuda() && chunk.scalar_type() == torch::kBFloat16 &&
                   chunk.dim() == 2 && chunk.stride(1) == 1,
               "chunk must be a bf16 [rows, cols] view, contiguous cols");
   check_f32(m_run, "m_run"); check_f32(s_run, "s_run");
+  check_f32(tlogit, "tlogit");
   launch_ce_chunk(bfp(chunk), chunk.stride(0), chunk.size(0), chunk.size(1),
-                  m_run.data_ptr<float>(), s_run.data_ptr<float>(),
-                  stream());
+                  targets.data_ptr<int64_t>(), v0,
+                  tlogit.data_ptr<float>(), m_run.data_ptr<float>(),
+                  s_run.data_ptr<float>(), stream());
 }
 
-std::vector<Tensor> ce_finalize(Tensor logits, Tensor targets, Tensor m_run,
-                                Tensor s_run, int64_t ignore_index) {
-  check_bf16(logits, "logits");
-  const int64_t rows = logits.size(0), vocab = logits.size(1);
-  auto lse = torch::empty({rows}, logits.options().dtype(torch::kFloat32));
-  auto loss = torch::zeros({1}, logits.options().dtype(torch::kFloat32));
-  auto count = torch::zeros({1}, logits.options().dtype(torch::kInt32));
-  launch_ce_finalize(bfp(logits), targets.data_ptr<int64_t>(),
-                     m_run.data_ptr<float>(), s_run.data_ptr<float>(), rows,
-                     vocab, ignore_index, lse.data_ptr<float>(),
+std::vector<Tensor> ce_finalize(Tensor targets, Tensor m_run, Tensor s_run,
+                                Tensor tlogit, int64_t ignore_index) {
+  const int64_t rows = m_run.numel();
+  auto lse = torch::empty({rows}, m_run.options());
+  auto loss = torch::zeros({1}, m_run.options());
+  auto count = torch::zeros({1}, m_run.options().dtype(torch::kInt32));
+  launch_ce_finalize(targets.data_ptr<int64_t>(), m_run.data_ptr<float>(),
+                     s_run.data_ptr<float>(), tlogit.data_ptr<float>(),
+                     rows, ignore_index, lse.data_ptr<float>(),
                      loss.data_ptr<float>(), count.data_ptr<int>(),
                      stream());
   return {loss.squeeze(0), lse, count.squeeze(0)};
@@ -383,15 +386,16 @@ std::vector<Tensor> ce_finalize(Tensor logits, Tensor targets, Tensor m_run,
 
 // scale_dev: empty tensor => host 'scale' scalar; else a 0/1-dim fp32
 // device scalar (dloss/count), keeping backward free of host syncs.
+// v0/nt: vocab-tile offset + store policy for the tiled pipeline.
 Tensor ce_bwd(Tensor logits, Tensor targets, Tensor lse, Tensor scale_dev,
-              double scale, int64_t ignore_index) {
+              double scale, int64_t ignore_index, int64_t v0, bool nt) {
   check_bf16(logits, "logits");
   const int64_t rows = logits.size(0), vocab = logits.size(1);
   auto dl = torch::empty_like(logits);
   const float* scp = scale_dev.numel() ? scale_dev.data_ptr<float>() : nullptr;
   launch_ce_bwd(bfp(logits), targets.data_ptr<int64_t>(),
-                lse.data_ptr<float>(), float(scale), scp, ignore_index,
-                bfp_mut(dl), rows, vocab, stream());
+                lse.data_ptr<float>(), float(scale), scp, ignore_index, v0,
+                nt, bfp_mut(dl), rows, vocab, stream());
   return dl;
 }
 

@@ -70,20 +70,27 @@ __global__ void ce_fwd_k(const ushort* __restrict__ logits,
 
 // scale_p: optional device pointer (dloss/count computed on device so the
 // whole backward is hipGraph-capturable with no host sync); if null the
-// host 'scale' scalar is used.
+// host 'scale' scalar is used. v0: vocab-tile offset (the onehot column
+// is targets[row] - v0). NT: nontemporal dlogits stores — right for the
+// one-shot 6.5 GB pass (keeps the logits read stream L2-resident), wrong
+// for the tiled pipeline (the tile is re-read by the two head GEMMs
+// immediately, so it should STAY in cache).
+template <bool NT>
 __global__ void ce_bwd_k(const ushort* __restrict__ logits,
                          const int64_t* __restrict__ targets,
                          const float* __restrict__ lse, float scale,
                          const float* __restrict__ scale_p,
-                         int64_t ignore_index, ushort* __restrict__ dlogits,
-                         int64_t rows, int64_t vocab) {
+                         int64_t ignore_index, int64_t v0,
+                         ushort* __restrict__ dlogits, int64_t rows,
+                         int64_t vocab) {
   const float sc_base = scale_p ? *scale_p : scale;
   for (int64_t row = blockIdx.x; row < rows; row += gridDim.x) {
     const ushort* xr = logits + row * vocab;
     ushort* dxr = dlogits + row * vocab;
-    const int64_t tgt = targets[row];
+    const int64_t tgt_raw = targets[row];
+    const int64_t tgt = tgt_raw - v0;
     const float lb = lse[row] * LOG2E;
-    const float sc = (tgt == ignore_index) ? 0.f : sc_base;
+    const float sc = (tgt_raw == ignore_index) ? 0.f : sc_base;
     // two 16 B vectors in flight per iteration (load/store ILP)
     int64_t i = int64_t(threadIdx.x) * 16;
     const int64_t stride = int64_t(CE_BLOCK) * 16;
@@ -98,11 +105,14 @@ __global__ void ce_bwd_k(const ushort* __restrict__ logits,
         oa[j] = f2bf(sc * (pa - ((i + j) == tgt ? 1.f : 0.f)));
         ob[j] = f2bf(sc * (pb - ((i + 8 + j) == tgt ? 1.f : 0.f)));
       }
-      // nontemporal: dlogits is consumed by the big head GEMMs, keep it
-      // out of L2 so the logits read stream stays resident
-      __builtin_nontemporal_store(oa, reinterpret_cast<s16x8*>(dxr + i));
-      __builtin_nontemporal_store(ob,
-                                  reinterpret_cast<s16x8*>(dxr + i + 8));
+      if (NT) {
+        __builtin_nontemporal_store(oa, reinterpret_cast<s16x8*>(dxr + i));
+        __builtin_nontemporal_store(ob,
+                                    reinterpret_cast<s16x8*>(dxr + i + 8));
+      } else {
+        *reinterpret_cast<s16x8*>(dxr + i) = oa;
+        *reinterpret_cast<s16x8*>(dxr + i + 8) = ob;
+      }
     }
     if (i < vocab)
       for (; i < vocab; ++i) {
@@ -118,8 +128,13 @@ __global__ void ce_bwd_k(const ushort* __restrict__ logits,
 // online-softmax contribution IMMEDIATELY after its GEMM (while the tile
 // is L2/Infinity-Cache resident) removes the full-logits HBM re-read of
 // the one-shot ce_fwd. State: per-row running (m, s) in exp2 units.
+// targets/v0/tlogit: the tile holding a row's target column records the
+// target logit into tlogit[row] (exactly one tile matches per row), so
+// finalize never re-touches the logits.
 __global__ void ce_chunk_k(const ushort* __restrict__ chunk, int64_t ld,
                            int64_t rows, int64_t cols,
+                           const int64_t* __restrict__ targets, int64_t v0,
+                           float* __restrict__ tlogit,
                            float* __restrict__ m_run,
                            float* __restrict__ s_run) {
   __shared__ float lds[16];
@@ -159,18 +174,20 @@ __global__ void ce_chunk_k(const ushort* __restrict__ chunk, int64_t ld,
       s_run[row] = s0 * __builtin_exp2f(m0 - mn) +
                    S * __builtin_exp2f(M - mn);
       m_run[row] = mn;
+      const int64_t tc = targets[row] - v0;
+      if (tc >= 0 && tc < cols) tlogit[row] = bf2f(xr[tc]);
     }
     __syncthreads();
   }
 }
 
 // lse[t] = ln-units lse from (m, s); loss_sum/count accumulated over
-// non-ignored rows with one target-logit gather per row.
-__global__ void ce_finalize_k(const ushort* __restrict__ logits,
-                              const int64_t* __restrict__ targets,
+// non-ignored rows using the tlogit recorded by the chunks.
+__global__ void ce_finalize_k(const int64_t* __restrict__ targets,
                               const float* __restrict__ m_run,
-                              const float* __restrict__ s_run, int64_t rows,
-                              int64_t vocab, int64_t ignore_index,
+                              const float* __restrict__ s_run,
+                              const float* __restrict__ tlogit, int64_t rows,
+                              int64_t ignore_index,
                               float* __restrict__ lse,
                               float* __restrict__ loss_sum,
                               int* __restrict__ count) {
@@ -182,7 +199,7 @@ __global__ void ce_finalize_k(const ushort* __restrict__ logits,
     lse[t] = l;
     const int64_t tg = targets[t];
     if (tg != ignore_index) {
-      part += l - bf2f(logits[t * vocab + tg]);
+      part += l - tlogit[t];
       ++c;
     }
   }
@@ -198,20 +215,21 @@ __global__ void ce_finalize_k(const ushort* __restrict__ logits,
 }  // namespace
 
 void launch_ce_chunk(const bf16_t* chunk, int64_t ld, int64_t rows,
-                     int64_t cols, float* m_run, float* s_run,
+                     int64_t cols, const int64_t* targets, int64_t v0,
+                     float* tlogit, float* m_run, float* s_run,
                      hipStream_t s) {
   const int grid = int(rows < 4096 ? (rows > 0 ? rows : 1) : 4096);
-  ce_chunk_k<<<grid, CE_BLOCK, 0, s>>>(chunk, ld, rows, cols, m_run, s_run);
+  ce_chunk_k<<<grid, CE_BLOCK, 0, s>>>(chunk, ld, rows, cols, targets, v0,
+                                       tlogit, m_run, s_run);
 }
 
-void launch_ce_finalize(const bf16_t* logits, const int64_t* targets,
-                        const float* m_run, const float* s_run, int64_t rows,
-                        int64_t vocab, int64_t ignore_index, float* lse,
+void launch_ce_finalize(const int64_t* targets, const float* m_run,
+                        const float* s_run, const float* tlogit,
+                        int64_t rows, int64_t ignore_index, float* lse,
                         float* loss_sum, int* count, hipStream_t s) {
   const int grid = elementwise_grid(rows, 256, 1);
-  ce_finalize_k<<<grid, 256, 0, s>>>(logits, targets, m_run, s_run, rows,
-                                     vocab, ignore_index, lse, loss_sum,
-                                     count);
+  ce_finalize_k<<<grid, 256, 0, s>>>(targets, m_run, s_run, tlogit, rows,
+                                     ignore_index, lse, loss_sum, count);
 }
 
 void launch_ce_fwd(const bf16_t* logits, const int64_t* targets, int64_t rows,
@@ -224,9 +242,16 @@ void launch_ce_fwd(const bf16_t* logits, const int64_t* targets, int64_t rows,
 
 void launch_ce_bwd(const bf16_t* logits, const int64_t* targets,
                    const float* lse, float scale, const float* scale_p,
-                   int64_t ignore_index, bf16_t* dlogits, int64_t rows,
-                   int64_t vocab, hipStream_t s) {
+                   int64_t ignore_index, int64_t v0, bool nt,
+                   bf16_t* dlogits, int64_t rows, int64_t vocab,
+                   hipStream_t s) {
   const int grid = int(rows < 4096 ? (rows > 0 ? rows : 1) : 4096);
-  ce_bwd_k<<<grid, CE_BLOCK, 0, s>>>(logits, targets, lse, scale, scale_p,
-                                     ignore_index, dlogits, rows, vocab);
+  if (nt)
+    ce_bwd_k<true><<<grid, CE_BLOCK, 0, s>>>(logits, targets, lse, scale,
+                                             scale_p, ignore_index, v0,
+                                             dlogits, rows, vocab);
+  else
+    ce_bwd_k<false><<<grid, CE_BLOCK, 0, s>>>(logits, targets, lse, scale,
+                                              scale_p, ignore_index, v0,
+                                              dlogits, rows, vocab);
 }

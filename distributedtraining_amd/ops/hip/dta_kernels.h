@@ -117,17 +117,21 @@ void launch_rmsnorm_bwd(const bf16_t* dy, const bf16_t* ds,
 void launch_ce_fwd(const bf16_t* logits, const int64_t* targets,
                    int64_t rows, int64_t vocab, int64_t ignore_index,
                    float* lse, float* loss_sum, int* count, hipStream_t s);
+// v0: vocab-tile offset for tiled dlogits; nt=false keeps the tile in
+// cache for the immediately following head GEMMs
 void launch_ce_bwd(const bf16_t* logits, const int64_t* targets,
                    const float* lse, float scale, const float* scale_p,
-                   int64_t ignore_index, bf16_t* dlogits, int64_t rows,
-                   int64_t vocab, hipStream_t s);
+                   int64_t ignore_index, int64_t v0, bool nt,
+                   bf16_t* dlogits, int64_t rows, int64_t vocab,
+                   hipStream_t s);
 // pipelined head-GEMM+CE: per-tile online-softmax update + finalize
 void launch_ce_chunk(const bf16_t* chunk, int64_t ld, int64_t rows,
-                     int64_t cols, float* m_run, float* s_run,
+                     int64_t cols, const int64_t* targets, int64_t v0,
+                     float* tlogit, float* m_run, float* s_run,
                      hipStream_t s);
-void launch_ce_finalize(const bf16_t* logits, const int64_t* targets,
-                        const float* m_run, const float* s_run, int64_t rows,
-                        int64_t vocab, int64_t ignore_index, float* lse,
+void launch_ce_finalize(const int64_t* targets, const float* m_run,
+                        const float* s_run, const float* tlogit,
+                        int64_t rows, int64_t ignore_index, float* lse,
                         float* loss_sum, int* count, hipStream_t s);
 
 // ---- embedding ------------------------------------------------------------
