@@ -124,6 +124,12 @@ def main() -> int:
     pool = [{k: t.to(device) for k, t in next(node.miner.data).items()}
             for _ in range(8)]
     graphed = None
+    if args.model == "llama3-8b":
+        # 8B: the hipGraph private memory pool (~80 GB for the captured
+        # fwd+bwd) pushes the 128 GB param/optimizer plane + base + delta
+        # past the 288 GB budget, and at ~0.5 s/step launch overhead is
+        # irrelevant — run eager.
+        args.no_graph = True
     if use_gpu and not args.no_graph:
         try:
             from distributedtraining_amd.parallel.graphstep import (

@@ -112,11 +112,11 @@ class LocalSGDNode:
             self._validator = DeltaValidator(self.model, self.fp,
                                              self.val_batches,
                                              self.cfg.validate)
-            self._validator_base = self.miner.base
-        elif self._validator_base is not self.miner.base:
+            self._validator_base = self.miner.base_version
+        elif self._validator_base != self.miner.base_version:
             v = self._validator
             v.base_loss, v.base_perplexity = v.evaluate_model()
-            self._validator_base = self.miner.base
+            self._validator_base = self.miner.base_version
         return self._validator
 
     def validation_round(self) -> Dict[str, float]:

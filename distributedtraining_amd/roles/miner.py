@@ -51,6 +51,7 @@ class DeltaLoop:
         self._loss_acc = None   # device-resident Σ(loss·batch)
         self.total_examples = 0
         self.base = fp.snapshot()
+        self.base_version = 0    # bumped on every install_base
         self._base_hash = None   # lazy: SHA-256 of 500 MB costs ~300 ms on host
         self.last_push_step = 0
         self.last_base_time = time.time()
@@ -72,11 +73,19 @@ class DeltaLoop:
 
     def install_base(self, flat_fp32: torch.Tensor) -> None:
         """New shared base: load, reinit optimizer, re-snapshot
-        (reference deliberately re-creates optimizer state, :371-373)."""
+        (reference deliberately re-creates optimizer state, :371-373).
+        The snapshot REUSES the existing base buffer (a fresh 32 GB clone
+        per merge round pushed Llama-3-8B over the HBM budget);
+        ``base_version`` tracks identity for cached-base-loss consumers."""
         self.fp.load_flat_master(flat_fp32)
         self.opt.reset_state()
         self.opt.zero_grad()
-        self.base = self.fp.snapshot()
+        if (self.base.shape == self.fp.master.shape
+                and self.base.device == self.fp.master.device):
+            self.base.copy_(self.fp.master)
+        else:
+            self.base = self.fp.snapshot()
+        self.base_version += 1
         self._base_hash = None
         self.last_base_time = time.time()
 

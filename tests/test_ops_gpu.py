@@ -202,10 +202,10 @@ def test_cross_entropy(rows, vocab):
     torch.nn.functional.cross_entropy(lr, targets, ignore_index=-100).backward()
     dl = m.ce_bwd(logits, targets, lse,
                   torch.empty(0, device=DEV, dtype=torch.float32),
-                  1.0 / n_valid, -100)
+                  1.0 / n_valid, -100, 0, True)
     # device-scale path must agree with the host-scalar path
     sdev = torch.tensor([1.0 / n_valid], device=DEV)
-    dl2 = m.ce_bwd(logits, targets, lse, sdev, 0.0, -100)
+    dl2 = m.ce_bwd(logits, targets, lse, sdev, 0.0, -100, 0, True)
     torch.testing.assert_close(dl2, dl)
     torch.testing.assert_close(dl.float(), lr.grad, rtol=5e-2,
                                atol=1e-4)
