@@ -716,12 +716,14 @@ def _ce_side_stream() -> torch.cuda.Stream:
 
 class _LMHeadCEFn(torch.autograd.Function):
     @staticmethod
-    def forward(ctx, x2, w, targets, ignore_index, need_logits):
+    def forward(ctx, x2, w, targets, ignore_index, allow_pipe):
+        # NOTE: grad mode is always OFF inside Function.forward — the
+        # caller decides pipe eligibility (allow_pipe) where grad state
+        # is visible.
         m = require_ext()
         T, _K = x2.shape
         V = w.shape[0]
-        pipe = (_CE_PIPE and not need_logits and T >= 2 * _CE_TT
-                and torch.is_grad_enabled())
+        pipe = _CE_PIPE and allow_pipe and T >= 2 * _CE_TT
         if pipe:
             # tiled: per-(token, vocab) tile CONTIGUOUS buffers — the GEMM
             # writes a tile, the side stream folds its online-softmax
@@ -815,8 +817,9 @@ def lm_head_ce(x: torch.Tensor, w: torch.Tensor, targets: torch.Tensor,
     if use_hip(x):
         if not x2.is_contiguous():
             x2 = x2.contiguous()
+        allow_pipe = not need_logits and torch.is_grad_enabled()
         loss, logits = _LMHeadCEFn.apply(x2, w, targets, ignore_index,
-                                         need_logits)
+                                         allow_pipe)
         return loss, (logits if logits.numel() else None)
     logits = F.linear(x2, w)
     loss = F.cross_entropy(logits.float(), targets,
