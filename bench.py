@@ -54,6 +54,11 @@ def main() -> int:
                    help="CPU smoke mode (tiny model, tests only)")
     p.add_argument("--no-preflight", action="store_true",
                    help="skip the RCCL collective-shape pre-flight")
+    p.add_argument("--pdrop", type=float, default=None,
+                   help="override the model's dropout probabilities "
+                        "(default: the config's reference-faithful values "
+                        "— GPT-2 trains with transformers pdrop 0.1; pass "
+                        "0 for a dropout-free ablation)")
     args = p.parse_args()
 
     from distributedtraining_amd.config import Config, ModelConfig
@@ -72,6 +77,10 @@ def main() -> int:
         cfg.model = ModelConfig.llama3_8b()
     else:
         cfg.model = ModelConfig.gpt2_small()
+    if args.pdrop is not None:
+        cfg.model.resid_pdrop = args.pdrop
+        cfg.model.embd_pdrop = args.pdrop
+        cfg.model.attn_pdrop = args.pdrop
     cfg.train.batch_size = args.batch_size
     cfg.train.seq_len = args.seq_len
     cfg.train.send_interval_steps = 10**9   # exchange via merge_round below
@@ -201,6 +210,7 @@ def main() -> int:
                 "merge_every": args.merge_every,
                 "merges_timed": merges_timed,
                 "merge_strategy": args.merge_strategy,
+                "pdrop": cfg.model.resid_pdrop,
                 "optimizer": "fused AdamW (lr 5e-4, reference config)",
                 "final_train_loss": node.miner.average_loss(),
             },

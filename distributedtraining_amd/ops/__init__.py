@@ -701,7 +701,15 @@ def cross_entropy_loss(logits: torch.Tensor, targets: torch.Tensor,
 # Logits stay fully materialized (backward's ce_bwd + head GEMMs need
 # them). Tile sizes keep two tiles inside the 256 MiB Infinity Cache.
 # --------------------------------------------------------------------------
-_CE_PIPE = _os.environ.get("DTA_CE_PIPELINE", "1") == "1"
+# NEGATIVE RESULT (r02, measured): the tiled pipeline loses 5.6 ms/step
+# at the flagship shape (80.4 vs 74.9 ms) — the cache-resident tiles that
+# make the CE read free also shrink the backward GEMMs to ~50-600
+# workgroups (vs 256 CU x 8 slots), and the GEMM-efficiency loss dwarfs
+# the ~1 ms of saved logits traffic. A strided-out variant writing tiles
+# of ONE [T,V] buffer (big one-shot backward GEMMs kept) is blocked by
+# TunableOp rejecting non-contiguous out (hipErrorInvalidValue at
+# ld=50257). Default OFF; the machinery stays correct/tested/opt-in.
+_CE_PIPE = _os.environ.get("DTA_CE_PIPELINE", "0") == "1"
 _CE_TT = int(_os.environ.get("DTA_CE_TT", "8192"))
 _CE_VC = int(_os.environ.get("DTA_CE_VC", "4224"))
 _ce_stream: Optional[torch.cuda.Stream] = None

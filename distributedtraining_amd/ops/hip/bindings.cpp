@@ -104,7 +104,7 @@ std::vector<Tensor> layernorm_bwd(Tensor dy, Tensor ds, Tensor x, Tensor w,
   auto f32 = x.options().dtype(torch::kFloat32);
   auto dw32 = torch::zeros({cols}, f32);
   auto db32 = torch::zeros({cols}, f32);
-  const int stripes = dta_colred_stripes(rows, cols);
+  const int stripes = dta_norm_bwd_stripes(rows, cols);
   auto part = torch::empty({2, stripes, cols}, f32);
   launch_layernorm_bwd(bfp(dy), has_ds ? bfp(ds) : nullptr, bfp(x), bfp(w),
                        mean.data_ptr<float>(), rstd.data_ptr<float>(),
@@ -152,7 +152,7 @@ std::vector<Tensor> rmsnorm_bwd(Tensor dy, Tensor ds, Tensor x, Tensor w,
                        : torch::empty({0}, x.options());
   auto f32 = x.options().dtype(torch::kFloat32);
   auto dw32 = torch::zeros({cols}, f32);
-  const int stripes = dta_colred_stripes(rows, cols);
+  const int stripes = dta_norm_bwd_stripes(rows, cols);
   auto part = torch::empty({stripes, cols}, f32);
   launch_rmsnorm_bwd(bfp(dy), has_ds ? bfp(ds) : nullptr, bfp(x), bfp(w),
                      rstd.data_ptr<float>(), bfp_mut(dx),

@@ -77,6 +77,20 @@ void launch_grad_merge_weights(const void* g, bool g_is_bf16,
                                int n_segs, hipStream_t s);
 
 // ---- norms ----------------------------------------------------------------
+// dγ/dβ partial-panel geometry: for cols <= 1024 the partials come from
+// the dx kernel itself (register accumulation, one panel row per
+// (block, wave)); wider rows use the standalone column-reduction kernel.
+constexpr int DTA_NORM_ROW_WAVES = 4;
+inline bool dta_norm_fused_dwdb(int cols) { return cols <= 1024; }
+inline int dta_norm_bwd_grid(int64_t rows) {
+  int64_t want = (rows + DTA_NORM_ROW_WAVES - 1) / DTA_NORM_ROW_WAVES;
+  return int(want < 4096 ? (want > 0 ? want : 1) : 4096);
+}
+inline int dta_norm_bwd_stripes(int64_t rows, int cols) {
+  return dta_norm_fused_dwdb(cols)
+             ? dta_norm_bwd_grid(rows) * DTA_NORM_ROW_WAVES
+             : dta_colred_stripes(rows, cols);
+}
 // res/sum_out: optional fused residual (sum = bf16(x+res) feeds both the
 // statistics and the ongoing stream); ds: optional additive gradient on
 // the sum stream folded into dx.

@@ -126,7 +126,13 @@ __global__ void norm_fwd_k(const ushort* __restrict__ x,
 // total sum-gradient gated by the regenerated mask: dres = dx ⊙ M/(1-p)
 // — one extra streaming write here instead of a separate dropout-bwd
 // pass over dx.
-template <int ITERS, bool RMS, bool DS, bool DROP>
+// FDW: dγ/dβ partials accumulated IN-REGISTER here (each lane owns fixed
+// columns across all its rows) and spilled once per (block, wave) to the
+// partial panel — removing the separate dwdb_part kernel's full re-read
+// of dy and x (~0.9 ms/step at the flagship shape). Register cost is
+// 16·ITERS VGPRs, so the fusion is gated to cols ≤ 1024 (the GPT-2
+// family); wider rows keep the two-kernel path.
+template <int ITERS, bool RMS, bool DS, bool DROP, bool FDW>
 __global__ void norm_bwd_dx_k(const ushort* __restrict__ dy,
                               const ushort* __restrict__ ds,
                               const ushort* __restrict__ x,
@@ -134,7 +140,9 @@ __global__ void norm_bwd_dx_k(const ushort* __restrict__ dy,
                               const float* __restrict__ mean,
                               const float* __restrict__ rstd,
                               ushort* __restrict__ dx,
-                              ushort* __restrict__ dres, int64_t rows,
+                              ushort* __restrict__ dres,
+                              float* __restrict__ pdw,
+                              float* __restrict__ pdb, int64_t rows,
                               int cols,
                               const unsigned long long* __restrict__ rng,
                               unsigned long long site, unsigned int thr16,
@@ -142,6 +150,16 @@ __global__ void norm_bwd_dx_k(const ushort* __restrict__ dy,
   const int lane = threadIdx.x & 63, wid = threadIdx.x >> 6;
   const int nchunk = cols >> 3;
   const uint64_t s1d = DROP ? sm64(*rng + site * DTA_RNG_SITE_K) : 0;
+  float dwp[FDW ? ITERS : 1][8], dbp[(FDW && !RMS) ? ITERS : 1][8];
+  if (FDW) {
+#pragma unroll
+    for (int it = 0; it < (FDW ? ITERS : 1); ++it)
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        dwp[it][j] = 0.f;
+        if (!RMS) dbp[it][j] = 0.f;
+      }
+  }
   float wv[ITERS][8];
 #pragma unroll
   for (int it = 0; it < ITERS; ++it) {
@@ -169,11 +187,16 @@ __global__ void norm_bwd_dx_k(const ushort* __restrict__ dy,
 #pragma unroll
         for (int j = 0; j < 8; ++j) {
           float xhj = (bf2f(ushort(vx[j])) - mu) * rs;
-          float dywj = bf2f(ushort(vdy[j])) * wv[it][j];
+          const float dyv = bf2f(ushort(vdy[j]));
+          float dywj = dyv * wv[it][j];
           xh[it][j] = xhj;
           dyw[it][j] = dywj;
           s1 += dywj;
           s2 = fmaf(dywj, xhj, s2);
+          if (FDW) {
+            dwp[it][j] = fmaf(dyv, xhj, dwp[it][j]);
+            if (!RMS) dbp[it][j] += dyv;
+          }
         }
       } else {
 #pragma unroll
@@ -213,6 +236,23 @@ __global__ void norm_bwd_dx_k(const ushort* __restrict__ dy,
         }
         *reinterpret_cast<s16x8*>(dxr + c * 8) = o;
         if (DROP) *reinterpret_cast<s16x8*>(drr + c * 8) = od;
+      }
+    }
+  }
+  if (FDW) {
+    // one partial row per (block, wave); dwdb_reduce folds them
+    const int64_t prow = int64_t(blockIdx.x) * ROW_WAVES + wid;
+    float* pw = pdw + prow * cols;
+    float* pb = (!RMS) ? pdb + prow * cols : nullptr;
+#pragma unroll
+    for (int it = 0; it < ITERS; ++it) {
+      const int c = lane + it * 64;
+      if (c < nchunk) {
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          pw[c * 8 + j] = dwp[it][j];
+          if (!RMS) pb[c * 8 + j] = dbp[it][j];
+        }
       }
     }
   }
@@ -321,29 +361,34 @@ void dispatch_bwd(const ushort* dy, const ushort* ds, const ushort* x,
                   unsigned int thr16, float ik, hipStream_t s) {
   const int nchunk = cols >> 3;
   const int iters = (nchunk + 63) / 64;
-  int64_t want = (rows + ROW_WAVES - 1) / ROW_WAVES;
-  const int grid = int(want < 4096 ? (want > 0 ? want : 1) : 4096);
+  const bool fdw = dta_norm_fused_dwdb(cols);
+  const int grid = dta_norm_bwd_grid(rows);
   const dim3 blk(64 * ROW_WAVES);
 #define CASE_B(I)                                                            \
   case I:                                                                    \
-    norm_bwd_dx_k<I, RMS, DS, DROP><<<grid, blk, 0, s>>>(                    \
-        dy, ds, x, w, mean, rstd, dx, dres, rows, cols, rng, site, thr16,    \
-        ik);                                                                 \
+    if (fdw)                                                                 \
+      norm_bwd_dx_k<I, RMS, DS, DROP, true><<<grid, blk, 0, s>>>(            \
+          dy, ds, x, w, mean, rstd, dx, dres, pdw, pdb, rows, cols, rng,     \
+          site, thr16, ik);                                                  \
+    else                                                                     \
+      norm_bwd_dx_k<I, RMS, DS, DROP, false><<<grid, blk, 0, s>>>(           \
+          dy, ds, x, w, mean, rstd, dx, dres, nullptr, nullptr, rows, cols,  \
+          rng, site, thr16, ik);                                             \
     break;
   switch (iters) {
     CASE_B(1) CASE_B(2) CASE_B(3) CASE_B(4) CASE_B(6) CASE_B(8)
     default:
-      norm_bwd_dx_k<8, RMS, DS, DROP><<<grid, blk, 0, s>>>(
-          dy, ds, x, w, mean, rstd, dx, dres, rows, cols, rng, site, thr16,
-          ik);
+      norm_bwd_dx_k<8, RMS, DS, DROP, false><<<grid, blk, 0, s>>>(
+          dy, ds, x, w, mean, rstd, dx, dres, nullptr, nullptr, rows, cols,
+          rng, site, thr16, ik);
   }
 #undef CASE_B
-  const ColRedCfg cfg = dta_colred_cfg(rows, cols);
-  dim3 g2(cfg.gx, unsigned(stripes));
-  norm_bwd_dwdb_part_k<RMS><<<g2, cfg.threads, 0, s>>>(dy, x, mean, rstd,
-                                                       pdw,
-                                                       RMS ? nullptr : pdb,
-                                                       rows, cols);
+  if (!fdw) {
+    const ColRedCfg cfg = dta_colred_cfg(rows, cols);
+    dim3 g2(cfg.gx, unsigned(stripes));
+    norm_bwd_dwdb_part_k<RMS><<<g2, cfg.threads, 0, s>>>(
+        dy, x, mean, rstd, pdw, RMS ? nullptr : pdb, rows, cols);
+  }
   const int g3 = (cols / 4 + 255) / 256;
   const int ry = stripes < 32 ? stripes : 32;
   dwdb_reduce_k<<<dim3(g3, ry), 256, 0, s>>>(pdw, RMS ? nullptr : pdb, dw,
