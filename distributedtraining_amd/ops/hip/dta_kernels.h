@@ -4,6 +4,7 @@
 
 #include <hip/hip_runtime.h>
 #include <cstdint>
+#include <cstdlib>
 
 using bf16_t = unsigned short;  // raw bf16 bits at the ABI boundary
 
@@ -81,7 +82,14 @@ void launch_grad_merge_weights(const void* g, bool g_is_bf16,
 // the dx kernel itself (register accumulation, one panel row per
 // (block, wave)); wider rows use the standalone column-reduction kernel.
 constexpr int DTA_NORM_ROW_WAVES = 4;
-inline bool dta_norm_fused_dwdb(int cols) { return cols <= 1024; }
+inline bool dta_norm_fused_dwdb(int cols) {
+  // DTA_NORM_FDW=0 restores the two-kernel path (A/B measurement)
+  static const bool en = [] {
+    const char* e = ::getenv("DTA_NORM_FDW");
+    return !(e && e[0] == '0');
+  }();
+  return en && cols <= 1024;
+}
 inline int dta_norm_bwd_grid(int64_t rows) {
   int64_t want = (rows + DTA_NORM_ROW_WAVES - 1) / DTA_NORM_ROW_WAVES;
   return int(want < 4096 ? (want > 0 ? want : 1) : 4096);
