@@ -95,9 +95,9 @@ inline int dta_norm_bwd_grid(int64_t rows) {
   return int(want < 4096 ? (want > 0 ? want : 1) : 4096);
 }
 inline int dta_norm_bwd_stripes(int64_t rows, int cols) {
-  return dta_norm_fused_dwdb(cols)
-             ? dta_norm_bwd_grid(rows) * DTA_NORM_ROW_WAVES
-             : dta_colred_stripes(rows, cols);
+  // fused path: one panel row per BLOCK (waves LDS-combined in-kernel)
+  return dta_norm_fused_dwdb(cols) ? dta_norm_bwd_grid(rows)
+                                   : dta_colred_stripes(rows, cols);
 }
 // res/sum_out: optional fused residual (sum = bf16(x+res) feeds both the
 // statistics and the ongoing stream); ds: optional additive gradient on
