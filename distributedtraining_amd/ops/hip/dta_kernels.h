@@ -83,10 +83,14 @@ void launch_grad_merge_weights(const void* g, bool g_is_bf16,
 // (block, wave)); wider rows use the standalone column-reduction kernel.
 constexpr int DTA_NORM_ROW_WAVES = 4;
 inline bool dta_norm_fused_dwdb(int cols) {
-  // DTA_NORM_FDW=0 restores the two-kernel path (A/B measurement)
+  // NEGATIVE RESULT (r02, same-box A/B): folding the dγ/dβ partials into
+  // the dx kernel removes the 0.9 ms/step standalone partial pass but the
+  // 16·ITERS accumulator VGPRs double the dx kernel itself (75 → 195 µs
+  // per call; whole step 76.9 vs 75.2 ms). Register pressure beats the
+  // saved read on this streaming kernel. Opt-in via DTA_NORM_FDW=1.
   static const bool en = [] {
     const char* e = ::getenv("DTA_NORM_FDW");
-    return !(e && e[0] == '0');
+    return e && e[0] == '1';
   }();
   return en && cols <= 1024;
 }
