@@ -184,3 +184,67 @@ def test_distributed_validation_round_sharded():
     assert all(ok for _, ok, *_ in results), results
     assert results[0][2] == results[1][2], "ranks disagree on scores"
     assert results[0][3] == results[1][3]
+
+
+def _bf16_wire_worker(rank, world, port, q):
+    os.environ.update({
+        "RANK": str(rank), "WORLD_SIZE": str(world),
+        "LOCAL_RANK": str(rank), "MASTER_ADDR": "127.0.0.1",
+        "MASTER_PORT": str(port),
+    })
+    import torch
+    from distributedtraining_amd.config import Config, ModelConfig, TrainConfig
+    from distributedtraining_amd.models import build_model
+    from distributedtraining_amd.parallel.comm import CommPlane
+    from distributedtraining_amd.parallel.flat import FlatParams
+    from distributedtraining_amd.parallel.local_sgd import LocalSGDNode
+    from distributedtraining_amd.utils.data import synthetic_batches
+    try:
+        cfg = Config()
+        cfg.model = ModelConfig.gpt2_tiny()
+        cfg.train = TrainConfig(batch_size=2, seq_len=16,
+                                send_interval_steps=10**9,
+                                pull_interval_steps=0)
+        cfg.comm.exchange_dtype = "bf16"   # halved-wire gather (Llama-scale)
+        torch.manual_seed(100 + rank)
+        model = build_model(cfg.model)
+        fp = FlatParams(model)
+        comm = CommPlane(backend="gloo", device=torch.device("cpu"))
+        data = synthetic_batches(cfg.model.vocab_size, 2, 16, seed=rank)
+        node = LocalSGDNode(model, fp, data, cfg, comm,
+                            merge_strategy="score_weighted")
+        node.sync_initial_base()
+        base0 = fp.master.clone()
+        node.train_steps(2)
+        delta = fp.make_delta(node.miner.base)
+        gathered = comm.all_gather_flat(delta.flat, torch.bfloat16)
+        ok_dtype = gathered.dtype == torch.bfloat16
+        node.merge_round(scores=[1.0] * world)
+        # uniform scores => bf16-rounded mean of deltas
+        expected = base0 + gathered.to(torch.float32).mean(dim=0)
+        ok = torch.allclose(fp.master, expected, rtol=1e-2, atol=1e-2)
+        q.put((rank, bool(ok_dtype and ok), float(fp.master.sum())))
+        comm.close()
+    except Exception as e:  # pragma: no cover
+        q.put((rank, False, repr(e)))
+        raise
+
+
+@pytest.mark.timeout(300)
+def test_bf16_wire_gather_merge():
+    """score_weighted merge over the bf16 wire dtype (the mandatory
+    exchange mode for Llama-scale gathers) — world 2, identical results
+    on every rank."""
+    world = 2
+    port = 29781
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    procs = [ctx.Process(target=_bf16_wire_worker, args=(r, world, port, q))
+             for r in range(world)]
+    for p in procs:
+        p.start()
+    results = [q.get(timeout=240) for _ in range(world)]
+    for p in procs:
+        p.join(timeout=60)
+    assert all(ok for _, ok, _ in results), results
+    assert results[0][2] == pytest.approx(results[1][2], rel=1e-6)
