@@ -18,6 +18,40 @@ from typing import Optional
 import torch
 
 
+class GraphedDecoder:
+    """hipGraph-captured decode step: ~100 eager kernel launches per token
+    collapse into ONE graph replay (the round-1 serving path was
+    launch-bound at batch 1). Requires a graphable KVCache (device
+    position counter) — the captured i32_inc advances it every replay."""
+
+    def __init__(self, model, cache, batch: int, device):
+        assert cache.pos_dev is not None
+        self.model = model
+        self.cache = cache
+        self.tok = torch.zeros(batch, 1, dtype=torch.long, device=device)
+        len0 = cache.len
+        s = torch.cuda.Stream()
+        s.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(s):
+            for _ in range(2):     # allocator warmup (mutates, rewound)
+                model.decode_step(self.tok, cache)
+        torch.cuda.current_stream().wait_stream(s)
+        torch.cuda.synchronize()
+        cache.len = len0
+        cache.pos_dev.fill_(len0)
+        self.graph = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(self.graph):
+            self.logits = model.decode_step(self.tok, cache)
+        cache.len = len0           # capture records, does not execute
+        cache.pos_dev.fill_(len0)
+
+    def step(self, tok: torch.Tensor) -> torch.Tensor:
+        self.tok.copy_(tok, non_blocking=True)
+        self.graph.replay()        # device pos advances inside the graph
+        self.cache.len += 1
+        return self.logits
+
+
 def _sample(logits, temperature, top_k, generator):
     if temperature <= 0:
         return logits.argmax(dim=-1)
@@ -48,8 +82,10 @@ def generate(model, input_ids: torch.Tensor, max_new_tokens: int,
     try:
         if use_cache and hasattr(model, "prefill"):
             total = min(ids.shape[1] + max_new_tokens, n_pos)
-            cache = model.new_cache(ids.shape[0], total, ids.device)
+            cache = model.new_cache(ids.shape[0], total, ids.device,
+                                    graphable=ids.is_cuda)
             logits = model.prefill(ids, cache).float()
+            dec = None
             for _ in range(max_new_tokens):
                 nxt = _sample(logits, temperature, top_k, generator)
                 if eos_token_id is not None:
@@ -61,7 +97,12 @@ def generate(model, input_ids: torch.Tensor, max_new_tokens: int,
                 if ((eos_token_id is not None and bool(done.all()))
                         or cache.len >= cache.max_len):
                     break
-                logits = model.decode_step(ids[:, -1:], cache).float()
+                if dec is None and cache.pos_dev is not None:
+                    dec = GraphedDecoder(model, cache, ids.shape[0],
+                                         ids.device)
+                step = dec.step if dec is not None else (
+                    lambda t: model.decode_step(t, cache))
+                logits = step(ids[:, -1:]).float()
             return ids
         for _ in range(max_new_tokens):
             ctx = ids[:, -n_pos:]

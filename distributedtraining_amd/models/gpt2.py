@@ -169,10 +169,17 @@ class GPT2LM(nn.Module):
 
 class KVCache:
     """Preallocated per-layer KV cache for serving ([B, H, max_len, D];
-    the decode kernel reads the first ``len`` positions)."""
+    the decode kernel reads the first ``len`` positions).
+
+    ``graphable=True`` (GPU): the write position also lives in a device
+    int32 counter (``pos_dev``) consumed by the append/embedding/decode
+    kernels, so a whole decode step is hipGraph-replayable — one graph
+    launch per token instead of ~100 eager launches (the round-1
+    launch-bound serving path)."""
 
     def __init__(self, n_layer: int, batch: int, n_kv_head: int,
-                 max_len: int, head_dim: int, device, dtype=None):
+                 max_len: int, head_dim: int, device, dtype=None,
+                 graphable: bool = False):
         dtype = dtype or torch.bfloat16
         self.k = [torch.zeros(batch, n_kv_head, max_len, head_dim,
                               device=device, dtype=dtype)
@@ -180,6 +187,9 @@ class KVCache:
         self.v = [torch.zeros_like(self.k[0]) for _ in range(n_layer)]
         self.max_len = max_len
         self.len = 0
+        self.pos_dev = (torch.zeros(1, dtype=torch.int32, device=device)
+                        if graphable
+                        and torch.device(device).type == "cuda" else None)
 
     def append(self, layer: int, k_new: torch.Tensor,
                v_new: torch.Tensor) -> None:
@@ -189,6 +199,14 @@ class KVCache:
         if self.len + S > self.max_len:
             raise ValueError(
                 f"KV cache overflow: {self.len}+{S} > {self.max_len}")
+        if self.pos_dev is not None and S == 1 and k_new.is_cuda:
+            from ..ops.backend import require_ext
+            B = k_new.shape[0]
+            require_ext().kv_append(k_new.reshape(B, 1, -1),
+                                    v_new.reshape(B, 1, -1),
+                                    self.k[layer], self.v[layer],
+                                    self.pos_dev)
+            return
         self.k[layer][:, :, self.len:self.len + S].copy_(
             k_new.transpose(1, 2))
         self.v[layer][:, :, self.len:self.len + S].copy_(
@@ -196,6 +214,12 @@ class KVCache:
 
     def advance(self, n: int) -> None:
         self.len += n
+        if self.pos_dev is not None:
+            if n == 1:   # capturable in-graph increment
+                from ..ops.backend import require_ext
+                require_ext().i32_inc(self.pos_dev)
+            else:        # prefill (eager): host-driven sync
+                self.pos_dev.fill_(self.len)
 
 
 def _gpt2_block_attn_cached(blk: GPT2Block, x: torch.Tensor,
@@ -211,7 +235,8 @@ def _gpt2_block_attn_cached(blk: GPT2Block, x: torch.Tensor,
     cache.append(i, k.view(B, S, H, D), v.view(B, S, H, D))
     if S == 1:
         o = ops.decode_attention(q.reshape(B, H, D), cache.k[i],
-                                 cache.v[i], cache.len + 1)
+                                 cache.v[i], cache.len + 1,
+                                 kv_len_dev=cache.pos_dev)
         o = o.view(B, 1, E)
     else:
         assert cache.len == 0, "prefill must start an empty cache"
@@ -230,9 +255,14 @@ def _gpt2_cached_forward(model: "GPT2LM", input_ids: torch.Tensor,
                          cache: KVCache) -> torch.Tensor:
     """Shared prefill/decode body: returns last-position logits [B, V]."""
     S = input_ids.shape[1]
-    off = cache.len
-    wpe_slice = model.wpe[off:off + S].contiguous()
-    x = ops.embedding_fwd(input_ids, model.wte, wpe_slice)
+    if cache.pos_dev is not None and S == 1:
+        # device position offset: the step stays graph-replayable
+        x = ops.embedding_fwd(input_ids, model.wte, model.wpe,
+                              pos=cache.pos_dev)
+    else:
+        off = cache.len
+        wpe_slice = model.wpe[off:off + S].contiguous()
+        x = ops.embedding_fwd(input_ids, model.wte, wpe_slice)
     for i, blk in enumerate(model.blocks):
         x = _gpt2_block_attn_cached(blk, x, cache, i)
     cache.advance(S)
@@ -241,11 +271,12 @@ def _gpt2_cached_forward(model: "GPT2LM", input_ids: torch.Tensor,
 
 
 def _gpt2_new_cache(self: "GPT2LM", batch: int, max_len: int, device,
-                    dtype=None) -> KVCache:
+                    dtype=None, graphable: bool = False) -> KVCache:
     cfg = self.cfg
     return KVCache(cfg.n_layer, batch, cfg.n_head, max_len,
                    cfg.n_embd // cfg.n_head, device,
-                   dtype or next(self.parameters()).dtype)
+                   dtype or next(self.parameters()).dtype,
+                   graphable=graphable)
 
 
 def _gpt2_prefill(self: "GPT2LM", input_ids: torch.Tensor,

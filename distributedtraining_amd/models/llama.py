@@ -185,7 +185,9 @@ def _llama_cached_forward(model: "LlamaLM", input_ids: torch.Tensor,
 
 
 def _llama_new_cache(self: "LlamaLM", batch: int, max_len: int, device,
-                     dtype=None):
+                     dtype=None, graphable: bool = False):
+    # graphable decode needs a device-position rope variant — not wired
+    # for the Llama family yet, so the flag is accepted and ignored
     from .gpt2 import KVCache
     cfg = self.cfg
     return KVCache(cfg.n_layer, batch, cfg.n_kv_head or cfg.n_head,

@@ -634,15 +634,23 @@ def qkv_attention(qkv: torch.Tensor, n_head: int,
 
 def decode_attention(q: torch.Tensor, k_cache: torch.Tensor,
                      v_cache: torch.Tensor, kv_len: int,
-                     scale: Optional[float] = None) -> torch.Tensor:
+                     scale: Optional[float] = None,
+                     kv_len_dev: Optional[torch.Tensor] = None
+                     ) -> torch.Tensor:
     """Serving decode step: one new query per head attends the whole KV
     cache. q [B,H,D]; caches [B,Hk,Lmax,D] with the first kv_len rows
-    valid. Inference-only (no autograd)."""
+    valid. ``kv_len_dev`` (int32 [1] device counter holding the cache
+    position BEFORE this token; effective length = *kv_len_dev + 1) makes
+    the step hipGraph-replayable. Inference-only (no autograd)."""
     if scale is None:
         scale = 1.0 / math.sqrt(q.shape[-1])
     if use_hip(q):
-        return require_ext().attn_decode(q.contiguous(), k_cache, v_cache,
-                                         kv_len, scale)
+        return require_ext().attn_decode(
+            q.contiguous(), k_cache, v_cache, kv_len, scale,
+            kv_len_dev if kv_len_dev is not None
+            else torch.empty(0, dtype=torch.int32, device=q.device))
+    if kv_len_dev is not None:
+        kv_len = int(kv_len_dev.item()) + 1
     qf = q.float().unsqueeze(2)                       # [B,H,1,D]
     B, Hk, _, D = k_cache.shape
     H = q.shape[1]
@@ -852,7 +860,7 @@ class _EmbeddingFn(torch.autograd.Function):
         ctx.vocab, ctx.npos = wte.shape[0], (wpe.shape[0] if wpe is not None else 0)
         ctx.dim = wte.shape[1]
         ctx.dtype = wte.dtype
-        return m.embedding_fwd(ids, wte, wpe if wpe is not None else torch.empty(0, dtype=wte.dtype, device=wte.device))
+        return m.embedding_fwd(ids, wte, wpe if wpe is not None else torch.empty(0, dtype=wte.dtype, device=wte.device), torch.empty(0, dtype=torch.int32, device=wte.device))
 
     @staticmethod
     def backward(ctx, dy):
@@ -863,13 +871,21 @@ class _EmbeddingFn(torch.autograd.Function):
 
 
 def embedding_fwd(ids: torch.Tensor, wte: torch.Tensor,
-                  wpe: Optional[torch.Tensor] = None) -> torch.Tensor:
-    """x[b,s,:] = wte[ids[b,s]] (+ wpe[s] if given)."""
+                  wpe: Optional[torch.Tensor] = None,
+                  pos: Optional[torch.Tensor] = None) -> torch.Tensor:
+    """x[b,s,:] = wte[ids[b,s]] (+ wpe[pos + s] if given). ``pos``: int32
+    [1] DEVICE offset — graph-replayable decode (inference-only path)."""
     if use_hip(wte):
+        if pos is not None:   # serving decode: no autograd needed
+            return require_ext().embedding_fwd(
+                ids.contiguous(), wte,
+                wpe if wpe is not None else torch.empty(
+                    0, dtype=wte.dtype, device=wte.device), pos)
         return _EmbeddingFn.apply(ids, wte, wpe)
     x = F.embedding(ids, wte)
     if wpe is not None:
-        x = x + wpe[: ids.shape[-1]].unsqueeze(0)
+        off = int(pos.item()) if pos is not None else 0
+        x = x + wpe[off:off + ids.shape[-1]].unsqueeze(0)
     return x
 
 

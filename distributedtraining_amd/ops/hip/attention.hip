@@ -586,6 +586,7 @@ __global__ void attn_decode_k(const ushort* __restrict__ q,   // [B,H,D]
                               const ushort* __restrict__ vc,
                               ushort* __restrict__ o,         // [B,H,D]
                               int B, int H, int grp, int kvlen,
+                              const int* __restrict__ len_p,  // device pos
                               int64_t cb, int64_t ch,  // cache strides
                               float scale) {
   constexpr int D = 16 * DTILES;
@@ -593,6 +594,9 @@ __global__ void attn_decode_k(const ushort* __restrict__ q,   // [B,H,D]
   const int lane = threadIdx.x & 63, wid = threadIdx.x >> 6;
   const int bh = blockIdx.x * 4 + wid;
   if (bh >= B * H) return;
+  // hipGraph-replayable serving: the valid cache length comes from a
+  // device counter (the append wrote row *len_p, so attend *len_p + 1)
+  if (len_p) kvlen = *len_p + 1;
   const int b = bh / H, h = bh % H;
   const int hk = h / grp;
   const ushort* qp = q + (int64_t(b) * H + h) * D;
@@ -765,13 +769,14 @@ void launch_mfma_probe_32(const bf16_t* A, const bf16_t* B, float* D,
 }
 
 void launch_attn_decode(const bf16_t* q, const bf16_t* kc, const bf16_t* vc,
-                        bf16_t* o, int B, int H, int grp, int kvlen, int hd,
-                        int64_t cb, int64_t ch, float scale, hipStream_t s) {
+                        bf16_t* o, int B, int H, int grp, int kvlen,
+                        const int* len_p, int hd, int64_t cb, int64_t ch,
+                        float scale, hipStream_t s) {
   const int grid = (B * H + 3) / 4;
   if (hd == 64)
     attn_decode_k<4><<<grid, 256, 0, s>>>(q, kc, vc, o, B, H, grp, kvlen,
-                                          cb, ch, scale);
+                                          len_p, cb, ch, scale);
   else if (hd == 128)
     attn_decode_k<8><<<grid, 256, 0, s>>>(q, kc, vc, o, B, H, grp, kvlen,
-                                          cb, ch, scale);
+                                          len_p, cb, ch, scale);
 }

@@ -400,7 +400,8 @@ Tensor ce_bwd(Tensor logits, Tensor targets, Tensor lse, Tensor scale_dev,
 }
 
 // ---- embedding -------------------------------------------------------------
-Tensor embedding_fwd(Tensor ids, Tensor wte, Tensor wpe) {
+// pos: optional int32 [1] device offset for the wpe row (graph decode)
+Tensor embedding_fwd(Tensor ids, Tensor wte, Tensor wpe, Tensor pos) {
   check_bf16(wte, "wte");
   TORCH_CHECK(ids.scalar_type() == torch::kInt64, "ids int64");
   const bool has_wpe = wpe.numel() > 0;
@@ -410,10 +411,33 @@ Tensor embedding_fwd(Tensor ids, Tensor wte, Tensor wpe) {
   auto out_sizes = ids.sizes().vec();
   out_sizes.push_back(dim);
   auto out = torch::empty(out_sizes, wte.options());
+  const int* posp = nullptr;
+  if (pos.numel()) {
+    TORCH_CHECK(pos.scalar_type() == torch::kInt32 && pos.is_cuda(),
+                "pos must be an int32 GPU scalar");
+    posp = pos.data_ptr<int>();
+  }
   launch_embedding_fwd(ids.data_ptr<int64_t>(), bfp(wte),
                        has_wpe ? bfp(wpe) : nullptr, bfp_mut(out), n_tok,
-                       seq, dim, has_wpe, stream());
+                       seq, dim, has_wpe, posp, stream());
   return out;
+}
+
+// new k/v rows ([B, 1, Hk*D] strided slices of the qkv projection) into
+// the caches at device row *pos; pos += 1 via i32_inc afterwards
+void kv_append(Tensor kn, Tensor vn, Tensor kc, Tensor vc, Tensor pos) {
+  TORCH_CHECK(kn.dim() == 3 && kn.size(1) == 1 && kn.stride(2) == 1,
+              "kn must be [B,1,F] with contiguous F");
+  TORCH_CHECK(pos.scalar_type() == torch::kInt32 && pos.is_cuda(), "pos");
+  const int B = int(kc.size(0)), Hk = int(kc.size(1)), D = int(kc.size(3));
+  launch_kv_append(bfp(kn), bfp(vn), kn.stride(0), bfp_mut(kc), bfp_mut(vc),
+                   pos.data_ptr<int>(), B, Hk, D, kc.stride(0),
+                   kc.stride(1), stream());
+}
+
+void i32_inc(Tensor t) {
+  TORCH_CHECK(t.scalar_type() == torch::kInt32 && t.is_cuda(), "int32 gpu");
+  launch_i32_inc(t.data_ptr<int>(), stream());
 }
 
 std::vector<Tensor> embedding_bwd(Tensor dy, Tensor ids, int64_t vocab,
@@ -620,16 +644,22 @@ void attn_bwd_packed(Tensor dout, Tensor q, Tensor k, Tensor v,
 // Decode attention over a KV cache: q [B,H,D], k/v caches [B,Hk,Lmax,D]
 // contiguous, first kv_len positions valid -> o [B,H,D].
 Tensor attn_decode(Tensor q, Tensor kc, Tensor vc, int64_t kv_len,
-                   double scale) {
+                   double scale, Tensor len_dev) {
   check_bf16(q, "q"); check_bf16(kc, "kcache"); check_bf16(vc, "vcache");
   TORCH_CHECK(q.dim() == 3 && kc.dim() == 4, "q [B,H,D], cache [B,Hk,L,D]");
   const int B = int(q.size(0)), H = int(q.size(1)), hd = int(q.size(2));
   const int Hk = int(kc.size(1));
   TORCH_CHECK(hd == 64 || hd == 128, "decode head dim must be 64/128");
   TORCH_CHECK(H % Hk == 0 && kv_len <= kc.size(2), "bad cache geometry");
+  const int* lp = nullptr;
+  if (len_dev.numel()) {
+    TORCH_CHECK(len_dev.scalar_type() == torch::kInt32 && len_dev.is_cuda(),
+                "len_dev must be an int32 GPU scalar");
+    lp = len_dev.data_ptr<int>();
+  }
   auto o = torch::empty_like(q);
   launch_attn_decode(bfp(q), bfp(kc), bfp(vc), bfp_mut(o), B, H, H / Hk,
-                     int(kv_len), hd, kc.stride(0), kc.stride(1),
+                     int(kv_len), lp, hd, kc.stride(0), kc.stride(1),
                      float(scale), stream());
   return o;
 }
@@ -679,6 +709,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("ce_finalize", &ce_finalize);
   m.def("embedding_fwd", &embedding_fwd);
   m.def("embedding_bwd", &embedding_bwd);
+  m.def("kv_append", &kv_append);
+  m.def("i32_inc", &i32_inc);
   m.def("rope_fwd", &rope_fwd);
   m.def("rope_bwd", &rope_bwd);
   m.def("attn_fwd", &attn_fwd);

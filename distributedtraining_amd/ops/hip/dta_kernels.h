@@ -161,9 +161,16 @@ void launch_ce_finalize(const int64_t* targets, const float* m_run,
                         float* loss_sum, int* count, hipStream_t s);
 
 // ---- embedding ------------------------------------------------------------
+// pos_p: device position offset for wpe (graph-replayable decode)
 void launch_embedding_fwd(const int64_t* ids, const bf16_t* wte,
                           const bf16_t* wpe, bf16_t* out, int64_t n_tok,
-                          int seq_len, int dim, bool has_wpe, hipStream_t s);
+                          int seq_len, int dim, bool has_wpe,
+                          const int* pos_p, hipStream_t s);
+// KV-cache append at device position *pos_p + device int increment
+void launch_kv_append(const bf16_t* kn, const bf16_t* vn, int64_t knb,
+                      bf16_t* kc, bf16_t* vc, const int* pos_p, int B,
+                      int Hk, int D, int64_t cb, int64_t ch, hipStream_t s);
+void launch_i32_inc(int* p, hipStream_t s);
 void launch_embedding_bwd(const bf16_t* dy, const int64_t* ids,
                           float* dwte_f32, float* dwpe_f32, int64_t n_tok,
                           int seq_len, int dim, bool has_wpe, hipStream_t s);
@@ -214,9 +221,12 @@ void launch_attn_bwd_dkv(const bf16_t* dout, const bf16_t* q,
                          bf16_t* dk, bf16_t* dv, const AttnGeom& geo,
                          hipStream_t s);
 
+// len_p: device cache-length counter (effective kvlen = *len_p + 1) for
+// hipGraph-replayable decode; null = host kvlen scalar
 void launch_attn_decode(const bf16_t* q, const bf16_t* kc, const bf16_t* vc,
-                        bf16_t* o, int B, int H, int grp, int kvlen, int hd,
-                        int64_t cb, int64_t ch, float scale, hipStream_t s);
+                        bf16_t* o, int B, int H, int grp, int kvlen,
+                        const int* len_p, int hd, int64_t cb, int64_t ch,
+                        float scale, hipStream_t s);
 
 // ---- mfma layout self-test ------------------------------------------------
 // D[32,32] = A[32,16] x B[16,32] and D[16,16] = A[16,32] x B[32,16]
