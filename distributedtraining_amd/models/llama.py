@@ -144,22 +144,24 @@ class LlamaLM(nn.Module):
 
 
 def _llama_block_attn_cached(blk: LlamaBlock, x: torch.Tensor, cos, sin,
-                             cache, i: int) -> torch.Tensor:
+                             cache, i: int, pos=None) -> torch.Tensor:
     """One Llama block with the KV cache (GQA-aware): prefill (S>1) or
-    decode (S==1 through ops.decode_attention). cos/sin are already
-    position-offset slices."""
+    decode (S==1 through ops.decode_attention). cos/sin are
+    position-offset slices, OR the full tables with ``pos`` a device
+    int32 offset (graph-replayable decode)."""
     B, S, E = x.shape
     D = blk.head_dim
     h = ops.rms_norm(x, blk.attn_norm_w, blk.norm_eps)
     q = ops.linear(h, blk.q_w).view(B, S, blk.n_head, D).transpose(1, 2)
     k = ops.linear(h, blk.k_w).view(B, S, blk.n_kv, D).transpose(1, 2)
     v = ops.linear(h, blk.v_w).view(B, S, blk.n_kv, D).transpose(1, 2)
-    q = ops.rope(q, cos, sin)
-    k = ops.rope(k, cos, sin)
+    q = ops.rope(q, cos, sin, pos=pos)
+    k = ops.rope(k, cos, sin, pos=pos)
     cache.append(i, k.transpose(1, 2), v.transpose(1, 2))
     if S == 1:
         o = ops.decode_attention(q.reshape(B, blk.n_head, D), cache.k[i],
-                                 cache.v[i], cache.len + 1)
+                                 cache.v[i], cache.len + 1,
+                                 kv_len_dev=cache.pos_dev)
         o = o.view(B, 1, E)
     else:
         assert cache.len == 0, "prefill must start an empty cache"
@@ -173,12 +175,17 @@ def _llama_block_attn_cached(blk: LlamaBlock, x: torch.Tensor, cos, sin,
 def _llama_cached_forward(model: "LlamaLM", input_ids: torch.Tensor,
                           cache) -> torch.Tensor:
     S = input_ids.shape[1]
-    off = cache.len
-    cos = model.rope_cos[off:off + S].contiguous()
-    sin = model.rope_sin[off:off + S].contiguous()
+    if cache.pos_dev is not None and S == 1:
+        # full tables + device offset: the step stays graph-replayable
+        cos, sin, pos = model.rope_cos, model.rope_sin, cache.pos_dev
+    else:
+        off = cache.len
+        cos = model.rope_cos[off:off + S].contiguous()
+        sin = model.rope_sin[off:off + S].contiguous()
+        pos = None
     x = ops.embedding_fwd(input_ids, model.tok_emb, None)
     for i, blk in enumerate(model.blocks):
-        x = _llama_block_attn_cached(blk, x, cos, sin, cache, i)
+        x = _llama_block_attn_cached(blk, x, cos, sin, cache, i, pos=pos)
     cache.advance(S)
     x = ops.rms_norm(x[:, -1:], model.final_norm_w, model.cfg.norm_eps)
     return ops.linear(x, model._head()).squeeze(1)
@@ -186,13 +193,12 @@ def _llama_cached_forward(model: "LlamaLM", input_ids: torch.Tensor,
 
 def _llama_new_cache(self: "LlamaLM", batch: int, max_len: int, device,
                      dtype=None, graphable: bool = False):
-    # graphable decode needs a device-position rope variant — not wired
-    # for the Llama family yet, so the flag is accepted and ignored
     from .gpt2 import KVCache
     cfg = self.cfg
     return KVCache(cfg.n_layer, batch, cfg.n_kv_head or cfg.n_head,
                    max_len, cfg.n_embd // cfg.n_head, device,
-                   dtype or next(self.parameters()).dtype)
+                   dtype or next(self.parameters()).dtype,
+                   graphable=graphable)
 
 
 def _llama_prefill(self: "LlamaLM", input_ids: torch.Tensor, cache):

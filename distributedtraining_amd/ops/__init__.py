@@ -897,28 +897,36 @@ class _RopeFn(torch.autograd.Function):
     def forward(ctx, x, cos, sin):
         m = require_ext()
         ctx.save_for_backward(cos, sin)
-        return m.rope_fwd(x.contiguous(), cos, sin)
+        e = torch.empty(0, dtype=torch.int32, device=x.device)
+        ctx.e = e
+        return m.rope_fwd(x.contiguous(), cos, sin, e)
 
     @staticmethod
     def backward(ctx, dy):
         m = require_ext()
         cos, sin = ctx.saved_tensors
-        return m.rope_bwd(dy.contiguous(), cos, sin), None, None
+        return m.rope_bwd(dy.contiguous(), cos, sin, ctx.e), None, None
 
 
-def rope(x: torch.Tensor, cos: torch.Tensor, sin: torch.Tensor) -> torch.Tensor:
+def rope(x: torch.Tensor, cos: torch.Tensor, sin: torch.Tensor,
+         pos: Optional[torch.Tensor] = None) -> torch.Tensor:
     """Apply rotary embedding to x [B,H,S,D] with cos/sin [S,D/2] (fp32;
-    other dtypes are upcast — Module.to(bf16) casts registered buffers)."""
+    other dtypes are upcast — Module.to(bf16) casts registered buffers).
+    ``pos``: int32 [1] DEVICE offset into the tables (graph-replayable
+    decode; inference-only path)."""
     if cos.dtype != torch.float32:
         cos = cos.float()
         sin = sin.float()
     if use_hip(x):
+        if pos is not None:   # serving decode: no autograd needed
+            return require_ext().rope_fwd(x.contiguous(), cos, sin, pos)
         return _RopeFn.apply(x, cos, sin)
+    off = int(pos.item()) if pos is not None else 0
     xf = x.float()
     d2 = x.shape[-1] // 2
     x1, x2 = xf[..., :d2], xf[..., d2:]
-    c = cos[: x.shape[-2]].view(1, 1, x.shape[-2], d2)
-    s = sin[: x.shape[-2]].view(1, 1, x.shape[-2], d2)
+    c = cos[off:off + x.shape[-2]].view(1, 1, x.shape[-2], d2)
+    s = sin[off:off + x.shape[-2]].view(1, 1, x.shape[-2], d2)
     out = torch.cat([x1 * c - x2 * s, x1 * s + x2 * c], dim=-1)
     return out.to(x.dtype)
 

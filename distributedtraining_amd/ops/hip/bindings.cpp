@@ -469,21 +469,32 @@ std::vector<Tensor> embedding_bwd(Tensor dy, Tensor ids, int64_t vocab,
 }
 
 // ---- rope ------------------------------------------------------------------
-Tensor rope_apply(Tensor x, Tensor cos_t, Tensor sin_t, bool backward) {
+Tensor rope_apply(Tensor x, Tensor cos_t, Tensor sin_t, bool backward,
+                  const Tensor& pos) {
   check_bf16(x, "x");
   // x: [B,H,S,D] or [BH,S,D]
   const int hd = int(x.size(-1));
   const int seq = int(x.size(-2));
   const int64_t bh = x.numel() / (int64_t(seq) * hd);
+  const int* posp = nullptr;
+  if (pos.numel()) {
+    TORCH_CHECK(pos.scalar_type() == torch::kInt32 && pos.is_cuda(),
+                "pos must be an int32 GPU scalar");
+    posp = pos.data_ptr<int>();
+  }
   auto y = torch::empty_like(x);
   auto c = cos_t.contiguous();
   auto sn = sin_t.contiguous();
   launch_rope(bfp(x), c.data_ptr<float>(), sn.data_ptr<float>(), bfp_mut(y),
-              bh, seq, hd, backward, stream());
+              bh, seq, hd, backward, posp, stream());
   return y;
 }
-Tensor rope_fwd(Tensor x, Tensor c, Tensor s) { return rope_apply(x, c, s, false); }
-Tensor rope_bwd(Tensor x, Tensor c, Tensor s) { return rope_apply(x, c, s, true); }
+Tensor rope_fwd(Tensor x, Tensor c, Tensor s, Tensor pos) {
+  return rope_apply(x, c, s, false, pos);
+}
+Tensor rope_bwd(Tensor x, Tensor c, Tensor s, Tensor pos) {
+  return rope_apply(x, c, s, true, pos);
+}
 
 // ---- attention -------------------------------------------------------------
 // q/k/v: [B,H(,Hk),S,D] views, innermost stride 1 (checked). Output o (and

@@ -177,9 +177,10 @@ void launch_embedding_bwd(const bf16_t* dy, const int64_t* ids,
 void launch_f32_to_bf16(const float* x, bf16_t* y, int64_t n, hipStream_t s);
 
 // ---- rope -----------------------------------------------------------------
+// pos_p: device position offset for graph-replayable decode
 void launch_rope(const bf16_t* x, const float* cos_t, const float* sin_t,
                  bf16_t* y, int64_t bh, int seq, int hd, bool backward,
-                 hipStream_t s);
+                 const int* pos_p, hipStream_t s);
 
 // ---- attention ------------------------------------------------------------
 // Strided geometry: q/k/v/o/dout are [B,H,S,D]-shaped views with arbitrary

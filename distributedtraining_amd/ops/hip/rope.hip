@@ -10,17 +10,21 @@ namespace {
 
 constexpr int ROW_WAVES = 4;
 
+// pos_p: device position offset (graph-replayable decode — the table row
+// is *pos_p + in-sequence index, so one captured graph serves every step)
 template <bool BWD>
 __global__ void rope_k(const ushort* __restrict__ x,
                        const float* __restrict__ cost,
                        const float* __restrict__ sint,
-                       ushort* __restrict__ y, int64_t bh, int seq, int hd) {
+                       ushort* __restrict__ y, int64_t bh, int seq, int hd,
+                       const int* __restrict__ pos_p) {
   const int lane = threadIdx.x & 63, wid = threadIdx.x >> 6;
   const int d2 = hd >> 1;
+  const int pos0 = pos_p ? *pos_p : 0;
   const int64_t rows = bh * seq;
   for (int64_t r = int64_t(blockIdx.x) * ROW_WAVES + wid; r < rows;
        r += int64_t(gridDim.x) * ROW_WAVES) {
-    const int pos = int(r % seq);
+    const int pos = pos0 + int(r % seq);
     const ushort* xr = x + r * hd;
     ushort* yr = y + r * hd;
     const float* c = cost + int64_t(pos) * d2;
@@ -53,12 +57,14 @@ __global__ void rope_k(const ushort* __restrict__ x,
 
 void launch_rope(const bf16_t* x, const float* cos_t, const float* sin_t,
                  bf16_t* y, int64_t bh, int seq, int hd, bool backward,
-                 hipStream_t s) {
+                 const int* pos_p, hipStream_t s) {
   int64_t rows = bh * seq;
   int64_t want = (rows + ROW_WAVES - 1) / ROW_WAVES;
   const int grid = int(want < 4096 ? (want > 0 ? want : 1) : 4096);
   if (backward)
-    rope_k<true><<<grid, 256, 0, s>>>(x, cos_t, sin_t, y, bh, seq, hd);
+    rope_k<true><<<grid, 256, 0, s>>>(x, cos_t, sin_t, y, bh, seq, hd,
+                                      pos_p);
   else
-    rope_k<false><<<grid, 256, 0, s>>>(x, cos_t, sin_t, y, bh, seq, hd);
+    rope_k<false><<<grid, 256, 0, s>>>(x, cos_t, sin_t, y, bh, seq, hd,
+                                       pos_p);
 }

@@ -1003,3 +1003,25 @@ def test_graphed_decode_matches_eager_decode():
     eager = generate(m, ids, max_new_tokens=12, use_cache=True)
     m.new_cache = orig
     assert torch.equal(graphed, eager)
+
+
+def test_graphed_decode_matches_eager_llama():
+    """Graph-replayed Llama decode (device-position rope + GQA decode
+    kernel) produces the same greedy tokens as the eager path."""
+    from distributedtraining_amd.config import ModelConfig
+    from distributedtraining_amd.models import build_model, generate
+    cfg = ModelConfig(family="llama", vocab_size=512, n_layer=2, n_head=4,
+                      n_kv_head=2, n_embd=256, n_positions=128,
+                      intermediate_size=512, rope_theta=10000.0,
+                      tie_word_embeddings=False)
+    torch.manual_seed(6)
+    m = build_model(cfg).to(DEV, torch.bfloat16).eval()
+    ids = torch.randint(0, 512, (2, 8), device=DEV)
+    graphed = generate(m, ids, max_new_tokens=10, use_cache=True)
+    orig = m.new_cache
+    m.new_cache = lambda *a, **kw: orig(*a, **{**kw, "graphable": False})
+    eager = generate(m, ids, max_new_tokens=10, use_cache=True)
+    m.new_cache = orig
+    assert torch.equal(graphed, eager)
+    uncached = generate(m, ids, max_new_tokens=10, use_cache=False)
+    assert torch.equal(graphed, uncached)
