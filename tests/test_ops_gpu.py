@@ -245,7 +245,8 @@ def test_rope_fwd_bwd():
     cos, sin = rope_tables(S, D, 10000.0)
     cos, sin = cos.to(DEV), sin.to(DEV)
     x = _rand_bf16(B, H, S, D, seed=20)
-    y = m.rope_fwd(x, cos, sin)
+    y = m.rope_fwd(x, cos, sin,
+                   torch.empty(0, dtype=torch.int32, device=DEV))
     xf = x.float()
     d2 = D // 2
     x1, x2 = xf[..., :d2], xf[..., d2:]
@@ -254,7 +255,8 @@ def test_rope_fwd_bwd():
     ref = torch.cat([x1 * c - x2 * s, x1 * s + x2 * c], dim=-1)
     torch.testing.assert_close(y.float(), ref, rtol=2e-2, atol=2e-2)
     # bwd is the inverse rotation: rope_bwd(rope_fwd(x)) == x
-    x_rt = m.rope_bwd(y, cos, sin)
+    x_rt = m.rope_bwd(y, cos, sin,
+                      torch.empty(0, dtype=torch.int32, device=DEV))
     torch.testing.assert_close(x_rt.float(), xf, rtol=3e-2, atol=3e-2)
 
 
