@@ -1,6 +1,6 @@
 #!/usr/bin/env python3
 """Serving decode throughput: tokens/s of the KV-cache decode path
-(hipGraph-replayed step vs eager), GPT-2-small, greedy.
+(hipGraph-replayed step vs eager), greedy.
 
 Usage: python benchmarks/bench_serving.py [--batch 1] [--prompt 64]
        [--new 128] [--no-graph]
@@ -12,6 +12,8 @@ import torch
 
 def main():
     ap = argparse.ArgumentParser()
+    ap.add_argument("--model", default="gpt2-small",
+                    choices=["gpt2-small", "llama3-8b"])
     ap.add_argument("--batch", type=int, default=1)
     ap.add_argument("--prompt", type=int, default=64)
     ap.add_argument("--new", type=int, default=128)
@@ -22,7 +24,8 @@ def main():
     from distributedtraining_amd.models import generate as gen_mod
     dev = "cuda:0"
     torch.manual_seed(0)
-    cfg = ModelConfig.gpt2_small()
+    cfg = (ModelConfig.gpt2_small() if args.model == "gpt2-small"
+           else ModelConfig.llama3_8b())
     model = build_model(cfg).to(dev, torch.bfloat16).eval()
     ids = torch.randint(0, cfg.vocab_size, (args.batch, args.prompt),
                         device=dev)
@@ -39,7 +42,7 @@ def main():
     dt = time.perf_counter() - t0
     ntok = (out.shape[1] - args.prompt) * args.batch
     print(json.dumps({
-        "metric": "serving decode tokens/s (GPT-2-small, KV cache)",
+        "metric": f"serving decode tokens/s ({args.model}, KV cache)",
         "value": ntok / dt, "batch": args.batch, "prompt": args.prompt,
         "new_tokens": out.shape[1] - args.prompt,
         "graph": not args.no_graph,
