@@ -6,7 +6,16 @@ Usage: python benchmarks/bench_serving.py [--batch 1] [--prompt 64]
        [--new 128] [--no-graph]
 """
 import argparse, json, os, sys, time
-sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+_root = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+sys.path.insert(0, _root)
+# decode GEMV shapes (M = batch) want tuned algorithms too — reuse the
+# committed TunableOp selections and tune unknown shapes on first use
+if os.path.exists(os.path.join(_root, "tuning", "tunableop0.csv")):
+    os.environ.setdefault("PYTORCH_TUNABLEOP_ENABLED", "1")
+    os.environ.setdefault("PYTORCH_TUNABLEOP_TUNING", "1")
+    os.environ.setdefault("PYTORCH_TUNABLEOP_FILENAME",
+                          os.path.join(_root, "tuning", "tunableop.csv"))
+    os.environ.setdefault("PYTORCH_TUNABLEOP_VERBOSE", "0")
 import torch
 
 
