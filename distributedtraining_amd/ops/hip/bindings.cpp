@@ -468,6 +468,25 @@ std::vector<Tensor> embedding_bwd(Tensor dy, Tensor ids, int64_t vocab,
           npos ? dwpe.to(torch::kBFloat16) : dwpe};
 }
 
+// ---- serving gemv ----------------------------------------------------------
+Tensor gemv(Tensor x, Tensor w, Tensor bias) {
+  check_bf16(x, "x"); check_bf16(w, "w");
+  TORCH_CHECK(x.dim() == 2 && w.dim() == 2 && x.size(1) == w.size(1),
+              "x [M,K], w [N,K]");
+  const int M = int(x.size(0));
+  TORCH_CHECK(M >= 1 && M <= 16, "gemv is for M <= 16");
+  const int64_t N = w.size(0);
+  const int K = int(x.size(1));
+  const bf16_t* bp = nullptr;
+  if (bias.numel()) {
+    check_bf16(bias, "bias");
+    bp = bfp(bias);
+  }
+  auto y = torch::empty({M, N}, x.options());
+  launch_gemv(bfp(x), bfp(w), bp, bfp_mut(y), M, N, K, stream());
+  return y;
+}
+
 // ---- rope ------------------------------------------------------------------
 Tensor rope_apply(Tensor x, Tensor cos_t, Tensor sin_t, bool backward,
                   const Tensor& pos) {
@@ -722,6 +741,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("embedding_bwd", &embedding_bwd);
   m.def("kv_append", &kv_append);
   m.def("i32_inc", &i32_inc);
+  m.def("gemv", &gemv);
   m.def("rope_fwd", &rope_fwd);
   m.def("rope_bwd", &rope_bwd);
   m.def("attn_fwd", &attn_fwd);

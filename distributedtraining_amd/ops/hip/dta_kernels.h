@@ -176,6 +176,11 @@ void launch_embedding_bwd(const bf16_t* dy, const int64_t* ids,
                           int seq_len, int dim, bool has_wpe, hipStream_t s);
 void launch_f32_to_bf16(const float* x, bf16_t* y, int64_t n, hipStream_t s);
 
+// ---- serving gemv ---------------------------------------------------------
+// y[M,N] = x[M,K] @ W[N,K]^T (+ bias), M <= 16 weight-streaming decode
+void launch_gemv(const bf16_t* x, const bf16_t* w, const bf16_t* bias,
+                 bf16_t* y, int M, int64_t N, int K, hipStream_t s);
+
 // ---- rope -----------------------------------------------------------------
 // pos_p: device position offset for graph-replayable decode
 void launch_rope(const bf16_t* x, const float* cos_t, const float* sin_t,

@@ -23,6 +23,7 @@ SRC = [
     "distributedtraining_amd/ops/hip/ce.hip",
     "distributedtraining_amd/ops/hip/embedding.hip",
     "distributedtraining_amd/ops/hip/rope.hip",
+    "distributedtraining_amd/ops/hip/gemv.hip",
     "distributedtraining_amd/ops/hip/merge.hip",
     "distributedtraining_amd/ops/hip/attention.hip",
 ]

@@ -1032,3 +1032,21 @@ def test_graphed_decode_matches_eager_llama():
     # horizon is the stable invariant (test_generate_on_gpu keeps it)
     uncached = generate(m, ids, max_new_tokens=4, use_cache=False)
     assert torch.equal(graphed[:, :12], uncached)
+
+
+@pytest.mark.parametrize("M,N,K", [(1, 4096, 4096), (1, 3000, 768),
+                                   (4, 1024, 4096), (16, 768, 3072),
+                                   (2, 50264, 768)])
+def test_gemv_matches_torch(M, N, K):
+    """The weight-streaming serving GEMV vs torch fp32 (with bias)."""
+    m = _ext()
+    x = _rand_bf16(M, K, seed=80, scale=0.5)
+    w = _rand_bf16(N, K, seed=81, scale=0.05)
+    b = _rand_bf16(N, seed=82)
+    y = m.gemv(x, w, b)
+    ref = x.float() @ w.float().t() + b.float()
+    torch.testing.assert_close(y.float(), ref, rtol=2e-2,
+                               atol=2e-2 * math.sqrt(K / 768))
+    y2 = m.gemv(x, w, torch.empty(0, dtype=torch.bfloat16, device=DEV))
+    torch.testing.assert_close(y2.float(), ref - b.float(), rtol=2e-2,
+                               atol=2e-2 * math.sqrt(K / 768))
