@@ -104,13 +104,13 @@ def linear(x: torch.Tensor, w: torch.Tensor,
            b: Optional[torch.Tensor] = None) -> torch.Tensor:
     """y = x @ w.T (+ b). hipBLASLt GEMM with fused bias epilogue forward;
     backward: wgrad+bias-grad in one BGRADB-epilogue GEMM accumulated into
-    the flat-grad plane (colsum fallback). Inference at M <= 16 (serving
+    the flat-grad plane (colsum fallback). Inference at M <= 4 (serving
     decode) routes to the hand-written weight-streaming GEMV kernel —
     hipBLASLt's small-M path measured 1.7 TB/s of weight stream vs the
     ~6.3 TB/s HBM rate this shape is bound by."""
     if use_hip(x):
         rows = x.numel() // x.shape[-1]
-        if rows <= 16 and not torch.is_grad_enabled():
+        if rows <= 4 and not torch.is_grad_enabled():
             x2 = x.reshape(rows, x.shape[-1])
             if not x2.is_contiguous():
                 x2 = x2.contiguous()
@@ -184,7 +184,7 @@ def mlp_gelu(x: torch.Tensor, w1: torch.Tensor, b1: torch.Tensor,
     once per shape)."""
     if use_hip(x):
         rows = x.numel() // x.shape[-1]
-        if rows <= 16 and not torch.is_grad_enabled():
+        if rows <= 4 and not torch.is_grad_enabled():
             # serving decode: composed path over the streaming GEMV
             return linear(gelu(linear(x, w1, b1)), w2, b2)
         key = (rows, w1.shape[0], w1.shape[1])

@@ -474,7 +474,7 @@ Tensor gemv(Tensor x, Tensor w, Tensor bias) {
   TORCH_CHECK(x.dim() == 2 && w.dim() == 2 && x.size(1) == w.size(1),
               "x [M,K], w [N,K]");
   const int M = int(x.size(0));
-  TORCH_CHECK(M >= 1 && M <= 16, "gemv is for M <= 16");
+  TORCH_CHECK(M >= 1 && M <= 4, "gemv is for M <= 4");
   const int64_t N = w.size(0);
   const int K = int(x.size(1));
   const bf16_t* bp = nullptr;

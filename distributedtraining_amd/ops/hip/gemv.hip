@@ -1,22 +1,24 @@
-// Weight-streaming GEMV / skinny GEMM for serving decode (M <= 16).
+// Weight-streaming GEMV / skinny GEMM for serving decode (M <= 4).
 //
 // y[M,N] = x[M,K] @ W[N,K]^T (+ bias) — the decode-step projection
 // shapes (M = batch). hipBLASLt's small-M path measured only ~1.7 TB/s
 // of weight stream on the Llama-3-8B decode step (9.4 ms/token against
 // a ~2.5 ms HBM floor); this kernel is a pure streaming design:
-//   * x (tiny) staged fp32 in LDS, transposed [k][m] so the inner loop
-//     reads per-k column vectors;
-//   * one WAVE per output column n: lanes stride K in 16 B bf16x8 loads
-//     of W's row (rows are contiguous — coalesced within the wave);
-//   * fp32 accumulators per m; wave-reduce at the end; fused bias.
-// Weights are read exactly once per token — the whole model streams at
-// HBM rate instead of hipBLASLt's tile machinery built for big M.
+//   * ONE BLOCK per output column n (grid fills the chip even at the
+//     small kv-proj N=1024 — a wave-per-n layout left 256 blocks there),
+//     its 4 waves splitting the K-contiguous weight row in 16 B chunks
+//     with an explicit 2-deep load pipeline;
+//   * x (tiny) staged fp32 in LDS, transposed [k][m];
+//   * fp32 accumulators per m, wave-reduce + LDS cross-wave combine,
+//     fused bias. Weights are read exactly once per token.
+// M > 4 falls back to hipBLASLt in ops.linear (at batch >= 16 the tile
+// machinery is competitive again).
 #include "dta_common.h"
 #include "dta_kernels.h"
 
 namespace {
 
-// LDS budget 32 KiB fp32: KC * MT = 8192
+// LDS budget 32 KiB fp32 for the x chunk: KC * MT = 8192
 template <int MT>
 __global__ void gemv_k(const ushort* __restrict__ x,
                        const ushort* __restrict__ w,
@@ -24,23 +26,33 @@ __global__ void gemv_k(const ushort* __restrict__ x,
                        ushort* __restrict__ y, int M, int64_t N, int K) {
   constexpr int KC = 8192 / MT;
   __shared__ float xs[8192];
-  const int wave = threadIdx.x >> 6, lane = threadIdx.x & 63;
-  const int64_t n0 = int64_t(blockIdx.x) * 4 + wave;
-  float acc[MT];
+  __shared__ float red[4][MT <= 4 ? 4 : MT];
+  const int lane = threadIdx.x & 63, wave = threadIdx.x >> 6;
+  for (int64_t n = blockIdx.x; n < N; n += gridDim.x) {
+    float acc[MT];
 #pragma unroll
-  for (int mm = 0; mm < MT; ++mm) acc[mm] = 0.f;
-  for (int kc = 0; kc < K; kc += KC) {
-    const int kn = (K - kc) < KC ? (K - kc) : KC;
-    __syncthreads();
-    for (int i = threadIdx.x; i < kn * MT; i += 256) {
-      const int k = i / MT, mm = i % MT;
-      xs[i] = (mm < M) ? bf2f(x[int64_t(mm) * K + kc + k]) : 0.f;
-    }
-    __syncthreads();
-    if (n0 < N) {
-      const ushort* wr = w + n0 * int64_t(K) + kc;
-      for (int k8 = lane * 8; k8 + 8 <= kn; k8 += 64 * 8) {
-        const s16x8 wv = *reinterpret_cast<const s16x8*>(wr + k8);
+    for (int mm = 0; mm < MT; ++mm) acc[mm] = 0.f;
+    for (int kc = 0; kc < K; kc += KC) {
+      const int kn = (K - kc) < KC ? (K - kc) : KC;
+      __syncthreads();
+      for (int i = threadIdx.x; i < kn * MT; i += 256) {
+        const int k = i / MT, mm = i % MT;
+        xs[i] = (mm < M) ? bf2f(x[int64_t(mm) * K + kc + k]) : 0.f;
+      }
+      __syncthreads();
+      const ushort* wr = w + n * int64_t(K) + kc;
+      // 256 lanes stride the row; 2-deep pipeline keeps a second load in
+      // flight while the previous vector's FMAs retire
+      int k8 = threadIdx.x * 8;
+      const int step = 256 * 8;
+      bool have = k8 + 8 <= kn;
+      s16x8 wv{};
+      if (have) wv = *reinterpret_cast<const s16x8*>(wr + k8);
+      while (have) {
+        const int nk8 = k8 + step;
+        const bool hn = nk8 + 8 <= kn;
+        s16x8 nxt{};
+        if (hn) nxt = *reinterpret_cast<const s16x8*>(wr + nk8);
 #pragma unroll
         for (int j = 0; j < 8; ++j) {
           const float wf = bf2f(ushort(wv[j]));
@@ -49,23 +61,32 @@ __global__ void gemv_k(const ushort* __restrict__ x,
           for (int mm = 0; mm < MT; ++mm)
             acc[mm] = fmaf(wf, xk[mm], acc[mm]);
         }
+        k8 = nk8;
+        wv = nxt;
+        have = hn;
       }
       // ragged tail (K % 8 != 0 never occurs at our shapes, kept safe)
-      for (int k = (kn & ~7) + lane; k < kn; k += 64) {
+      for (int k = (kn & ~7) + threadIdx.x; k < kn; k += 256) {
         const float wf = bf2f(wr[k]);
 #pragma unroll
         for (int mm = 0; mm < MT; ++mm)
           acc[mm] = fmaf(wf, xs[k * MT + mm], acc[mm]);
       }
     }
-  }
-  if (n0 >= N) return;
 #pragma unroll
-  for (int mm = 0; mm < MT; ++mm) acc[mm] = wave_sum(acc[mm]);
-  if (lane == 0) {
-    const float bv = bias ? bf2f(bias[n0]) : 0.f;
-    for (int mm = 0; mm < M; ++mm)
-      y[int64_t(mm) * N + n0] = f2bf(acc[mm] + bv);
+    for (int mm = 0; mm < MT; ++mm) acc[mm] = wave_sum(acc[mm]);
+    if (lane == 0) {
+#pragma unroll
+      for (int mm = 0; mm < MT; ++mm) red[wave][mm] = acc[mm];
+    }
+    __syncthreads();
+    if (threadIdx.x == 0) {
+      const float bv = bias ? bf2f(bias[n]) : 0.f;
+      for (int mm = 0; mm < M; ++mm)
+        y[int64_t(mm) * N + n] = f2bf(red[0][mm] + red[1][mm] +
+                                      red[2][mm] + red[3][mm] + bv);
+    }
+    __syncthreads();
   }
 }
 
@@ -73,13 +94,11 @@ __global__ void gemv_k(const ushort* __restrict__ x,
 
 void launch_gemv(const bf16_t* x, const bf16_t* w, const bf16_t* bias,
                  bf16_t* y, int M, int64_t N, int K, hipStream_t s) {
-  const int grid = int((N + 3) / 4);
+  const int grid = int(N < 16384 ? N : 16384);
 #define GEMV(MT)                                                          \
   gemv_k<MT><<<grid, 256, 0, s>>>(x, w, bias, y, M, N, K)
   if (M == 1) GEMV(1);
   else if (M <= 2) GEMV(2);
-  else if (M <= 4) GEMV(4);
-  else if (M <= 8) GEMV(8);
-  else GEMV(16);
+  else GEMV(4);
 #undef GEMV
 }

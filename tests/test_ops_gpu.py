@@ -1035,8 +1035,8 @@ def test_graphed_decode_matches_eager_llama():
 
 
 @pytest.mark.parametrize("M,N,K", [(1, 4096, 4096), (1, 3000, 768),
-                                   (4, 1024, 4096), (16, 768, 3072),
-                                   (2, 50264, 768)])
+                                   (4, 1024, 4096), (3, 768, 3072),
+                                   (1, 14336, 4096), (2, 50264, 768)])
 def test_gemv_matches_torch(M, N, K):
     """The weight-streaming serving GEMV vs torch fp32 (with bias)."""
     m = _ext()
