@@ -575,11 +575,13 @@ __global__ void attn_bwd_dkv_k(const ushort* __restrict__ dout,
 
 // ---------------- decode (KV-cache serving path) ----------------
 // One new query per (b,h) against a cached K/V of length kvlen (no mask:
-// every cached key is attended). One wave per (b,h), flash-decoding
-// shape: per 64-key chunk, phase 1 computes the chunk's scores (lane =
-// key; full q register-resident), phase 2 rescales the per-lane output
-// columns online (lane = head-dim column, CPL columns each). Wave-private
-// LDS carries the chunk probabilities between phases.
+// every cached key is attended). ONE BLOCK per (b,h): the 4 waves split
+// the cache into interleaved 64-key chunks (4x the scan parallelism of
+// the round-1 wave-per-head layout, which left batch-1 GPT-2/Llama with
+// 8 workgroups on 256 CUs and measured 2 ms/token in the llama decode),
+// each wave online-softmaxing its subset; a flash-decoding LDS combine
+// (m*, rescaled l and acc) merges the four partials. Phase 2 unrolls
+// the V-row walk 2-deep to keep two row loads in flight.
 template <int DTILES>  // D = 16*DTILES; CPL = D/64 columns per lane
 __global__ void attn_decode_k(const ushort* __restrict__ q,   // [B,H,D]
                               const ushort* __restrict__ kc,  // [B,Hk,L,D]
@@ -592,7 +594,7 @@ __global__ void attn_decode_k(const ushort* __restrict__ q,   // [B,H,D]
   constexpr int D = 16 * DTILES;
   constexpr int CPL = D / 64;
   const int lane = threadIdx.x & 63, wid = threadIdx.x >> 6;
-  const int bh = blockIdx.x * 4 + wid;
+  const int bh = blockIdx.x;
   if (bh >= B * H) return;
   // hipGraph-replayable serving: the valid cache length comes from a
   // device counter (the append wrote row *len_p, so attend *len_p + 1)
@@ -614,13 +616,16 @@ __global__ void attn_decode_k(const ushort* __restrict__ q,   // [B,H,D]
   }
 
   __shared__ float p_all[4][64];
+  __shared__ float comb_m[4], comb_l[4];
+  __shared__ float comb_acc[4][D];
   float* p = p_all[wid];
   float m_run = -INFINITY, l_run = 0.f;
   float acc[CPL];
 #pragma unroll
   for (int c = 0; c < CPL; ++c) acc[c] = 0.f;
 
-  for (int s0 = 0; s0 < kvlen; s0 += 64) {
+  // wave wid owns interleaved chunks s0 = (wid + 4i) * 64
+  for (int s0 = wid * 64; s0 < kvlen; s0 += 4 * 64) {
     // phase 1: this lane's key
     const int srow = s0 + lane;
     float sc = -INFINITY;
@@ -650,11 +655,26 @@ __global__ void attn_decode_k(const ushort* __restrict__ q,   // [B,H,D]
     m_run = m_new;
     p[lane] = pj;
     __threadfence_block();   // wave-local LDS visibility
-    // phase 2: lane owns CPL output columns
+    // phase 2: lane owns CPL output columns; 2-deep V-row pipeline
     const int n = min(64, kvlen - s0);
 #pragma unroll
     for (int c = 0; c < CPL; ++c) acc[c] *= alpha;
-    for (int s_ = 0; s_ < n; ++s_) {
+    int s_ = 0;
+    for (; s_ + 2 <= n; s_ += 2) {
+      const ushort* vr0 = vp + int64_t(s0 + s_) * D;
+      const ushort* vr1 = vr0 + D;
+      float va0[CPL], va1[CPL];
+#pragma unroll
+      for (int c = 0; c < CPL; ++c) {
+        va0[c] = bf2f(vr0[64 * c + lane]);
+        va1[c] = bf2f(vr1[64 * c + lane]);
+      }
+      const float p0 = p[s_], p1 = p[s_ + 1];
+#pragma unroll
+      for (int c = 0; c < CPL; ++c)
+        acc[c] = fmaf(p1, va1[c], fmaf(p0, va0[c], acc[c]));
+    }
+    if (s_ < n) {
       const ushort* vr = vp + int64_t(s0 + s_) * D;
       const float ps_ = p[s_];
 #pragma unroll
@@ -663,10 +683,37 @@ __global__ void attn_decode_k(const ushort* __restrict__ q,   // [B,H,D]
     }
     __threadfence_block();   // p reads done before next chunk overwrites
   }
-  const float inv = l_run > 0.f ? 1.0f / l_run : 0.f;
-  ushort* op = o + (int64_t(b) * H + h) * D;
+  // flash-decoding combine of the 4 wave-partials via LDS
+  if (lane == 0) {
+    comb_m[wid] = m_run;
+    comb_l[wid] = l_run;
+  }
 #pragma unroll
-  for (int c = 0; c < CPL; ++c) op[64 * c + lane] = f2bf(acc[c] * inv);
+  for (int c = 0; c < CPL; ++c) comb_acc[wid][64 * c + lane] = acc[c];
+  __syncthreads();
+  if (wid == 0) {
+    float mg = -INFINITY;
+#pragma unroll
+    for (int w = 0; w < 4; ++w) mg = fmaxf(mg, comb_m[w]);
+    float lg = 0.f;
+    float r[4];
+#pragma unroll
+    for (int w = 0; w < 4; ++w) {
+      r[w] = (comb_m[w] == -INFINITY) ? 0.f
+                                      : __builtin_exp2f(comb_m[w] - mg);
+      lg += comb_l[w] * r[w];
+    }
+    const float inv = lg > 0.f ? 1.0f / lg : 0.f;
+    ushort* op = o + (int64_t(b) * H + h) * D;
+#pragma unroll
+    for (int c = 0; c < CPL; ++c) {
+      float s = 0.f;
+#pragma unroll
+      for (int w = 0; w < 4; ++w)
+        s = fmaf(comb_acc[w][64 * c + lane], r[w], s);
+      op[64 * c + lane] = f2bf(s * inv);
+    }
+  }
 }
 
 // ---------------- mfma layout probes ----------------
@@ -772,7 +819,7 @@ void launch_attn_decode(const bf16_t* q, const bf16_t* kc, const bf16_t* vc,
                         bf16_t* o, int B, int H, int grp, int kvlen,
                         const int* len_p, int hd, int64_t cb, int64_t ch,
                         float scale, hipStream_t s) {
-  const int grid = (B * H + 3) / 4;
+  const int grid = B * H;   // block per (b, h); 4 waves split the cache
   if (hd == 64)
     attn_decode_k<4><<<grid, 256, 0, s>>>(q, kc, vc, o, B, H, grp, kvlen,
                                           len_p, cb, ch, scale);
