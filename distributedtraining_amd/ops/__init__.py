@@ -104,13 +104,18 @@ def linear(x: torch.Tensor, w: torch.Tensor,
            b: Optional[torch.Tensor] = None) -> torch.Tensor:
     """y = x @ w.T (+ b). hipBLASLt GEMM with fused bias epilogue forward;
     backward: wgrad+bias-grad in one BGRADB-epilogue GEMM accumulated into
-    the flat-grad plane (colsum fallback). Inference at M <= 4 (serving
-    decode) routes to the hand-written weight-streaming GEMV kernel —
-    hipBLASLt's small-M path measured 1.7 TB/s of weight stream vs the
-    ~6.3 TB/s HBM rate this shape is bound by."""
+    the flat-grad plane (colsum fallback).
+
+    Serving decode (inference, tiny M): LATENCY-bound projections route
+    to the hand-written GEMV; the big weight-streaming shapes stay on
+    tuned hipBLASLt. Per-shape A/B at M=1 (tools/gemv_ab.py, MI355X):
+    gemv wins up to ~17M-element weights (q/o 4096x4096: 12.1 vs
+    22.6 us), tuned hipBLASLt wins the streaming shapes (down-proj
+    22.0 vs 53.3 us at 5.3 TB/s; untuned heuristics were the original
+    1.7 TB/s problem — tuning/ carries the decode-shape selections)."""
     if use_hip(x):
         rows = x.numel() // x.shape[-1]
-        if rows <= 4 and not torch.is_grad_enabled():
+        if rows <= 4 and not torch.is_grad_enabled() and _gemv_wins(rows, w):
             x2 = x.reshape(rows, x.shape[-1])
             if not x2.is_contiguous():
                 x2 = x2.contiguous()
@@ -120,6 +125,14 @@ def linear(x: torch.Tensor, w: torch.Tensor,
             return y.view(*x.shape[:-1], w.shape[0])
         return _LinearFn.apply(x, w, b)
     return F.linear(x, w, b)
+
+
+def _gemv_wins(rows: int, w: torch.Tensor) -> bool:
+    """Measured crossover (tools/gemv_ab.py): the hand-written GEMV beats
+    tuned hipBLASLt while the weight matrix is latency- not
+    bandwidth-bound."""
+    numel = w.shape[0] * w.shape[1]
+    return numel < (20_000_000 if rows == 1 else 5_000_000)
 
 
 # --------------------------------------------------------------------------
