@@ -43,6 +43,11 @@ class LocalSGDNode:
         self.model = model
         self.val_batches = val_batches or []
         self.merge_strategy = merge_strategy or cfg.average.strategy
+        # decorrelate the dropout streams across miner ranks (counters
+        # start at 0 everywhere; without an offset all miners would draw
+        # IDENTICAL mask sequences)
+        from ..ops import droprng
+        droprng.counter(fp.device).fill_(comm.rank << 32)
         self.miner = DeltaLoop(model, fp, data_iter, cfg.train,
                                hotkey=f"rank{comm.rank}")
         self.averager = ParameterizedAverager(model, fp, cfg.average)
