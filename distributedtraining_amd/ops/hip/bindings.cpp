@@ -428,6 +428,8 @@ Tensor embedding_fwd(Tensor ids, Tensor wte, Tensor wpe, Tensor pos) {
 void kv_append(Tensor kn, Tensor vn, Tensor kc, Tensor vc, Tensor pos) {
   TORCH_CHECK(kn.dim() == 3 && kn.size(1) == 1 && kn.stride(2) == 1,
               "kn must be [B,1,F] with contiguous F");
+  TORCH_CHECK(vn.stride(0) == kn.stride(0) && vn.stride(2) == 1,
+              "vn must share kn's layout (one qkv projection)");
   TORCH_CHECK(pos.scalar_type() == torch::kInt32 && pos.is_cuda(), "pos");
   const int B = int(kc.size(0)), Hk = int(kc.size(1)), D = int(kc.size(3));
   launch_kv_append(bfp(kn), bfp(vn), kn.stride(0), bfp_mut(kc), bfp_mut(vc),
