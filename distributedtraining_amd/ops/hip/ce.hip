@@ -24,33 +24,51 @@ __global__ void ce_fwd_k(const ushort* __restrict__ logits,
   __shared__ float lds[16];
   for (int64_t row = blockIdx.x; row < rows; row += gridDim.x) {
     const ushort* xr = logits + row * vocab;
-    // online pass: per-thread running (m, s) in exp2 units, one rescale
-    // per 8-wide vector
-    float m_run = -INFINITY, s_run = 0.f;
-    int64_t i = int64_t(threadIdx.x) * 8;
-    const int64_t stride = int64_t(CE_BLOCK) * 8;
-    for (; i + 8 <= vocab; i += stride) {
-      s16x8 vx = *reinterpret_cast<const s16x8*>(xr + i);
-      float f[8];
+    // online pass: TWO independent per-thread (m, s) streams (the single
+    // running pair is a serial rescale chain — no ILP across vectors;
+    // splitting even/odd vectors measured the fwd at 1.45x SOL before)
+    float m0 = -INFINITY, s0 = 0.f, m1 = -INFINITY, s1 = 0.f;
+    int64_t i = int64_t(threadIdx.x) * 16;
+    const int64_t stride = int64_t(CE_BLOCK) * 16;
+    for (; i + 16 <= vocab; i += stride) {
+      s16x8 va = *reinterpret_cast<const s16x8*>(xr + i);
+      s16x8 vb = *reinterpret_cast<const s16x8*>(xr + i + 8);
+      float fa[8], fb[8];
 #pragma unroll
-      for (int j = 0; j < 8; ++j) f[j] = bf2f(ushort(vx[j])) * LOG2E;
-      float mx = fmaxf(fmaxf(fmaxf(f[0], f[1]), fmaxf(f[2], f[3])),
-                       fmaxf(fmaxf(f[4], f[5]), fmaxf(f[6], f[7])));
-      const float m_new = fmaxf(m_run, mx);
-      float ps = 0.f;
+      for (int j = 0; j < 8; ++j) {
+        fa[j] = bf2f(ushort(va[j])) * LOG2E;
+        fb[j] = bf2f(ushort(vb[j])) * LOG2E;
+      }
+      const float mxa = fmaxf(fmaxf(fmaxf(fa[0], fa[1]), fmaxf(fa[2], fa[3])),
+                              fmaxf(fmaxf(fa[4], fa[5]), fmaxf(fa[6], fa[7])));
+      const float mxb = fmaxf(fmaxf(fmaxf(fb[0], fb[1]), fmaxf(fb[2], fb[3])),
+                              fmaxf(fmaxf(fb[4], fb[5]), fmaxf(fb[6], fb[7])));
+      const float na = fmaxf(m0, mxa), nb = fmaxf(m1, mxb);
+      float pa = 0.f, pb = 0.f;
 #pragma unroll
-      for (int j = 0; j < 8; ++j) ps += __builtin_exp2f(f[j] - m_new);
-      s_run = s_run * __builtin_exp2f(m_run - m_new) + ps;
-      m_run = m_new;
+      for (int j = 0; j < 8; ++j) {
+        pa += __builtin_exp2f(fa[j] - na);
+        pb += __builtin_exp2f(fb[j] - nb);
+      }
+      s0 = s0 * __builtin_exp2f(m0 - na) + pa;
+      s1 = s1 * __builtin_exp2f(m1 - nb) + pb;
+      m0 = na;
+      m1 = nb;
     }
-    if (i < vocab && i + 8 > vocab)
+    if (i < vocab)
       for (; i < vocab; ++i) {
         const float f = bf2f(xr[i]) * LOG2E;
-        const float m_new = fmaxf(m_run, f);
-        s_run = s_run * __builtin_exp2f(m_run - m_new) +
-                __builtin_exp2f(f - m_new);
-        m_run = m_new;
+        const float m_new = fmaxf(m0, f);
+        s0 = s0 * __builtin_exp2f(m0 - m_new) +
+             __builtin_exp2f(f - m_new);
+        m0 = m_new;
       }
+    // merge the two streams (guard empty ones: exp2(-inf - -inf) is NaN
+    // for threads whose slice starts past a small vocab)
+    const float m_run = fmaxf(m0, m1);
+    float s_run = 0.f;
+    if (m0 != -INFINITY) s_run += s0 * __builtin_exp2f(m0 - m_run);
+    if (m1 != -INFINITY) s_run += s1 * __builtin_exp2f(m1 - m_run);
     // combine the per-thread (m, s) pairs across the block
     const float M = block_max<16>(m_run, lds);
     const float s_adj = s_run * __builtin_exp2f(m_run - M);
@@ -91,27 +109,29 @@ __global__ void ce_bwd_k(const ushort* __restrict__ logits,
     const int64_t tgt = tgt_raw - v0;
     const float lb = lse[row] * LOG2E;
     const float sc = (tgt_raw == ignore_index) ? 0.f : sc_base;
-    // two 16 B vectors in flight per iteration (load/store ILP)
-    int64_t i = int64_t(threadIdx.x) * 16;
-    const int64_t stride = int64_t(CE_BLOCK) * 16;
-    for (; i + 16 <= vocab; i += stride) {
-      s16x8 va = *reinterpret_cast<const s16x8*>(xr + i);
-      s16x8 vb = *reinterpret_cast<const s16x8*>(xr + i + 8);
-      s16x8 oa, ob;
+    // four 16 B vectors in flight per iteration (two measured 1.55x SOL;
+    // the extra ILP covers the ~900-cycle HBM latency of the row stream)
+    int64_t i = int64_t(threadIdx.x) * 32;
+    const int64_t stride = int64_t(CE_BLOCK) * 32;
+    for (; i + 32 <= vocab; i += stride) {
+      s16x8 v[4], o[4];
 #pragma unroll
-      for (int j = 0; j < 8; ++j) {
-        float pa = __builtin_exp2f(bf2f(ushort(va[j])) * LOG2E - lb);
-        float pb = __builtin_exp2f(bf2f(ushort(vb[j])) * LOG2E - lb);
-        oa[j] = f2bf(sc * (pa - ((i + j) == tgt ? 1.f : 0.f)));
-        ob[j] = f2bf(sc * (pb - ((i + 8 + j) == tgt ? 1.f : 0.f)));
-      }
-      if (NT) {
-        __builtin_nontemporal_store(oa, reinterpret_cast<s16x8*>(dxr + i));
-        __builtin_nontemporal_store(ob,
-                                    reinterpret_cast<s16x8*>(dxr + i + 8));
-      } else {
-        *reinterpret_cast<s16x8*>(dxr + i) = oa;
-        *reinterpret_cast<s16x8*>(dxr + i + 8) = ob;
+      for (int u = 0; u < 4; ++u)
+        v[u] = *reinterpret_cast<const s16x8*>(xr + i + 8 * u);
+#pragma unroll
+      for (int u = 0; u < 4; ++u)
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          float p = __builtin_exp2f(bf2f(ushort(v[u][j])) * LOG2E - lb);
+          o[u][j] = f2bf(sc * (p - ((i + 8 * u + j) == tgt ? 1.f : 0.f)));
+        }
+#pragma unroll
+      for (int u = 0; u < 4; ++u) {
+        if (NT)
+          __builtin_nontemporal_store(
+              o[u], reinterpret_cast<s16x8*>(dxr + i + 8 * u));
+        else
+          *reinterpret_cast<s16x8*>(dxr + i + 8 * u) = o[u];
       }
     }
     if (i < vocab)
