@@ -25,14 +25,16 @@ __global__ void ce_fwd_k(const ushort* __restrict__ logits,
   for (int64_t row = blockIdx.x; row < rows; row += gridDim.x) {
     const ushort* xr = logits + row * vocab;
     // online pass: TWO independent per-thread (m, s) streams (the single
-    // running pair is a serial rescale chain — no ILP across vectors;
-    // splitting even/odd vectors measured the fwd at 1.45x SOL before)
+    // running pair is a serial rescale chain — no ILP across vectors),
+    // each reading a WAVE-CONTIGUOUS 1 KiB pass (stream B one full
+    // block-sweep ahead — per-thread window widening shatters wave
+    // coalescing, see ce_bwd_k)
     float m0 = -INFINITY, s0 = 0.f, m1 = -INFINITY, s1 = 0.f;
-    int64_t i = int64_t(threadIdx.x) * 16;
-    const int64_t stride = int64_t(CE_BLOCK) * 16;
-    for (; i + 16 <= vocab; i += stride) {
+    const int64_t CB = int64_t(CE_BLOCK) * 8;
+    int64_t i = int64_t(threadIdx.x) * 8;
+    for (; i + CB + 8 <= vocab; i += 2 * CB) {
       s16x8 va = *reinterpret_cast<const s16x8*>(xr + i);
-      s16x8 vb = *reinterpret_cast<const s16x8*>(xr + i + 8);
+      s16x8 vb = *reinterpret_cast<const s16x8*>(xr + i + CB);
       float fa[8], fb[8];
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
@@ -54,6 +56,20 @@ __global__ void ce_fwd_k(const ushort* __restrict__ logits,
       s1 = s1 * __builtin_exp2f(m1 - nb) + pb;
       m0 = na;
       m1 = nb;
+    }
+    for (; i + 8 <= vocab; i += CB) {
+      s16x8 va = *reinterpret_cast<const s16x8*>(xr + i);
+      float fa[8];
+#pragma unroll
+      for (int j = 0; j < 8; ++j) fa[j] = bf2f(ushort(va[j])) * LOG2E;
+      const float mxa = fmaxf(fmaxf(fmaxf(fa[0], fa[1]), fmaxf(fa[2], fa[3])),
+                              fmaxf(fmaxf(fa[4], fa[5]), fmaxf(fa[6], fa[7])));
+      const float na = fmaxf(m0, mxa);
+      float pa = 0.f;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) pa += __builtin_exp2f(fa[j] - na);
+      s0 = s0 * __builtin_exp2f(m0 - na) + pa;
+      m0 = na;
     }
     if (i < vocab)
       for (; i < vocab; ++i) {
@@ -109,30 +125,45 @@ __global__ void ce_bwd_k(const ushort* __restrict__ logits,
     const int64_t tgt = tgt_raw - v0;
     const float lb = lse[row] * LOG2E;
     const float sc = (tgt_raw == ignore_index) ? 0.f : sc_base;
-    // four 16 B vectors in flight per iteration (two measured 1.55x SOL;
-    // the extra ILP covers the ~900-cycle HBM latency of the row stream)
-    int64_t i = int64_t(threadIdx.x) * 32;
-    const int64_t stride = int64_t(CE_BLOCK) * 32;
-    for (; i + 32 <= vocab; i += stride) {
-      s16x8 v[4], o[4];
+    // two WAVE-CONTIGUOUS 1 KiB passes in flight per iteration: lane l
+    // reads 16 B at 16·l — perfectly coalesced — and the second pass sits
+    // one whole block-sweep (CB elements) ahead. (Widening the per-THREAD
+    // window instead measured 2.3x SLOWER: 32-element windows stride the
+    // wave's lanes 64 B apart and shatter coalescing.)
+    const int64_t CB = int64_t(CE_BLOCK) * 8;
+    int64_t i = int64_t(threadIdx.x) * 8;
+    for (; i + CB + 8 <= vocab; i += 2 * CB) {
+      s16x8 va = *reinterpret_cast<const s16x8*>(xr + i);
+      s16x8 vb = *reinterpret_cast<const s16x8*>(xr + i + CB);
+      s16x8 oa, ob;
 #pragma unroll
-      for (int u = 0; u < 4; ++u)
-        v[u] = *reinterpret_cast<const s16x8*>(xr + i + 8 * u);
-#pragma unroll
-      for (int u = 0; u < 4; ++u)
-#pragma unroll
-        for (int j = 0; j < 8; ++j) {
-          float p = __builtin_exp2f(bf2f(ushort(v[u][j])) * LOG2E - lb);
-          o[u][j] = f2bf(sc * (p - ((i + 8 * u + j) == tgt ? 1.f : 0.f)));
-        }
-#pragma unroll
-      for (int u = 0; u < 4; ++u) {
-        if (NT)
-          __builtin_nontemporal_store(
-              o[u], reinterpret_cast<s16x8*>(dxr + i + 8 * u));
-        else
-          *reinterpret_cast<s16x8*>(dxr + i + 8 * u) = o[u];
+      for (int j = 0; j < 8; ++j) {
+        float pa = __builtin_exp2f(bf2f(ushort(va[j])) * LOG2E - lb);
+        float pb = __builtin_exp2f(bf2f(ushort(vb[j])) * LOG2E - lb);
+        oa[j] = f2bf(sc * (pa - ((i + j) == tgt ? 1.f : 0.f)));
+        ob[j] = f2bf(sc * (pb - ((i + CB + j) == tgt ? 1.f : 0.f)));
       }
+      if (NT) {
+        __builtin_nontemporal_store(oa, reinterpret_cast<s16x8*>(dxr + i));
+        __builtin_nontemporal_store(ob,
+                                    reinterpret_cast<s16x8*>(dxr + i + CB));
+      } else {
+        *reinterpret_cast<s16x8*>(dxr + i) = oa;
+        *reinterpret_cast<s16x8*>(dxr + i + CB) = ob;
+      }
+    }
+    for (; i + 8 <= vocab; i += CB) {
+      s16x8 va = *reinterpret_cast<const s16x8*>(xr + i);
+      s16x8 oa;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        float pa = __builtin_exp2f(bf2f(ushort(va[j])) * LOG2E - lb);
+        oa[j] = f2bf(sc * (pa - ((i + j) == tgt ? 1.f : 0.f)));
+      }
+      if (NT)
+        __builtin_nontemporal_store(oa, reinterpret_cast<s16x8*>(dxr + i));
+      else
+        *reinterpret_cast<s16x8*>(dxr + i) = oa;
     }
     if (i < vocab)
       for (; i < vocab; ++i) {
