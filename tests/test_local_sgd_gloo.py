@@ -248,3 +248,26 @@ def test_bf16_wire_gather_merge():
         p.join(timeout=60)
     assert all(ok for _, ok, _ in results), results
     assert results[0][2] == pytest.approx(results[1][2], rel=1e-6)
+
+
+@pytest.mark.timeout(600)
+def test_eight_rank_local_sgd_allreduce(tmp_path):
+    """World 8 — the driver's SCALE topology (8 miner ranks, all-reduce
+    mean merge): every rank must hold the identical merged base."""
+    world = 8
+    port = 29841
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    procs = [ctx.Process(target=_worker,
+                         args=(r, world, port, str(tmp_path), q))
+             for r in range(world)]
+    for p in procs:
+        p.start()
+    results = [q.get(timeout=500) for _ in range(world)]
+    for p in procs:
+        p.join(timeout=60)
+    assert all(ok for _, ok, _ in results), results
+    digests = [d for _, _, d in results]
+    for d in digests[1:]:
+        assert digests[0] == pytest.approx(d, rel=1e-6), \
+            "ranks diverged after merge"
