@@ -89,3 +89,17 @@ def test_textdataset_with_bpe(tok):
                     attention_mask=b["attention_mask"],
                     labels=b["labels"])
     assert torch.isfinite(out.loss)
+
+
+def test_roundtrip_property_random_unicode(tok):
+    """Property: decode(encode(t)) == t for arbitrary unicode (byte-level
+    BPE never loses information)."""
+    hypothesis = pytest.importorskip("hypothesis")
+    from hypothesis import given, settings, strategies as st
+
+    @settings(max_examples=200, deadline=None)
+    @given(st.text(max_size=64))
+    def check(t):
+        assert tok.decode(tok.encode(t)) == t
+
+    check()
