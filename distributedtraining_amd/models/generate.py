@@ -97,7 +97,10 @@ def generate(model, input_ids: torch.Tensor, max_new_tokens: int,
                 if ((eos_token_id is not None and bool(done.all()))
                         or cache.len >= cache.max_len):
                     break
-                if dec is None and cache.pos_dev is not None:
+                # capture costs ~2 warmup decodes + the graph build —
+                # only worth amortizing over a long enough generation
+                if (dec is None and cache.pos_dev is not None
+                        and max_new_tokens >= 16):
                     dec = GraphedDecoder(model, cache, ids.shape[0],
                                          ids.device)
                 step = dec.step if dec is not None else (

@@ -999,10 +999,10 @@ def test_graphed_decode_matches_eager_decode():
     torch.manual_seed(4)
     m = build_model(cfg).to(DEV, torch.bfloat16).eval()
     ids = torch.randint(0, 512, (3, 10), device=DEV)
-    graphed = generate(m, ids, max_new_tokens=12, use_cache=True)
+    graphed = generate(m, ids, max_new_tokens=16, use_cache=True)
     orig = m.new_cache
     m.new_cache = lambda *a, **kw: orig(*a, **{**kw, "graphable": False})
-    eager = generate(m, ids, max_new_tokens=12, use_cache=True)
+    eager = generate(m, ids, max_new_tokens=16, use_cache=True)
     m.new_cache = orig
     assert torch.equal(graphed, eager)
 
@@ -1019,10 +1019,10 @@ def test_graphed_decode_matches_eager_llama():
     torch.manual_seed(6)
     m = build_model(cfg).to(DEV, torch.bfloat16).eval()
     ids = torch.randint(0, 512, (2, 8), device=DEV)
-    graphed = generate(m, ids, max_new_tokens=10, use_cache=True)
+    graphed = generate(m, ids, max_new_tokens=16, use_cache=True)
     orig = m.new_cache
     m.new_cache = lambda *a, **kw: orig(*a, **{**kw, "graphable": False})
-    eager = generate(m, ids, max_new_tokens=10, use_cache=True)
+    eager = generate(m, ids, max_new_tokens=16, use_cache=True)
     m.new_cache = orig
     assert torch.equal(graphed, eager)
     # vs the full re-forward: same kernelset only for the first tokens —
