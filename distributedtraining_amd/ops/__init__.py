@@ -65,6 +65,7 @@ class _LinearFn(torch.autograd.Function):
         ctx.has_b = b is not None
         ctx.xshape = x.shape
         ctx.wgrad = getattr(w, "main_grad", None)
+        ctx.bgrad = getattr(b, "main_grad", None) if b is not None else None
         y = torch.addmm(b, x2, w.t()) if b is not None else x2.mm(w.t())
         return y.view(*x.shape[:-1], w.shape[0])
 
@@ -96,7 +97,8 @@ class _LinearFn(torch.autograd.Function):
             dw = None
         else:
             dw = dy2.t().mm(x2)
-        db = m.colsum(dy2).to(w.dtype) if ctx.has_b else None
+        db = (_grad_out(m.colsum(dy2), ctx.bgrad, w.dtype)
+              if ctx.has_b else None)
         return dx, dw, db
 
 
@@ -164,6 +166,7 @@ class _FusedMLPFn(torch.autograd.Function):
         ctx.xshape = x.shape
         ctx.w1grad = getattr(w1, "main_grad", None)
         ctx.w2grad = getattr(w2, "main_grad", None)
+        ctx.b2grad = getattr(b2, "main_grad", None)
         return y.view(*x.shape[:-1], w2.shape[0])
 
     @staticmethod
@@ -173,7 +176,7 @@ class _FusedMLPFn(torch.autograd.Function):
         dy2 = dy.reshape(-1, dy.shape[-1])
         if dy2.stride(-1) != 1:
             dy2 = dy2.contiguous()
-        db2 = m.colsum(dy2).to(w2.dtype)
+        db2 = _grad_out(m.colsum(dy2), ctx.b2grad, w2.dtype)
         if ctx.w2grad is not None:
             ctx.w2grad.addmm_(dy2.t(), h)
             dw2 = None
@@ -227,6 +230,16 @@ def _rng0(t: torch.Tensor) -> torch.Tensor:
     return droprng.counter(t.device)
 
 
+def _grad_out(fp32_grad, main_grad, dtype):
+    """Deliver a parameter gradient: accumulate the fp32 reduction into
+    the flat bf16 grad plane when the param carries one (single fused
+    kernel), else hand autograd a cast tensor (plain-module models)."""
+    if main_grad is not None:
+        require_ext().accum_f32_into_bf16(main_grad, fp32_grad)
+        return None
+    return fp32_grad.to(dtype)
+
+
 class _LayerNormFn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, w, b, eps):
@@ -236,6 +249,8 @@ class _LayerNormFn(torch.autograd.Function):
                                            _empty_like0(x2), w, b, eps,
                                            _rng0(x), 0, 0.0)
         ctx.save_for_backward(x2, w, mean, rstd)
+        ctx.wg = getattr(w, "main_grad", None)
+        ctx.bg = getattr(b, "main_grad", None)
         return y.view_as(x2)
 
     @staticmethod
@@ -243,10 +258,11 @@ class _LayerNormFn(torch.autograd.Function):
         m = require_ext()
         x, w, mean, rstd = ctx.saved_tensors
         N = x.shape[-1]
-        dx, dw, db, _ = m.layernorm_bwd(dy.contiguous().view(-1, N),
-                                        _empty_like0(x), x.view(-1, N), w,
-                                        mean, rstd, _rng0(x), 0, 0.0)
-        return dx.view_as(x), dw, db, None
+        dx, dw32, db32, _ = m.layernorm_bwd(dy.contiguous().view(-1, N),
+                                            _empty_like0(x), x.view(-1, N),
+                                            w, mean, rstd, _rng0(x), 0, 0.0)
+        return (dx.view_as(x), _grad_out(dw32, ctx.wg, w.dtype),
+                _grad_out(db32, ctx.bg, w.dtype), None)
 
 
 def layer_norm(x: torch.Tensor, w: torch.Tensor, b: torch.Tensor,
@@ -279,6 +295,8 @@ class _AddLayerNormFn(torch.autograd.Function):
         ctx.save_for_backward(s, w, mean, rstd)
         ctx.shape = x.shape
         ctx.drop = (p_drop, site)
+        ctx.wg = getattr(w, "main_grad", None)
+        ctx.bg = getattr(b, "main_grad", None)
         return s.view(x.shape), y.view(x.shape)
 
     @staticmethod
@@ -289,12 +307,13 @@ class _AddLayerNormFn(torch.autograd.Function):
         N = s.shape[-1]
         ds2 = (ds.contiguous().view(-1, N) if ds is not None
                else _empty_like0(s))
-        dx, dw, db, dres = m.layernorm_bwd(dy.contiguous().view(-1, N),
-                                           ds2, s, w, mean, rstd, _rng0(s),
-                                           site, p_drop)
+        dx, dw32, db32, dres = m.layernorm_bwd(dy.contiguous().view(-1, N),
+                                               ds2, s, w, mean, rstd,
+                                               _rng0(s), site, p_drop)
         dx = dx.view(ctx.shape)
         dr = dres.view(ctx.shape) if p_drop > 0.0 else dx
-        return dx, dr, dw, db, None, None, None
+        return (dx, dr, _grad_out(dw32, ctx.wg, w.dtype),
+                _grad_out(db32, ctx.bg, w.dtype), None, None, None)
 
 
 def add_layer_norm(x: torch.Tensor, res: torch.Tensor, w: torch.Tensor,
@@ -328,6 +347,7 @@ class _RMSNormFn(torch.autograd.Function):
                                    _empty_like0(x2), w, eps, _rng0(x), 0,
                                    0.0)
         ctx.save_for_backward(x2, w, rstd)
+        ctx.wg = getattr(w, "main_grad", None)
         return y.view_as(x2)
 
     @staticmethod
@@ -335,10 +355,10 @@ class _RMSNormFn(torch.autograd.Function):
         m = require_ext()
         x, w, rstd = ctx.saved_tensors
         N = x.shape[-1]
-        dx, dw, _ = m.rmsnorm_bwd(dy.contiguous().view(-1, N),
-                                  _empty_like0(x), x.view(-1, N), w, rstd,
-                                  _rng0(x), 0, 0.0)
-        return dx.view_as(x), dw, None
+        dx, dw32, _ = m.rmsnorm_bwd(dy.contiguous().view(-1, N),
+                                    _empty_like0(x), x.view(-1, N), w,
+                                    rstd, _rng0(x), 0, 0.0)
+        return dx.view_as(x), _grad_out(dw32, ctx.wg, w.dtype), None
 
 
 class _AddRMSNormFn(torch.autograd.Function):
@@ -356,6 +376,7 @@ class _AddRMSNormFn(torch.autograd.Function):
         ctx.save_for_backward(s, w, rstd)
         ctx.shape = x.shape
         ctx.drop = (p_drop, site)
+        ctx.wg = getattr(w, "main_grad", None)
         return s.view(x.shape), y.view(x.shape)
 
     @staticmethod
@@ -366,11 +387,11 @@ class _AddRMSNormFn(torch.autograd.Function):
         N = s.shape[-1]
         ds2 = (ds.contiguous().view(-1, N) if ds is not None
                else _empty_like0(s))
-        dx, dw, dres = m.rmsnorm_bwd(dy.contiguous().view(-1, N), ds2, s,
-                                     w, rstd, _rng0(s), site, p_drop)
+        dx, dw32, dres = m.rmsnorm_bwd(dy.contiguous().view(-1, N), ds2,
+                                       s, w, rstd, _rng0(s), site, p_drop)
         dx = dx.view(ctx.shape)
         dr = dres.view(ctx.shape) if p_drop > 0.0 else dx
-        return dx, dr, dw, None, None, None
+        return (dx, dr, _grad_out(dw32, ctx.wg, w.dtype), None, None, None)
 
 
 def add_rms_norm(x: torch.Tensor, res: torch.Tensor, w: torch.Tensor,
@@ -889,14 +910,21 @@ class _EmbeddingFn(torch.autograd.Function):
         ctx.vocab, ctx.npos = wte.shape[0], (wpe.shape[0] if wpe is not None else 0)
         ctx.dim = wte.shape[1]
         ctx.dtype = wte.dtype
+        ctx.wteg = getattr(wte, "main_grad", None)
+        ctx.wpeg = (getattr(wpe, "main_grad", None)
+                    if wpe is not None else None)
         return m.embedding_fwd(ids, wte, wpe if wpe is not None else torch.empty(0, dtype=wte.dtype, device=wte.device), torch.empty(0, dtype=torch.int32, device=wte.device))
 
     @staticmethod
     def backward(ctx, dy):
         m = require_ext()
         (ids,) = ctx.saved_tensors
-        dwte, dwpe = m.embedding_bwd(dy.contiguous(), ids, ctx.vocab, ctx.npos)
-        return None, dwte, (dwpe if ctx.npos else None)
+        dwte32, dwpe32 = m.embedding_bwd(dy.contiguous(), ids, ctx.vocab,
+                                         ctx.npos)
+        dwte = _grad_out(dwte32, ctx.wteg, ctx.dtype)
+        dwpe = (_grad_out(dwpe32, ctx.wpeg, ctx.dtype) if ctx.npos
+                else None)
+        return None, dwte, dwpe
 
 
 def embedding_fwd(ids: torch.Tensor, wte: torch.Tensor,

@@ -113,7 +113,7 @@ std::vector<Tensor> layernorm_bwd(Tensor dy, Tensor ds, Tensor x, Tensor w,
                        part[0].data_ptr<float>(), part[1].data_ptr<float>(),
                        stripes, rows, cols, da.rng, da.site, da.thr16,
                        da.ik, stream());
-  return {dx, dw32.to(torch::kBFloat16), db32.to(torch::kBFloat16), dres};
+  return {dx, dw32, db32, dres};  // fp32 grads: ops layer accumulates or casts
 }
 
 std::vector<Tensor> rmsnorm_fwd(Tensor x, Tensor res, Tensor w,
@@ -160,7 +160,7 @@ std::vector<Tensor> rmsnorm_bwd(Tensor dy, Tensor ds, Tensor x, Tensor w,
                      dw32.data_ptr<float>(), part.data_ptr<float>(),
                      stripes, rows, cols, da.rng, da.site, da.thr16, da.ik,
                      stream());
-  return {dx, dw32.to(torch::kBFloat16), dres};
+  return {dx, dw32, dres};  // fp32 grad
 }
 
 // ---- elementwise ----------------------------------------------------------
@@ -210,6 +210,14 @@ Tensor dropout_apply(Tensor x, Tensor rng, int64_t site, double p) {
                      rng.data_ptr<int64_t>()),
                  (unsigned long long)site, thr, inv_keep, stream());
   return y;
+}
+
+// dst must be a CONTIGUOUS bf16 view (a flat-grad slice); src fp32
+void accum_f32_into_bf16(Tensor dst, Tensor src) {
+  check_bf16(dst, "dst"); check_f32(src, "src");
+  TORCH_CHECK(dst.numel() == src.numel(), "size mismatch");
+  launch_accum_f32_bf16(bfp_mut(dst), src.data_ptr<float>(), dst.numel(),
+                        stream());
 }
 
 void rng_tick(Tensor ctr) {
@@ -466,8 +474,7 @@ std::vector<Tensor> embedding_bwd(Tensor dy, Tensor ids, int64_t vocab,
     launch_colsum(bfp(dy), part.data_ptr<float>(), dwpe.data_ptr<float>(),
                   B, pcols, stripes, stream());
   }
-  return {dwte.to(torch::kBFloat16),
-          npos ? dwpe.to(torch::kBFloat16) : dwpe};
+  return {dwte, dwpe};  // fp32: ops layer accumulates or casts
 }
 
 // ---- serving gemv ----------------------------------------------------------
@@ -726,6 +733,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("swiglu_bwd", &swiglu_bwd);
   m.def("dropout_apply", &dropout_apply);
   m.def("rng_tick", &rng_tick);
+  m.def("accum_f32_into_bf16", &accum_f32_into_bf16);
   m.def("delta_sub", &delta_sub);
   m.def("axpy", &axpy);
   m.def("has_nan", &has_nan);

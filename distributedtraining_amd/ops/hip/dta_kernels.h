@@ -25,6 +25,9 @@ void launch_dropout(const bf16_t* x, bf16_t* y, int64_t n,
                     const unsigned long long* rng, unsigned long long site,
                     unsigned int thr16, float inv_keep, hipStream_t s);
 void launch_rng_tick(unsigned long long* ctr, hipStream_t s);
+// dst (bf16 grad view) += src (fp32): replaces cast + autograd-add pairs
+void launch_accum_f32_bf16(bf16_t* dst, const float* src, int64_t n,
+                           hipStream_t s);
 
 // ---- flat-plane ops -------------------------------------------------------
 void launch_delta_sub(const float* w, const float* base, float* out,

@@ -304,6 +304,30 @@ __global__ void dropout_k(const ushort* __restrict__ x,
 
 __global__ void rng_tick_k(unsigned long long* ctr) { ++(*ctr); }
 
+// dst (bf16 flat-grad view) += src (fp32 reduction result): one pass
+// replacing the fp32->bf16 cast kernel + autograd's separate add kernel
+// per parameter (the norm/bias/embedding grads were ~150 such kernel
+// pairs per step)
+__global__ void accum_f32_bf16_k(ushort* __restrict__ dst,
+                                 const float* __restrict__ src, int64_t n) {
+  int64_t i = (int64_t(blockIdx.x) * blockDim.x + threadIdx.x) * 8;
+  const int64_t stride = int64_t(gridDim.x) * blockDim.x * 8;
+  for (; i + 8 <= n; i += stride) {
+    s16x8 vd = *reinterpret_cast<const s16x8*>(dst + i);
+    f32x4 sa = *reinterpret_cast<const f32x4*>(src + i);
+    f32x4 sb = *reinterpret_cast<const f32x4*>(src + i + 4);
+    s16x8 o;
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      o[j] = f2bf(bf2f(ushort(vd[j])) + sa[j]);
+      o[j + 4] = f2bf(bf2f(ushort(vd[j + 4])) + sb[j]);
+    }
+    *reinterpret_cast<s16x8*>(dst + i) = o;
+  }
+  if (i < n)
+    for (; i < n; ++i) dst[i] = f2bf(bf2f(dst[i]) + src[i]);
+}
+
 }  // namespace
 
 #define LAUNCH_EW(kernel, n, ...)                                          \
@@ -367,4 +391,8 @@ void launch_dropout(const bf16_t* x, bf16_t* y, int64_t n,
 }
 void launch_rng_tick(unsigned long long* ctr, hipStream_t s) {
   rng_tick_k<<<1, 1, 0, s>>>(ctr);
+}
+void launch_accum_f32_bf16(bf16_t* dst, const float* src, int64_t n,
+                           hipStream_t s) {
+  LAUNCH_EW(accum_f32_bf16_k, n, dst, src, n);
 }
