@@ -86,3 +86,23 @@ def test_llama_loss_and_grad_match_transformers():
     g_ref = hf.model.embed_tokens.weight.grad
     g_got = ours.tok_emb.grad
     torch.testing.assert_close(g_got, g_ref, rtol=5e-3, atol=1e-5)
+
+
+def test_llama_padded_parity_vs_transformers():
+    """Right-padded batch through the Llama mask path matches transformers
+    with the same attention_mask and -100 pad labels (round-1 verdict #1
+    extended to the GQA family)."""
+    hf, ours = _hf_and_ours(seed=2)
+    torch.manual_seed(4)
+    B, S = 3, 20
+    ids = torch.randint(0, 512, (B, S))
+    lens = torch.tensor([11, 20, 7])
+    am = (torch.arange(S)[None, :] < lens[:, None]).long()
+    with torch.no_grad():
+        ref = hf(input_ids=ids, attention_mask=am,
+                 labels=ids.masked_fill(am == 0, -100))
+        got = ours(input_ids=ids, attention_mask=am, labels=ids)
+    torch.testing.assert_close(got.loss, ref.loss, rtol=2e-4, atol=2e-4)
+    valid = am[:, :-1].bool() & am[:, 1:].bool()
+    torch.testing.assert_close(got.logits[valid], ref.logits[:, :-1][valid],
+                               rtol=2e-4, atol=2e-4)
