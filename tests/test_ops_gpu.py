@@ -1050,3 +1050,47 @@ def test_gemv_matches_torch(M, N, K):
     y2 = m.gemv(x, w, torch.empty(0, dtype=torch.bfloat16, device=DEV))
     torch.testing.assert_close(y2.float(), ref - b.float(), rtol=2e-2,
                                atol=2e-2 * math.sqrt(K / 768))
+
+
+def test_graphed_step_matches_eager_with_padded_batches():
+    """hipGraph-captured step with STAGED attention masks: the kvlen
+    derivation + masked attention + CE-ignore all replay per batch with
+    different padding patterns."""
+    from distributedtraining_amd.config import Config, ModelConfig
+    from distributedtraining_amd.models import build_model
+    from distributedtraining_amd.parallel.flat import FlatParams
+    from distributedtraining_amd.parallel.graphstep import GraphedMinerStep
+    from distributedtraining_amd.roles.miner import DeltaLoop
+    from distributedtraining_amd.utils.data import synthetic_batches
+
+    cfg = Config()
+    cfg.model = ModelConfig.gpt2_tiny()
+
+    def mk():
+        torch.manual_seed(0)
+        model = build_model(cfg.model).to(DEV)
+        fp = FlatParams(model)
+        data = synthetic_batches(cfg.model.vocab_size, 4, 32, seed=1)
+        return DeltaLoop(model, fp, data, cfg.train)
+
+    g = torch.Generator().manual_seed(7)
+    batches = []
+    for i in range(4):
+        ids = torch.randint(0, cfg.model.vocab_size, (4, 32), generator=g)
+        lens = torch.randint(5, 33, (4,), generator=g)
+        am = (torch.arange(32)[None, :] < lens[:, None]).long()
+        batches.append({"input_ids": ids.to(DEV), "labels": ids.to(DEV),
+                        "attention_mask": am.to(DEV)})
+
+    graphed = mk()
+    eager = mk()
+    for _ in range(3):
+        eager.train_step(batches[0])
+    gs = GraphedMinerStep(graphed, batches, warmup=3)
+    assert gs.mask is not None   # the mask IS staged
+    for i in range(5):
+        gs.step(batches[i % 4])
+        eager.train_step(batches[i % 4])
+    torch.cuda.synchronize()
+    torch.testing.assert_close(graphed.fp.master, eager.fp.master,
+                               rtol=1e-4, atol=1e-4)
