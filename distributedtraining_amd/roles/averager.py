@@ -104,7 +104,11 @@ class ParameterizedAverager:
                 self.fp.zero_grad()
                 ids = batch["input_ids"].to(self.fp.device)
                 labels = batch.get("labels", batch["input_ids"]).to(self.fp.device)
-                out = self.model(input_ids=ids, labels=labels)
+                am = batch.get("attention_mask")
+                if am is not None:
+                    am = am.to(self.fp.device)
+                out = self.model(input_ids=ids, attention_mask=am,
+                                 labels=labels)
                 if not bool(torch.isfinite(out.loss)):
                     # a bad delta can make some merge diverge mid-search;
                     # skip the update rather than poisoning W (the
@@ -162,7 +166,12 @@ class ParameterizedAverager:
                 for batch in val_batches:
                     ids = batch["input_ids"].to(self.fp.device)
                     labels = batch.get("labels", batch["input_ids"]).to(self.fp.device)
-                    total += float(self.model(input_ids=ids, labels=labels).loss) * ids.shape[0]
+                    am = batch.get("attention_mask")
+                    if am is not None:
+                        am = am.to(self.fp.device)
+                    total += float(self.model(input_ids=ids,
+                                              attention_mask=am,
+                                              labels=labels).loss) * ids.shape[0]
                     cnt += ids.shape[0]
             return -total / max(cnt, 1)  # reference fitness = −val loss (:895-910)
 

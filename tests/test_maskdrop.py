@@ -278,3 +278,37 @@ def test_llama3_8b_hbm_budget():
     # and 8 fp32 deltas would NOT fit a gather — the bf16 wire option or
     # the all-reduce path is mandatory at 8 ranks (comm.py docstring)
     assert 8 * params * 4 / GB > 288 - (work_grad + master_opt)
+
+
+def test_averager_meta_learning_ignores_padding():
+    """The meta-learned merge's val evaluation must be invariant to PAD
+    token content (attention_mask honored in the averager too)."""
+    from distributedtraining_amd.config import AverageConfig
+    from distributedtraining_amd.parallel.flat import FlatParams
+    from distributedtraining_amd.roles.averager import ParameterizedAverager
+    cfg = ModelConfig.gpt2_tiny()
+    torch.manual_seed(2)
+    model = GPT2LM(cfg)
+    fp = FlatParams(model)
+    base = fp.snapshot()
+    deltas = torch.randn(2, fp.numel) * 1e-3
+    S = 16
+    ids = torch.randint(0, cfg.vocab_size, (2, S))
+    am = torch.ones(2, S, dtype=torch.long)
+    am[:, 10:] = 0
+    batch = {"input_ids": ids, "labels": ids.clone(), "attention_mask": am}
+    av = ParameterizedAverager(model, fp,
+                               AverageConfig(meta_epochs=2, meta_lr=0.01))
+    m1 = av.meta_learning(base, deltas.clone(), [batch]).clone()
+    w1 = av.weights.clone()
+    # mutate ONLY the padded positions
+    ids2 = ids.clone()
+    ids2[:, 10:] = (ids2[:, 10:] + 3) % cfg.vocab_size
+    batch2 = {"input_ids": ids2, "labels": ids2.clone(),
+              "attention_mask": am}
+    av2 = ParameterizedAverager(model, fp,
+                                AverageConfig(meta_epochs=2, meta_lr=0.01))
+    fp.load_flat_master(base)
+    m2 = av2.meta_learning(base, deltas.clone(), [batch2])
+    torch.testing.assert_close(av2.weights, w1, rtol=0, atol=0)
+    torch.testing.assert_close(m2, m1, rtol=0, atol=0)
