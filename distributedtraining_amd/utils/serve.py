@@ -70,8 +70,8 @@ class InferenceServer:
     def handle(self, req: dict) -> dict:
         ids = torch.tensor(req["ids"], dtype=torch.long,
                            device=self.device)
-        if ids.dim() != 2:
-            raise ValueError("ids must be [batch, seq]")
+        if ids.dim() != 2 or ids.numel() == 0:
+            raise ValueError("ids must be a non-empty [batch, seq]")
         vocab = self.model.cfg.vocab_size
         if int(ids.min()) < 0 or int(ids.max()) >= vocab:
             raise ValueError("token id out of range")

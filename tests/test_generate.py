@@ -122,15 +122,16 @@ def test_inference_server_roundtrip():
         # matches direct generation (greedy, cached)
         ref = generate(model, torch.tensor(ids), 4, use_cache=True)
         assert out == ref.tolist()
-        # bad request -> 400, not a crash
-        req = urllib.request.Request(url + "/generate", data=b"{}",
-                                     headers={"Content-Type":
-                                              "application/json"})
-        try:
-            urllib.request.urlopen(req, timeout=10)
-            assert False, "expected 400"
-        except urllib.error.HTTPError as e:
-            assert e.code == 400
+        # bad requests -> 400, not a crash (missing ids; empty sequence)
+        for bad in (b"{}", b'{"ids": [[]]}'):
+            req = urllib.request.Request(url + "/generate", data=bad,
+                                         headers={"Content-Type":
+                                                  "application/json"})
+            try:
+                urllib.request.urlopen(req, timeout=10)
+                assert False, "expected 400"
+            except urllib.error.HTTPError as e:
+                assert e.code == 400
         with urllib.request.urlopen(url + "/health", timeout=10) as r:
             h = json.loads(r.read())
         assert h["status"] == "ok" and h["served"] == 1
