@@ -805,6 +805,56 @@ def test_attention_dropout_fwd_bwd(B, H, Hkv, S, D, p):
                                rtol=6e-2, atol=6e-2)
 
 
+@pytest.mark.parametrize("B,H,Hkv,S,D,p", [(3, 2, 2, 64, 64, 0.1),
+                                           (2, 4, 2, 96, 128, 0.25)])
+def test_attention_masked_dropout_fwd_bwd(B, H, Hkv, S, D, p):
+    """kvlen padding mask AND attention-prob dropout TOGETHER (the actual
+    training configuration on ragged real-text batches): the dropout draws
+    are index-based, so the mask must not shift them."""
+    from distributedtraining_amd import ops
+    from distributedtraining_amd.ops import droprng
+    scale = 1.0 / math.sqrt(D)
+    torch.manual_seed(50)
+    droprng.counter(DEV).fill_(99)
+    site = 11
+    kvlen = torch.randint(1, S + 1, (B,), dtype=torch.int32)
+    kvlen[0] = S                      # one unpadded row
+    kvlen[-1] = 1                     # one maximally-padded row
+    q = _rand_bf16(B, S, H * D, seed=51).view(B, S, H, D) \
+        .transpose(1, 2).requires_grad_(True)
+    k = _rand_bf16(B, S, Hkv * D, seed=52).view(B, S, Hkv, D) \
+        .transpose(1, 2).requires_grad_(True)
+    v = _rand_bf16(B, S, Hkv * D, seed=53).view(B, S, Hkv, D) \
+        .transpose(1, 2).requires_grad_(True)
+    o = ops.causal_attention(q, k, v, scale, kvlen=kvlen.to(DEV),
+                             p_drop=p, site=site)
+    ref = _attn_ref_cpu(q.detach().cpu(), k.detach().cpu(),
+                        v.detach().cpu(), scale, kvlen, p_drop=p,
+                        site=site, ctr=99)
+    torch.testing.assert_close(o.float().cpu(), ref, rtol=4e-2, atol=4e-2)
+
+    do = _rand_bf16(B, H, S, D, seed=54)
+    qmask = (torch.arange(S)[None, :] < kvlen[:, None].long()) \
+        .to(DEV).view(B, 1, S, 1)
+    do = (do * qmask).to(torch.bfloat16)
+    o.backward(do)
+    qr = q.detach().cpu().float().requires_grad_(True)
+    kr = k.detach().cpu().float().requires_grad_(True)
+    vr = v.detach().cpu().float().requires_grad_(True)
+    _attn_ref_cpu(qr, kr, vr, scale, kvlen, p_drop=p, site=site,
+                  ctr=99).backward(do.float().cpu())
+    torch.testing.assert_close(q.grad.float().cpu(), qr.grad,
+                               rtol=6e-2, atol=6e-2)
+    torch.testing.assert_close(k.grad.float().cpu(), kr.grad,
+                               rtol=6e-2, atol=8e-2)
+    torch.testing.assert_close(v.grad.float().cpu(), vr.grad,
+                               rtol=6e-2, atol=6e-2)
+    for b in range(B):
+        L = int(kvlen[b])
+        assert torch.all(k.grad[b, :, L:] == 0)
+        assert torch.all(v.grad[b, :, L:] == 0)
+
+
 def test_dropout_kernel_matches_host_gold():
     from distributedtraining_amd import ops
     from distributedtraining_amd.ops import droprng
