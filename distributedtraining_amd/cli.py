@@ -132,8 +132,12 @@ def main(argv=None) -> int:
             store.push_model({"format": "dta-base-v1",
                               "flat_master": fp.master.cpu(),
                               "spec": fp.spec})
-        data = synthetic_batches(cfg.model.vocab_size, cfg.train.batch_size,
-                                 cfg.train.seq_len, seed=cfg.seed + hash(hotkey) % 1000)
+        # stable per-hotkey seed (builtin hash() is salted per process and
+        # would make a miner's data stream irreproducible across runs)
+        import zlib
+        data = synthetic_batches(
+            cfg.model.vocab_size, cfg.train.batch_size, cfg.train.seq_len,
+            seed=cfg.seed + zlib.crc32(hotkey.encode()) % 1000)
         if ns.gradient_mode:
             from .roles.gradient_loop import GradientLoop
             miner = GradientLoop(model, fp, data, cfg.train, store=store,
