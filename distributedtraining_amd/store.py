@@ -37,9 +37,10 @@ DELTA_FILE = "weight_diff.pt"        # reference: training_manager.py:413-423
 def tensor_sha256(t: torch.Tensor) -> str:
     """SHA-256 over a tensor's raw bytes (reference hashes params the same
     way: calculate_model_hash, training_manager.py:198-203)."""
-    t = t.detach().contiguous().cpu()
-    return hashlib.sha256(t.view(torch.uint8).numpy().tobytes()).hexdigest() \
-        if t.dtype == torch.uint8 else hashlib.sha256(t.numpy().tobytes()).hexdigest()
+    t = t.detach().contiguous().cpu().reshape(-1)
+    # byte reinterpretation works for every dtype incl. bf16 (which has no
+    # numpy equivalent); identical bytes to numpy().tobytes() for fp32
+    return hashlib.sha256(t.view(torch.uint8).numpy().tobytes()).hexdigest()
 
 
 def spec_of(named_params) -> List[Tuple[str, tuple, int]]:
