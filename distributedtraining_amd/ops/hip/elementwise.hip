@@ -1,4 +1,9 @@
-// Elementwise kernels: GELU (tanh approx), SwiGLU, flat-plane delta ops.
+// Elementwise kernels: GELU (tanh approx), SwiGLU, flat-plane delta ops,
+// counter-based dropout. Covers the implicit elementwise surface of the
+// reference's transformers GPT-2 step (GELU/dropout inside
+// `self.model(...)`, training_manager.py:380-385) and its flat-plane
+// tensor loops (delta = theta - base, training_manager.py:417-421;
+// apply theta += delta, validation_logic.py:251-259).
 // All memory-bound: vectorized bf16x8 / float4 loads (Guideline 13),
 // grid-stride with capped grid (Guideline 11).
 #include "dta_common.h"

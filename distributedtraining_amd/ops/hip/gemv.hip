@@ -1,4 +1,6 @@
 // Weight-streaming GEMV / skinny GEMM for serving decode (M <= 4).
+// (Serving is beyond-parity: the reference trains only; this backs the
+// deployed-model inference endpoint, utils/serve.py.)
 //
 // y[M,N] = x[M,K] @ W[N,K]^T (+ bias) — the decode-step projection
 // shapes (M = batch). hipBLASLt's small-M path measured only ~1.7 TB/s

@@ -1,4 +1,6 @@
-// Rotary position embedding (Llama), fwd and bwd (bwd = inverse rotation).
+// Rotary position embedding (Llama family — beyond the reference's GPT-2
+// learned positions, neurons/miner.py:60-62), fwd and bwd (bwd = inverse
+// rotation).
 // Host-precomputed cos/sin tables [S, D/2] fp32 (Appendix B: never call
 // trig per element on-device — it turns a memory-bound op VALU-bound).
 // Layout: x [BH, S, D] with the half-split convention
