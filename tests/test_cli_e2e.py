@@ -71,6 +71,45 @@ def test_three_role_pipeline(tmp_path):
 
 
 @pytest.mark.timeout(600)
+def test_serve_role_smoke(tmp_path):
+    """`cli serve` over a store published by a miner: the server loads the
+    current base and answers /generate."""
+    import urllib.request
+
+    root = str(tmp_path / "sx")
+    common = ["--tiny", "--comm.root", root,
+              "--metrics-dir", str(tmp_path / "metrics"),
+              "--train.batch-size", "2", "--train.seq-len", "16"]
+    _run(["miner", "--hotkey", "m0", "--steps", "2", *common])
+
+    proc = subprocess.Popen(
+        [sys.executable, "-u", "-m", "distributedtraining_amd.cli", "serve",
+         "--tiny", "--comm.root", root, "--port", "0"],
+        cwd=REPO, stdout=subprocess.PIPE, stderr=subprocess.STDOUT,
+        text=True)
+    try:
+        port = None
+        for _ in range(60):
+            line = proc.stdout.readline()
+            if "inference server on :" in line:
+                port = int(line.rsplit(":", 1)[1].strip().split()[0])
+                break
+            assert proc.poll() is None, "serve process exited early"
+        assert port, "server never reported its port"
+        req = urllib.request.Request(
+            f"http://127.0.0.1:{port}/generate",
+            data=json.dumps({"ids": [[1, 2, 3]],
+                             "max_new_tokens": 3}).encode(),
+            headers={"Content-Type": "application/json"})
+        with urllib.request.urlopen(req, timeout=60) as r:
+            out = json.loads(r.read())
+        assert len(out["ids"]) == 1 and len(out["ids"][0]) == 6
+    finally:
+        proc.terminate()
+        proc.wait(timeout=30)
+
+
+@pytest.mark.timeout(600)
 def test_gradient_mode_pipeline(tmp_path):
     """Legacy gradient-publication protocol over the CLI (reference
     TrainingLoop gradients.pt + Averager alpha-apply)."""
