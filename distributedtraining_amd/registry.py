@@ -19,11 +19,13 @@ to hotkeys "rank{r}".
 
 from __future__ import annotations
 
+import fcntl
 import json
 import os
 import tempfile
 import threading
 import time
+from contextlib import contextmanager
 from typing import Dict, List, Optional
 
 import numpy as np
@@ -158,12 +160,15 @@ class Registry:
     def set_weights(self, scores: Dict[str, float]) -> Dict[str, float]:
         """EMA-fold new scores into the registry, alpha=0.333 (T=5)."""
         with self._lock:
-            a = self.ema_alpha
-            for hk in set(list(self._scores) + list(scores)):
-                prev = self._scores.get(hk, 0.0)
-                self._scores[hk] = a * scores.get(hk, 0.0) + (1 - a) * prev
-            self._last_weight_set = self.current_block()
-            return dict(self._scores)
+            return self._fold_weights_unlocked(scores)
+
+    def _fold_weights_unlocked(self, scores: Dict[str, float]) -> Dict[str, float]:
+        a = self.ema_alpha
+        for hk in set(list(self._scores) + list(scores)):
+            prev = self._scores.get(hk, 0.0)
+            self._scores[hk] = a * scores.get(hk, 0.0) + (1 - a) * prev
+        self._last_weight_set = self.current_block()
+        return dict(self._scores)
 
     def get_weights(self) -> Dict[str, float]:
         with self._lock:
@@ -184,7 +189,11 @@ class FileRegistry(Registry):
 
     Mirrors the reference's LocalAddressStore (chain_manager.py:124-168) and
     LocalBittensorNetwork JSON metagraph (btt_connector.py:558-571), unified
-    behind one interface.
+    behind one interface. Every mutation is a load-modify-save under an
+    fcntl file lock — without it, two processes registering concurrently
+    could each load, add their key and save, with the last save silently
+    dropping the other's registration (caught by the concurrent-writer
+    stress test).
     """
 
     def __init__(self, root: str, epoch_length: int = 100,
@@ -192,7 +201,17 @@ class FileRegistry(Registry):
         super().__init__(epoch_length, ema_alpha)
         os.makedirs(root, exist_ok=True)
         self.path = os.path.join(root, "registry.json")
+        self.lock_path = os.path.join(root, "registry.lock")
         self._load()
+
+    @contextmanager
+    def _file_lock(self):
+        with open(self.lock_path, "w") as lf:
+            fcntl.flock(lf, fcntl.LOCK_EX)
+            try:
+                yield
+            finally:
+                fcntl.flock(lf, fcntl.LOCK_UN)
 
     def _load(self) -> None:
         if os.path.exists(self.path):
@@ -216,7 +235,7 @@ class FileRegistry(Registry):
         os.replace(tmp, self.path)  # atomic on POSIX
 
     def store_address(self, hotkey: str, address: str) -> None:
-        with self._lock:
+        with self._lock, self._file_lock():
             self._load_unlocked()
             self._addresses[hotkey] = address
             self._save()
@@ -251,22 +270,22 @@ class FileRegistry(Registry):
             return list(self._addresses.keys())
 
     def set_weights(self, scores: Dict[str, float]) -> Dict[str, float]:
-        with self._lock:          # fold the EMA into the LATEST persisted
-            self._load_unlocked()  # scores, not this process's stale copy
-        out = super().set_weights(scores)
-        with self._lock:
+        with self._lock, self._file_lock():
+            self._load_unlocked()  # fold the EMA into the LATEST persisted
+            out = self._fold_weights_unlocked(scores)
             self._save()
         return out
 
     def set_stake(self, hotkey: str, stake: float) -> None:
-        super().set_stake(hotkey, stake)
-        with self._lock:
+        with self._lock, self._file_lock():
+            self._load_unlocked()
+            self._stakes[hotkey] = float(stake)
             self._save()
 
     def deregister(self, hotkey: str) -> None:
         # must persist the removal BEFORE any reload: _load_unlocked merges
         # the file back in, which would resurrect the hotkey otherwise
-        with self._lock:
+        with self._lock, self._file_lock():
             self._load_unlocked()
             self._addresses.pop(hotkey, None)
             self._scores.pop(hotkey, None)
