@@ -232,3 +232,66 @@ def test_periodic_wrappers(tmp_path):
                                store=store, registry=registry)
     av.run_periodic_averaging(ev, interval_s=0.0, max_rounds=2)
     assert store.pull_model() is not None
+
+
+def test_concurrent_roles_soak(tmp_path):
+    """All three roles racing over one shared store+registry (threads —
+    the deployment shape is independent processes, same file contracts):
+    miners push deltas and pull bases mid-train while the validator scores
+    and the averager merges+publishes. No exceptions may escape and the
+    final published base must be complete and finite."""
+    import threading
+
+    from distributedtraining_amd.config import ValidateConfig
+
+    errors = []
+
+    def guard(fn):
+        def run():
+            try:
+                fn()
+            except Exception as e:  # pragma: no cover - failure path
+                errors.append(e)
+        return run
+
+    def miner_role(hotkey, seed):
+        cfg, model, fp, store, registry = _mk(tmp_path, hotkey)
+        cfg.train.pull_interval_steps = 3   # pull new bases mid-train
+        data = synthetic_batches(cfg.model.vocab_size, 2, 16, seed=seed)
+        miner = DeltaLoop(model, fp, data, cfg.train, store=store,
+                          registry=registry, hotkey=hotkey)
+        miner.train(24)
+
+    def validator_role():
+        cfg, model, fp, store, registry = _mk(tmp_path, "val")
+        ev = synthetic_eval_set(cfg.model.vocab_size, 1, 2, 16)
+        v = DeltaValidator(model, fp, ev, ValidateConfig(), store=store,
+                           registry=registry)
+        for _ in range(3):
+            v.validate_and_score()
+
+    def averager_role():
+        cfg, model, fp, store, registry = _mk(tmp_path, "avg")
+        ev = synthetic_eval_set(cfg.model.vocab_size, 1, 2, 16)
+        a = ParameterizedAverager(model, fp, AverageConfig(strategy="mean"),
+                                  store=store, registry=registry)
+        for _ in range(3):
+            a.run_round(ev)
+
+    # publish an initial base first (the first miner does this in cli.py)
+    cfg, model, fp, store, _ = _mk(tmp_path, "seed")
+    store.push_model({"format": "dta-base-v1", "flat_master": fp.master.cpu(),
+                      "spec": fp.spec})
+
+    threads = [threading.Thread(target=guard(lambda: miner_role("m0", 1))),
+               threading.Thread(target=guard(lambda: miner_role("m1", 2))),
+               threading.Thread(target=guard(validator_role)),
+               threading.Thread(target=guard(averager_role))]
+    for t in threads:
+        t.start()
+    for t in threads:
+        t.join(timeout=240)
+    assert not errors, errors
+    final = FileStore(str(tmp_path)).pull_model()
+    assert final is not None and "flat_master" in final
+    assert bool(torch.isfinite(final["flat_master"]).all())
