@@ -75,6 +75,11 @@ class GraphedMinerStep:
         lbl = batch.get("labels", batch["input_ids"])
         if self.labels is not self.ids:
             self.labels.copy_(lbl, non_blocking=True)
+        elif lbl is not batch["input_ids"]:
+            # captured with labels aliasing input_ids: a batch with distinct
+            # labels would be silently mis-trained through the alias
+            raise ValueError("graph captured with labels == input_ids; "
+                             "re-capture to feed distinct labels")
         if self.mask is not None:
             self.mask.copy_(batch["attention_mask"], non_blocking=True)
         self.graph.replay()
