@@ -3,6 +3,7 @@ ALL inputs, not just the fixtures: merge algebra, EMA folding, checkpoint
 round-trips, rate-limiter invariants."""
 
 import hypothesis.strategies as st
+import pytest
 import torch
 from hypothesis import given, settings
 
@@ -96,3 +97,58 @@ def test_grad_merge_weights_matches_dense_math(n, p):
             ref = float((g[lo:hi] * (base[lo:hi] + deltas[i, lo:hi]
                                      - merged[lo:hi])).sum())
             assert abs(float(gw[i, j]) - ref) < 1e-3 + 1e-3 * abs(ref)
+
+
+# ---------------------------------------------------------------------------
+# Counter-RNG dropout statistics (host mirror == kernel draws, so these
+# properties hold for the GPU masks bit-for-bit)
+# ---------------------------------------------------------------------------
+@pytest.mark.parametrize("p", [0.1, 0.25, 0.5])
+def test_droprng_keep_rate_concentration(p):
+    """Empirical keep rate within 5 sigma of the quantized target."""
+    import numpy as np
+    from distributedtraining_amd.ops import droprng
+    n = 200_000
+    keep = droprng.elem_keep_mask(n, ctr=3, site=5, p=p)
+    target = 1.0 - droprng.thr16(p) / 65536.0
+    sigma = (target * (1 - target) / n) ** 0.5
+    assert abs(float(np.mean(keep)) - target) < 5 * sigma
+
+
+def test_droprng_site_and_counter_decorrelation():
+    """Masks drawn at different sites (and different counter values) agree
+    at ~the independence rate (1-p)^2 + p^2, never near 1."""
+    import numpy as np
+    from distributedtraining_amd.ops import droprng
+    n, p = 200_000, 0.25
+    a = droprng.elem_keep_mask(n, ctr=3, site=5, p=p)
+    for other in (droprng.elem_keep_mask(n, ctr=3, site=6, p=p),
+                  droprng.elem_keep_mask(n, ctr=4, site=5, p=p)):
+        agree = float(np.mean(a == other))
+        q = 1.0 - droprng.thr16(p) / 65536.0
+        expect = q * q + (1 - q) * (1 - q)
+        sigma = (expect * (1 - expect) / n) ** 0.5
+        assert abs(agree - expect) < 6 * sigma
+    # identical inputs reproduce exactly
+    assert np.array_equal(a, droprng.elem_keep_mask(n, ctr=3, site=5, p=p))
+
+
+def test_droprng_attn_adjacent_key_independence():
+    """Adjacent attention keys share one hash word (4x16-bit slices): the
+    slices must still be pairwise independent in aggregate."""
+    import numpy as np
+    from distributedtraining_amd.ops import droprng
+    bh, S, p = 4, 256, 0.25
+    keep = droprng.attn_keep_mask(bh, S, S, ctr=9, site=2, p=p)  # [bh,S,S]
+    k = keep.reshape(bh * S, S)
+    a, b = k[:, :-1].ravel(), k[:, 1:].ravel()   # adjacent-key pairs
+    agree = float(np.mean(a == b))
+    q = 1.0 - droprng.thr16(p) / 65536.0
+    expect = q * q + (1 - q) * (1 - q)
+    n = a.size
+    sigma = (expect * (1 - expect) / n) ** 0.5
+    assert abs(agree - expect) < 6 * sigma
+    # per-row keep rate has no row-index drift: first vs second half
+    row_rate = k.mean(axis=1)
+    assert abs(float(row_rate[:bh * S // 2].mean())
+               - float(row_rate[bh * S // 2:].mean())) < 0.01
