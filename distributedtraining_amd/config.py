@@ -140,7 +140,9 @@ class CommConfig:
     # wire dtype for the gather-based exchanges (score_weighted /
     # parameterized strategies need all deltas resident): "fp32" is exact;
     # "bf16" halves xGMI bytes + gather footprint (needed for Llama-scale
-    # models at 8 ranks). mean/nesterov all-reduce stays fp32 always.
+    # models at 8 ranks); "int8" quarters them via blockwise-absmax
+    # quantization (worst-case element error absmax(4096-block)/127).
+    # mean/nesterov all-reduce stays fp32 always.
     exchange_dtype: str = "fp32"
 
 

@@ -59,6 +59,14 @@ class LocalSGDNode:
         return (torch.bfloat16 if self.cfg.comm.exchange_dtype == "bf16"
                 else None)
 
+    def _gather_deltas(self, flat: torch.Tensor) -> torch.Tensor:
+        """All-gather the flat delta at the configured wire dtype; always
+        returns [world, P] fp32 for the merge math."""
+        if self.cfg.comm.exchange_dtype == "int8":
+            return self.comm.all_gather_flat_int8(flat)
+        return self.comm.all_gather_flat(flat, self._wire_dtype()) \
+            .to(torch.float32)
+
     def sync_initial_base(self) -> None:
         """Rank 0's random init becomes the shared base on all ranks."""
         self.comm.broadcast_flat(self.fp.master, src=0)
@@ -90,12 +98,10 @@ class LocalSGDNode:
                 merged = d.add_(base)        # in place: d becomes merged
         elif strat == "score_weighted":
             assert scores is not None
-            deltas = self.comm.all_gather_flat(delta.flat, self._wire_dtype())
             merged = self.averager.score_weighted_merge(
-                base, deltas.to(torch.float32), scores)
+                base, self._gather_deltas(delta.flat), scores)
         elif strat == "parameterized":
-            deltas = self.comm.all_gather_flat(delta.flat, self._wire_dtype())
-            deltas = deltas.to(torch.float32)
+            deltas = self._gather_deltas(delta.flat)
             if self.comm.rank == 0:
                 merged = self.averager.meta_learning(base, deltas,
                                                      self.val_batches)

@@ -271,3 +271,73 @@ def test_eight_rank_local_sgd_allreduce(tmp_path):
     for d in digests[1:]:
         assert digests[0] == pytest.approx(d, rel=1e-6), \
             "ranks diverged after merge"
+
+
+def _int8_wire_worker(rank, world, port, q):
+    os.environ.update({
+        "RANK": str(rank), "WORLD_SIZE": str(world),
+        "LOCAL_RANK": str(rank), "MASTER_ADDR": "127.0.0.1",
+        "MASTER_PORT": str(port),
+    })
+    import torch
+    from distributedtraining_amd.config import (Config, ModelConfig,
+                                                TrainConfig)
+    from distributedtraining_amd.models import build_model
+    from distributedtraining_amd.parallel.comm import (
+        CommPlane, dequantize_blockwise_int8, quantize_blockwise_int8)
+    from distributedtraining_amd.parallel.flat import FlatParams
+    from distributedtraining_amd.parallel.local_sgd import LocalSGDNode
+    from distributedtraining_amd.utils.data import synthetic_batches
+    try:
+        cfg = Config()
+        cfg.model = ModelConfig.gpt2_tiny()
+        cfg.train = TrainConfig(batch_size=2, seq_len=16,
+                                send_interval_steps=10**9,
+                                pull_interval_steps=0)
+        cfg.comm.exchange_dtype = "int8"   # quartered-wire gather
+        torch.manual_seed(100 + rank)
+        model = build_model(cfg.model)
+        fp = FlatParams(model)
+        comm = CommPlane(backend="gloo", device=torch.device("cpu"))
+        data = synthetic_batches(cfg.model.vocab_size, 2, 16, seed=rank)
+        node = LocalSGDNode(model, fp, data, cfg, comm,
+                            merge_strategy="score_weighted")
+        node.sync_initial_base()
+        base0 = fp.master.clone()
+        node.train_steps(2)
+        delta = fp.make_delta(node.miner.base)
+        # exact expectation: mean of the int8-roundtripped deltas
+        qc, sc = quantize_blockwise_int8(delta.flat)
+        rt = dequantize_blockwise_int8(qc, sc, delta.flat.numel())
+        gathered = comm.all_gather_flat(rt)
+        node.merge_round(scores=[1.0] * world)
+        expected = base0 + gathered.mean(dim=0)
+        ok = torch.allclose(fp.master, expected, rtol=1e-5, atol=1e-6)
+        # and the quantization stayed close to the true deltas
+        true = comm.all_gather_flat(delta.flat)
+        ok_err = float((gathered - true).abs().max()) < 1e-3
+        q.put((rank, bool(ok and ok_err), float(fp.master.sum())))
+        comm.close()
+    except Exception as e:  # pragma: no cover
+        q.put((rank, False, repr(e)))
+        raise
+
+
+@pytest.mark.timeout(300)
+def test_int8_wire_gather_merge():
+    """score_weighted merge over the int8-quantized wire (4x fewer bytes
+    than fp32): world 2, bit-identical merged base on every rank, and the
+    dequantized deltas within the blockwise error bound of the originals."""
+    world = 2
+    port = 29861
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    procs = [ctx.Process(target=_int8_wire_worker, args=(r, world, port, q))
+             for r in range(world)]
+    for p in procs:
+        p.start()
+    results = [q.get(timeout=240) for _ in range(world)]
+    for p in procs:
+        p.join(timeout=60)
+    assert all(ok for _, ok, _ in results), results
+    assert results[0][2] == pytest.approx(results[1][2], rel=1e-6)

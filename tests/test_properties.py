@@ -152,3 +152,25 @@ def test_droprng_attn_adjacent_key_independence():
     row_rate = k.mean(axis=1)
     assert abs(float(row_rate[:bh * S // 2].mean())
                - float(row_rate[bh * S // 2:].mean())) < 0.01
+
+
+@settings(max_examples=20, deadline=None)
+@given(st.integers(10, 9000), st.integers(1, 10**6))
+def test_int8_delta_quantization_error_bound(n, seed):
+    """Blockwise-int8 roundtrip: per-element error <= absmax(block)/127
+    (+ float eps), exact zeros preserved, shape preserved."""
+    from distributedtraining_amd.parallel.comm import (
+        dequantize_blockwise_int8, quantize_blockwise_int8)
+    g = torch.Generator().manual_seed(seed)
+    x = torch.randn(n, generator=g) * 1e-2     # delta-like magnitudes
+    x[::7] = 0.0
+    block = 256
+    q, s = quantize_blockwise_int8(x, block)
+    y = dequantize_blockwise_int8(q, s, n, block)
+    assert y.shape == x.shape
+    nb = (n + block - 1) // block
+    xp = torch.nn.functional.pad(x, (0, nb * block - n)).view(nb, block)
+    bound = (xp.abs().amax(dim=1) / 127.0 + 1e-7).unsqueeze(1) \
+        .expand(nb, block).reshape(-1)[:n]
+    assert bool((y - x).abs().le(bound * 1.001).all())
+    assert bool((y[::7] == 0).all())
