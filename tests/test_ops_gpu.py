@@ -1144,3 +1144,26 @@ def test_graphed_step_matches_eager_with_padded_batches():
     torch.cuda.synchronize()
     torch.testing.assert_close(graphed.fp.master, eager.fp.master,
                                rtol=1e-4, atol=1e-4)
+
+
+def test_int8_delta_wire_roundtrip_gpu():
+    """Blockwise-int8 delta compression on device (the opt-in wire mode):
+    error bound holds and the world-1 gather path returns fp32 [1, P]."""
+    from distributedtraining_amd.parallel.comm import (
+        CommPlane, dequantize_blockwise_int8, quantize_blockwise_int8)
+    torch.manual_seed(3)
+    x = (torch.randn(100_003, device=DEV) * 1e-2).float()
+    q, s = quantize_blockwise_int8(x)
+    assert q.dtype == torch.int8 and q.is_cuda and s.is_cuda
+    y = dequantize_blockwise_int8(q, s, x.numel())
+    block = 4096
+    nb = (x.numel() + block - 1) // block
+    xp = torch.nn.functional.pad(x, (0, nb * block - x.numel())) \
+        .view(nb, block)
+    bound = (xp.abs().amax(dim=1) / 127.0 + 1e-7).unsqueeze(1) \
+        .expand(nb, block).reshape(-1)[:x.numel()]
+    assert bool((y - x).abs().le(bound * 1.001).all())
+    comm = CommPlane(device=DEV)        # world 1 here
+    g = comm.all_gather_flat_int8(x)
+    assert g.shape == (1, x.numel()) and g.dtype == torch.float32
+    assert bool((g[0] - x).abs().le(bound * 1.001).all())
