@@ -57,7 +57,7 @@ def main(argv=None) -> int:
     logging.basicConfig(level=logging.INFO)
     top = argparse.ArgumentParser("distributedtraining_amd")
     top.add_argument("role", choices=["miner", "validator", "averager",
-                                      "bootstrap", "serve"])
+                                      "bootstrap", "serve", "convert"])
     top.add_argument("--hotkey", default=None)
     top.add_argument("--steps", type=int, default=100)
     top.add_argument("--rounds", type=int, default=1)
@@ -74,11 +74,33 @@ def main(argv=None) -> int:
     top.add_argument("--metrics-dir", default="metrics")
     top.add_argument("--port", type=int, default=8500,
                      help="bootstrap: HTTP port")
+    top.add_argument("--src", default=None,
+                     help="convert: HF checkpoint directory "
+                          "(save_pretrained layout: config.json + "
+                          "model.safetensors / pytorch_model.bin)")
     ns, rest = top.parse_known_args(argv)
     cfg = cfg_mod.from_args(rest)
     if ns.tiny:
         cfg.model = ModelConfig.gpt2_tiny()
     hotkey = ns.hotkey or ns.role
+
+    if ns.role == "convert":
+        # migrate a reference-style HF checkpoint into the exchange store
+        # as the shared base model (the reference's from_pretrained step,
+        # neurons/miner.py:60-62, done once offline)
+        assert ns.src, "convert needs --src <hf_checkpoint_dir>"
+        from .models.convert import hf_dir_to_native
+        from .parallel.flat import FlatParams as _FP
+        from .store import FileStore as _FS
+        model, mcfg = hf_dir_to_native(ns.src)
+        fp = _FP(model)
+        _FS(cfg.comm.root, hotkey="convert").push_model(
+            {"format": "dta-base-v1", "flat_master": fp.master.cpu(),
+             "spec": fp.spec, "meta": {"source": ns.src,
+                                       "family": mcfg.family}})
+        print(f"converted {mcfg.family} checkpoint ({fp.numel} params) "
+              f"-> {cfg.comm.root}/model/averaged_model.pt")
+        return 0
 
     if ns.role == "serve":
         import time as _time

@@ -102,6 +102,47 @@ def export_gpt2_to_hf_state_dict(model: GPT2LM) -> Dict[str, torch.Tensor]:
     return sd
 
 
+def hf_dir_to_native(path: str):
+    """Load an HF ``save_pretrained`` checkpoint DIRECTORY (config.json +
+    model.safetensors / pytorch_model.bin) into a native model — no
+    transformers import needed, fully offline. Returns (model, cfg).
+    This is the migration entry for reference users (`cli.py convert`)."""
+    import json
+    import os
+
+    with open(os.path.join(path, "config.json")) as f:
+        hc = json.load(f)
+
+    class _NS:
+        def __init__(self, d):
+            self.__dict__.update(d)
+
+        def __getattr__(self, k):   # missing key -> AttributeError
+            raise AttributeError(k)
+
+    hf_cfg = _NS(hc)
+    sd = None
+    st_path = os.path.join(path, "model.safetensors")
+    pt_path = os.path.join(path, "pytorch_model.bin")
+    if os.path.exists(st_path):
+        from safetensors.torch import load_file
+        sd = load_file(st_path)
+    elif os.path.exists(pt_path):
+        sd = torch.load(pt_path, map_location="cpu", weights_only=True)
+    else:
+        raise FileNotFoundError(
+            f"no model.safetensors / pytorch_model.bin under {path}")
+
+    mt = hc.get("model_type", "")
+    if mt == "gpt2":
+        cfg = gpt2_config_from_hf(hf_cfg)
+        return load_gpt2_from_hf(sd, cfg), cfg
+    if mt in ("llama", "mistral"):
+        cfg = llama_config_from_hf(hf_cfg)
+        return load_llama_from_hf(sd, cfg), cfg
+    raise ValueError(f"unsupported model_type {mt!r} (gpt2/llama)")
+
+
 def _rope_theta(hf_config) -> float:
     # newer transformers moves rope_theta into per-layer rope_parameters
     # and raises on global attribute access
