@@ -114,3 +114,28 @@ def test_cli_convert_hf_dir(tmp_path):
         ref = hf(input_ids=ids).logits
         got = native(input_ids=ids).logits
     torch.testing.assert_close(got, ref, rtol=2e-4, atol=2e-4)
+
+
+def test_hf_dir_to_native_llama(tmp_path):
+    """hf_dir_to_native on a Llama save_pretrained directory: config
+    detection + weight load + logits parity."""
+    from transformers import LlamaConfig, LlamaForCausalLM
+
+    from distributedtraining_amd.models.convert import hf_dir_to_native
+    hf_cfg = LlamaConfig(vocab_size=256, hidden_size=64,
+                         intermediate_size=128, num_hidden_layers=2,
+                         num_attention_heads=4, num_key_value_heads=2,
+                         max_position_embeddings=64, rms_norm_eps=1e-5,
+                         tie_word_embeddings=False)
+    torch.manual_seed(1)
+    hf = LlamaForCausalLM(hf_cfg).eval()
+    src = tmp_path / "llama_ckpt"
+    hf.save_pretrained(src)
+    native, cfg = hf_dir_to_native(str(src))
+    native.eval()
+    assert cfg.family == "llama" and cfg.n_kv_head == 2
+    ids = torch.randint(0, 256, (2, 12))
+    with torch.no_grad():
+        ref = hf(input_ids=ids).logits
+        got = native(input_ids=ids).logits
+    torch.testing.assert_close(got, ref, rtol=5e-4, atol=5e-4)
