@@ -52,13 +52,21 @@ class GraphedDecoder:
         return self.logits
 
 
-def _sample(logits, temperature, top_k, generator):
+def _sample(logits, temperature, top_k, generator, top_p=0.0):
     if temperature <= 0:
         return logits.argmax(dim=-1)
     logits = logits / temperature
     if top_k > 0 and top_k < logits.shape[-1]:
         kth = torch.topk(logits, top_k, dim=-1).values[:, -1:]
         logits = logits.masked_fill(logits < kth, float("-inf"))
+    if 0.0 < top_p < 1.0:
+        # nucleus: keep the smallest prefix of sorted probs summing > p
+        sorted_logits, idx = torch.sort(logits, descending=True, dim=-1)
+        cum = torch.softmax(sorted_logits, dim=-1).cumsum(dim=-1)
+        cut = cum - torch.softmax(sorted_logits, dim=-1) >= top_p
+        sorted_logits = sorted_logits.masked_fill(cut, float("-inf"))
+        logits = torch.full_like(logits, float("-inf")) \
+            .scatter_(-1, idx, sorted_logits)
     probs = torch.softmax(logits, dim=-1)
     return torch.multinomial(probs, 1, generator=generator).squeeze(-1)
 
@@ -66,13 +74,16 @@ def _sample(logits, temperature, top_k, generator):
 @torch.no_grad()
 def generate(model, input_ids: torch.Tensor, max_new_tokens: int,
              temperature: float = 0.0, top_k: int = 0,
+             top_p: float = 0.0,
              eos_token_id: Optional[int] = None,
              generator: Optional[torch.Generator] = None,
              use_cache: bool = False) -> torch.Tensor:
     """Extend ``input_ids`` [B, S] by up to ``max_new_tokens``.
 
     temperature 0 = greedy; top_k > 0 restricts sampling to the k highest
-    logits. Stops early when every sequence has produced eos_token_id.
+    logits; 0 < top_p < 1 restricts to the smallest nucleus of tokens
+    whose probability mass exceeds top_p (applied after top_k).
+    Stops early when every sequence has produced eos_token_id.
     """
     model_was_training = model.training
     model.eval()
@@ -87,7 +98,8 @@ def generate(model, input_ids: torch.Tensor, max_new_tokens: int,
             logits = model.prefill(ids, cache).float()
             dec = None
             for _ in range(max_new_tokens):
-                nxt = _sample(logits, temperature, top_k, generator)
+                nxt = _sample(logits, temperature, top_k, generator,
+                              top_p=top_p)
                 if eos_token_id is not None:
                     nxt = torch.where(done,
                                       torch.full_like(nxt, eos_token_id),
@@ -110,7 +122,8 @@ def generate(model, input_ids: torch.Tensor, max_new_tokens: int,
         for _ in range(max_new_tokens):
             ctx = ids[:, -n_pos:]
             logits = model(input_ids=ctx).logits[:, -1].float()
-            nxt = _sample(logits, temperature, top_k, generator)
+            nxt = _sample(logits, temperature, top_k, generator,
+                          top_p=top_p)
             if eos_token_id is not None:
                 nxt = torch.where(done, torch.full_like(nxt, eos_token_id),
                                   nxt)

@@ -83,10 +83,11 @@ class InferenceServer:
                 f"the model context length {npos}")
         temperature = float(req.get("temperature", 0.0))
         top_k = int(req.get("top_k", 0))
+        top_p = float(req.get("top_p", 0.0))
         eos = req.get("eos_token_id")
         with self._lock, torch.no_grad():
             out = generate(self.model, ids, mnt, temperature=temperature,
-                           top_k=top_k,
+                           top_k=top_k, top_p=top_p,
                            eos_token_id=None if eos is None else int(eos),
                            use_cache=hasattr(self.model, "prefill"))
         self.served += 1
