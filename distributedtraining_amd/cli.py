@@ -57,7 +57,8 @@ def main(argv=None) -> int:
     logging.basicConfig(level=logging.INFO)
     top = argparse.ArgumentParser("distributedtraining_amd")
     top.add_argument("role", choices=["miner", "validator", "averager",
-                                      "bootstrap", "serve", "convert"])
+                                      "bootstrap", "serve", "convert",
+                                      "export"])
     top.add_argument("--hotkey", default=None)
     top.add_argument("--steps", type=int, default=100)
     top.add_argument("--rounds", type=int, default=1)
@@ -78,6 +79,9 @@ def main(argv=None) -> int:
                      help="convert: HF checkpoint directory "
                           "(save_pretrained layout: config.json + "
                           "model.safetensors / pytorch_model.bin)")
+    top.add_argument("--out", default=None,
+                     help="export: output directory for the HF-layout "
+                          "checkpoint of the store's current base model")
     ns, rest = top.parse_known_args(argv)
     cfg = cfg_mod.from_args(rest)
     if ns.tiny:
@@ -100,6 +104,28 @@ def main(argv=None) -> int:
                                        "family": mcfg.family}})
         print(f"converted {mcfg.family} checkpoint ({fp.numel} params) "
               f"-> {cfg.comm.root}/model/averaged_model.pt")
+        return 0
+
+    if ns.role == "export":
+        # store base -> HF save_pretrained directory (the reverse of
+        # convert: trained bases flow back to transformers users).
+        # Model architecture comes from the config flags (--tiny /
+        # --model.* overrides), like serve.
+        assert ns.out, "export needs --out <dir>"
+        from .models.convert import native_to_hf_dir
+        from .parallel.flat import FlatParams as _FP
+        from .store import FileStore as _FS
+        torch.manual_seed(cfg.seed)
+        model = build_model(cfg.model)
+        st = _FS(cfg.comm.root, hotkey="export").pull_model()
+        if st is not None and "flat_master" in st:
+            fp = _FP(model)
+            fp.load_flat_master(st["flat_master"])
+        else:
+            print("WARNING: no base model in the store; exporting "
+                  "random init", file=sys.stderr)
+        native_to_hf_dir(model, cfg.model, ns.out)
+        print(f"exported {cfg.model.family} base -> {ns.out}")
         return 0
 
     if ns.role == "serve":

@@ -102,6 +102,65 @@ def export_gpt2_to_hf_state_dict(model: GPT2LM) -> Dict[str, torch.Tensor]:
     return sd
 
 
+@torch.no_grad()
+def export_llama_to_hf_state_dict(model: LlamaLM) -> Dict[str, torch.Tensor]:
+    """Native LlamaLM → transformers-layout state dict (inverse of
+    load_llama_from_hf)."""
+    sd: Dict[str, torch.Tensor] = {}
+    p = "model."
+    sd[p + "embed_tokens.weight"] = model.tok_emb.detach().clone()
+    if model.lm_head_w is not None:
+        sd["lm_head.weight"] = model.lm_head_w.detach().clone()
+    sd[p + "norm.weight"] = model.final_norm_w.detach().clone()
+    for i, blk in enumerate(model.blocks):
+        h = f"{p}layers.{i}."
+        sd[h + "input_layernorm.weight"] = blk.attn_norm_w.detach().clone()
+        sd[h + "self_attn.q_proj.weight"] = blk.q_w.detach().clone()
+        sd[h + "self_attn.k_proj.weight"] = blk.k_w.detach().clone()
+        sd[h + "self_attn.v_proj.weight"] = blk.v_w.detach().clone()
+        sd[h + "self_attn.o_proj.weight"] = blk.o_w.detach().clone()
+        sd[h + "post_attention_layernorm.weight"] = \
+            blk.mlp_norm_w.detach().clone()
+        sd[h + "mlp.gate_proj.weight"] = blk.gate_w.detach().clone()
+        sd[h + "mlp.up_proj.weight"] = blk.up_w.detach().clone()
+        sd[h + "mlp.down_proj.weight"] = blk.down_w.detach().clone()
+    return sd
+
+
+def native_to_hf_dir(model, cfg: ModelConfig, out_dir: str) -> None:
+    """Write an HF ``save_pretrained``-layout directory (config.json +
+    pytorch_model.bin) loadable by transformers ``from_pretrained`` —
+    the export half of the reference-migration story (`cli.py export`)."""
+    import json
+    import os
+
+    os.makedirs(out_dir, exist_ok=True)
+    if cfg.family == "gpt2":
+        hc = {"model_type": "gpt2", "architectures": ["GPT2LMHeadModel"],
+              "vocab_size": cfg.vocab_size, "n_layer": cfg.n_layer,
+              "n_head": cfg.n_head, "n_embd": cfg.n_embd,
+              "n_positions": cfg.n_positions, "n_ctx": cfg.n_positions}
+        sd = export_gpt2_to_hf_state_dict(model)
+    elif cfg.family == "llama":
+        hc = {"model_type": "llama", "architectures": ["LlamaForCausalLM"],
+              "vocab_size": cfg.vocab_size,
+              "hidden_size": cfg.n_embd,
+              "intermediate_size": cfg.intermediate_size,
+              "num_hidden_layers": cfg.n_layer,
+              "num_attention_heads": cfg.n_head,
+              "num_key_value_heads": cfg.n_kv_head,
+              "max_position_embeddings": cfg.n_positions,
+              "rms_norm_eps": cfg.norm_eps,
+              "rope_theta": cfg.rope_theta,
+              "tie_word_embeddings": cfg.tie_word_embeddings}
+        sd = export_llama_to_hf_state_dict(model)
+    else:
+        raise ValueError(f"unsupported family {cfg.family!r}")
+    with open(os.path.join(out_dir, "config.json"), "w") as f:
+        json.dump(hc, f, indent=1)
+    torch.save(sd, os.path.join(out_dir, "pytorch_model.bin"))
+
+
 def hf_dir_to_native(path: str):
     """Load an HF ``save_pretrained`` checkpoint DIRECTORY (config.json +
     model.safetensors / pytorch_model.bin) into a native model — no
