@@ -11,6 +11,8 @@ Usage:
   python -m distributedtraining_amd.cli averager  --comm.root /tmp/ex --rounds 1
   python -m distributedtraining_amd.cli bootstrap --port 8500
   python -m distributedtraining_amd.cli serve     --comm.root /tmp/ex --port 8600
+  python -m distributedtraining_amd.cli convert   --comm.root /tmp/ex --src ./hf_ckpt
+  python -m distributedtraining_amd.cli export    --comm.root /tmp/ex --out ./hf_out
 
 Every role writes a metrics JSONL under --metrics-dir (utils/metrics.py);
 --resume restores a miner's saved train state (base + step counter) from
