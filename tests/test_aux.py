@@ -59,12 +59,20 @@ def _fast(x):
     return x * 2
 
 
+def _boom():
+    raise ValueError("child failure")
+
+
 def test_run_with_ttl():
     assert run_with_ttl(_fast, 10.0, 21) == 42
     t0 = time.time()
     with pytest.raises(TTLTimeout):
         run_with_ttl(_slow, 1.0)
     assert time.time() - t0 < 10
+    # a child exception is re-raised in the parent (reference semantics:
+    # chain_manager.run_in_subprocess propagates failures)
+    with pytest.raises(ValueError, match="child failure"):
+        run_with_ttl(_boom, 10.0)
 
 
 # ---------------------------------------------------------------- keys
