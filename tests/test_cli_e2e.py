@@ -39,7 +39,8 @@ def test_three_role_pipeline(tmp_path):
     _run(["miner", "--hotkey", "m1", "--steps", "6", *common])
 
     # both miners registered + deltas present
-    reg = json.load(open(os.path.join(root, "registry.json")))
+    with open(os.path.join(root, "registry.json")) as f:
+        reg = json.load(f)
     assert set(reg["addresses"]) >= {"m0", "m1"}
     assert os.path.exists(os.path.join(root, "grads", "m0",
                                        "weight_diff.pt"))
