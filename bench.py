@@ -47,7 +47,10 @@ def main() -> int:
     p.add_argument("--merge-every", type=int, default=25)
     p.add_argument("--model", default="gpt2-small",
                    choices=["gpt2-small", "gpt2-tiny", "llama3-8b"])
-    p.add_argument("--merge-strategy", default="mean")
+    p.add_argument("--merge-strategy", default="mean",
+                   choices=["mean", "nesterov", "parameterized"])
+    # (score_weighted needs validator scores, which the throughput bench
+    # does not produce — benchmarks/bench_roles.py measures that path)
     p.add_argument("--no-graph", action="store_true",
                    help="disable hipGraph step capture (eager launches)")
     p.add_argument("--cpu", action="store_true",
