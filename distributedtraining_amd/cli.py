@@ -136,8 +136,12 @@ def main(argv=None) -> int:
         from .parallel.flat import FlatParams as _FP
         from .store import FileStore as _FS
         from .utils.serve import InferenceServer
-        device = torch.device(cfg.comm.device if not torch.cuda.is_available()
-                              else "cuda:0")
+        # honor an explicit --comm.device (e.g. cuda:1 to serve beside a
+        # training rank); default to cuda:0 when a GPU exists
+        if cfg.comm.device.startswith("cuda") or not torch.cuda.is_available():
+            device = torch.device(cfg.comm.device)
+        else:
+            device = torch.device("cuda:0")
         torch.manual_seed(cfg.seed)
         model = _bm(cfg.model).to(device)
         if device.type == "cuda":
