@@ -30,7 +30,7 @@ Initialization is explicit — importing this package has no side effects
 training_manager.py:22-24; we deliberately do not).
 """
 
-__version__ = "0.1.0"
-__spec_version__ = 100 * 0 + 10 * 1 + 0  # reference: hivetrain/__init__.py:1-10
+__version__ = "0.2.0"
+__spec_version__ = 100 * 0 + 10 * 2 + 0  # reference: hivetrain/__init__.py:1-10
 
 from . import config  # noqa: F401

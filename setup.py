@@ -30,7 +30,7 @@ SRC = [
 
 setup(
     name="distributedtraining_amd",
-    version="0.1.0",
+    version="0.2.0",
     packages=find_packages(include=["distributedtraining_amd*"]),
     ext_modules=[
         CUDAExtension(
