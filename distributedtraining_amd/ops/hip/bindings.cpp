@@ -69,6 +69,8 @@ std::vector<Tensor> layernorm_fwd(Tensor x, Tensor res, Tensor w, Tensor b,
   const int64_t rows = x.size(0);
   const int cols = int(x.size(1));
   TORCH_CHECK(cols % 8 == 0, "cols must be a multiple of 8");
+  TORCH_CHECK(cols <= 8192,
+              "norm kernels support cols <= 8192 (register-resident row)");
   const bool has_res = res.numel() > 0;
   if (has_res) check_bf16(res, "res");
   TORCH_CHECK(p == 0.0 || has_res, "dropout needs the residual branch");
@@ -95,6 +97,8 @@ std::vector<Tensor> layernorm_bwd(Tensor dy, Tensor ds, Tensor x, Tensor w,
   const int64_t rows = x.size(0);
   const int cols = int(x.size(1));
   TORCH_CHECK(cols % 8 == 0, "cols must be a multiple of 8");
+  TORCH_CHECK(cols <= 8192,
+              "norm kernels support cols <= 8192 (register-resident row)");
   const bool has_ds = ds.numel() > 0;
   if (has_ds) check_bf16(ds, "ds");
   const DropArgs da = drop_args(rng, site, p);
@@ -123,6 +127,8 @@ std::vector<Tensor> rmsnorm_fwd(Tensor x, Tensor res, Tensor w,
   const int64_t rows = x.size(0);
   const int cols = int(x.size(1));
   TORCH_CHECK(cols % 8 == 0, "cols must be a multiple of 8");
+  TORCH_CHECK(cols <= 8192,
+              "norm kernels support cols <= 8192 (register-resident row)");
   const bool has_res = res.numel() > 0;
   if (has_res) check_bf16(res, "res");
   TORCH_CHECK(p == 0.0 || has_res, "dropout needs the residual branch");
@@ -144,6 +150,8 @@ std::vector<Tensor> rmsnorm_bwd(Tensor dy, Tensor ds, Tensor x, Tensor w,
   const int64_t rows = x.size(0);
   const int cols = int(x.size(1));
   TORCH_CHECK(cols % 8 == 0, "cols must be a multiple of 8");
+  TORCH_CHECK(cols <= 8192,
+              "norm kernels support cols <= 8192 (register-resident row)");
   const bool has_ds = ds.numel() > 0;
   if (has_ds) check_bf16(ds, "ds");
   const DropArgs da = drop_args(rng, site, p);
