@@ -494,6 +494,7 @@ Tensor gemv(Tensor x, Tensor w, Tensor bias) {
   TORCH_CHECK(M >= 1 && M <= 4, "gemv is for M <= 4");
   const int64_t N = w.size(0);
   const int K = int(x.size(1));
+  TORCH_CHECK(K % 8 == 0, "gemv needs K % 8 == 0 (16 B row alignment)");
   const bf16_t* bp = nullptr;
   if (bias.numel()) {
     check_bf16(bias, "bias");
