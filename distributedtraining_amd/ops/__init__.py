@@ -133,6 +133,8 @@ def _gemv_wins(rows: int, w: torch.Tensor) -> bool:
     """Measured crossover (tools/gemv_ab.py): the hand-written GEMV beats
     tuned hipBLASLt while the weight matrix is latency- not
     bandwidth-bound."""
+    if w.shape[1] % 8 != 0:      # 16 B row-alignment contract of gemv.hip
+        return False
     numel = w.shape[0] * w.shape[1]
     return numel < (20_000_000 if rows == 1 else 5_000_000)
 
