@@ -78,6 +78,16 @@ class LlamaLM(nn.Module):
     def __init__(self, cfg: ModelConfig):
         super().__init__()
         assert cfg.family == "llama"
+        if cfg.resid_pdrop or cfg.embd_pdrop or cfg.attn_pdrop:
+            # the Llama family trains dropout-free (Llama-2/3 recipes);
+            # the pdrop fields are GPT-2 semantics and are NOT applied
+            # here — say so instead of silently ignoring them
+            import logging
+            logging.getLogger(__name__).warning(
+                "LlamaLM ignores pdrop settings (%.2f/%.2f/%.2f): the "
+                "llama family trains dropout-free; dropout is wired for "
+                "the gpt2 family only",
+                cfg.resid_pdrop, cfg.embd_pdrop, cfg.attn_pdrop)
         self.cfg = cfg
         self.tok_emb = nn.Parameter(torch.empty(cfg.vocab_size, cfg.n_embd))
         self.blocks = nn.ModuleList(LlamaBlock(cfg) for _ in range(cfg.n_layer))
