@@ -14,11 +14,12 @@ import torch
 
 _ext = None
 _tried = False
+_import_error: Exception | None = None
 
 
 def ext():
     """Return the HIP extension module, importing it on first use."""
-    global _ext, _tried
+    global _ext, _tried, _import_error
     if _ext is None and not _tried:
         _tried = True
         try:
@@ -36,7 +37,8 @@ def require_ext():
             "distributedtraining_amd._dta_hip (the gfx950 HIP extension) is not "
             "built. Run `python setup.py build_ext --inplace` (or "
             "__graft_entry__.build()) — GPU execution without the native "
-            "kernels is not supported by design.")
+            "kernels is not supported by design. "
+            f"(import error: {_import_error!r})")
     return m
 
 
