@@ -130,9 +130,15 @@ class DeltaValidator:
                      ppl_score)
             if self.registry is not None:
                 self.registry.report_metric(hotkey, loss)
-        for hotkey in deltas:
-            self.normalized_scores[hotkey] = (
-                max(0.0, self.scores[hotkey] / total) if total > 0 else 0.0)
+        # normalize over THIS round's scored set only — a dict that kept
+        # every hotkey ever scored would feed stale entries for
+        # deregistered miners into the registry's EMA fold forever
+        # (caught by the membership-churn test); self.scores keeps the
+        # cumulative history like the reference's ModelValidator
+        self.normalized_scores = {
+            hotkey: (max(0.0, self.scores[hotkey] / total)
+                     if total > 0 else 0.0)
+            for hotkey in deltas}
         if self.registry is not None and self.registry.should_set_weights():
             self.registry.set_weights(self.normalized_scores)
         return dict(self.normalized_scores)

@@ -295,3 +295,53 @@ def test_concurrent_roles_soak(tmp_path):
     final = FileStore(str(tmp_path)).pull_model()
     assert final is not None and "flat_master" in final
     assert bool(torch.isfinite(final["flat_master"]).all())
+
+
+def test_membership_churn_elasticity(tmp_path):
+    """The protocol's elasticity contract (reference: the averager merges
+    whatever deltas exist, vanished miners score 0): over several rounds,
+    miners randomly join, push, go silent, or deregister; every round the
+    validator scores and the averager merges. Nothing may crash, scores
+    must cover exactly the registered hotkeys, and the base must stay
+    finite through every merge."""
+    import random
+
+    from distributedtraining_amd.config import ValidateConfig
+
+    rng = random.Random(7)
+    cfg, model, fp, store, registry = _mk(tmp_path, "seed")
+    store.push_model({"format": "dta-base-v1", "flat_master": fp.master.cpu(),
+                      "spec": fp.spec})
+    ev = synthetic_eval_set(cfg.model.vocab_size, 1, 2, 16)
+    validator = DeltaValidator(model, fp, ev, ValidateConfig(),
+                               store=store, registry=registry)
+    averager = ParameterizedAverager(model, fp, AverageConfig(strategy="mean"),
+                                     store=store, registry=registry)
+    pool = [f"miner{i}" for i in range(6)]
+    active: set = set()
+    for rnd in range(6):
+        # churn: some join, some leave
+        for hk in pool:
+            if hk not in active and rng.random() < 0.5:
+                active.add(hk)
+            elif hk in active and rng.random() < 0.25:
+                active.discard(hk)
+                registry.deregister(hk)
+        # a random subset of active miners trains + pushes this round
+        for hk in sorted(active):
+            if rng.random() < 0.7:
+                st = FileStore(str(tmp_path), hotkey=hk)
+                data = synthetic_batches(cfg.model.vocab_size, 2, 16,
+                                         seed=hash(hk) % 1000 + rnd)
+                loop = DeltaLoop(model, fp, data, cfg.train, store=st,
+                                 registry=registry, hotkey=hk)
+                loop.maybe_pull_base()
+                loop.train(2)
+                loop.last_push_step = -10**9
+                loop.maybe_push_delta()
+        scores = validator.validate_and_score()
+        assert set(scores) == set(registry.hotkeys)
+        assert all(0.0 <= v <= 1.0 for v in scores.values())
+        averager.run_round(ev)
+        base = store.pull_model()["flat_master"]
+        assert bool(torch.isfinite(base).all())
