@@ -47,6 +47,7 @@ class DeltaValidator:
         self.registry = registry
         self.scores: Dict[str, float] = {}
         self.normalized_scores: Dict[str, float] = {}
+        self._base_hash: Optional[str] = None
         self.base_loss, self.base_perplexity = self.evaluate_model()
 
     # -- evaluation (reference :78-97) --------------------------------------
@@ -69,7 +70,13 @@ class DeltaValidator:
 
     def refresh_base(self, flat_fp32: torch.Tensor) -> None:
         self.fp.load_flat_master(flat_fp32)
+        self._base_hash = None     # recomputed lazily on first stale check
         self.base_loss, self.base_perplexity = self.evaluate_model()
+
+    def _current_base_hash(self) -> str:
+        if self._base_hash is None:
+            self._base_hash = self.fp.master_hash()
+        return self._base_hash
 
     def maybe_pull_base(self) -> bool:
         if self.store is None or not self.store.check_for_new_model():
@@ -92,6 +99,13 @@ class DeltaValidator:
         if ops.has_nan(delta):
             log.warning("NaN delta rejected")
             return 1e8, 1e8, 0.0, 0.0
+        if ckpt.base_hash and ckpt.base_hash != self._current_base_hash():
+            # stale delta (computed against an older base): still scored —
+            # local-SGD tolerates staleness by design (the reference never
+            # even pins base identity) — but surfaced for operators
+            log.warning("delta base_hash %s != current base %s (stale "
+                        "delta, scoring anyway)", ckpt.base_hash[:12],
+                        self._current_base_hash()[:12])
         saved = self.fp.master.clone()
         ops.axpy_(self.fp.master, delta, 1.0)   # θ += δ (reference :252-259)
         self.fp.sync_work_from_master()

@@ -345,3 +345,23 @@ def test_membership_churn_elasticity(tmp_path):
         averager.run_round(ev)
         base = store.pull_model()["flat_master"]
         assert bool(torch.isfinite(base).all())
+
+
+def test_validator_warns_on_stale_base_hash(tmp_path, caplog):
+    """A delta pinned to an older base is still scored (local-SGD
+    staleness tolerance) but surfaced in the logs."""
+    import logging
+
+    cfg, model, fp, store, registry = _mk(tmp_path, "v")
+    ev = synthetic_eval_set(cfg.model.vocab_size, 1, 2, 16)
+    from distributedtraining_amd.config import ValidateConfig
+    v = DeltaValidator(model, fp, ev, ValidateConfig())
+    fresh = DeltaCheckpoint(torch.randn(fp.numel) * 1e-3, fp.spec,
+                            base_hash=fp.master_hash())
+    stale = DeltaCheckpoint(torch.randn(fp.numel) * 1e-3, fp.spec,
+                            base_hash="deadbeef" * 8)
+    with caplog.at_level(logging.WARNING):
+        v.score_delta(fresh)
+        assert not any("stale" in r.message for r in caplog.records)
+        v.score_delta(stale)
+        assert any("stale" in r.message for r in caplog.records)
