@@ -106,3 +106,19 @@ def test_file_registry_concurrent_writers(tmp_path):
     assert hks >= expect
     for w in range(N_WRITERS):
         assert final.retrieve_address(f"hk{w}_0") == f"addr{w}_0"
+
+
+def test_file_registry_survives_corrupt_json(tmp_path):
+    """A half-written or garbage registry.json must not crash readers or
+    writers; the next clean save repairs the file."""
+    a = FileRegistry(str(tmp_path))
+    a.store_address("hk1", "addr1")
+    with open(a.path, "w") as f:
+        f.write('{"addresses": {"hk1"')     # truncated JSON
+    b = FileRegistry(str(tmp_path))          # load of garbage -> tolerated
+    assert b.retrieve_address("hk_missing") is None
+    b.store_address("hk2", "addr2")          # write-through repairs
+    c = FileRegistry(str(tmp_path))
+    assert c.retrieve_address("hk2") == "addr2"
+    # a still holds its in-memory copy and re-merges cleanly
+    assert a.retrieve_address("hk1") == "addr1"
