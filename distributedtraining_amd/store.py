@@ -144,8 +144,14 @@ class FileStore:
         if not os.path.exists(self.model_path):
             return None
         self._last_seen_hash = self._file_hash(self.model_path)
-        return torch.load(self.model_path, map_location=map_location,
-                          weights_only=False)
+        try:
+            return torch.load(self.model_path, map_location=map_location,
+                              weights_only=False)
+        except Exception:
+            # corrupt/foreign file: treated like absence (the delta channel
+            # already does — receive_delta). The recorded hash prevents a
+            # retry loop; a later clean push changes the hash and re-pulls.
+            return None
 
     # -- delta side (reference hf_manager.py:91-114,186-197) -----------------
     def my_address(self) -> str:

@@ -103,3 +103,21 @@ def test_file_store_concurrent_push_pull(tmp_path):
     for t in trs:
         t.join()
     assert not errors
+
+
+def test_pull_model_survives_corrupt_file(tmp_path):
+    """A corrupt model file reads as absence (like the delta channel);
+    a subsequent clean push is picked up again."""
+    import torch
+
+    from distributedtraining_amd.store import FileStore
+
+    store = FileStore(str(tmp_path))
+    with open(store.model_path, "wb") as f:
+        f.write(b"not a torch file")
+    assert store.pull_model() is None
+    assert not store.check_for_new_model()    # corrupt hash recorded
+    store.push_model({"format": "dta-base-v1",
+                      "flat_master": torch.ones(4)})
+    assert store.check_for_new_model()        # new content -> re-pull
+    assert store.pull_model()["flat_master"].sum() == 4
