@@ -31,9 +31,9 @@ while :; do
   start=$(date +%s)
   echo "[supervise] starting role=$ROLE (restart #$restarts)"
   if [ "$ROLE" = bench ]; then
-    (cd "$HERE" && python bench.py "$@") &
+    (cd "$HERE" && exec python bench.py "$@") &
   else
-    (cd "$HERE" && python -m distributedtraining_amd.cli "$ROLE" "$@") &
+    (cd "$HERE" && exec python -m distributedtraining_amd.cli "$ROLE" "$@") &
   fi
   child=$!
 
