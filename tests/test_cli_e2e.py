@@ -59,6 +59,15 @@ def test_three_role_pipeline(tmp_path):
                             weights_only=False)["flat_master"]
     assert not torch.equal(base_before, base_after)
 
+    # score-weighted merge consumes the validator's EMA-folded registry
+    # scores (C7 in SURVEY §2.4): the full validator -> averager loop
+    out_sw = _run(["averager", "--rounds", "1",
+                   "--average.strategy", "score_weighted", *common])
+    assert "merged + published" in out_sw
+    base_sw = torch.load(os.path.join(root, "model", "averaged_model.pt"),
+                         weights_only=False)["flat_master"]
+    assert not torch.equal(base_after, base_sw)
+
     # resumed miner picks up its saved train state
     out2 = _run(["miner", "--hotkey", "m0", "--steps", "2", "--resume",
                  *common])
