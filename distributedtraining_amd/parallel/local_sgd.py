@@ -8,9 +8,9 @@ installed on every rank — replacing the reference's
 miner→HF-hub→averager→HF-hub→miner round trip.
 
 Collective choice per strategy (see merge_round): mean/nesterov use an
-all-reduce (O(P) resident — mandatory at Llama scale); score-weighted and
-meta-learned use an all-gather (every delta HBM-resident, optionally bf16
-on the wire). Deterministic merges are computed redundantly by every rank
+all-reduce (O(P) resident — mandatory at Llama scale); score-weighted,
+meta-learned and genetic use an all-gather (every delta HBM-resident,
+optionally bf16/int8 on the wire). Deterministic merges are computed redundantly by every rank
 (identical bases, no broadcast); the meta-learned merge runs on rank 0
 (it needs val-loss backward passes) and is broadcast (C2/C5).
 """
@@ -84,9 +84,9 @@ class LocalSGDNode:
         """Exchange + merge. Memory shapes:
         * mean/nesterov: ONE all-reduce of the flat delta — O(P) resident
           (required for Llama-3-8B x 8 ranks: a gather would need 256 GB);
-        * score_weighted/parameterized: all-gather, all deltas HBM-resident
-          (8 fp32 GPT-2 deltas ≈ 4 GB — the scoring/meta paths need every
-          delta individually)."""
+        * score_weighted/parameterized/genetic: all-gather, all deltas
+          HBM-resident (8 fp32 GPT-2 deltas ≈ 4 GB — the scoring/meta/
+          evolutionary paths need every delta individually)."""
         base = self.miner.base
         delta = self.fp.make_delta(base)     # fused θ−θ_base
         strat = self.merge_strategy
@@ -100,11 +100,15 @@ class LocalSGDNode:
             assert scores is not None
             merged = self.averager.score_weighted_merge(
                 base, self._gather_deltas(delta.flat), scores)
-        elif strat == "parameterized":
+        elif strat in ("parameterized", "genetic"):
+            # both need val-loss evaluations: run on rank 0, broadcast
             deltas = self._gather_deltas(delta.flat)
             if self.comm.rank == 0:
-                merged = self.averager.meta_learning(base, deltas,
-                                                     self.val_batches)
+                merged = (self.averager.meta_learning(base, deltas,
+                                                      self.val_batches)
+                          if strat == "parameterized" else
+                          self.averager.genetic_merge(base, deltas,
+                                                      self.val_batches))
             else:
                 merged = torch.empty_like(base)
             self.comm.broadcast_flat(merged, src=0)

@@ -54,6 +54,12 @@ def _worker(rank, world, port, tmpdir, q):
         node.merge_strategy = "parameterized"
         node.train_steps(2)
         node.merge_round()
+        # genetic merge path (rank-0 evolutionary search + broadcast)
+        node.merge_strategy = "genetic"
+        node.averager.cfg.population_size = 4
+        node.averager.cfg.generations = 1
+        node.train_steps(1)
+        node.merge_round()
         digest = float(fp.master.sum())
         q.put((rank, bool(ok_merge), digest))
         comm.close()
