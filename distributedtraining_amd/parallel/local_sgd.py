@@ -10,9 +10,10 @@ miner→HF-hub→averager→HF-hub→miner round trip.
 Collective choice per strategy (see merge_round): mean/nesterov use an
 all-reduce (O(P) resident — mandatory at Llama scale); score-weighted,
 meta-learned and genetic use an all-gather (every delta HBM-resident,
-optionally bf16/int8 on the wire). Deterministic merges are computed redundantly by every rank
-(identical bases, no broadcast); the meta-learned merge runs on rank 0
-(it needs val-loss backward passes) and is broadcast (C2/C5).
+optionally bf16/int8 on the wire). Deterministic merges are computed
+redundantly by every rank (identical bases, no broadcast); the
+meta-learned and genetic merges run on rank 0 (they need val-loss
+evaluations) and are broadcast (C2/C5).
 """
 
 from __future__ import annotations
