@@ -38,6 +38,10 @@ class ModelConfig:
     rope_theta: float = 500000.0
     norm_eps: float = 1e-5
     tie_word_embeddings: bool = True
+    # qwen2-only knob: biases on the q/k/v projections (the sole
+    # architectural difference from llama — family "qwen2" is LlamaLM
+    # with this on)
+    attention_bias: bool = False
     # training dropout — the reference trains transformers GPT-2 with its
     # defaults (resid/embd/attn pdrop 0.1, from_pretrained at
     # neurons/miner.py:60-62); Llama-3 trains dropout-free
@@ -62,6 +66,15 @@ class ModelConfig:
         return ModelConfig(family="llama", vocab_size=128256, n_layer=32,
                            n_head=32, n_kv_head=8, n_embd=4096,
                            intermediate_size=14336, n_positions=8192,
+                           tie_word_embeddings=False, resid_pdrop=0.0,
+                           embd_pdrop=0.0, attn_pdrop=0.0)
+
+    @staticmethod
+    def qwen2_tiny() -> "ModelConfig":
+        return ModelConfig(family="qwen2", vocab_size=512, n_layer=2,
+                           n_head=4, n_kv_head=2, n_embd=64,
+                           intermediate_size=176, n_positions=128,
+                           rope_theta=10000.0, attention_bias=True,
                            tie_word_embeddings=False, resid_pdrop=0.0,
                            embd_pdrop=0.0, attn_pdrop=0.0)
 

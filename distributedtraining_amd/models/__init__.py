@@ -12,7 +12,7 @@ __all__ = ["build_model", "generate", "GPT2LM", "LlamaLM", "FeedforwardNN",
 def build_model(cfg: ModelConfig):
     if cfg.family == "gpt2":
         return GPT2LM(cfg)
-    if cfg.family == "llama":
+    if cfg.family in ("llama", "qwen2"):
         return LlamaLM(cfg)
     if cfg.family == "mlp":
         return FeedforwardNN()

@@ -9,7 +9,8 @@ from a transformers model object or a plain state dict (no network
 needed: any locally saved checkpoint works).
 
 Covers GPT-2 (Conv1D [in,out] → linear [out,in], c_attn packed qkv kept
-packed) and the Llama family (q/k/v/o/gate/up/down + norms).
+packed), the Llama family (q/k/v/o/gate/up/down + norms) and Qwen2
+(llama layout + q/k/v biases).
 """
 
 from __future__ import annotations
@@ -118,6 +119,10 @@ def export_llama_to_hf_state_dict(model: LlamaLM) -> Dict[str, torch.Tensor]:
         sd[h + "self_attn.q_proj.weight"] = blk.q_w.detach().clone()
         sd[h + "self_attn.k_proj.weight"] = blk.k_w.detach().clone()
         sd[h + "self_attn.v_proj.weight"] = blk.v_w.detach().clone()
+        if blk.q_b is not None:
+            sd[h + "self_attn.q_proj.bias"] = blk.q_b.detach().clone()
+            sd[h + "self_attn.k_proj.bias"] = blk.k_b.detach().clone()
+            sd[h + "self_attn.v_proj.bias"] = blk.v_b.detach().clone()
         sd[h + "self_attn.o_proj.weight"] = blk.o_w.detach().clone()
         sd[h + "post_attention_layernorm.weight"] = \
             blk.mlp_norm_w.detach().clone()
@@ -141,8 +146,10 @@ def native_to_hf_dir(model, cfg: ModelConfig, out_dir: str) -> None:
               "n_head": cfg.n_head, "n_embd": cfg.n_embd,
               "n_positions": cfg.n_positions, "n_ctx": cfg.n_positions}
         sd = export_gpt2_to_hf_state_dict(model)
-    elif cfg.family == "llama":
-        hc = {"model_type": "llama", "architectures": ["LlamaForCausalLM"],
+    elif cfg.family in ("llama", "qwen2"):
+        arch = ("Qwen2ForCausalLM" if cfg.family == "qwen2"
+                else "LlamaForCausalLM")
+        hc = {"model_type": cfg.family, "architectures": [arch],
               "vocab_size": cfg.vocab_size,
               "hidden_size": cfg.n_embd,
               "intermediate_size": cfg.intermediate_size,
@@ -153,6 +160,8 @@ def native_to_hf_dir(model, cfg: ModelConfig, out_dir: str) -> None:
               "rms_norm_eps": cfg.norm_eps,
               "rope_theta": cfg.rope_theta,
               "tie_word_embeddings": cfg.tie_word_embeddings}
+        if cfg.family == "qwen2":
+            hc["attention_bias"] = cfg.attention_bias
         sd = export_llama_to_hf_state_dict(model)
     else:
         raise ValueError(f"unsupported family {cfg.family!r}")
@@ -199,7 +208,10 @@ def hf_dir_to_native(path: str):
     if mt in ("llama", "mistral"):
         cfg = llama_config_from_hf(hf_cfg)
         return load_llama_from_hf(sd, cfg), cfg
-    raise ValueError(f"unsupported model_type {mt!r} (gpt2/llama)")
+    if mt == "qwen2":
+        cfg = qwen2_config_from_hf(hf_cfg)
+        return load_llama_from_hf(sd, cfg), cfg
+    raise ValueError(f"unsupported model_type {mt!r} (gpt2/llama/qwen2)")
 
 
 def _rope_theta(hf_config) -> float:
@@ -216,6 +228,13 @@ def _rope_theta(hf_config) -> float:
     except AttributeError:
         pass
     return 10000.0
+
+
+def qwen2_config_from_hf(hf_config) -> ModelConfig:
+    cfg = llama_config_from_hf(hf_config)
+    cfg.family = "qwen2"
+    cfg.attention_bias = bool(getattr(hf_config, "attention_bias", True))
+    return cfg
 
 
 def llama_config_from_hf(hf_config) -> ModelConfig:
@@ -257,6 +276,10 @@ def load_llama_from_hf(src, cfg: Optional[ModelConfig] = None) -> LlamaLM:
         blk.q_w.copy_(_get(sd, h + "self_attn.q_proj.weight"))
         blk.k_w.copy_(_get(sd, h + "self_attn.k_proj.weight"))
         blk.v_w.copy_(_get(sd, h + "self_attn.v_proj.weight"))
+        if blk.q_b is not None:          # qwen2 attention biases
+            blk.q_b.copy_(_get(sd, h + "self_attn.q_proj.bias"))
+            blk.k_b.copy_(_get(sd, h + "self_attn.k_proj.bias"))
+            blk.v_b.copy_(_get(sd, h + "self_attn.v_proj.bias"))
         blk.o_w.copy_(_get(sd, h + "self_attn.o_proj.weight"))
         blk.mlp_norm_w.copy_(_get(sd, h + "post_attention_layernorm.weight"))
         blk.gate_w.copy_(_get(sd, h + "mlp.gate_proj.weight"))

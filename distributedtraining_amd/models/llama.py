@@ -4,6 +4,9 @@ The reference has no Llama path (it trains GPT-2 only); this is the scale
 model called for by BASELINE.json config #4 (Llama-3-8B bf16, 8 miners,
 sized for 288 GB HBM3E). Same kernel stack as GPT-2 with the Llama op
 variants: RMSNorm, RoPE, SwiGLU, grouped-query attention.
+
+Also serves the Qwen2 family (cfg.family == "qwen2"): architecturally
+llama plus biases on the q/k/v projections (cfg.attention_bias).
 """
 
 from __future__ import annotations
@@ -43,6 +46,12 @@ class LlamaBlock(nn.Module):
         self.q_w = nn.Parameter(torch.empty(E, E))
         self.k_w = nn.Parameter(torch.empty(kv_dim, E))
         self.v_w = nn.Parameter(torch.empty(kv_dim, E))
+        if cfg.attention_bias:           # qwen2: biases on q/k/v only
+            self.q_b = nn.Parameter(torch.zeros(E))
+            self.k_b = nn.Parameter(torch.zeros(kv_dim))
+            self.v_b = nn.Parameter(torch.zeros(kv_dim))
+        else:
+            self.q_b = self.k_b = self.v_b = None
         self.o_w = nn.Parameter(torch.empty(E, E))
         self.mlp_norm_w = nn.Parameter(torch.ones(E))
         self.gate_w = nn.Parameter(torch.empty(I, E))
@@ -61,9 +70,12 @@ class LlamaBlock(nn.Module):
         else:
             s, h = ops.add_rms_norm(x, pending, self.attn_norm_w,
                                     self.norm_eps)
-        q = ops.linear(h, self.q_w).view(B, S, self.n_head, D).transpose(1, 2)
-        k = ops.linear(h, self.k_w).view(B, S, self.n_kv, D).transpose(1, 2)
-        v = ops.linear(h, self.v_w).view(B, S, self.n_kv, D).transpose(1, 2)
+        q = ops.linear(h, self.q_w, self.q_b) \
+            .view(B, S, self.n_head, D).transpose(1, 2)
+        k = ops.linear(h, self.k_w, self.k_b) \
+            .view(B, S, self.n_kv, D).transpose(1, 2)
+        v = ops.linear(h, self.v_w, self.v_b) \
+            .view(B, S, self.n_kv, D).transpose(1, 2)
         q = ops.rope(q, cos, sin)
         k = ops.rope(k, cos, sin)
         o = ops.causal_attention(q, k, v, kvlen=kvlen)
@@ -77,7 +89,7 @@ class LlamaBlock(nn.Module):
 class LlamaLM(nn.Module):
     def __init__(self, cfg: ModelConfig):
         super().__init__()
-        assert cfg.family == "llama"
+        assert cfg.family in ("llama", "qwen2")
         if cfg.resid_pdrop or cfg.embd_pdrop or cfg.attn_pdrop:
             # the Llama family trains dropout-free (Llama-2/3 recipes);
             # the pdrop fields are GPT-2 semantics and are NOT applied
@@ -111,6 +123,8 @@ class LlamaLM(nn.Module):
             for b in self.blocks:
                 for w in (b.q_w, b.k_w, b.v_w, b.gate_w, b.up_w):
                     w.normal_(0, std)
+                if b.q_b is not None:
+                    b.q_b.zero_(); b.k_b.zero_(); b.v_b.zero_()
                 for w in (b.o_w, b.down_w):
                     w.normal_(0, std / math.sqrt(2 * self.cfg.n_layer))
 
@@ -162,9 +176,12 @@ def _llama_block_attn_cached(blk: LlamaBlock, x: torch.Tensor, cos, sin,
     B, S, E = x.shape
     D = blk.head_dim
     h = ops.rms_norm(x, blk.attn_norm_w, blk.norm_eps)
-    q = ops.linear(h, blk.q_w).view(B, S, blk.n_head, D).transpose(1, 2)
-    k = ops.linear(h, blk.k_w).view(B, S, blk.n_kv, D).transpose(1, 2)
-    v = ops.linear(h, blk.v_w).view(B, S, blk.n_kv, D).transpose(1, 2)
+    q = ops.linear(h, blk.q_w, blk.q_b) \
+        .view(B, S, blk.n_head, D).transpose(1, 2)
+    k = ops.linear(h, blk.k_w, blk.k_b) \
+        .view(B, S, blk.n_kv, D).transpose(1, 2)
+    v = ops.linear(h, blk.v_w, blk.v_b) \
+        .view(B, S, blk.n_kv, D).transpose(1, 2)
     q = ops.rope(q, cos, sin, pos=pos)
     k = ops.rope(k, cos, sin, pos=pos)
     cache.append(i, k.transpose(1, 2), v.transpose(1, 2))
