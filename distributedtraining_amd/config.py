@@ -185,10 +185,14 @@ def _add_dataclass_args(parser: argparse.ArgumentParser, prefix: str, dc) -> Non
         name = f"--{prefix}.{f.name}".replace("_", "-")
         typ = f.type if callable(f.type) and f.type in (int, float, str) else None
         if typ is None:
-            # dataclasses store types as strings under future annotations
+            # dataclasses store types as strings under future annotations —
+            # both "Optional[int]" and "typing.Optional[int]" spellings
+            tname = str(f.type)
+            for pre in ("typing.Optional[", "Optional["):
+                if tname.startswith(pre):
+                    tname = tname[len(pre):].rstrip("]")
             typ = {"int": int, "float": float, "str": str,
-                   "bool": _str2bool}.get(
-                str(f.type).replace("typing.Optional[", "").rstrip("]"), str)
+                   "bool": _str2bool}.get(tname, str)
         parser.add_argument(name, type=typ, default=None, dest=f"{prefix}__{f.name}")
 
 

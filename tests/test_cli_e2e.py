@@ -140,3 +140,30 @@ def test_gradient_mode_pipeline(tmp_path):
                                          "averaged_model.pt"),
                             weights_only=False)["flat_master"]
     assert not torch.equal(base_before, base_after)
+
+
+@pytest.mark.timeout(600)
+def test_pipeline_with_qwen2_family(tmp_path):
+    """The protocol is model-family agnostic: run miner -> validator ->
+    averager over the CLI with the qwen2 config selected via flags."""
+    root = str(tmp_path / "qx")
+    common = ["--comm.root", root, "--metrics-dir", str(tmp_path / "m"),
+              "--model.family", "qwen2", "--model.vocab-size", "512",
+              "--model.n-layer", "2", "--model.n-head", "4",
+              "--model.n-kv-head", "2", "--model.n-embd", "64",
+              "--model.n-positions", "64", "--model.intermediate-size",
+              "176", "--model.rope-theta", "10000.0",
+              "--model.attention-bias", "true",
+              "--model.tie-word-embeddings", "false",
+              "--model.resid-pdrop", "0", "--model.embd-pdrop", "0",
+              "--model.attn-pdrop", "0",
+              "--train.batch-size", "2", "--train.seq-len", "16",
+              "--validate.batch-size", "2", "--validate.seq-len", "16",
+              "--validate.n-eval-batches", "1"]
+    out = _run(["miner", "--hotkey", "q0", "--steps", "4", *common])
+    assert "avg loss" in out
+    out_v = _run(["validator", "--rounds", "1", *common])
+    assert "scores:" in out_v and "q0" in out_v
+    out_a = _run(["averager", "--rounds", "1", "--average.strategy",
+                  "mean", *common])
+    assert "merged + published" in out_a
